@@ -1,0 +1,324 @@
+"""Scheduler in-memory domain model: TaskInfo / JobInfo / NodeInfo / QueueInfo.
+
+Analog of the reference's ``pkg/scheduler/api/{job_info,node_info,
+queue_info}.go`` — but reshaped for tensor batching: tasks with identical
+requests *and* identical scheduling constraints are grouped into a
+``TaskClass`` so one HIP kernel pass places a whole class (the reference
+predicates/scores every (task, node) pair one task at a time,
+util/predicate_helper.go:45).
+"""
+
+from __future__ import annotations
+
+import hashlib
+import json
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Tuple
+
+from .objects import (ANN_PREEMPTABLE, LBL_TASK_SPEC, Pod, Node, PodGroup,
+                      Queue, Toleration)
+from .resource import Resource
+from .types import ALLOCATED_STATUSES, PodGroupPhase, TaskStatus
+
+
+@dataclass
+class TaskInfo:
+    """One schedulable pod (reference api/job_info.go:118-175)."""
+
+    uid: str
+    name: str
+    namespace: str
+    job_key: str
+    role: str                      # task-spec name (volcano.sh/task-spec)
+    request: Resource
+    status: TaskStatus = TaskStatus.PENDING
+    node_name: str = ""
+    priority: int = 0
+    best_effort: bool = False
+    preemptable: bool = False
+    revocable_zone: str = ""
+    pod: Optional[Pod] = None
+
+    @property
+    def key(self) -> str:
+        return f"{self.namespace}/{self.name}"
+
+    def class_signature(self) -> str:
+        """Tasks with equal signatures are scheduled as one batch."""
+        p = self.pod
+        sig = {
+            "role": self.role,
+            "req": sorted(self.request.q.items()),
+            "sel": sorted((p.node_selector or {}).items()) if p else [],
+            "tol": [(t.key, t.operator, t.value, t.effect) for t in (p.tolerations if p else [])],
+            "aff": p.affinity if p else None,
+            "prio": self.priority,
+        }
+        return hashlib.md5(json.dumps(sig, sort_keys=True, default=str).encode()).hexdigest()
+
+    @classmethod
+    def from_pod(cls, pod: Pod, job_key: str) -> "TaskInfo":
+        status = {
+            "Pending": TaskStatus.PENDING if not pod.node_name else TaskStatus.BOUND,
+            "Running": TaskStatus.RUNNING,
+            "Succeeded": TaskStatus.SUCCEEDED,
+            "Failed": TaskStatus.FAILED,
+        }.get(pod.phase, TaskStatus.UNKNOWN)
+        return cls(
+            uid=pod.meta.uid or pod.meta.key,
+            name=pod.meta.name,
+            namespace=pod.meta.namespace,
+            job_key=job_key,
+            role=pod.meta.labels.get(LBL_TASK_SPEC, ""),
+            request=pod.request.clone(),
+            status=status,
+            node_name=pod.node_name,
+            priority=pod.priority,
+            best_effort=pod.best_effort,
+            preemptable=pod.meta.annotations.get(ANN_PREEMPTABLE, "") == "true",
+            pod=pod,
+        )
+
+
+@dataclass
+class TaskClass:
+    """A batch of identical pending tasks of one job — the kernel launch unit."""
+
+    signature: str
+    role: str
+    request: Resource
+    tasks: List[TaskInfo] = field(default_factory=list)
+    priority: int = 0
+    # filled by the snapshot packer:
+    class_id: int = -1
+
+    @property
+    def count(self) -> int:
+        return len(self.tasks)
+
+
+class JobInfo:
+    """A PodGroup plus its tasks (reference api/job_info.go:399-445)."""
+
+    def __init__(self, key: str, podgroup: Optional[PodGroup] = None):
+        self.key = key                       # namespace/name of the podgroup
+        self.podgroup = podgroup
+        self.tasks: Dict[str, TaskInfo] = {}           # task key → info
+        self.task_status_index: Dict[TaskStatus, Dict[str, TaskInfo]] = {}
+        self.job_id = -1                     # dense index assigned by snapshot
+
+    # -- basic accessors ----------------------------------------------------
+    @property
+    def name(self) -> str:
+        return self.key.split("/", 1)[1]
+
+    @property
+    def namespace(self) -> str:
+        return self.key.split("/", 1)[0]
+
+    @property
+    def queue(self) -> str:
+        return self.podgroup.spec.queue if self.podgroup else "default"
+
+    @property
+    def min_available(self) -> int:
+        return self.podgroup.spec.min_member if self.podgroup else 1
+
+    @property
+    def min_task_member(self) -> Dict[str, int]:
+        return self.podgroup.spec.min_task_member if self.podgroup else {}
+
+    @property
+    def priority(self) -> int:
+        if self.podgroup and self.podgroup.meta.annotations.get("priority"):
+            return int(self.podgroup.meta.annotations["priority"])
+        return max((t.priority for t in self.tasks.values()), default=0)
+
+    @property
+    def creation_timestamp(self) -> float:
+        return self.podgroup.meta.creation_timestamp if self.podgroup else 0.0
+
+    @property
+    def phase(self) -> str:
+        return self.podgroup.status.phase if self.podgroup else PodGroupPhase.PENDING.value
+
+    # -- task bookkeeping ---------------------------------------------------
+    def add_task(self, task: TaskInfo) -> None:
+        self.tasks[task.key] = task
+        self.task_status_index.setdefault(task.status, {})[task.key] = task
+
+    def remove_task(self, task_key: str) -> Optional[TaskInfo]:
+        t = self.tasks.pop(task_key, None)
+        if t is not None:
+            self.task_status_index.get(t.status, {}).pop(task_key, None)
+        return t
+
+    def update_task_status(self, task: TaskInfo, status: TaskStatus) -> None:
+        self.task_status_index.get(task.status, {}).pop(task.key, None)
+        task.status = status
+        self.task_status_index.setdefault(status, {})[task.key] = task
+
+    def tasks_with_status(self, *statuses: TaskStatus) -> List[TaskInfo]:
+        out: List[TaskInfo] = []
+        for s in statuses:
+            out.extend(self.task_status_index.get(s, {}).values())
+        return out
+
+    # -- derived quantities -------------------------------------------------
+    @property
+    def pending_tasks(self) -> List[TaskInfo]:
+        return self.tasks_with_status(TaskStatus.PENDING)
+
+    @property
+    def occupied_count(self) -> int:
+        """Tasks holding or promised resources (reference ReadyTaskNum:
+        Bound+Binding+Running+Allocated+Succeeded)."""
+        n = len(self.tasks_with_status(*ALLOCATED_STATUSES))
+        n += len(self.task_status_index.get(TaskStatus.SUCCEEDED, {}))
+        return n
+
+    @property
+    def waiting_count(self) -> int:
+        return len(self.task_status_index.get(TaskStatus.PIPELINED, {}))
+
+    def is_ready(self) -> bool:
+        return self.occupied_count >= self.min_available
+
+    def is_pipelined(self) -> bool:
+        return self.occupied_count + self.waiting_count >= self.min_available
+
+    def is_starving(self) -> bool:
+        return self.occupied_count + self.waiting_count < self.min_available
+
+    def role_occupied(self, role: str) -> int:
+        return sum(1 for t in self.tasks.values()
+                   if t.role == role and (t.status.occupies_node or t.status == TaskStatus.SUCCEEDED))
+
+    def roles_ready(self) -> bool:
+        """Per-role minimums (minTaskMember) — reference CheckTaskReady."""
+        for role, need in self.min_task_member.items():
+            if self.role_occupied(role) < need:
+                return False
+        return True
+
+    def pending_classes(self) -> List[TaskClass]:
+        """Group pending tasks into batching classes (stable order: by role
+        then signature)."""
+        groups: Dict[str, TaskClass] = {}
+        for t in sorted(self.pending_tasks, key=lambda x: (x.role, x.name)):
+            sig = t.class_signature()
+            g = groups.get(sig)
+            if g is None:
+                g = groups[sig] = TaskClass(signature=sig, role=t.role,
+                                            request=t.request.clone(),
+                                            priority=t.priority)
+            g.tasks.append(t)
+        return sorted(groups.values(), key=lambda g: (-g.priority, g.role, g.signature))
+
+    def total_request(self) -> Resource:
+        r = Resource()
+        for t in self.tasks.values():
+            r.add(t.request)
+        return r
+
+    def allocated_resource(self) -> Resource:
+        r = Resource()
+        for t in self.tasks.values():
+            if t.status.occupies_node:
+                r.add(t.request)
+        return r
+
+    def clone_shell(self) -> "JobInfo":
+        j = JobInfo(self.key, self.podgroup)
+        j.job_id = self.job_id
+        return j
+
+
+class NodeInfo:
+    """Per-node state (reference api/node_info.go:52-97)."""
+
+    def __init__(self, node: Node):
+        self.node = node
+        self.name = node.meta.name
+        self.node_id = -1                    # dense index
+        self.allocatable = node.allocatable.clone()
+        self.used = Resource()
+        self.releasing = Resource()
+        self.pipelined = Resource()
+        self.tasks: Dict[str, TaskInfo] = {}
+
+    @property
+    def idle(self) -> Resource:
+        return self.allocatable.clone().sub(self.used)
+
+    @property
+    def future_idle(self) -> Resource:
+        """idle + releasing − pipelined (reference FutureIdle)."""
+        return self.allocatable.clone().sub(self.used).add(self.releasing).sub(self.pipelined)
+
+    @property
+    def ready(self) -> bool:
+        return self.node.ready and not self.node.unschedulable
+
+    def add_task(self, task: TaskInfo) -> None:
+        self.tasks[task.key] = task
+        if task.status.occupies_node:
+            self.used.add(task.request)
+        elif task.status == TaskStatus.RELEASING:
+            self.used.add(task.request)
+            self.releasing.add(task.request)
+        elif task.status == TaskStatus.PIPELINED:
+            self.pipelined.add(task.request)
+
+    def remove_task(self, task: TaskInfo) -> None:
+        if task.key not in self.tasks:
+            return
+        del self.tasks[task.key]
+        if task.status.occupies_node:
+            self.used.sub(task.request)
+        elif task.status == TaskStatus.RELEASING:
+            self.used.sub(task.request)
+            self.releasing.sub(task.request)
+        elif task.status == TaskStatus.PIPELINED:
+            self.pipelined.sub(task.request)
+
+
+class QueueInfo:
+    """Per-queue state (reference api/queue_info.go)."""
+
+    def __init__(self, queue: Queue):
+        self.queue = queue
+        self.name = queue.meta.name
+        self.queue_id = -1
+
+    @property
+    def weight(self) -> int:
+        return max(int(self.queue.spec.weight), 1)
+
+    @property
+    def capability(self) -> Resource:
+        return self.queue.spec.capability
+
+    @property
+    def guarantee(self) -> Resource:
+        return self.queue.spec.guarantee
+
+    @property
+    def deserved_spec(self) -> Resource:
+        return self.queue.spec.deserved
+
+    @property
+    def reclaimable(self) -> bool:
+        return self.queue.spec.reclaimable
+
+    @property
+    def parent(self) -> str:
+        return self.queue.spec.parent
+
+    @property
+    def priority(self) -> int:
+        return self.queue.spec.priority
+
+    @property
+    def is_open(self) -> bool:
+        return self.queue.status.state == "Open"

@@ -1,0 +1,83 @@
+// Whole-cycle launcher: replays an allocate cycle's kernel sequence from a
+// host-built plan with one library call (the Python side would otherwise
+// pay per-launch ctypes overhead per job; here the per-job cost is two
+// hipLaunchKernelGGL enqueues).  Host code only — kernels live in
+// scheduler_kernels.hip.
+//
+// Gang semantics (reference actions/allocate/allocate.go:719-866 +
+// framework/statement.go): jobs run in the host-given priority order;
+// classes of a job see each other's staged usage; a failed gang reverts
+// before the next job's first kernel, so the sequential-cycle semantics
+// are preserved exactly while the whole cycle stays on-device.
+
+#include "vamd_api.h"
+
+extern "C" void vamd_run_cycle(
+    const VamdClassDesc* classes, int n_classes,
+    const VamdJobDesc* jobs, int n_jobs,
+    const float* alloc, float* used, const float* extra,
+    const uint8_t* ready, const int64_t* taints, const int64_t* planes,
+    const float* bias,
+    const float* class_req, const int64_t* class_tol,
+    const int64_t* class_require, const int64_t* class_forbid,
+    const int32_t* class_min, const float* dim_w,
+    float* queue_alloc, const float* queue_limit,
+    float* score_scratch, int* cap_scratch,
+    int* log_nodes, int* log_counts, int* log_len,
+    int* class_placed, int* job_placed, uint8_t* job_flag,
+    int N, int R, int W, hipStream_t stream)
+{
+    (void)n_classes;
+    for (int j = 0; j < n_jobs; ++j) {
+        const VamdJobDesc& job = jobs[j];
+        int nc = job.class_end - job.class_begin;
+        bool single = (nc == 1);
+
+        for (int c = job.class_begin; c < job.class_end; ++c) {
+            const VamdClassDesc& cd = classes[c];
+            const float* ext = (cd.flags & 1) ? extra : nullptr;
+
+            vamd_score_cap(alloc, used, ext, ready, taints, planes,
+                           class_req + (size_t)c * R, class_tol[c],
+                           class_require + (size_t)c * W,
+                           class_forbid + (size_t)c * W,
+                           cd.w_least, cd.w_most, cd.w_bal, dim_w, bias,
+                           score_scratch, cap_scratch, N, R, W, stream);
+
+            // single-class jobs: gang check fused into the commit.  The
+            // effective minimum is how many more tasks the job needs to
+            // become ready (min_available - occupied), but never less than
+            // the class's own role minimum.
+            int fuse_min = -1;
+            if (single) {
+                int need = job.min_available - job.occupied;
+                if (cd.min_needed > need) need = cd.min_needed;
+                fuse_min = need > 0 ? need : 0;
+            }
+
+            vamd_select_commit(score_scratch, cap_scratch,
+                               class_req + (size_t)c * R, cd.ntasks, used,
+                               queue_alloc + (size_t)cd.queue_idx * R,
+                               queue_limit + (size_t)cd.queue_idx * R,
+                               log_nodes + cd.log_off, log_counts + cd.log_off,
+                               log_len + c, class_placed + c, job_placed + j,
+                               fuse_min, N, R, cd.log_cap, stream);
+        }
+
+        if (!single) {
+            vamd_finalize_job(job_placed + j, job.occupied, job.min_available,
+                              class_placed + job.class_begin,
+                              class_min + job.class_begin, job_flag + j, nc,
+                              stream);
+            for (int c = job.class_begin; c < job.class_end; ++c) {
+                const VamdClassDesc& cd = classes[c];
+                vamd_cond_revert(job_flag + j, log_nodes + cd.log_off,
+                                 log_counts + cd.log_off, log_len + c,
+                                 class_req + (size_t)c * R, used,
+                                 queue_alloc + (size_t)cd.queue_idx * R,
+                                 class_placed + c, job_placed + j, N, R,
+                                 stream);
+            }
+        }
+    }
+}
