@@ -1,0 +1,21 @@
+"""Per-cycle algorithm pipeline (reference ``pkg/scheduler/actions/``).
+
+Registry mirrors ``actions/factory.go``.
+"""
+
+from .allocate import AllocateAction
+from .backfill import BackfillAction
+from .enqueue import EnqueueAction
+
+ACTION_REGISTRY = {
+    "enqueue": EnqueueAction,
+    "allocate": AllocateAction,
+    "backfill": BackfillAction,
+}
+
+
+def new_action(name: str):
+    try:
+        return ACTION_REGISTRY[name]()
+    except KeyError:
+        raise KeyError(f"unknown action {name!r}; known: {sorted(ACTION_REGISTRY)}")

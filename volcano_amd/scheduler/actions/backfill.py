@@ -1,0 +1,68 @@
+"""Backfill action (reference ``actions/backfill/backfill.go:58-120``).
+
+Places BestEffort (zero-request) pending tasks of Inqueue/Running jobs on
+any feasible node after allocate — fills fragments.  Reuses the same
+score/select kernels: a zero request makes every ready node capacity-
+unbounded, so one select pass spreads the class by score.
+"""
+
+from __future__ import annotations
+
+from typing import List
+
+import numpy as np
+
+from ...api.types import PodGroupPhase, TaskStatus
+from ..plan import ClassPlan, CyclePlan, run_plan_hip, run_plan_torch
+
+
+class BackfillAction:
+    name = "backfill"
+
+    def execute(self, ssn) -> None:
+        nt = ssn.node_tensors
+        if nt is None or nt.n == 0:
+            return
+        plan = CyclePlan(nt, ssn.queue_limit, ssn.queue_alloc)
+        plan.dim_w = ssn.dim_weight_vector()
+        predicates = getattr(ssn, "predicates", None)
+
+        for job in ssn.jobs.values():
+            if job.phase not in (PodGroupPhase.INQUEUE.value,
+                                 PodGroupPhase.RUNNING.value):
+                continue
+            classes: List[ClassPlan] = []
+            qi = ssn.queue_index.get(job.queue)
+            if qi is None:
+                continue
+            for tc in job.pending_classes():
+                if not tc.tasks[0].best_effort:
+                    continue
+                req = np.zeros(nt.r, dtype=np.float32)
+                if predicates is not None:
+                    tol, require, forbid = predicates.class_constraints(tc)
+                else:
+                    tol, require, forbid = -1, \
+                        np.zeros(max(nt.labels.words, 1), dtype=np.int64), \
+                        np.zeros(max(nt.labels.words, 1), dtype=np.int64)
+                classes.append(ClassPlan(
+                    tclass=tc, job_key=job.key, queue_idx=qi, req=req,
+                    tolerated=tol, require=require, forbid=forbid,
+                    min_needed=0, w_least=1.0, w_most=0.0, w_bal=0.0))
+            if classes:
+                # backfill is not gang-gated: min 0 via a shim job desc
+                from ..plan import JobPlan
+                begin = len(plan.classes)
+                plan.classes.extend(classes)
+                plan.jobs.append(JobPlan(job_key=job.key, class_begin=begin,
+                                         class_end=len(plan.classes),
+                                         occupied=0, min_available=0))
+
+        if plan.n_classes == 0:
+            return
+        plan.finalize()
+        use_hip = getattr(ssn.config, "use_hip", False)
+        result = (run_plan_hip if use_hip else run_plan_torch)(plan)
+
+        from .allocate import AllocateAction
+        AllocateAction._apply(AllocateAction(), ssn, plan, result)

@@ -1,0 +1,240 @@
+"""Scheduler cache: incremental mirror of the store + bind pipeline.
+
+Analog of ``pkg/scheduler/cache/`` (SchedulerCache, event_handlers.go,
+interface.go Binder/Evictor).  Key MI355X-first departure: the reference
+deep-clones the whole world every cycle (cache.go:1481 Snapshot — its
+dominant fixed cost, SURVEY.md §7 hard-parts); here the cache applies
+store watch deltas incrementally to persistent infos, and the per-cycle
+"snapshot" is just a tensor re-pack of node planes (the session borrows
+the live infos — the cycle is single-threaded by design, so there is
+nothing to clone; the reference needs the clone because its informers
+mutate concurrently).
+"""
+
+from __future__ import annotations
+
+import threading
+from typing import Dict, List, Optional
+
+import torch
+
+from ..api.info import JobInfo, NodeInfo, QueueInfo, TaskInfo
+from ..api.objects import ObjectMeta, PodGroup, Queue
+from ..api.resource import ResourceDims
+from ..api.types import TaskStatus
+from ..store import EventType, ObjectStore
+from .tensors import NodeTensors
+
+
+class Binder:
+    """Commit interface (reference cache/interface.go:113-120)."""
+
+    def bind(self, tasks: List[TaskInfo]) -> None:  # pragma: no cover
+        raise NotImplementedError
+
+    def evict(self, task: TaskInfo, reason: str = "") -> None:  # pragma: no cover
+        raise NotImplementedError
+
+
+class FakeBinder(Binder):
+    """Records binds/evictions (reference util/test_utils.go:536 FakeBinder)
+    — also the bench-mode sink so the GPU loop, not a cluster API, is
+    measured (SURVEY.md §7 'binding throughput')."""
+
+    def __init__(self):
+        self.binds: Dict[str, str] = {}
+        self.evictions: List[str] = []
+
+    def bind(self, tasks: List[TaskInfo]) -> None:
+        for t in tasks:
+            self.binds[t.key] = t.node_name
+
+    def evict(self, task: TaskInfo, reason: str = "") -> None:
+        self.evictions.append(task.key)
+
+
+class StoreBinder(Binder):
+    """Writes the binding back to the object store (the apiserver analog)."""
+
+    def __init__(self, store: ObjectStore):
+        self.store = store
+
+    def bind(self, tasks: List[TaskInfo]) -> None:
+        for t in tasks:
+            pod = self.store.get("Pod", t.namespace, t.name)
+            if pod is not None:
+                pod.node_name = t.node_name
+                self.store.update("Pod", pod)
+
+    def evict(self, task: TaskInfo, reason: str = "") -> None:
+        pod = self.store.get("Pod", task.namespace, task.name)
+        if pod is not None:
+            pod.phase = "Failed"
+            pod.meta.annotations["volcano.sh/evicted"] = reason or "preempted"
+            self.store.update("Pod", pod)
+
+
+class SchedulerCache:
+    def __init__(self, store: Optional[ObjectStore] = None,
+                 binder: Optional[Binder] = None, device: str = "cpu",
+                 scheduler_name: str = "volcano"):
+        self.store = store
+        self.binder = binder or (StoreBinder(store) if store else FakeBinder())
+        self.scheduler_name = scheduler_name
+        self.dims = ResourceDims()
+        self.device = device
+        self.node_tensors = NodeTensors(self.dims, device=device)
+
+        self.jobs: Dict[str, JobInfo] = {}
+        self.nodes: Dict[str, NodeInfo] = {}
+        self.queues: Dict[str, QueueInfo] = {}
+        self._task_node: Dict[str, str] = {}    # task key -> node name
+        self._task_job: Dict[str, str] = {}     # task key -> job key
+        self._lock = threading.RLock()
+        self._watch = store.watch("Pod", "Node", "PodGroup", "Queue") \
+            if store else None
+        self._tensors_dirty = True
+
+    # -- event ingestion (reference cache/event_handlers.go) -----------------
+    def sync(self) -> int:
+        """Drain pending store events into the infos; returns event count."""
+        if self._watch is None:
+            return 0
+        evs = self._watch.drain()
+        # within a batch, structure before pods (a replayed Pod may refer to
+        # a Node/PodGroup whose ADDED event is later in the same drain)
+        kind_rank = {"Node": 0, "Queue": 1, "PodGroup": 2, "Pod": 3}
+        evs.sort(key=lambda e: kind_rank.get(e.kind, 4))
+        with self._lock:
+            for ev in evs:
+                handler = getattr(self, f"_on_{ev.kind.lower()}", None)
+                if handler:
+                    handler(ev)
+            if evs:
+                self._tensors_dirty = True
+        return len(evs)
+
+    def _job_for(self, pg_key: str) -> JobInfo:
+        job = self.jobs.get(pg_key)
+        if job is None:
+            job = self.jobs[pg_key] = JobInfo(pg_key)
+        return job
+
+    def _on_pod(self, ev) -> None:
+        pod = ev.obj
+        if pod.scheduler_name != self.scheduler_name:
+            return
+        pg = pod.podgroup_name
+        if not pg:
+            # normal pod without a group: implicit single-pod group
+            pg = f"pod-{pod.meta.name}"
+        key = f"{pod.meta.namespace}/{pg}"
+        tkey = pod.meta.key
+
+        # remove any previous incarnation
+        old_node = self._task_node.pop(tkey, None)
+        old_job = self._task_job.pop(tkey, None)
+        if old_job is not None and old_job in self.jobs:
+            t = self.jobs[old_job].remove_task(tkey)
+            if t is not None and old_node and old_node in self.nodes:
+                self.nodes[old_node].remove_task(t)
+
+        if ev.type == EventType.DELETED:
+            return
+        task = TaskInfo.from_pod(pod, key)
+        self._job_for(key).add_task(task)
+        self._task_job[tkey] = key
+        if task.node_name and task.node_name in self.nodes:
+            self.nodes[task.node_name].add_task(task)
+            self._task_node[tkey] = task.node_name
+
+    def _on_node(self, ev) -> None:
+        name = ev.obj.meta.name
+        if ev.type == EventType.DELETED:
+            self.nodes.pop(name, None)
+            return
+        old = self.nodes.get(name)
+        ni = NodeInfo(ev.obj)
+        if old is not None:
+            for t in old.tasks.values():
+                ni.add_task(t)
+        self.nodes[name] = ni
+
+    def _on_podgroup(self, ev) -> None:
+        pg: PodGroup = ev.obj
+        key = pg.meta.key
+        if ev.type == EventType.DELETED:
+            self.jobs.pop(key, None)
+            return
+        self._job_for(key).podgroup = pg
+
+    def _on_queue(self, ev) -> None:
+        q: Queue = ev.obj
+        if ev.type == EventType.DELETED:
+            self.queues.pop(q.meta.name, None)
+            return
+        self.queues[q.meta.name] = QueueInfo(q)
+
+    # -- direct population (uthelper-style tests / bench) ---------------------
+    def add_node_info(self, ni: NodeInfo) -> None:
+        self.nodes[ni.name] = ni
+        self._tensors_dirty = True
+
+    def add_job_info(self, job: JobInfo) -> None:
+        self.jobs[job.key] = job
+        for t in job.tasks.values():
+            if t.node_name and t.node_name in self.nodes:
+                self.nodes[t.node_name].add_task(t)
+        self._tensors_dirty = True
+
+    def add_queue_info(self, qi: QueueInfo) -> None:
+        self.queues[qi.name] = qi
+
+    # -- snapshot -------------------------------------------------------------
+    def snapshot_into(self, ssn) -> None:
+        self.sync()
+        if "default" not in self.queues:
+            self.queues["default"] = QueueInfo(
+                Queue(meta=ObjectMeta(name="default")))
+        nodes = sorted(self.nodes.values(), key=lambda n: n.name)
+        if self._tensors_dirty or self.node_tensors.alloc_t is None:
+            self.node_tensors.pack(nodes)
+            self._tensors_dirty = False
+        for i, ni in enumerate(nodes):
+            ni.node_id = i
+        ssn.jobs = self.jobs
+        ssn.nodes = self.nodes
+        ssn.queues = self.queues
+        ssn.node_tensors = self.node_tensors
+        ssn.total_resource = self.node_tensors.alloc_t.sum(dim=1).to("cpu")
+
+    # -- commit pipeline ------------------------------------------------------
+    def bind_tasks(self, tasks: List[TaskInfo]) -> None:
+        """Async in the reference (cache.go:1343 AddBindTask → 20 ms drain);
+        here a batched call — the binder itself may thread if it wants."""
+        for t in tasks:
+            self._task_node[t.key] = t.node_name
+        self.binder.bind(tasks)
+        for t in tasks:
+            job = self.jobs.get(t.job_key)
+            if job is not None:
+                job.update_task_status(t, TaskStatus.BOUND)
+
+    def evict_task(self, task: TaskInfo, reason: str = "") -> None:
+        self.binder.evict(task, reason)
+        job = self.jobs.get(task.job_key)
+        if job is not None:
+            job.update_task_status(task, TaskStatus.RELEASING)
+        node = self.nodes.get(task.node_name)
+        if node is not None:
+            # flip accounting used -> releasing
+            node.remove_task(task)
+            node.add_task(task)
+        self._tensors_dirty = True
+
+    def update_podgroup(self, job: JobInfo) -> None:
+        if self.store is not None and job.podgroup is not None:
+            try:
+                self.store.update("PodGroup", job.podgroup)
+            except KeyError:
+                pass

@@ -1,0 +1,276 @@
+"""Cycle plan: the host-built launch schedule for one allocate pass.
+
+The reference's allocate action walks queues→jobs→tasks with Go callbacks
+per (task, node) (`actions/allocate/allocate.go:719-866`).  Here the host
+builds a *plan* — an ordered array of task-class descriptors and job gang
+descriptors — and the whole cycle executes as one enqueue:
+
+* GPU path: a single ``vamd_run_cycle`` library call replays the plan as
+  back-to-back HIP kernels on one stream, zero host syncs mid-cycle
+  (``ops/csrc/cycle_runner.hip``); one D2H readback returns every
+  placement.
+* CPU path: the same plan interpreted over the torch oracle ops
+  (``ops/reference.py``) — bit-identical decisions, used by the non-GPU
+  test tier and as the golden reference for the HIP kernels.
+"""
+
+from __future__ import annotations
+
+import ctypes
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Tuple
+
+import numpy as np
+import torch
+
+from ..api.info import JobInfo, TaskClass
+from ..ops import reference as ref
+from .tensors import NodeTensors
+
+BIG_LIMIT = 1.0e18     # "no queue limit" sentinel (finite: kernel does int cast)
+
+
+@dataclass
+class ClassPlan:
+    tclass: TaskClass
+    job_key: str
+    queue_idx: int
+    req: np.ndarray            # [R] f32
+    tolerated: int
+    require: np.ndarray        # [W] i64
+    forbid: np.ndarray         # [W] i64
+    min_needed: int            # per-role minimum still missing
+    w_least: float = 1.0
+    w_most: float = 0.0
+    w_bal: float = 0.0
+    use_future: bool = False   # score against future-idle (pipelining)
+    log_off: int = 0
+    log_cap: int = 0
+
+
+@dataclass
+class JobPlan:
+    job_key: str
+    class_begin: int
+    class_end: int
+    occupied: int
+    min_available: int
+
+
+@dataclass
+class ClassResult:
+    placements: List[Tuple[int, int]]   # (node_id, count)
+    placed: int
+
+
+@dataclass
+class CycleResult:
+    class_results: List[ClassResult]
+    job_placed: Dict[str, int]
+    job_committed: Dict[str, bool]
+
+
+class CyclePlan:
+    """Ordered classes + gang jobs + packed per-class constraint tensors."""
+
+    def __init__(self, nt: NodeTensors, queue_limit: torch.Tensor,
+                 queue_alloc: torch.Tensor):
+        self.nt = nt
+        self.classes: List[ClassPlan] = []
+        self.jobs: List[JobPlan] = []
+        self.queue_limit = queue_limit          # [Q, R] f32 (host)
+        self.queue_alloc = queue_alloc          # [Q, R] f32 (host)
+        self.bias: Optional[torch.Tensor] = None
+        self.dim_w = torch.ones(nt.r, dtype=torch.float32)
+        self.log_total = 0
+
+    def add_job(self, job: JobInfo, classes: List[ClassPlan]) -> None:
+        if not classes:
+            return
+        begin = len(self.classes)
+        self.classes.extend(classes)
+        self.jobs.append(JobPlan(
+            job_key=job.key, class_begin=begin, class_end=len(self.classes),
+            occupied=job.occupied_count, min_available=job.min_available))
+
+    def finalize(self) -> None:
+        """Assign undo-log slots (one contiguous region per class)."""
+        off = 0
+        n = self.nt.n
+        for cp in self.classes:
+            cp.log_off = off
+            cp.log_cap = max(1, min(cp.tclass.count, n))
+            off += cp.log_cap
+        self.log_total = off
+
+    @property
+    def n_classes(self) -> int:
+        return len(self.classes)
+
+
+def _flag_of(cp: ClassPlan) -> int:
+    return 1 if cp.use_future else 0
+
+
+def run_plan_torch(plan: CyclePlan) -> CycleResult:
+    """Interpret the plan over the torch oracle ops — mirrors
+    ``cycle_runner.hip`` statement for statement (CPU test tier)."""
+    nt = plan.nt
+    N, R = nt.n, nt.r
+    dev = nt.alloc_t.device
+    C = plan.n_classes
+    if C == 0:
+        return CycleResult([], {}, {})
+
+    alloc = nt.alloc_t.t()          # [N, R] views over the [R, N] planes
+    used = nt.used_t.t()
+    extra_full = nt.extra_t.t()
+    zeros_extra = torch.zeros_like(extra_full)
+    planes = nt.planes_t.t()        # [N, W]
+
+    dim_w = plan.dim_w.to(dev)
+    score = torch.empty(N, dtype=torch.float32, device=dev)
+    cap = torch.empty(N, dtype=torch.int32, device=dev)
+    log_nodes = torch.zeros(plan.log_total, dtype=torch.int32, device=dev)
+    log_counts = torch.zeros(plan.log_total, dtype=torch.int32, device=dev)
+    log_len = torch.zeros(C, dtype=torch.int32, device=dev)
+    class_placed = torch.zeros(C, dtype=torch.int32, device=dev)
+    job_placed = torch.zeros(len(plan.jobs), dtype=torch.int32, device=dev)
+    job_flag = torch.ones(len(plan.jobs), dtype=torch.uint8, device=dev)
+    class_min = torch.tensor([cp.min_needed for cp in plan.classes],
+                             dtype=torch.int32, device=dev)
+
+    for j, jp in enumerate(plan.jobs):
+        nc = jp.class_end - jp.class_begin
+        single = nc == 1
+        for c in range(jp.class_begin, jp.class_end):
+            cp = plan.classes[c]
+            req = torch.from_numpy(cp.req).to(dev)
+            require = torch.from_numpy(cp.require).to(dev)
+            forbid = torch.from_numpy(cp.forbid).to(dev)
+            extra = extra_full if cp.use_future else zeros_extra
+            ref.score_cap(alloc, used, extra, nt.ready.bool(), nt.taint_mask,
+                          planes, req, cp.tolerated, require, forbid,
+                          cp.w_least, cp.w_most, cp.w_bal, dim_w, plan.bias,
+                          score, cap)
+            sl = slice(cp.log_off, cp.log_off + cp.log_cap)
+            ref.select_commit(score, cap, req, cp.tclass.count, used,
+                              plan.queue_alloc[cp.queue_idx],
+                              plan.queue_limit[cp.queue_idx],
+                              log_nodes[sl], log_counts[sl], log_len[c],
+                              class_placed[c], job_placed[j])
+            if single:
+                need = max(jp.min_available - jp.occupied, cp.min_needed, 0)
+                flag = (class_placed[c] >= need).to(torch.uint8)
+                job_flag[j] = flag
+                if int(flag) == 0:
+                    ref.cond_revert(flag, log_nodes[sl], log_counts[sl],
+                                    log_len[c], req, used,
+                                    plan.queue_alloc[cp.queue_idx],
+                                    class_placed[c], job_placed[j])
+        if not single:
+            ref.finalize_job(job_placed[j], jp.occupied, jp.min_available,
+                             class_placed[jp.class_begin:jp.class_end],
+                             class_min[jp.class_begin:jp.class_end],
+                             job_flag[j])
+            for c in range(jp.class_begin, jp.class_end):
+                cp = plan.classes[c]
+                req = torch.from_numpy(cp.req).to(dev)
+                sl = slice(cp.log_off, cp.log_off + cp.log_cap)
+                ref.cond_revert(job_flag[j], log_nodes[sl], log_counts[sl],
+                                log_len[c], req, used,
+                                plan.queue_alloc[cp.queue_idx],
+                                class_placed[c], job_placed[j])
+
+    return _collect(plan, log_nodes, log_counts, log_len, class_placed,
+                    job_placed, job_flag)
+
+
+def run_plan_hip(plan: CyclePlan) -> CycleResult:
+    """One library call → whole cycle on the GPU (no host syncs mid-cycle)."""
+    from ..ops import hip
+
+    nt = plan.nt
+    N, R = nt.n, nt.r
+    W = nt.planes_t.shape[0]
+    dev = nt.alloc_t.device
+    C = plan.n_classes
+    if C == 0:
+        return CycleResult([], {}, {})
+
+    cds = (hip.VamdClassDesc * C)()
+    job_index = {jp.job_key: j for j, jp in enumerate(plan.jobs)}
+    for c, cp in enumerate(plan.classes):
+        d = cds[c]
+        d.job_idx = job_index[cp.job_key]
+        d.queue_idx = cp.queue_idx
+        d.ntasks = cp.tclass.count
+        d.min_needed = cp.min_needed
+        d.log_off = cp.log_off
+        d.log_cap = cp.log_cap
+        d.flags = _flag_of(cp)
+        d.w_least, d.w_most, d.w_bal = cp.w_least, cp.w_most, cp.w_bal
+    jds = (hip.VamdJobDesc * len(plan.jobs))()
+    for j, jp in enumerate(plan.jobs):
+        d = jds[j]
+        d.class_begin, d.class_end = jp.class_begin, jp.class_end
+        d.occupied, d.min_available = jp.occupied, jp.min_available
+
+    class_req = torch.from_numpy(
+        np.stack([cp.req for cp in plan.classes])).to(dev)          # [C,R]
+    class_tol = torch.tensor([cp.tolerated for cp in plan.classes],
+                             dtype=torch.int64).to(dev)
+    class_require = torch.from_numpy(
+        np.stack([cp.require for cp in plan.classes])).to(dev)      # [C,W]
+    class_forbid = torch.from_numpy(
+        np.stack([cp.forbid for cp in plan.classes])).to(dev)
+    class_min = torch.tensor([cp.min_needed for cp in plan.classes],
+                             dtype=torch.int32).to(dev)
+    dim_w = plan.dim_w.to(dev, torch.float32)
+    q_alloc = plan.queue_alloc.to(dev).contiguous()
+    q_limit = plan.queue_limit.to(dev).contiguous()
+    bias = plan.bias.to(dev) if plan.bias is not None else None
+
+    score = torch.empty(N, dtype=torch.float32, device=dev)
+    cap = torch.empty(N, dtype=torch.int32, device=dev)
+    log_nodes = torch.zeros(plan.log_total, dtype=torch.int32, device=dev)
+    log_counts = torch.zeros(plan.log_total, dtype=torch.int32, device=dev)
+    log_len = torch.zeros(C, dtype=torch.int32, device=dev)
+    class_placed = torch.zeros(C, dtype=torch.int32, device=dev)
+    job_placed = torch.zeros(len(plan.jobs), dtype=torch.int32, device=dev)
+    job_flag = torch.ones(len(plan.jobs), dtype=torch.uint8, device=dev)
+
+    hip.run_cycle(
+        ctypes.cast(cds, ctypes.c_void_p), C,
+        ctypes.cast(jds, ctypes.c_void_p), len(plan.jobs),
+        nt.alloc_t, nt.used_t, nt.extra_t, nt.ready, nt.taint_mask,
+        nt.planes_t, bias, class_req, class_tol, class_require, class_forbid,
+        class_min, dim_w, q_alloc, q_limit, score, cap, log_nodes, log_counts,
+        log_len, class_placed, job_placed, job_flag)
+
+    res = _collect(plan, log_nodes.cpu(), log_counts.cpu(), log_len.cpu(),
+                   class_placed.cpu(), job_placed.cpu(), job_flag.cpu())
+    # the cycle mutated device queue_alloc; reflect back to the host copy
+    plan.queue_alloc.copy_(q_alloc.cpu())
+    return res
+
+
+def _collect(plan, log_nodes, log_counts, log_len, class_placed, job_placed,
+             job_flag) -> CycleResult:
+    ln = log_nodes.cpu().numpy()
+    lc = log_counts.cpu().numpy()
+    ll = log_len.cpu().numpy()
+    cpn = class_placed.cpu().numpy()
+    jpn = job_placed.cpu().numpy()
+    jf = job_flag.cpu().numpy()
+
+    class_results: List[ClassResult] = []
+    for c, cp in enumerate(plan.classes):
+        entries = []
+        for e in range(cp.log_off, cp.log_off + int(ll[c])):
+            if lc[e] > 0:
+                entries.append((int(ln[e]), int(lc[e])))
+        class_results.append(ClassResult(entries, int(cpn[c])))
+    job_placed_map = {jp.job_key: int(jpn[j]) for j, jp in enumerate(plan.jobs)}
+    job_committed = {jp.job_key: bool(jf[j]) for j, jp in enumerate(plan.jobs)}
+    return CycleResult(class_results, job_placed_map, job_committed)

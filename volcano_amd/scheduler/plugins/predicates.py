@@ -1,0 +1,44 @@
+"""Predicates plugin (reference ``plugins/predicates/predicates.go``).
+
+The reference wraps the k8s in-tree filter plugins (nodeunschedulable,
+nodeaffinity, tainttoleration, ...) as per-(task,node) Go callbacks run
+16-way parallel over a 5 % node sample (util/predicate_helper.go:45,
+scheduler_helper.go:56).  The MI355X design precompiles the same
+constraints to per-node bit planes and masks at session open; the fused
+score kernel (K1, ops/csrc/scheduler_kernels.hip) evaluates them for ALL
+nodes per class in one pass — no sampling, strictly better placements.
+
+This plugin owns turning each task class's pod constraints into
+(tolerated_mask, require_bits, forbid_bits); the allocate action calls
+``class_constraints``.
+"""
+
+from __future__ import annotations
+
+from typing import Tuple
+
+import numpy as np
+
+from .base import Plugin, register
+
+
+@register("predicates")
+class PredicatesPlugin(Plugin):
+    def on_session_open(self, ssn) -> None:
+        ssn.predicates = self
+
+        self._nt = ssn.node_tensors
+
+    def class_constraints(self, tclass) -> Tuple[int, np.ndarray, np.ndarray]:
+        """(tolerated taint mask, require planes, forbid planes) for a class.
+
+        All tasks of a class share constraints by construction
+        (TaskInfo.class_signature)."""
+        nt = self._nt
+        t = tclass.tasks[0]
+        pod = t.pod
+        tolerations = pod.tolerations if pod else []
+        tolerated = nt.tolerated_mask(tolerations)
+        require, forbid = nt.selector_bits(
+            pod.node_selector if pod else {}, pod.affinity if pod else None)
+        return tolerated, require, forbid

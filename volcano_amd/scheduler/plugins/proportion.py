@@ -1,0 +1,115 @@
+"""Proportion plugin (reference ``plugins/proportion/proportion.go``).
+
+Queue fair share: iterative weighted water-filling of ``deserved``
+capped by capability and demand (proportion.go:90-260), computed as one
+tensor pass (ops.reference.waterfill, float64 accumulators for
+reproducibility).  Registers: QueueOrder by share (:268), Overused
+(:321), Allocatable → the per-queue ``queue_limit`` row the select/commit
+kernel enforces ON DEVICE (:359 — in the reference this is a host
+callback per allocation; here the quota is a tensor bound the kernel
+clamps against, so enforcement costs nothing per task), JobEnqueueable
+(:404), Reclaimable (:288).
+"""
+
+from __future__ import annotations
+
+import torch
+
+from ...ops import reference as ref
+from ..plan import BIG_LIMIT
+from ..session import ABSTAIN, PERMIT, REJECT
+from .base import Plugin, register
+
+
+@register("proportion")
+class ProportionPlugin(Plugin):
+    def on_session_open(self, ssn) -> None:
+        nt = ssn.node_tensors
+        queues = [ssn.queues[name] for name in sorted(ssn.queues)]
+        Q, R = len(queues), nt.r
+        if Q == 0:
+            return
+        total = ssn.total_resource if ssn.total_resource is not None else \
+            torch.zeros(R)
+
+        weight = torch.tensor([q.weight for q in queues], dtype=torch.float32)
+        request = torch.zeros((Q, R), dtype=torch.float32)
+        guarantee = torch.zeros((Q, R), dtype=torch.float32)
+        capability = torch.full((Q, R), BIG_LIMIT, dtype=torch.float32)
+        for i, q in enumerate(queues):
+            guarantee[i] = torch.from_numpy(nt.resource_vector(q.guarantee))
+            cap_vec = nt.resource_vector(q.capability)
+            for r in range(R):
+                if cap_vec[r] > 0:
+                    capability[i, r] = cap_vec[r]
+        for job in ssn.jobs.values():
+            qi = ssn.queue_index.get(job.queue)
+            if qi is not None:
+                request[qi] += torch.from_numpy(
+                    nt.resource_vector(job.total_request()))
+
+        deserved = ref.waterfill(weight, request, guarantee, capability, total)
+        ssn.queue_deserved = deserved
+        # Allocatable enforcement: the kernel clamps each placement against
+        # queue_limit - queue_alloc; deserved IS the limit (proportion
+        # semantics: a queue may not allocate past its deserved share).
+        # Dims with zero cluster total stay unlimited (count-less dims).
+        limit = torch.where(total.unsqueeze(0) > 0, deserved,
+                            torch.full_like(deserved, BIG_LIMIT))
+        ssn.queue_limit = torch.maximum(limit, ssn.queue_alloc)
+        self.deserved = deserved
+        self.queues = queues
+
+        tot64 = torch.clamp(total.to(torch.float64), min=1.0)
+
+        def q_share(qi: int) -> float:
+            return float((ssn.queue_alloc[qi].to(torch.float64) / tot64).amax())
+
+        def queue_order(a, b) -> int:
+            sa, sb = q_share(ssn.queue_index[a.name]), q_share(ssn.queue_index[b.name])
+            if abs(sa - sb) < 1e-12:
+                return 0
+            return -1 if sa < sb else 1
+
+        def overused(q) -> bool:
+            qi = ssn.queue_index[q.name]
+            alloc = ssn.queue_alloc[qi]
+            des = deserved[qi]
+            mask = total > 0
+            return bool((alloc[mask] > des[mask] + 0.1).any())
+
+        def allocatable(q, job) -> bool:
+            # coarse host-side gate; the exact bound is enforced in-kernel
+            return not overused(q)
+
+        def job_enqueueable(job) -> int:
+            qi = ssn.queue_index.get(job.queue)
+            if qi is None:
+                return REJECT
+            minres = torch.from_numpy(
+                nt.resource_vector(job.podgroup.spec.min_resources)) \
+                if job.podgroup else torch.zeros(R)
+            mask = total > 0
+            fits = (ssn.queue_alloc[qi] + minres)[mask] <= deserved[qi][mask] + 0.1
+            return PERMIT if bool(fits.all()) else REJECT
+
+        def reclaimable(reclaimer, candidates):
+            # a victim whose queue is over its deserved share may be
+            # reclaimed (proportion.go:288), if the queue allows it
+            out = []
+            for v in candidates:
+                job = ssn.jobs.get(v.job_key)
+                if job is None:
+                    continue
+                q = ssn.queues.get(job.queue)
+                if q is None or not q.reclaimable:
+                    continue
+                if overused(q):
+                    out.append(v)
+            return out
+
+        ssn.add_queue_order_fn(queue_order)
+        ssn.overused_fns.append(overused)
+        ssn.allocatable_fns.append(allocatable)
+        ssn.job_enqueueable_fns.append(job_enqueueable)
+        ssn.reclaimable_fns.append(reclaimable)

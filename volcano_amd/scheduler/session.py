@@ -1,0 +1,237 @@
+"""Per-cycle Session: snapshot + plugin callback registries.
+
+Analog of the reference's ``pkg/scheduler/framework/session.go`` (the
+Session struct with ~35 extension-point registries and the tiered
+aggregation in ``session_plugins.go``), re-shaped for the tensor decision
+plane:
+
+* callbacks that *order* work (queue/job/task order) or *vote* (enqueueable,
+  pipelined) stay host-side functions — they run over Q/J-sized data;
+* callbacks that in the reference run per (task, node) — predicates and
+  node scores — become **tensor configuration**: plugins contribute label
+  bit-planes, taint masks, score weights and queue limit rows that the HIP
+  kernels consume.
+
+Tier semantics (reference framework/session_plugins.go): order functions
+are compared tier-by-tier, first non-zero wins; JobReady is AND over all
+registered fns (:483); enqueueable/pipelined are votes — any reject wins,
+else any permit, else default (:582/:505); victim functions intersect
+per tier (:225/:274).
+"""
+
+from __future__ import annotations
+
+import time
+from typing import Callable, Dict, List, Optional, Tuple
+
+import torch
+
+from ..api.info import JobInfo, NodeInfo, QueueInfo, TaskInfo
+from .plan import BIG_LIMIT
+from .tensors import NodeTensors
+
+# vote values (reference util.Permit/Abstain/Reject)
+PERMIT, ABSTAIN, REJECT = 1, 0, -1
+
+CmpFn = Callable[[object, object], int]
+
+
+class Session:
+    def __init__(self, cache, config):
+        self.cache = cache
+        self.config = config
+        self.uid = f"ssn-{time.time_ns()}"
+
+        # snapshot views (built by cache.snapshot_into)
+        self.jobs: Dict[str, JobInfo] = {}
+        self.nodes: Dict[str, NodeInfo] = {}
+        self.queues: Dict[str, QueueInfo] = {}
+        self.node_tensors: Optional[NodeTensors] = None
+        self.queue_index: Dict[str, int] = {}
+        self.total_resource: Optional[torch.Tensor] = None  # [R] f32
+
+        # host-side registries, keyed per tier: List[List[fn]]
+        self.job_order_fns: List[List[CmpFn]] = []
+        self.queue_order_fns: List[List[CmpFn]] = []
+        self.task_order_fns: List[List[CmpFn]] = []
+        self.job_valid_fns: List[Callable[[JobInfo], bool]] = []
+        self.job_ready_fns: List[Callable[[JobInfo], bool]] = []
+        self.job_pipelined_fns: List[Callable[[JobInfo], int]] = []
+        self.job_enqueueable_fns: List[Callable[[JobInfo], int]] = []
+        self.job_starving_fns: List[Callable[[JobInfo], bool]] = []
+        self.overused_fns: List[Callable[[QueueInfo], bool]] = []
+        self.allocatable_fns: List[Callable[[QueueInfo, JobInfo], bool]] = []
+        self.preemptable_fns: List[
+            Callable[[TaskInfo, List[TaskInfo]], List[TaskInfo]]] = []
+        self.reclaimable_fns: List[
+            Callable[[TaskInfo, List[TaskInfo]], List[TaskInfo]]] = []
+        self.victim_tasks_fns: List[
+            Callable[[List[TaskInfo]], List[TaskInfo]]] = []
+        self.event_handlers: List[object] = []   # objects w/ allocate/evict hooks
+
+        # tensor-plane configuration contributed by plugins
+        self.score_weights = {"least": 1.0, "most": 0.0, "bal": 0.0}
+        self.dim_weights: Dict[str, float] = {}
+        self.queue_limit: Optional[torch.Tensor] = None   # [Q, R]
+        self.queue_alloc: Optional[torch.Tensor] = None   # [Q, R]
+        self.queue_deserved: Optional[torch.Tensor] = None  # [Q, R]
+
+        self._tier_open = False
+        self.plugins: List[object] = []
+
+    # -- tier plumbing (plugins of one tier register into the same slot) -----
+    def open_tier(self) -> None:
+        self.job_order_fns.append([])
+        self.queue_order_fns.append([])
+        self.task_order_fns.append([])
+
+    def add_job_order_fn(self, fn: CmpFn) -> None:
+        self.job_order_fns[-1].append(fn)
+
+    def add_queue_order_fn(self, fn: CmpFn) -> None:
+        self.queue_order_fns[-1].append(fn)
+
+    def add_task_order_fn(self, fn: CmpFn) -> None:
+        self.task_order_fns[-1].append(fn)
+
+    # -- aggregations (reference session_plugins.go) --------------------------
+    @staticmethod
+    def _tiered_cmp(tiers: List[List[CmpFn]], a, b) -> int:
+        for tier in tiers:
+            for fn in tier:
+                r = fn(a, b)
+                if r != 0:
+                    return r
+        return 0
+
+    def job_order(self, a: JobInfo, b: JobInfo) -> int:
+        r = self._tiered_cmp(self.job_order_fns, a, b)
+        if r != 0:
+            return r
+        # FIFO fallback (reference: creation time then UID)
+        if a.creation_timestamp != b.creation_timestamp:
+            return -1 if a.creation_timestamp < b.creation_timestamp else 1
+        return -1 if a.key < b.key else (1 if a.key > b.key else 0)
+
+    def queue_order(self, a: QueueInfo, b: QueueInfo) -> int:
+        r = self._tiered_cmp(self.queue_order_fns, a, b)
+        if r != 0:
+            return r
+        return -1 if a.name < b.name else (1 if a.name > b.name else 0)
+
+    def task_order(self, a: TaskInfo, b: TaskInfo) -> int:
+        return self._tiered_cmp(self.task_order_fns, a, b)
+
+    def job_valid(self, job: JobInfo) -> bool:
+        return all(fn(job) for fn in self.job_valid_fns)
+
+    def job_ready(self, job: JobInfo) -> bool:
+        return all(fn(job) for fn in self.job_ready_fns)
+
+    def job_pipelined(self, job: JobInfo) -> bool:
+        """First non-abstain vote wins, registration order = tier order
+        (reference session_plugins.go:505)."""
+        for fn in self.job_pipelined_fns:
+            v = fn(job)
+            if v != ABSTAIN:
+                return v == PERMIT
+        return True
+
+    def job_enqueueable(self, job: JobInfo) -> bool:
+        """First non-abstain vote wins (reference session_plugins.go:582)."""
+        for fn in self.job_enqueueable_fns:
+            v = fn(job)
+            if v != ABSTAIN:
+                return v == PERMIT
+        return True   # default permit (reference enqueue falls through)
+
+    def job_starving(self, job: JobInfo) -> bool:
+        if not self.job_starving_fns:
+            return job.is_starving()
+        return any(fn(job) for fn in self.job_starving_fns)
+
+    def queue_overused(self, q: QueueInfo) -> bool:
+        return any(fn(q) for fn in self.overused_fns)
+
+    def allocatable(self, q: QueueInfo, job: JobInfo) -> bool:
+        return all(fn(q, job) for fn in self.allocatable_fns)
+
+    def preemptable(self, preemptor: TaskInfo,
+                    candidates: List[TaskInfo]) -> List[TaskInfo]:
+        """Intersection of victim sets across registered fns (:274)."""
+        victims = candidates
+        for fn in self.preemptable_fns:
+            allowed = {t.uid for t in fn(preemptor, victims)}
+            victims = [t for t in victims if t.uid in allowed]
+            if not victims:
+                return []
+        return victims if self.preemptable_fns else []
+
+    def reclaimable(self, reclaimer: TaskInfo,
+                    candidates: List[TaskInfo]) -> List[TaskInfo]:
+        victims = candidates
+        for fn in self.reclaimable_fns:
+            allowed = {t.uid for t in fn(reclaimer, victims)}
+            victims = [t for t in victims if t.uid in allowed]
+            if not victims:
+                return []
+        return victims if self.reclaimable_fns else []
+
+    def victim_tasks(self, tasks: List[TaskInfo]) -> List[TaskInfo]:
+        out: List[TaskInfo] = []
+        seen = set()
+        for fn in self.victim_tasks_fns:
+            for t in fn(tasks):
+                if t.uid not in seen:
+                    seen.add(t.uid)
+                    out.append(t)
+        return out
+
+    # -- sorted views ---------------------------------------------------------
+    def sorted_queues(self, queues: Optional[List[QueueInfo]] = None) -> List[QueueInfo]:
+        import functools
+        qs = queues if queues is not None else list(self.queues.values())
+        return sorted(qs, key=functools.cmp_to_key(self.queue_order))
+
+    def sorted_jobs(self, jobs: List[JobInfo]) -> List[JobInfo]:
+        import functools
+        return sorted(jobs, key=functools.cmp_to_key(self.job_order))
+
+    # -- queue tensor rows ----------------------------------------------------
+    def build_queue_tensors(self) -> None:
+        """queue_alloc from currently-allocated tasks; limit defaults open.
+        Plugins (proportion/capacity) overwrite limit rows at session open."""
+        nt = self.node_tensors
+        Q, R = len(self.queues), nt.r
+        self.queue_index = {name: i for i, name in enumerate(sorted(self.queues))}
+        alloc = torch.zeros((Q, R), dtype=torch.float32)
+        for job in self.jobs.values():
+            qi = self.queue_index.get(job.queue)
+            if qi is None:
+                continue
+            vec = nt.resource_vector(job.allocated_resource())
+            alloc[qi] += torch.from_numpy(vec)
+        self.queue_alloc = alloc
+        self.queue_limit = torch.full((Q, R), BIG_LIMIT, dtype=torch.float32)
+
+    def dim_weight_vector(self) -> torch.Tensor:
+        nt = self.node_tensors
+        w = torch.ones(nt.r, dtype=torch.float32)
+        for name, val in self.dim_weights.items():
+            i = nt.dims.index.get(name)
+            if i is not None:
+                w[i] = val
+        return w
+
+    # -- events ---------------------------------------------------------------
+    def fire_allocate(self, task_class, node_ids, counts) -> None:
+        for h in self.event_handlers:
+            fn = getattr(h, "on_allocate", None)
+            if fn:
+                fn(task_class, node_ids, counts)
+
+    def fire_evict(self, task: TaskInfo) -> None:
+        for h in self.event_handlers:
+            fn = getattr(h, "on_evict", None)
+            if fn:
+                fn(task)
