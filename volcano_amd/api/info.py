@@ -44,7 +44,11 @@ class TaskInfo:
         return f"{self.namespace}/{self.name}"
 
     def class_signature(self) -> str:
-        """Tasks with equal signatures are scheduled as one batch."""
+        """Tasks with equal signatures are scheduled as one batch.
+        Memoized — constraints are immutable for a task's lifetime."""
+        sig = getattr(self, "_sig", None)
+        if sig is not None:
+            return sig
         p = self.pod
         sig = {
             "role": self.role,
@@ -54,7 +58,10 @@ class TaskInfo:
             "aff": p.affinity if p else None,
             "prio": self.priority,
         }
-        return hashlib.md5(json.dumps(sig, sort_keys=True, default=str).encode()).hexdigest()
+        digest = hashlib.md5(
+            json.dumps(sig, sort_keys=True, default=str).encode()).hexdigest()
+        self._sig = digest
+        return digest
 
     @classmethod
     def from_pod(cls, pod: Pod, job_key: str) -> "TaskInfo":

@@ -77,7 +77,9 @@ __global__ void score_cap_kernel(
             float rq = req[r];
             if (rq > EPS) {
                 feasible = feasible && (avail + EPS >= rq);
-                long long c = (long long)floorf((avail + EPS) / rq);
+                float cc = (avail + EPS) / rq;     // clamp before the int
+                if (cc > (float)BIG_CAP) cc = (float)BIG_CAP;  // cast: UB on inf
+                long long c = (long long)floorf(cc);
                 cap = min(cap, max(c, 0ll));
             }
             float f = fminf((u + rq) / fmaxf(a, EPS), 1.0f);
@@ -156,6 +158,7 @@ select_commit_kernel(
     __shared__ ValIdx s_best;
     __shared__ int s_remaining;
     __shared__ int s_cursor;
+    __shared__ int s_take;
     __shared__ long long s_budget;
 
     const int tid = threadIdx.x;
@@ -170,7 +173,9 @@ select_commit_kernel(
             float rq = req[r];
             if (rq > EPS) {
                 float head = queue_limit[r] - queue_alloc[r];
-                long long q = (long long)floorf((head + EPS) / rq);
+                float qq = (head + EPS) / rq;
+                if (qq > (float)BIG_CAP) qq = (float)BIG_CAP;
+                long long q = (long long)floorf(qq);
                 quota = min(quota, max(q, 0ll));
             }
         }
@@ -203,13 +208,17 @@ select_commit_kernel(
         ValIdx best = s_best;
         if (best.v == NEG_INF || s_remaining <= 0 || s_cursor >= K) break;
 
-        int take = min(cap[best.i], s_remaining);
+        // take is published through LDS: every thread reading s_remaining
+        // directly would race with thread 0's decrement below
         if (tid == 0) {
+            s_take = min(cap[best.i], s_remaining);
             log_nodes[s_cursor] = best.i;
-            log_counts[s_cursor] = take;
+            log_counts[s_cursor] = s_take;
             s_cursor += 1;
-            s_remaining -= take;
+            s_remaining -= s_take;
         }
+        __syncthreads();
+        int take = s_take;
         if (tid < R) {
             used[(size_t)tid * N + best.i] += (float)take * req[tid];
         }

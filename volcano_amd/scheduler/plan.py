@@ -218,8 +218,10 @@ def run_plan_hip(plan: CyclePlan) -> CycleResult:
 
     class_req = torch.from_numpy(
         np.stack([cp.req for cp in plan.classes])).to(dev)          # [C,R]
+    # class_tol stays HOST-side: cycle_runner dereferences it per class on
+    # the CPU and passes the value into the launch (vamd_api.h)
     class_tol = torch.tensor([cp.tolerated for cp in plan.classes],
-                             dtype=torch.int64).to(dev)
+                             dtype=torch.int64)
     class_require = torch.from_numpy(
         np.stack([cp.require for cp in plan.classes])).to(dev)      # [C,W]
     class_forbid = torch.from_numpy(

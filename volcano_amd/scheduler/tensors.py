@@ -55,8 +55,14 @@ class BitRegistry:
         return max(1, (len(self.index) + 63) // 64)
 
 
+def _to_signed64(x: int) -> int:
+    """Wrap a Python uint64 bit pattern into the int64 value range."""
+    x &= (1 << 64) - 1
+    return x - (1 << 64) if x >= (1 << 63) else x
+
+
 def _set_bit(arr: np.ndarray, col: int, bit: int) -> None:
-    arr[bit // 64, col] |= np.int64(1 << (bit % 64))
+    arr[bit // 64, col] |= np.int64(_to_signed64(1 << (bit % 64)))
 
 
 class NodeTensors:
@@ -94,7 +100,8 @@ class NodeTensors:
         return self.taints.bit(f"{taint.key}={taint.value}:{taint.effect}")
 
     def tolerated_mask(self, tolerations: List[Toleration]) -> int:
-        """int64 mask of taint bits the given tolerations cover."""
+        """int64 mask (signed value range) of taint bits the tolerations
+        cover."""
         mask = 0
         for name, bit in self.taints.index.items():
             kv, _, effect = name.rpartition(":")
@@ -104,7 +111,7 @@ class NodeTensors:
                 if tol.tolerates(t):
                     mask |= 1 << bit
                     break
-        return mask
+        return _to_signed64(mask)
 
     def selector_bits(self, selector: Dict[str, str],
                       affinity: Optional[dict]) -> Tuple[np.ndarray, np.ndarray]:
@@ -178,7 +185,7 @@ class NodeTensors:
             ready[i] = 1 if ni.ready else 0
             for t in ni.node.taints:
                 if t.effect in _BLOCKING_EFFECTS:
-                    taints[i] |= np.int64(1 << self.taint_bit(t))
+                    taints[i] |= np.int64(_to_signed64(1 << self.taint_bit(t)))
             for k, v in ni.node.meta.labels.items():
                 _set_bit(planes, i, self.label_bit(k, v))
 
