@@ -91,6 +91,8 @@ def main():
     ap.add_argument("--pods-per-job", type=int, default=10)
     ap.add_argument("--cpu", action="store_true",
                     help="force the CPU torch-oracle path (no GPU)")
+    ap.add_argument("--timing", action="store_true",
+                    help="print per-phase timing summaries to stderr")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -166,6 +168,21 @@ def main():
     ms_per_step = elapsed / args.steps * 1000.0
     p99 = sorted(cycle_times)[max(0, int(len(cycle_times) * 0.99) - 1)] \
         if cycle_times else 0.0
+
+    if args.timing and rank == 0:
+        import sys
+        from volcano_amd.utils.metrics import METRICS
+        for name in ("open_session_duration",
+                     "action_scheduling_latency:enqueue",
+                     "action_scheduling_latency:allocate",
+                     "action_scheduling_latency:backfill",
+                     "allocate:plan_build", "allocate:plan_run",
+                     "allocate:apply", "e2e_scheduling_latency"):
+            s = METRICS.summary(name)
+            if s:
+                print(f"[timing] {name}: mean={s['mean']*1000:.1f}ms "
+                      f"max={s['max']*1000:.1f}ms n={s['count']}",
+                      file=sys.stderr, flush=True)
 
     if rank == 0:
         out = {

@@ -113,6 +113,9 @@ class JobInfo:
         self.tasks: Dict[str, TaskInfo] = {}           # task key → info
         self.task_status_index: Dict[TaskStatus, Dict[str, TaskInfo]] = {}
         self.job_id = -1                     # dense index assigned by snapshot
+        # dense-vector caches (invalidated on task/status mutation)
+        self._alloc_vec = None               # (r, np.ndarray)
+        self._total_vec = None
 
     # -- basic accessors ----------------------------------------------------
     @property
@@ -153,17 +156,59 @@ class JobInfo:
     def add_task(self, task: TaskInfo) -> None:
         self.tasks[task.key] = task
         self.task_status_index.setdefault(task.status, {})[task.key] = task
+        self._alloc_vec = self._total_vec = None
 
     def remove_task(self, task_key: str) -> Optional[TaskInfo]:
         t = self.tasks.pop(task_key, None)
         if t is not None:
             self.task_status_index.get(t.status, {}).pop(task_key, None)
+            self._alloc_vec = self._total_vec = None
         return t
 
     def update_task_status(self, task: TaskInfo, status: TaskStatus) -> None:
         self.task_status_index.get(task.status, {}).pop(task.key, None)
         task.status = status
         self.task_status_index.setdefault(status, {})[task.key] = task
+        self._alloc_vec = None
+
+    def move_tasks_status(self, tasks: List[TaskInfo],
+                          status: TaskStatus) -> None:
+        """Bulk status move (hot path: a whole class commits at once)."""
+        dst = self.task_status_index.setdefault(status, {})
+        for t in tasks:
+            self.task_status_index.get(t.status, {}).pop(t.key, None)
+            t.status = status
+            dst[t.key] = t
+        self._alloc_vec = None
+
+    # -- dense vector caches (hot: plugin tensor builds) ---------------------
+    def alloc_vec(self, nt):
+        """Allocated-resource vector over nt.dims (cached)."""
+        if self._alloc_vec is not None and self._alloc_vec[0] == nt.r:
+            return self._alloc_vec[1]
+        vec = nt.resource_vector(self.allocated_resource())
+        self._alloc_vec = (nt.r, vec)
+        return vec
+
+    def total_req_vec(self, nt):
+        """Total-request vector over nt.dims (cached)."""
+        if self._total_vec is not None and self._total_vec[0] == nt.r:
+            return self._total_vec[1]
+        vec = nt.resource_vector(self.total_request())
+        self._total_vec = (nt.r, vec)
+        return vec
+
+    def minres_vec(self, nt):
+        """PodGroup minResources vector (immutable per podgroup spec)."""
+        pg = self.podgroup
+        cached = getattr(self, "_minres_vec_c", None)
+        if cached is not None and cached[0] == nt.r:
+            return cached[1]
+        import numpy as np
+        vec = nt.resource_vector(pg.spec.min_resources) if pg is not None \
+            else np.zeros(nt.r, dtype=np.float32)
+        self._minres_vec_c = (nt.r, vec)
+        return vec
 
     def tasks_with_status(self, *statuses: TaskStatus) -> List[TaskInfo]:
         out: List[TaskInfo] = []
@@ -209,17 +254,21 @@ class JobInfo:
         return True
 
     def pending_classes(self) -> List[TaskClass]:
-        """Group pending tasks into batching classes (stable order: by role
-        then signature)."""
+        """Group pending tasks into batching classes.
+
+        Order: priority desc, then role/signature.  Tasks keep dict
+        insertion order inside a class (no per-task sort — hot path)."""
         groups: Dict[str, TaskClass] = {}
-        for t in sorted(self.pending_tasks, key=lambda x: (x.role, x.name)):
+        for t in self.task_status_index.get(TaskStatus.PENDING, {}).values():
             sig = t.class_signature()
             g = groups.get(sig)
             if g is None:
                 g = groups[sig] = TaskClass(signature=sig, role=t.role,
-                                            request=t.request.clone(),
+                                            request=t.request,
                                             priority=t.priority)
             g.tasks.append(t)
+        if len(groups) == 1:
+            return list(groups.values())
         return sorted(groups.values(), key=lambda g: (-g.priority, g.role, g.signature))
 
     def total_request(self) -> Resource:
@@ -276,6 +325,14 @@ class NodeInfo:
             self.releasing.add(task.request)
         elif task.status == TaskStatus.PIPELINED:
             self.pipelined.add(task.request)
+
+    def add_allocated_bulk(self, tasks: List[TaskInfo], request: Resource,
+                           count: int) -> None:
+        """Bulk add for freshly-allocated identical tasks: ONE Resource op
+        for the whole batch (hot path of the allocate commit)."""
+        for t in tasks:
+            self.tasks[t.key] = t
+        self.used.add(request.clone().multi(float(count)))
 
     def remove_task(self, task: TaskInfo) -> None:
         if task.key not in self.tasks:

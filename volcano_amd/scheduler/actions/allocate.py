@@ -18,12 +18,16 @@ same-share queues goes first; across cycles shares converge identically.
 
 from __future__ import annotations
 
+import time
 from typing import List
 
 import numpy as np
 
+from ...api.info import TaskClass
 from ...api.types import PodGroupPhase, TaskStatus
-from ..plan import ClassPlan, CyclePlan, run_plan_hip, run_plan_torch
+from ...utils.metrics import METRICS
+from ..plan import (BundleEntry, ClassPlan, CyclePlan, JobPlan, run_plan_hip,
+                    run_plan_torch)
 
 
 class AllocateAction:
@@ -43,6 +47,7 @@ class AllocateAction:
         nt = ssn.node_tensors
         if nt is None or nt.n == 0:
             return
+        t0 = time.perf_counter()
         plan = CyclePlan(nt, ssn.queue_limit, ssn.queue_alloc)
         plan.dim_w = ssn.dim_weight_vector()
         w = ssn.score_weights
@@ -69,6 +74,26 @@ class AllocateAction:
                 if not ssn.allocatable(q, job):
                     continue
                 ordered_jobs.append((q, job))
+
+        # Gang bundling: a run of consecutive jobs whose single pending
+        # class has identical (queue, signature) is fused into ONE kernel
+        # pass — ntasks summed, per-job gang minimums enforced at apply
+        # (tail-trim).  This is the dominant shape at scale (many identical
+        # gangs) and collapses 2 launches/job into 2 launches/run.
+        open_bundle: ClassPlan = None
+        open_key = None
+
+        def close_bundle():
+            nonlocal open_bundle, open_key
+            if open_bundle is not None:
+                plan.jobs.append(JobPlan(
+                    job_key=open_bundle.job_key,
+                    class_begin=len(plan.classes),
+                    class_end=len(plan.classes) + 1,
+                    occupied=0,
+                    min_available=open_bundle.bundle[0].min_needed))
+                plan.classes.append(open_bundle)
+            open_bundle, open_key = None, None
 
         for q, job in ordered_jobs:
             qi = ssn.queue_index[q.name]
@@ -97,49 +122,159 @@ class AllocateAction:
                     c.tclass.count for c in classes) + job.occupied_count:
                 # gang can never be satisfied this cycle
                 continue
-            plan.add_job(job, classes)
+            if not classes:
+                continue
+
+            bundleable = (len(classes) == 1 and job.occupied_count == 0
+                          and classes[0].tclass.count == len(job.tasks))
+            if bundleable:
+                cp = classes[0]
+                gang_min = max(job.min_available, cp.min_needed)
+                key = (qi, cp.tclass.signature)
+                if open_bundle is not None and key == open_key:
+                    b = open_bundle
+                    b.bundle.append(BundleEntry(job.key, list(cp.tclass.tasks),
+                                                cp.tclass.count, gang_min))
+                    b.tclass.tasks.extend(cp.tclass.tasks)
+                    continue
+                close_bundle()
+                merged = TaskClass(signature=cp.tclass.signature,
+                                   role=cp.tclass.role,
+                                   request=cp.tclass.request,
+                                   tasks=list(cp.tclass.tasks),
+                                   priority=cp.tclass.priority)
+                cp = ClassPlan(
+                    tclass=merged, job_key=job.key, queue_idx=qi, req=cp.req,
+                    tolerated=cp.tolerated, require=cp.require,
+                    forbid=cp.forbid, min_needed=gang_min,
+                    w_least=cp.w_least, w_most=cp.w_most, w_bal=cp.w_bal,
+                    bundle=[BundleEntry(job.key, list(merged.tasks),
+                                        merged.count, gang_min)])
+                open_bundle, open_key = cp, key
+            else:
+                close_bundle()
+                plan.add_job(job, classes)
+        close_bundle()
 
         if plan.n_classes == 0:
             return
         plan.finalize()
+        t1 = time.perf_counter()
         result = self._runner(ssn)(plan)
+        t2 = time.perf_counter()
         self._apply(ssn, plan, result)
+        t3 = time.perf_counter()
+        METRICS.observe("allocate:plan_build", t1 - t0)
+        METRICS.observe("allocate:plan_run", t2 - t1)
+        METRICS.observe("allocate:apply", t3 - t2)
 
     # -- statement commit (host mirror of the device-side state) ------------
     def _apply(self, ssn, plan: CyclePlan, result) -> None:
-        node_by_id = {}
+        nt = ssn.node_tensors
         nodes_sorted = sorted(ssn.nodes.values(), key=lambda n: n.name)
-        for i, ni in enumerate(nodes_sorted):
-            node_by_id[i] = ni
 
         to_bind = []
-        for cp, cres in zip(plan.classes, result.class_results):
+        committed_jobs = set()
+
+        def commit_pieces(job, cp, pieces):
+            """Assign (node_id, count) pieces to the next tasks of `job`;
+            bulk accounting: one Resource op per piece, one index move per
+            batch."""
+            placed_tasks = []
+            for nid, count, tasks in pieces:
+                ni = nodes_sorted[nid]
+                for t in tasks:
+                    t.node_name = ni.name
+                ni.add_allocated_bulk(tasks, cp.tclass.request, count)
+                placed_tasks.extend(tasks)
+            job.move_tasks_status(placed_tasks, TaskStatus.ALLOCATED)
+            to_bind.extend(placed_tasks)
+            committed_jobs.add(job.key)
+
+        for c, (cp, cres) in enumerate(zip(plan.classes, result.class_results)):
             if not result.job_committed.get(cp.job_key, False):
                 continue
             if not cres.placements:
                 continue
-            job = ssn.jobs[cp.job_key]
-            tasks = iter(cp.tclass.tasks)
-            node_ids = [nid for nid, _ in cres.placements]
-            counts = [c for _, c in cres.placements]
-            for nid, count in cres.placements:
-                ni = node_by_id[nid]
-                for _ in range(count):
-                    task = next(tasks)
-                    task.node_name = ni.name
-                    job.update_task_status(task, TaskStatus.ALLOCATED)
-                    ni.add_task(task)
-                    to_bind.append(task)
-            ssn.fire_allocate(cp.tclass, node_ids, counts)
+            if cp.bundle is None:
+                job = ssn.jobs[cp.job_key]
+                tasks = iter(cp.tclass.tasks)
+                pieces = [(nid, cnt, [next(tasks) for _ in range(cnt)])
+                          for nid, cnt in cres.placements]
+                commit_pieces(job, cp, pieces)
+                ssn.fire_allocate(cp.tclass,
+                                  [nid for nid, _ in cres.placements],
+                                  [cnt for _, cnt in cres.placements])
+                continue
+
+            # -- bundle: walk jobs over the placement stream, trim the tail
+            stream = list(cres.placements)
+            ei, eoff = 0, 0
+            used_slots = 0
+            for be in cp.bundle:
+                pieces = []
+                need = be.ntasks
+                toff = 0
+                while need > 0 and ei < len(stream):
+                    nid, cnt = stream[ei]
+                    avail = cnt - eoff
+                    take = min(avail, need)
+                    pieces.append((nid, take, be.tasks[toff:toff + take]))
+                    toff += take
+                    need -= take
+                    eoff += take
+                    if eoff == cnt:
+                        ei += 1
+                        eoff = 0
+                got = be.ntasks - need
+                if got < be.min_needed:
+                    break   # gang unmet: trim this partial + stop the walk
+                job = ssn.jobs[be.job_key]
+                commit_pieces(job, cp, pieces)
+                used_slots += got
+                ssn.fire_allocate(cp.tclass, [p[0] for p in pieces],
+                                  [p[1] for p in pieces])
+            trim = cres.placed - used_slots
+            if trim > 0:
+                self._trim_tail(plan, cp, stream, used_slots, trim)
 
         if to_bind:
             ssn.cache.bind_tasks(to_bind)
 
         # flip gang-ready podgroups to Running (job_updater analog)
-        for jp in plan.jobs:
-            job = ssn.jobs[jp.job_key]
-            if result.job_committed.get(jp.job_key) and ssn.job_ready(job):
-                if job.podgroup is not None and \
+        seen = set()
+        for cp in plan.classes:
+            keys = [be.job_key for be in cp.bundle] if cp.bundle else [cp.job_key]
+            for key in keys:
+                if key in seen or key not in committed_jobs:
+                    continue
+                seen.add(key)
+                job = ssn.jobs[key]
+                if ssn.job_ready(job) and job.podgroup is not None and \
                         job.phase != PodGroupPhase.RUNNING.value:
                     job.podgroup.status.phase = PodGroupPhase.RUNNING.value
                     ssn.cache.update_podgroup(job)
+
+    @staticmethod
+    def _trim_tail(plan: CyclePlan, cp, stream, keep: int, trim: int) -> None:
+        """Undo the device-side staging for the trimmed tail of a bundle:
+        the kernel placed `keep + trim` instances; the last `trim` belong
+        to jobs whose gang minimum could not be met."""
+        import torch
+        nt = plan.nt
+        # walk the stream to find per-node trimmed counts
+        acc = 0
+        nids, cnts = [], []
+        for nid, cnt in stream:
+            lo = max(acc, keep)
+            hi = acc + cnt
+            if hi > lo:
+                nids.append(nid)
+                cnts.append(hi - lo)
+            acc = hi
+        req = torch.from_numpy(cp.req).to(nt.used_t.device)
+        idx = torch.tensor(nids, dtype=torch.long, device=nt.used_t.device)
+        cvec = torch.tensor(cnts, dtype=torch.float32,
+                            device=nt.used_t.device)
+        nt.used_t.index_add_(1, idx, -req.unsqueeze(1) * cvec.unsqueeze(0))
+        plan.queue_alloc[cp.queue_idx] -= float(trim) * torch.from_numpy(cp.req)

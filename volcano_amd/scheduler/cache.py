@@ -46,8 +46,7 @@ class FakeBinder(Binder):
         self.evictions: List[str] = []
 
     def bind(self, tasks: List[TaskInfo]) -> None:
-        for t in tasks:
-            self.binds[t.key] = t.node_name
+        self.binds.update((t.key, t.node_name) for t in tasks)
 
     def evict(self, task: TaskInfo, reason: str = "") -> None:
         self.evictions.append(task.key)
@@ -212,13 +211,15 @@ class SchedulerCache:
     def bind_tasks(self, tasks: List[TaskInfo]) -> None:
         """Async in the reference (cache.go:1343 AddBindTask → 20 ms drain);
         here a batched call — the binder itself may thread if it wants."""
-        for t in tasks:
-            self._task_node[t.key] = t.node_name
+        self._task_node.update((t.key, t.node_name) for t in tasks)
         self.binder.bind(tasks)
+        by_job: Dict[str, List[TaskInfo]] = {}
         for t in tasks:
-            job = self.jobs.get(t.job_key)
+            by_job.setdefault(t.job_key, []).append(t)
+        for key, ts in by_job.items():
+            job = self.jobs.get(key)
             if job is not None:
-                job.update_task_status(t, TaskStatus.BOUND)
+                job.move_tasks_status(ts, TaskStatus.BOUND)
 
     def evict_task(self, task: TaskInfo, reason: str = "") -> None:
         self.binder.evict(task, reason)

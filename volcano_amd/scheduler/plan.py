@@ -31,6 +31,15 @@ BIG_LIMIT = 1.0e18     # "no queue limit" sentinel (finite: kernel does int cast
 
 
 @dataclass
+class BundleEntry:
+    """One gang job inside a fused bundle class (see ClassPlan.bundle)."""
+    job_key: str
+    tasks: list                # the job's pending TaskInfos (class order)
+    ntasks: int
+    min_needed: int            # gang minimum for THIS job
+
+
+@dataclass
 class ClassPlan:
     tclass: TaskClass
     job_key: str
@@ -46,6 +55,12 @@ class ClassPlan:
     use_future: bool = False   # score against future-idle (pipelining)
     log_off: int = 0
     log_cap: int = 0
+    # Gang bundling: consecutive jobs with identical single-class shape are
+    # fused into ONE kernel pass (ntasks = sum); the per-job gang boundaries
+    # live here and are enforced at apply time (tail-trim).  The kernel-side
+    # fuse_min is the FIRST job's minimum: if even that gang can't place,
+    # the whole bundle reverts in-kernel.
+    bundle: Optional[List[BundleEntry]] = None
 
 
 @dataclass

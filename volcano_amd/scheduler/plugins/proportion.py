@@ -9,10 +9,15 @@ kernel enforces ON DEVICE (:359 — in the reference this is a host
 callback per allocation; here the quota is a tensor bound the kernel
 clamps against, so enforcement costs nothing per task), JobEnqueueable
 (:404), Reclaimable (:288).
+
+Host-side checks run on zero-copy numpy views of the queue tensors —
+the per-job vote is a handful of numpy scalar ops (hot: enqueue at 10k+
+jobs/cycle).
 """
 
 from __future__ import annotations
 
+import numpy as np
 import torch
 
 from ...ops import reference as ref
@@ -36,6 +41,7 @@ class ProportionPlugin(Plugin):
         request = torch.zeros((Q, R), dtype=torch.float32)
         guarantee = torch.zeros((Q, R), dtype=torch.float32)
         capability = torch.full((Q, R), BIG_LIMIT, dtype=torch.float32)
+        req_np = request.numpy()
         for i, q in enumerate(queues):
             guarantee[i] = torch.from_numpy(nt.resource_vector(q.guarantee))
             cap_vec = nt.resource_vector(q.capability)
@@ -45,8 +51,7 @@ class ProportionPlugin(Plugin):
         for job in ssn.jobs.values():
             qi = ssn.queue_index.get(job.queue)
             if qi is not None:
-                request[qi] += torch.from_numpy(
-                    nt.resource_vector(job.total_request()))
+                req_np[qi] += job.total_req_vec(nt)
 
         deserved = ref.waterfill(weight, request, guarantee, capability, total)
         ssn.queue_deserved = deserved
@@ -58,40 +63,58 @@ class ProportionPlugin(Plugin):
                             torch.full_like(deserved, BIG_LIMIT))
         ssn.queue_limit = torch.maximum(limit, ssn.queue_alloc)
         self.deserved = deserved
-        self.queues = queues
 
-        tot64 = torch.clamp(total.to(torch.float64), min=1.0)
+        # zero-copy numpy views for the hot host-side checks
+        alloc_np = ssn.queue_alloc.numpy()
+        des_np = deserved.numpy()
+        total_np = total.numpy()
+        mask = total_np > 0
+        tot64 = np.maximum(total_np.astype(np.float64), 1.0)
 
         def q_share(qi: int) -> float:
-            return float((ssn.queue_alloc[qi].to(torch.float64) / tot64).amax())
+            return float((alloc_np[qi] / tot64).max())
 
         def queue_order(a, b) -> int:
-            sa, sb = q_share(ssn.queue_index[a.name]), q_share(ssn.queue_index[b.name])
+            sa = q_share(ssn.queue_index[a.name])
+            sb = q_share(ssn.queue_index[b.name])
             if abs(sa - sb) < 1e-12:
                 return 0
             return -1 if sa < sb else 1
 
         def overused(q) -> bool:
             qi = ssn.queue_index[q.name]
-            alloc = ssn.queue_alloc[qi]
-            des = deserved[qi]
-            mask = total > 0
-            return bool((alloc[mask] > des[mask] + 0.1).any())
+            return bool((alloc_np[qi][mask] > des_np[qi][mask] + 0.1).any())
 
         def allocatable(q, job) -> bool:
             # coarse host-side gate; the exact bound is enforced in-kernel
             return not overused(q)
 
+        # inqueue accounting (proportion.go:404-441: enqueueable checks
+        # allocated + already-admitted-but-unscheduled against deserved)
+        from ...api.types import PodGroupPhase
+        inqueue_np = np.zeros((Q, R), dtype=np.float64)
+        for job in ssn.jobs.values():
+            if job.phase == PodGroupPhase.INQUEUE.value:
+                qi = ssn.queue_index.get(job.queue)
+                if qi is not None:
+                    inqueue_np[qi] += job.minres_vec(nt)
+
         def job_enqueueable(job) -> int:
             qi = ssn.queue_index.get(job.queue)
             if qi is None:
                 return REJECT
-            minres = torch.from_numpy(
-                nt.resource_vector(job.podgroup.spec.min_resources)) \
-                if job.podgroup else torch.zeros(R)
-            mask = total > 0
-            fits = (ssn.queue_alloc[qi] + minres)[mask] <= deserved[qi][mask] + 0.1
-            return PERMIT if bool(fits.all()) else REJECT
+            minres = job.minres_vec(nt)
+            head = (alloc_np[qi][mask].astype(np.float64)
+                    + inqueue_np[qi][mask] + minres[mask])
+            des = des_np[qi][mask].astype(np.float64)
+            # slack: absolute MIN_RESOURCE + f32 accumulation tolerance
+            return PERMIT if bool((head <= des + 0.1 + 1e-6 * des).all()) \
+                else REJECT
+
+        def job_enqueued(job) -> None:
+            qi = ssn.queue_index.get(job.queue)
+            if qi is not None:
+                inqueue_np[qi] += job.minres_vec(nt)
 
         def reclaimable(reclaimer, candidates):
             # a victim whose queue is over its deserved share may be
@@ -112,4 +135,6 @@ class ProportionPlugin(Plugin):
         ssn.overused_fns.append(overused)
         ssn.allocatable_fns.append(allocatable)
         ssn.job_enqueueable_fns.append(job_enqueueable)
+        ssn.job_enqueued_fns = getattr(ssn, "job_enqueued_fns", [])
+        ssn.job_enqueued_fns.append(job_enqueued)
         ssn.reclaimable_fns.append(reclaimable)

@@ -201,17 +201,16 @@ class Session:
     def build_queue_tensors(self) -> None:
         """queue_alloc from currently-allocated tasks; limit defaults open.
         Plugins (proportion/capacity) overwrite limit rows at session open."""
+        import numpy as np
         nt = self.node_tensors
         Q, R = len(self.queues), nt.r
         self.queue_index = {name: i for i, name in enumerate(sorted(self.queues))}
-        alloc = torch.zeros((Q, R), dtype=torch.float32)
+        alloc = np.zeros((Q, R), dtype=np.float32)
         for job in self.jobs.values():
             qi = self.queue_index.get(job.queue)
-            if qi is None:
-                continue
-            vec = nt.resource_vector(job.allocated_resource())
-            alloc[qi] += torch.from_numpy(vec)
-        self.queue_alloc = alloc
+            if qi is not None:
+                alloc[qi] += job.alloc_vec(nt)
+        self.queue_alloc = torch.from_numpy(alloc)
         self.queue_limit = torch.full((Q, R), BIG_LIMIT, dtype=torch.float32)
 
     def dim_weight_vector(self) -> torch.Tensor:
