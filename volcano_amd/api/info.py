@@ -13,6 +13,7 @@ from __future__ import annotations
 import hashlib
 import json
 from dataclasses import dataclass, field
+from functools import cached_property
 from typing import Dict, List, Optional, Tuple
 
 from .objects import (ANN_PREEMPTABLE, LBL_TASK_SPEC, Pod, Node, PodGroup,
@@ -39,7 +40,7 @@ class TaskInfo:
     revocable_zone: str = ""
     pod: Optional[Pod] = None
 
-    @property
+    @cached_property
     def key(self) -> str:
         return f"{self.namespace}/{self.name}"
 
@@ -225,13 +226,13 @@ class JobInfo:
     def occupied_count(self) -> int:
         """Tasks holding or promised resources (reference ReadyTaskNum:
         Bound+Binding+Running+Allocated+Succeeded)."""
-        n = len(self.tasks_with_status(*ALLOCATED_STATUSES))
-        n += len(self.task_status_index.get(TaskStatus.SUCCEEDED, {}))
-        return n
+        idx = self.task_status_index
+        n = sum(len(idx.get(s, ())) for s in ALLOCATED_STATUSES)
+        return n + len(idx.get(TaskStatus.SUCCEEDED, ()))
 
     @property
     def waiting_count(self) -> int:
-        return len(self.task_status_index.get(TaskStatus.PIPELINED, {}))
+        return len(self.task_status_index.get(TaskStatus.PIPELINED, ()))
 
     def is_ready(self) -> bool:
         return self.occupied_count >= self.min_available

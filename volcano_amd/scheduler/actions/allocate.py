@@ -178,17 +178,17 @@ class AllocateAction:
 
         def commit_pieces(job, cp, pieces):
             """Assign (node_id, count) pieces to the next tasks of `job`;
-            bulk accounting: one Resource op per piece, one index move per
-            batch."""
-            placed_tasks = []
+            bulk accounting: one Resource op per piece.  Status moves once,
+            PENDING→BOUND, inside bind_tasks (the reference's
+            Allocated→Binding→Bound pipeline compressed — nothing observes
+            the intermediate states between plan apply and bind here)."""
             for nid, count, tasks in pieces:
                 ni = nodes_sorted[nid]
+                name = ni.name
                 for t in tasks:
-                    t.node_name = ni.name
+                    t.node_name = name
                 ni.add_allocated_bulk(tasks, cp.tclass.request, count)
-                placed_tasks.extend(tasks)
-            job.move_tasks_status(placed_tasks, TaskStatus.ALLOCATED)
-            to_bind.extend(placed_tasks)
+                to_bind.extend(tasks)
             committed_jobs.add(job.key)
 
         for c, (cp, cres) in enumerate(zip(plan.classes, result.class_results)):
