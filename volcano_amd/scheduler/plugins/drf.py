@@ -45,5 +45,6 @@ class DrfPlugin(Plugin):
             return [v for v in candidates
                     if self.share.get(v.job_key, 0.0) > ps]
 
-        ssn.add_job_order_fn(job_order)
+        ssn.add_job_order_fn(
+            job_order, key=lambda j: self.share.get(j.key, 0.0))
         ssn.preemptable_fns.append(preemptable)

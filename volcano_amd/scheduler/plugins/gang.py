@@ -61,7 +61,7 @@ class GangPlugin(Plugin):
             return out
 
         ssn.job_valid_fns.append(job_valid)
-        ssn.add_job_order_fn(job_order)
+        ssn.add_job_order_fn(job_order, key=lambda j: j.is_ready())
         ssn.job_ready_fns.append(job_ready)
         ssn.job_pipelined_fns.append(job_pipelined)
         ssn.job_starving_fns.append(job_starving)

@@ -26,14 +26,18 @@ from .base import Plugin, register
 class PredicatesPlugin(Plugin):
     def on_session_open(self, ssn) -> None:
         ssn.predicates = self
-
         self._nt = ssn.node_tensors
+        self._memo = {}
 
     def class_constraints(self, tclass) -> Tuple[int, np.ndarray, np.ndarray]:
         """(tolerated taint mask, require planes, forbid planes) for a class.
 
         All tasks of a class share constraints by construction
-        (TaskInfo.class_signature)."""
+        (TaskInfo.class_signature); memoized per signature — thousands of
+        identical gangs share one entry."""
+        got = self._memo.get(tclass.signature)
+        if got is not None:
+            return got
         nt = self._nt
         t = tclass.tasks[0]
         pod = t.pod
@@ -41,4 +45,6 @@ class PredicatesPlugin(Plugin):
         tolerated = nt.tolerated_mask(tolerations)
         require, forbid = nt.selector_bits(
             pod.node_selector if pod else {}, pod.affinity if pod else None)
-        return tolerated, require, forbid
+        out = (tolerated, require, forbid)
+        self._memo[tclass.signature] = out
+        return out

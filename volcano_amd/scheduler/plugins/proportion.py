@@ -131,7 +131,8 @@ class ProportionPlugin(Plugin):
                     out.append(v)
             return out
 
-        ssn.add_queue_order_fn(queue_order)
+        ssn.add_queue_order_fn(
+            queue_order, key=lambda q: q_share(ssn.queue_index[q.name]))
         ssn.overused_fns.append(overused)
         ssn.allocatable_fns.append(allocatable)
         ssn.job_enqueueable_fns.append(job_enqueueable)

@@ -47,6 +47,6 @@ class SlaPlugin(Plugin):
         def job_pipelined(job) -> int:
             return PERMIT if overdue(job) else ABSTAIN
 
-        ssn.add_job_order_fn(job_order)
+        ssn.add_job_order_fn(job_order, key=lambda j: not overdue(j))
         ssn.job_enqueueable_fns.append(job_enqueueable)
         ssn.job_pipelined_fns.append(job_pipelined)
