@@ -6,11 +6,17 @@ Registry mirrors ``actions/factory.go``.
 from .allocate import AllocateAction
 from .backfill import BackfillAction
 from .enqueue import EnqueueAction
+from .preempt import PreemptAction
+from .reclaim import ReclaimAction
+from .shuffle import ShuffleAction
 
 ACTION_REGISTRY = {
     "enqueue": EnqueueAction,
     "allocate": AllocateAction,
     "backfill": BackfillAction,
+    "preempt": PreemptAction,
+    "reclaim": ReclaimAction,
+    "shuffle": ShuffleAction,
 }
 
 

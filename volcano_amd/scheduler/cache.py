@@ -192,6 +192,7 @@ class SchedulerCache:
     # -- snapshot -------------------------------------------------------------
     def snapshot_into(self, ssn) -> None:
         self.sync()
+        self._demote_pipelined()
         if "default" not in self.queues:
             self.queues["default"] = QueueInfo(
                 Queue(meta=ObjectMeta(name="default")))
@@ -206,6 +207,26 @@ class SchedulerCache:
         ssn.queues = self.queues
         ssn.node_tensors = self.node_tensors
         ssn.total_resource = self.node_tensors.alloc_t.sum(dim=1).to("cpu")
+
+    def _demote_pipelined(self) -> None:
+        """Pipelined reservations live one cycle: at the next snapshot they
+        re-enter Pending and compete with the capacity their evictions
+        freed.  (The reference carries Pipelined across cycles through
+        PodGroupOldState — session.go:77-79; the one-cycle reservation is
+        the plan-design equivalent: the gang stays pipelined through the
+        session in which its evictions were committed, and converges to
+        Bound the following cycle.)"""
+        for job in self.jobs.values():
+            pipelined = list(job.task_status_index.get(
+                TaskStatus.PIPELINED, {}).values())
+            for t in pipelined:
+                node = self.nodes.get(t.node_name)
+                if node is not None:
+                    node.remove_task(t)
+                t.node_name = ""
+                job.update_task_status(t, TaskStatus.PENDING)
+            if pipelined:
+                self._tensors_dirty = True
 
     # -- commit pipeline ------------------------------------------------------
     def bind_tasks(self, tasks: List[TaskInfo]) -> None:

@@ -190,6 +190,11 @@ class NodeTensors:
                 _set_bit(planes, i, self.label_bit(k, v))
 
         dev = self.device
+        # numpy mirrors for host-side rare paths (preempt/reclaim scans) —
+        # avoids per-element D2H reads when the planes live on the GPU
+        self.ready_np = ready
+        self.taints_np = taints
+        self.planes_np = planes
         # resource planes transposed to [R, N]: one wavefront reads 64
         # consecutive nodes of one dim (coalesced; see module docstring)
         self.alloc_t = torch.from_numpy(alloc.T.copy()).to(dev)
