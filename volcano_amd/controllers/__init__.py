@@ -1,0 +1,7 @@
+from .framework import Controller, ControllerManager
+from .job import JobController
+from .podgroup import PodGroupController
+from .queue import QueueController
+from .jobflow import JobFlowController
+from .cronjob import CronJobController
+from .garbagecollector import GarbageCollector
