@@ -80,11 +80,14 @@ class Watch:
 class ObjectStore:
     """Versioned, watchable object store keyed by (kind, namespace/name)."""
 
-    def __init__(self):
+    def __init__(self, journal_size: int = 200000):
         self._lock = threading.RLock()
         self._objects: Dict[str, Dict[str, object]] = {k: {} for k in KINDS}
         self._rv = 0
         self._watches: List[Watch] = []
+        # event journal for HTTP watch replay: (rv, type, kind, obj_snapshot)
+        from collections import deque
+        self._journal = deque(maxlen=journal_size)
 
     # -- internals -----------------------------------------------------------
     def _bump(self) -> int:
@@ -94,6 +97,14 @@ class ObjectStore:
     def _notify(self, ev: Event) -> None:
         for w in list(self._watches):
             w._push(ev)
+        self._journal.append((ev.resource_version, ev.type.value, ev.kind,
+                              to_dict(ev.obj)))
+
+    def journal_since(self, rv: int, kinds: Optional[Tuple[str, ...]] = None):
+        """Events with resource_version > rv (HTTP watch long-poll)."""
+        with self._lock:
+            return [(v, t, k, o) for (v, t, k, o) in self._journal
+                    if v > rv and (not kinds or k in kinds)]
 
     def _remove_watch(self, w: Watch) -> None:
         with self._lock:

@@ -1,0 +1,171 @@
+"""Admission handlers (reference ``pkg/webhooks/admission/``: jobs
+mutate/validate, queues mutate/validate, podgroups, pods, jobflows,
+cronjobs — SURVEY §2.5).
+"""
+
+from __future__ import annotations
+
+from ..api.objects import DEFAULT_QUEUE
+from ..api.types import Action, Event, QueueState
+from ..utils.cron import CronSchedule
+from .router import AdmissionChain, AdmissionError, AdmissionService
+
+VALID_EVENTS = {e.value for e in Event}
+VALID_ACTIONS = {a.value for a in Action}
+VALID_POLICY_EVENTS = VALID_EVENTS | {"*"}
+
+
+# -- jobs ---------------------------------------------------------------------
+
+def mutate_job(store, job, op) -> None:
+    """Defaulting (reference admission/jobs/mutate): queue, task names,
+    minAvailable."""
+    if not job.spec.queue:
+        job.spec.queue = DEFAULT_QUEUE
+    for i, ts in enumerate(job.spec.tasks):
+        if not ts.name:
+            ts.name = f"task-{i}"
+        if ts.replicas < 0:
+            ts.replicas = 0
+    if job.spec.min_available is None:
+        job.spec.min_available = sum(t.min_needed for t in job.spec.tasks)
+
+
+def validate_job(store, job, op) -> None:
+    """reference admission/jobs/validate (spec legality)."""
+    if not job.meta.name:
+        raise AdmissionError("job name required")
+    if not job.spec.tasks:
+        raise AdmissionError("job must define at least one task")
+    names = [t.name for t in job.spec.tasks]
+    if len(names) != len(set(names)):
+        raise AdmissionError(f"duplicated task names: {names}")
+    total = job.spec.total_replicas
+    if job.spec.min_available is not None and job.spec.min_available > total:
+        raise AdmissionError(
+            f"minAvailable {job.spec.min_available} > total replicas {total}")
+    for ts in job.spec.tasks:
+        if ts.min_available is not None and ts.min_available > ts.replicas:
+            raise AdmissionError(
+                f"task {ts.name}: minAvailable > replicas")
+        for pol in ts.policies:
+            _validate_policy(pol, f"task {ts.name}")
+    for pol in job.spec.policies:
+        _validate_policy(pol, "job")
+    if job.spec.max_retry < 0:
+        raise AdmissionError("maxRetry must be >= 0")
+    if store is not None and job.spec.queue:
+        q = store.get("Queue", "default", job.spec.queue)
+        if q is None:
+            raise AdmissionError(f"queue {job.spec.queue!r} does not exist")
+        if q.status.state != QueueState.OPEN.value:
+            raise AdmissionError(f"queue {job.spec.queue!r} is not open")
+
+
+def _validate_policy(pol, where: str) -> None:
+    evs = pol.events or ([pol.event] if pol.event else [])
+    for e in evs:
+        if e not in VALID_POLICY_EVENTS:
+            raise AdmissionError(f"{where}: invalid policy event {e!r}")
+    if pol.action and pol.action not in VALID_ACTIONS:
+        raise AdmissionError(f"{where}: invalid policy action {pol.action!r}")
+
+
+# -- queues -------------------------------------------------------------------
+
+def mutate_queue(store, queue, op) -> None:
+    if queue.spec.weight <= 0:
+        queue.spec.weight = 1
+
+
+def validate_queue(store, queue, op) -> None:
+    if not queue.meta.name:
+        raise AdmissionError("queue name required")
+    if queue.spec.parent and store is not None:
+        parent = store.get("Queue", "default", queue.spec.parent)
+        if parent is None:
+            raise AdmissionError(
+                f"parent queue {queue.spec.parent!r} does not exist")
+        # hierarchy legality: no cycles up the chain
+        seen = {queue.meta.name}
+        p = parent
+        while p is not None:
+            if p.meta.name in seen:
+                raise AdmissionError("queue hierarchy cycle")
+            seen.add(p.meta.name)
+            p = store.get("Queue", "default", p.spec.parent) \
+                if p.spec.parent else None
+
+
+# -- podgroups ----------------------------------------------------------------
+
+def validate_podgroup(store, pg, op) -> None:
+    if pg.spec.min_member < 0:
+        raise AdmissionError("minMember must be >= 0")
+    for role, n in pg.spec.min_task_member.items():
+        if n < 0:
+            raise AdmissionError(f"minTaskMember[{role}] must be >= 0")
+
+
+# -- pods ---------------------------------------------------------------------
+
+def mutate_pod(store, pod, op) -> None:
+    """reference admission/pods/mutate: annotate the scheduler."""
+    if not pod.scheduler_name:
+        pod.scheduler_name = "volcano"
+
+
+# -- jobflows -----------------------------------------------------------------
+
+def validate_jobflow(store, flow, op) -> None:
+    names = {s.name for s in flow.flows}
+    if len(names) != len(flow.flows):
+        raise AdmissionError("duplicated flow step names")
+    for s in flow.flows:
+        for d in s.depends_on:
+            if d not in names:
+                raise AdmissionError(
+                    f"step {s.name}: unknown dependency {d!r}")
+    # cycle check
+    state = {}
+
+    def visit(n):
+        if state.get(n) == 1:
+            raise AdmissionError("cycle in dependsOn graph")
+        if state.get(n) == 2:
+            return
+        state[n] = 1
+        step = next(s for s in flow.flows if s.name == n)
+        for d in step.depends_on:
+            visit(d)
+        state[n] = 2
+
+    for s in flow.flows:
+        visit(s.name)
+
+
+# -- cronjobs -----------------------------------------------------------------
+
+def validate_cronjob(store, cj, op) -> None:
+    try:
+        CronSchedule(cj.schedule)
+    except Exception as e:
+        raise AdmissionError(f"invalid cron schedule {cj.schedule!r}: {e}")
+    if cj.concurrency_policy not in ("Allow", "Forbid", "Replace"):
+        raise AdmissionError(
+            f"invalid concurrencyPolicy {cj.concurrency_policy!r}")
+
+
+def register_all(chain: AdmissionChain) -> None:
+    """reference webhooks/router/admission.go:32-53 RegisterAdmission."""
+    for svc in [
+        AdmissionService("/jobs/mutate", "Job", mutate_job),
+        AdmissionService("/jobs/validate", "Job", validate_job),
+        AdmissionService("/queues/mutate", "Queue", mutate_queue),
+        AdmissionService("/queues/validate", "Queue", validate_queue),
+        AdmissionService("/podgroups/validate", "PodGroup", validate_podgroup),
+        AdmissionService("/pods/mutate", "Pod", mutate_pod),
+        AdmissionService("/jobflows/validate", "JobFlow", validate_jobflow),
+        AdmissionService("/cronjobs/validate", "CronJob", validate_cronjob),
+    ]:
+        chain.register(svc)
