@@ -259,6 +259,10 @@ class QueueSpec:
     parent: str = ""                                       # hierarchy
     priority: int = 0
     type: str = ""
+    # nodegroup affinity (reference Queue.spec.affinity — plugins/nodegroup):
+    # {"affinity": {"required": [...], "preferred": [...]},
+    #  "antiAffinity": {"required": [...], "preferred": [...]}}
+    affinity: Optional[Dict[str, Any]] = None
 
 
 @dataclass
@@ -324,6 +328,16 @@ class NodeShard:
     nodes_in_use: List[str] = field(default_factory=list)
     nodes_to_add: List[str] = field(default_factory=list)
     nodes_to_remove: List[str] = field(default_factory=list)
+
+
+@dataclass
+class PodDisruptionBudget:
+    """policy/v1 PodDisruptionBudget as the pdb plugin consumes it."""
+
+    meta: ObjectMeta = field(default_factory=ObjectMeta)
+    selector: Dict[str, str] = field(default_factory=dict)   # label match
+    min_available: Optional[int] = None
+    max_unavailable: Optional[int] = None
 
 
 @dataclass
@@ -412,5 +426,5 @@ KINDS = {
     "Node": Node, "Pod": Pod, "Job": Job, "PodGroup": PodGroup,
     "Queue": Queue, "Command": Command, "HyperNode": HyperNode,
     "NodeShard": NodeShard, "JobFlow": JobFlow, "JobTemplate": JobTemplate,
-    "CronJob": CronJob,
+    "CronJob": CronJob, "PodDisruptionBudget": PodDisruptionBudget,
 }

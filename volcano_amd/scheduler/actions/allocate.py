@@ -50,6 +50,7 @@ class AllocateAction:
         t0 = time.perf_counter()
         plan = CyclePlan(nt, ssn.queue_limit, ssn.queue_alloc)
         plan.dim_w = ssn.dim_weight_vector()
+        plan.bias = getattr(ssn, "score_bias", None)
         w = ssn.score_weights
         predicates = getattr(ssn, "predicates", None)
 
@@ -105,7 +106,7 @@ class AllocateAction:
                     skipped = True    # asks for a resource no node offers
                     continue
                 if predicates is not None:
-                    tol, require, forbid = predicates.class_constraints(tc)
+                    tol, require, forbid = predicates.class_constraints(tc, job)
                 else:
                     tol = -1          # tolerate everything
                     W = max(nt.labels.words, 1)

@@ -25,6 +25,7 @@ class BackfillAction:
             return
         plan = CyclePlan(nt, ssn.queue_limit, ssn.queue_alloc)
         plan.dim_w = ssn.dim_weight_vector()
+        plan.bias = getattr(ssn, "score_bias", None)
         predicates = getattr(ssn, "predicates", None)
 
         for job in ssn.jobs.values():
@@ -40,7 +41,7 @@ class BackfillAction:
                     continue
                 req = np.zeros(nt.r, dtype=np.float32)
                 if predicates is not None:
-                    tol, require, forbid = predicates.class_constraints(tc)
+                    tol, require, forbid = predicates.class_constraints(tc, job)
                 else:
                     tol, require, forbid = -1, \
                         np.zeros(max(nt.labels.words, 1), dtype=np.int64), \

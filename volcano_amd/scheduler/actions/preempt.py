@@ -94,7 +94,7 @@ class PreemptAction:
             req_vec = nt.req_vector(tc.tasks[0])
             if req_vec is None:
                 continue
-            constraints = predicates.class_constraints(tc) if predicates \
+            constraints = predicates.class_constraints(tc, job) if predicates \
                 else (-1, np.zeros(max(nt.labels.words, 1), dtype=np.int64),
                       np.zeros(max(nt.labels.words, 1), dtype=np.int64))
             for task in tc.tasks:

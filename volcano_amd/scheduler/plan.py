@@ -109,13 +109,22 @@ class CyclePlan:
             occupied=job.occupied_count, min_available=job.min_available))
 
     def finalize(self) -> None:
-        """Assign undo-log slots (one contiguous region per class)."""
+        """Assign undo-log slots (one contiguous region per class) and
+        normalize constraint widths (the label-bit registry may have grown
+        while classes were built)."""
+        self.nt.ensure_plane_width()
+        W = self.nt.planes_t.shape[0] if self.nt.planes_t is not None \
+            else max(self.nt.labels.words, 1)
         off = 0
         n = self.nt.n
         for cp in self.classes:
             cp.log_off = off
             cp.log_cap = max(1, min(cp.tclass.count, n))
             off += cp.log_cap
+            if len(cp.require) < W:
+                cp.require = np.pad(cp.require, (0, W - len(cp.require)))
+            if len(cp.forbid) < W:
+                cp.forbid = np.pad(cp.forbid, (0, W - len(cp.forbid)))
         self.log_total = off
 
     @property

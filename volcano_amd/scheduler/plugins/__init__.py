@@ -6,8 +6,10 @@ mirrors the reference's ``plugins/factory.go``.
 """
 
 from .base import PLUGIN_REGISTRY, Plugin, register
-from . import (binpack, conformance, drf, gang, nodeorder, overcommit,
-               predicates, priority, proportion, sla)  # noqa: F401 (side-effect registration)
+from . import (binpack, capacity, cdp, conformance, drf, gang, nodegroup,
+               nodeorder, overcommit, pdb, predicates, priority, proportion,
+               rescheduling, resource_strategy_fit, sla, tdm,
+               usage)  # noqa: F401 (side-effect registration)
 
 
 def new_plugin(name: str, args=None) -> Plugin:

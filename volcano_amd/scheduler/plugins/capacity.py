@@ -1,0 +1,198 @@
+"""Capacity plugin (reference ``plugins/capacity/capacity.go``, design
+``docs/design/capacity-scheduling.md``) — the modern replacement for
+proportion: explicit per-queue ``deserved`` from the Queue spec, elastic
+capacity ``guarantee → deserved → capability``, and **hierarchical
+queues** (children split their parent's deserved share by weight;
+ancestor capability caps the subtree — buildHierarchicalQueueAttrs
+capacity.go:1212, ancestor checks :1452-1560).
+
+Tensorized: the hierarchy is resolved level by level with the same
+float64 water-filling kernel as proportion (ops.reference.waterfill) —
+one pass per tree level instead of per-queue Go recursion.
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List
+
+import numpy as np
+import torch
+
+from ...api.types import PodGroupPhase
+from ...ops import reference as ref
+from ..plan import BIG_LIMIT
+from ..session import ABSTAIN, PERMIT, REJECT
+from .base import Plugin, register
+
+
+@register("capacity")
+class CapacityPlugin(Plugin):
+    def on_session_open(self, ssn) -> None:
+        nt = ssn.node_tensors
+        queues = [ssn.queues[name] for name in sorted(ssn.queues)]
+        Q, R = len(queues), nt.r
+        if Q == 0:
+            return
+        total = ssn.total_resource if ssn.total_resource is not None else \
+            torch.zeros(R)
+        by_name = {q.name: q for q in queues}
+        qi_of = ssn.queue_index
+
+        # request per queue (demand)
+        request = torch.zeros((Q, R), dtype=torch.float32)
+        for job in ssn.jobs.values():
+            qi = qi_of.get(job.queue)
+            if qi is not None:
+                request[qi] += torch.from_numpy(job.total_req_vec(nt))
+
+        # per-queue vectors
+        guarantee = torch.zeros((Q, R), dtype=torch.float32)
+        capability = torch.full((Q, R), BIG_LIMIT, dtype=torch.float32)
+        deserved_spec = torch.zeros((Q, R), dtype=torch.float32)
+        has_deserved = [False] * Q
+        for q in queues:
+            i = qi_of[q.name]
+            guarantee[i] = torch.from_numpy(nt.resource_vector(q.guarantee))
+            cv = nt.resource_vector(q.capability)
+            for r in range(R):
+                if cv[r] > 0:
+                    capability[i, r] = cv[r]
+            dv = nt.resource_vector(q.deserved_spec)
+            if dv.any():
+                deserved_spec[i] = torch.from_numpy(dv)
+                has_deserved[i] = True
+
+        # hierarchy: children of each parent ("" = root level)
+        children: Dict[str, List[str]] = {}
+        for q in queues:
+            parent = q.parent if q.parent in by_name else ""
+            children.setdefault(parent, []).append(q.name)
+
+        # aggregate subtree demand bottom-up so parents see child demand
+        def subtree_request(name: str) -> torch.Tensor:
+            req = request[qi_of[name]].clone()
+            for c in children.get(name, []):
+                req += subtree_request(c)
+            return req
+
+        deserved = torch.zeros((Q, R), dtype=torch.float32)
+
+        # level-order: split each parent's pool among its children.
+        # Two passes: explicit spec.deserved dims are clamped first; the
+        # remaining pool water-fills by weight among the rest (so slack
+        # from a low explicit deserved flows to siblings).
+        def fill_level(parent: str, pool: torch.Tensor) -> None:
+            names = children.get(parent, [])
+            if not names:
+                return
+            idx = [qi_of[n] for n in names]
+            cap = {n: torch.minimum(capability[qi_of[n]], pool)
+                   for n in names}
+            fixed = torch.zeros_like(pool)
+            for n in names:
+                i = qi_of[n]
+                if has_deserved[i]:
+                    spec = torch.minimum(deserved_spec[i], cap[n])
+                    spec = torch.maximum(spec, guarantee[i])
+                    fixed = fixed + torch.where(deserved_spec[i] > 0, spec,
+                                                torch.zeros_like(spec))
+            rest_pool = torch.clamp(pool - fixed, min=0.0)
+            w = torch.tensor([by_name[n].weight for n in names],
+                             dtype=torch.float32)
+            req = torch.stack([subtree_request(n) for n in names])
+            # explicit-deserved dims don't compete in the water-fill
+            for k, n in enumerate(names):
+                i = qi_of[n]
+                if has_deserved[i]:
+                    req[k] = torch.where(deserved_spec[i] > 0,
+                                         torch.zeros_like(req[k]), req[k])
+            gua = guarantee[idx]
+            cap_m = torch.stack([cap[n] for n in names])
+            des = ref.waterfill(w, req, gua, cap_m, rest_pool)
+            for k, n in enumerate(names):
+                i = qi_of[n]
+                if has_deserved[i]:
+                    spec_k = torch.minimum(deserved_spec[i], cap[n])
+                    spec_k = torch.maximum(spec_k, guarantee[i])
+                    deserved[i] = torch.where(deserved_spec[i] > 0, spec_k,
+                                              des[k])
+                else:
+                    deserved[i] = des[k]
+                fill_level(n, deserved[i])
+
+        fill_level("", total.clone())
+        ssn.queue_deserved = deserved
+        limit = torch.where(total.unsqueeze(0) > 0, deserved,
+                            torch.full_like(deserved, BIG_LIMIT))
+        ssn.queue_limit = torch.maximum(limit, ssn.queue_alloc)
+        self.deserved = deserved
+
+        alloc_np = ssn.queue_alloc.numpy()
+        des_np = deserved.numpy()
+        total_np = total.numpy()
+        mask = total_np > 0
+        tot64 = np.maximum(total_np.astype(np.float64), 1.0)
+
+        def q_share(qi: int) -> float:
+            return float((alloc_np[qi] / tot64).max())
+
+        def queue_order(a, b) -> int:
+            # priority first (capacity honors queue priority), then share
+            if a.priority != b.priority:
+                return -1 if a.priority > b.priority else 1
+            sa, sb = q_share(qi_of[a.name]), q_share(qi_of[b.name])
+            if abs(sa - sb) < 1e-12:
+                return 0
+            return -1 if sa < sb else 1
+
+        def overused(q) -> bool:
+            qi = qi_of[q.name]
+            return bool((alloc_np[qi][mask] > des_np[qi][mask] + 0.1).any())
+
+        def allocatable(q, job) -> bool:
+            return not overused(q)
+
+        inqueue_np = np.zeros((Q, R), dtype=np.float64)
+        for job in ssn.jobs.values():
+            if job.phase == PodGroupPhase.INQUEUE.value:
+                qi = qi_of.get(job.queue)
+                if qi is not None:
+                    inqueue_np[qi] += job.minres_vec(nt)
+
+        def job_enqueueable(job) -> int:
+            qi = qi_of.get(job.queue)
+            if qi is None:
+                return REJECT
+            head = (alloc_np[qi][mask].astype(np.float64)
+                    + inqueue_np[qi][mask] + job.minres_vec(nt)[mask])
+            des = des_np[qi][mask].astype(np.float64)
+            return PERMIT if bool((head <= des + 0.1 + 1e-6 * des).all()) \
+                else REJECT
+
+        def job_enqueued(job) -> None:
+            qi = qi_of.get(job.queue)
+            if qi is not None:
+                inqueue_np[qi] += job.minres_vec(nt)
+
+        def reclaimable(reclaimer, candidates):
+            out = []
+            for v in candidates:
+                job = ssn.jobs.get(v.job_key)
+                if job is None:
+                    continue
+                q = ssn.queues.get(job.queue)
+                if q is None or not q.reclaimable:
+                    continue
+                if overused(q):
+                    out.append(v)
+            return out
+
+        ssn.add_queue_order_fn(
+            queue_order,
+            key=lambda q: (-q.priority, q_share(qi_of[q.name])))
+        ssn.overused_fns.append(overused)
+        ssn.allocatable_fns.append(allocatable)
+        ssn.job_enqueueable_fns.append(job_enqueueable)
+        ssn.job_enqueued_fns = getattr(ssn, "job_enqueued_fns", [])
+        ssn.job_enqueued_fns.append(job_enqueued)
+        ssn.reclaimable_fns.append(reclaimable)

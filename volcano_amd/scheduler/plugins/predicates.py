@@ -26,16 +26,22 @@ from .base import Plugin, register
 class PredicatesPlugin(Plugin):
     def on_session_open(self, ssn) -> None:
         ssn.predicates = self
+        self._ssn = ssn
         self._nt = ssn.node_tensors
         self._memo = {}
 
-    def class_constraints(self, tclass) -> Tuple[int, np.ndarray, np.ndarray]:
+    def class_constraints(self, tclass,
+                          job=None) -> Tuple[int, np.ndarray, np.ndarray]:
         """(tolerated taint mask, require planes, forbid planes) for a class.
 
         All tasks of a class share constraints by construction
-        (TaskInfo.class_signature); memoized per signature — thousands of
-        identical gangs share one entry."""
-        got = self._memo.get(tclass.signature)
+        (TaskInfo.class_signature); memoized per (signature, queue) —
+        thousands of identical gangs share one entry.  Other plugins
+        (nodegroup/tdm/usage) extend the bits through
+        ``ssn.class_constraint_hooks``."""
+        hooks = getattr(self._ssn, "class_constraint_hooks", [])
+        key = (tclass.signature, job.queue if (job and hooks) else None)
+        got = self._memo.get(key)
         if got is not None:
             return got
         nt = self._nt
@@ -45,6 +51,8 @@ class PredicatesPlugin(Plugin):
         tolerated = nt.tolerated_mask(tolerations)
         require, forbid = nt.selector_bits(
             pod.node_selector if pod else {}, pod.affinity if pod else None)
+        for hook in hooks:
+            hook(tclass, job, require, forbid)
         out = (tolerated, require, forbid)
-        self._memo[tclass.signature] = out
+        self._memo[key] = out
         return out

@@ -69,7 +69,12 @@ class Session:
             Callable[[TaskInfo, List[TaskInfo]], List[TaskInfo]]] = []
         self.victim_tasks_fns: List[
             Callable[[List[TaskInfo]], List[TaskInfo]]] = []
+        # filters applied over the nominated victim union (pdb/conformance)
+        self.victim_filter_fns: List[
+            Callable[[List[TaskInfo]], List[TaskInfo]]] = []
         self.event_handlers: List[object] = []   # objects w/ allocate/evict hooks
+        # fns(tclass, job, require, forbid) mutating per-class plane bits
+        self.class_constraint_hooks: List[Callable] = []
 
         # tensor-plane configuration contributed by plugins
         self.score_weights = {"least": 1.0, "most": 0.0, "bal": 0.0}
@@ -195,6 +200,8 @@ class Session:
                 if t.uid not in seen:
                     seen.add(t.uid)
                     out.append(t)
+        for filt in self.victim_filter_fns:
+            out = filt(out)
         return out
 
     # -- sorted views ---------------------------------------------------------

@@ -65,6 +65,11 @@ def _set_bit(arr: np.ndarray, col: int, bit: int) -> None:
     arr[bit // 64, col] |= np.int64(_to_signed64(1 << (bit % 64)))
 
 
+def set_plane_bit(words: np.ndarray, bit: int) -> None:
+    """Set one bit in a [W] require/forbid word vector (int64-safe)."""
+    words[bit // 64] |= np.int64(_to_signed64(1 << (bit % 64)))
+
+
 class NodeTensors:
     """Packed per-node planes; rebuilt (v1) or patched per cycle."""
 
@@ -215,6 +220,58 @@ class NodeTensors:
                     return None
                 continue
             out[i] = v
+        return out
+
+    def add_dynamic_bit(self, name: str, node_ids) -> int:
+        """Register (or reuse) a synthetic label bit and set it for the
+        given node ids — plugins project arbitrary node sets (nodegroup
+        membership, revocable zones, over-utilized nodes) into the SAME
+        bit planes the predicate kernel already tests, so a new node-set
+        predicate costs zero extra kernel work.  Returns the bit index."""
+        bit = self.labels.bit(f"dyn:{name}")
+        w, b = bit // 64, bit % 64
+        val = np.int64(_to_signed64(1 << b))
+        if self.planes_t is None:
+            return bit
+        if w >= self.planes_np.shape[0]:
+            pad_np = np.zeros((w + 1 - self.planes_np.shape[0], self.n),
+                              dtype=np.int64)
+            self.planes_np = np.concatenate([self.planes_np, pad_np])
+            pad_t = torch.zeros(
+                (w + 1 - self.planes_t.shape[0], self.planes_t.shape[1]),
+                dtype=torch.int64, device=self.planes_t.device)
+            self.planes_t = torch.cat([self.planes_t, pad_t])
+        ids = np.asarray(list(node_ids), dtype=np.int64)
+        if len(ids):
+            self.planes_np[w, ids] |= val
+            idx = torch.from_numpy(ids).to(self.planes_t.device)
+            row = self.planes_t[w]
+            row[idx] |= torch.tensor(int(val), dtype=torch.int64,
+                                     device=row.device)
+        return bit
+
+    def ensure_plane_width(self) -> None:
+        """Pad the packed planes to the registry's word count (a class
+        selector can register bits for labels no node carries — those
+        planes are all-zero and the class is simply infeasible there)."""
+        if self.planes_t is None:
+            return
+        W = max(self.labels.words, 1)
+        have = self.planes_t.shape[0]
+        if have >= W:
+            return
+        pad_np = np.zeros((W - have, self.n), dtype=np.int64)
+        self.planes_np = np.concatenate([self.planes_np, pad_np])
+        pad_t = torch.zeros((W - have, self.planes_t.shape[1]),
+                            dtype=torch.int64, device=self.planes_t.device)
+        self.planes_t = torch.cat([self.planes_t, pad_t])
+
+    def bit_words(self, bits) -> np.ndarray:
+        """Pack bit indices into a [W] require/forbid word vector."""
+        W = max(self.labels.words, 1)
+        out = np.zeros(W, dtype=np.int64)
+        for bit in bits:
+            out[bit // 64] |= np.int64(_to_signed64(1 << (bit % 64)))
         return out
 
     def resource_vector(self, resource) -> np.ndarray:

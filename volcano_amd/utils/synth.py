@@ -12,7 +12,8 @@ from __future__ import annotations
 import random
 from typing import Dict, List, Optional
 
-from ..api.objects import (ANN_PODGROUP, ANN_QUEUE, LBL_TASK_SPEC, Node,
+from ..api.objects import (ANN_PODGROUP, ANN_QUEUE, LBL_JOB_NAME,
+                           LBL_TASK_SPEC, Node,
                            ObjectMeta, Pod, PodGroup, PodGroupSpec, Queue,
                            QueueSpec, Taint, Toleration)
 from ..api.resource import CPU, MEMORY, PODS, Resource
@@ -74,6 +75,8 @@ def make_pod(name: str, podgroup: str, queue: str = "default",
     if extra:
         req.update(extra)
     labels = {LBL_TASK_SPEC: role} if role else {}
+    if podgroup:
+        labels[LBL_JOB_NAME] = podgroup
     return Pod(
         meta=ObjectMeta(name=name, namespace=namespace, labels=labels,
                         annotations={ANN_PODGROUP: podgroup, ANN_QUEUE: queue}),
