@@ -5,3 +5,4 @@ from .queue import QueueController
 from .jobflow import JobFlowController
 from .cronjob import CronJobController
 from .garbagecollector import GarbageCollector
+from .hypernode import HyperNodeController

@@ -74,7 +74,7 @@ class NodeTensors:
     """Packed per-node planes; rebuilt (v1) or patched per cycle."""
 
     def __init__(self, dims: ResourceDims, device: str = "cpu",
-                 label_words: int = 4):
+                 label_words: int = 16):
         self.dims = dims
         self.device = device
         self.labels = BitRegistry(label_words)
