@@ -1,0 +1,148 @@
+"""Deviceshare (GPU pools), numaaware, extender plugins."""
+
+import json
+import threading
+
+from volcano_amd.api.devices import (ANN_ASSIGNED, DeviceRequest,
+                                     GPUDevicePool)
+from volcano_amd.api.objects import Numatopology, NumaZone, ObjectMeta
+from volcano_amd.scheduler import (FakeBinder, Scheduler, SchedulerCache,
+                                   default_config)
+from volcano_amd.scheduler.config import PluginOption
+from volcano_amd.store import ObjectStore
+from volcano_amd.utils import synth
+
+GI = 1024 ** 3
+
+
+def mk(*plugins):
+    store = ObjectStore()
+    binder = FakeBinder()
+    cache = SchedulerCache(store=store, binder=binder)
+    config = default_config()
+    for name, args in plugins:
+        config.tiers[1].plugins.append(PluginOption(name, arguments=args))
+    sched = Scheduler(cache, config)
+    return store, binder, cache, sched
+
+
+def test_device_pool_packing():
+    node = synth.make_node("g", extra={})
+    node.meta.annotations["volcano.sh/gpu-count"] = "2"
+    node.meta.annotations["volcano.sh/gpu-memory-per-card"] = "1000"
+    pool = GPUDevicePool(node)
+    # two 600-MiB slices cannot share one 1000-MiB card
+    req = DeviceRequest(count=2, memory=600)
+    picked = pool.allocate(req)
+    assert picked is not None and len(set(picked)) == 2
+    # a third 600 slice no longer fits anywhere
+    assert pool.fit(DeviceRequest(count=1, memory=600)) is None
+    # but a 400 slice does
+    assert pool.fit(DeviceRequest(count=1, memory=400)) is not None
+    pool.release(req, picked)
+    assert pool.fit(DeviceRequest(count=1, memory=600)) is not None
+
+
+def test_deviceshare_scheduling():
+    store, binder, cache, sched = mk(("deviceshare", {}))
+    plain = synth.make_node("plain", cpu_milli=8000, mem=32 * GI)
+    gpu = synth.make_node("gpu", cpu_milli=8000, mem=32 * GI)
+    gpu.meta.annotations["volcano.sh/gpu-count"] = "1"
+    gpu.meta.annotations["volcano.sh/gpu-memory-per-card"] = "1000"
+    store.create("Node", plain)
+    store.create("Node", gpu)
+    store.create("Queue", synth.make_queue("default"))
+
+    pg = synth.make_podgroup("dj", min_member=1)
+    store.create("PodGroup", pg)
+    pod = synth.make_pod("dj-w-0", "dj", cpu_milli=500, mem=GI)
+    pod.meta.annotations["volcano.sh/gpu-number"] = "1"
+    pod.meta.annotations["volcano.sh/gpu-memory"] = "700"
+    store.create("Pod", pod)
+    sched.run_once()
+    assert binder.binds["default/dj-w-0"] == "gpu"
+    # card index was assigned
+    task = cache.jobs["default/dj"].tasks["default/dj-w-0"]
+    assert task.pod.meta.annotations[ANN_ASSIGNED] == "0"
+
+    # second 700-MiB slice can't fit the same card → unschedulable
+    pg2 = synth.make_podgroup("dj2", min_member=1)
+    store.create("PodGroup", pg2)
+    pod2 = synth.make_pod("dj2-w-0", "dj2", cpu_milli=500, mem=GI)
+    pod2.meta.annotations["volcano.sh/gpu-number"] = "1"
+    pod2.meta.annotations["volcano.sh/gpu-memory"] = "700"
+    store.create("Pod", pod2)
+    sched.run_once()
+    assert "default/dj2-w-0" not in binder.binds
+
+
+def test_numa_single_numa_node():
+    store, binder, cache, sched = mk(("numaaware", {}))
+    # node with 2 NUMA zones of 4 cpu each (8 total)
+    n = synth.make_node("numa", cpu_milli=8000, mem=32 * GI)
+    store.create("Node", n)
+    store.create("Numatopology", Numatopology(
+        meta=ObjectMeta(name="numa"),
+        zones=[NumaZone(0, 4000, 16 * GI), NumaZone(1, 4000, 16 * GI)]))
+    store.create("Queue", synth.make_queue("default"))
+
+    # 6-cpu single-numa pod cannot fit any one zone despite 8 free
+    pg = synth.make_podgroup("nj", min_member=1)
+    store.create("PodGroup", pg)
+    pod = synth.make_pod("nj-w-0", "nj", cpu_milli=6000, mem=GI)
+    pod.meta.annotations["volcano.sh/numa-topology-policy"] = "single-numa-node"
+    store.create("Pod", pod)
+    sched.run_once()
+    assert "default/nj-w-0" not in binder.binds
+
+    # a 3-cpu one fits and gets pinned
+    pg2 = synth.make_podgroup("nj2", min_member=1)
+    store.create("PodGroup", pg2)
+    pod2 = synth.make_pod("nj2-w-0", "nj2", cpu_milli=3000, mem=GI)
+    pod2.meta.annotations["volcano.sh/numa-topology-policy"] = "single-numa-node"
+    store.create("Pod", pod2)
+    sched.run_once()
+    assert binder.binds["default/nj2-w-0"] == "numa"
+    task = cache.jobs["default/nj2"].tasks["default/nj2-w-0"]
+    assert task.pod.meta.annotations["volcano.sh/numa-node"] in ("0", "1")
+
+
+def test_extender_predicate_http():
+    # tiny sidecar: only node "allowed" passes the predicate
+    from http.server import BaseHTTPRequestHandler, HTTPServer
+
+    class Handler(BaseHTTPRequestHandler):
+        def do_POST(self):
+            n = int(self.headers.get("Content-Length", 0))
+            self.rfile.read(n)
+            body = b"{}"
+            if self.path.endswith("/predicate"):
+                body = json.dumps({"nodes": ["allowed"]}).encode()
+            self.send_response(200)
+            self.send_header("Content-Type", "application/json")
+            self.send_header("Content-Length", str(len(body)))
+            self.end_headers()
+            self.wfile.write(body)
+
+        def log_message(self, *a):
+            pass
+
+    srv = HTTPServer(("127.0.0.1", 0), Handler)
+    port = srv.server_address[1]
+    t = threading.Thread(target=srv.serve_forever, daemon=True)
+    t.start()
+    try:
+        store, binder, cache, sched = mk(("extender", {
+            "extender.urlPrefix": f"http://127.0.0.1:{port}",
+            "extender.predicateVerb": "predicate"}))
+        store.create("Node", synth.make_node("allowed", cpu_milli=8000,
+                                             mem=32 * GI))
+        store.create("Node", synth.make_node("denied", cpu_milli=8000,
+                                             mem=32 * GI))
+        store.create("Queue", synth.make_queue("default"))
+        synth.make_gang(store, "ej", replicas=2, cpu_milli=1000, mem=GI)
+        sched.run_once()
+        assert len(binder.binds) == 2
+        assert set(binder.binds.values()) == {"allowed"}
+    finally:
+        srv.shutdown()

@@ -331,6 +331,23 @@ class NodeShard:
 
 
 @dataclass
+class NumaZone:
+    id: int = 0
+    cpu_milli: float = 0.0       # allocatable millicores in this NUMA node
+    memory: float = 0.0
+
+
+@dataclass
+class Numatopology:
+    """nodeinfo/v1alpha1 Numatopology — per-node CPU/NUMA layout written
+    by the node agent, consumed by the numaaware plugin."""
+
+    meta: ObjectMeta = field(default_factory=ObjectMeta)   # name == node name
+    zones: List[NumaZone] = field(default_factory=list)
+    policies: Dict[str, str] = field(default_factory=dict)  # e.g. topologyManager
+
+
+@dataclass
 class PodDisruptionBudget:
     """policy/v1 PodDisruptionBudget as the pdb plugin consumes it."""
 
@@ -427,4 +444,5 @@ KINDS = {
     "Queue": Queue, "Command": Command, "HyperNode": HyperNode,
     "NodeShard": NodeShard, "JobFlow": JobFlow, "JobTemplate": JobTemplate,
     "CronJob": CronJob, "PodDisruptionBudget": PodDisruptionBudget,
+    "Numatopology": Numatopology,
 }

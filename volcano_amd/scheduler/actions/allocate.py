@@ -205,7 +205,8 @@ class AllocateAction:
                 commit_pieces(job, cp, pieces)
                 ssn.fire_allocate(cp.tclass,
                                   [nid for nid, _ in cres.placements],
-                                  [cnt for _, cnt in cres.placements])
+                                  [cnt for _, cnt in cres.placements],
+                                  [t for _, _, ts in pieces for t in ts])
                 continue
 
             # -- bundle: walk jobs over the placement stream, trim the tail
@@ -234,7 +235,8 @@ class AllocateAction:
                 commit_pieces(job, cp, pieces)
                 used_slots += got
                 ssn.fire_allocate(cp.tclass, [p[0] for p in pieces],
-                                  [p[1] for p in pieces])
+                                  [p[1] for p in pieces],
+                                  [t for _, _, ts in pieces for t in ts])
             trim = cres.placed - used_slots
             if trim > 0:
                 self._trim_tail(plan, cp, stream, used_slots, trim)

@@ -137,6 +137,8 @@ class SchedulerCache:
             t = self.jobs[old_job].remove_task(tkey)
             if t is not None and old_node and old_node in self.nodes:
                 self.nodes[old_node].remove_task(t)
+            if ev.type == EventType.DELETED and t is not None:
+                self._release_devices(t, old_node)
 
         if ev.type == EventType.DELETED:
             return
@@ -146,6 +148,19 @@ class SchedulerCache:
         if task.node_name and task.node_name in self.nodes:
             self.nodes[task.node_name].add_task(task)
             self._task_node[tkey] = task.node_name
+
+    def _release_devices(self, task: TaskInfo, node_name: Optional[str]) -> None:
+        """Return GPU slices to the node pool when a device pod goes away
+        (deviceshare plugin owns allocation; see api/devices.py)."""
+        pools = getattr(self, "_device_pools", None)
+        if not pools or task.pod is None or not node_name:
+            return
+        from ..api.devices import ANN_ASSIGNED, DeviceRequest
+        assigned = task.pod.meta.annotations.get(ANN_ASSIGNED)
+        req = DeviceRequest.from_annotations(task.pod.meta.annotations)
+        pool = pools.get(node_name)
+        if assigned and req is not None and pool is not None:
+            pool.release(req, [int(x) for x in assigned.split(",") if x])
 
     def _on_node(self, ev) -> None:
         name = ev.obj.meta.name

@@ -101,7 +101,7 @@ class NetworkTopologyAwarePlugin(Plugin):
 
         domain_sets = {h: set(ids) for h, ids in self.domain_ids.items()}
 
-        def on_allocate(tclass, node_ids, counts):
+        def on_allocate(tclass, node_ids, counts, tasks=None):
             """Keep per-domain free capacity current within the cycle."""
             if not tclass.tasks:
                 return

@@ -249,11 +249,12 @@ class Session:
         return w
 
     # -- events ---------------------------------------------------------------
-    def fire_allocate(self, task_class, node_ids, counts) -> None:
+    def fire_allocate(self, task_class, node_ids, counts,
+                      tasks: Optional[List[TaskInfo]] = None) -> None:
         for h in self.event_handlers:
             fn = getattr(h, "on_allocate", None)
             if fn:
-                fn(task_class, node_ids, counts)
+                fn(task_class, node_ids, counts, tasks)
 
     def fire_evict(self, task: TaskInfo) -> None:
         for h in self.event_handlers:
