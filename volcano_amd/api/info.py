@@ -299,6 +299,19 @@ class NodeInfo:
         self.name = node.meta.name
         self.node_id = -1                    # dense index
         self.allocatable = node.allocatable.clone()
+        # node-agent oversubscription report (reference node_info.go:83-89:
+        # oversold capacity from real utilization extends allocatable)
+        self.oversubscription = node.oversubscription.clone()
+        ann = node.meta.annotations
+        if "volcano.sh/oversubscription-cpu" in ann:
+            from .resource import CPU
+            self.oversubscription.q[CPU] = float(
+                ann["volcano.sh/oversubscription-cpu"])
+        if "volcano.sh/oversubscription-memory" in ann:
+            from .resource import MEMORY
+            self.oversubscription.q[MEMORY] = float(
+                ann["volcano.sh/oversubscription-memory"])
+        self.allocatable.add(self.oversubscription)
         self.used = Resource()
         self.releasing = Resource()
         self.pipelined = Resource()

@@ -331,6 +331,22 @@ class NodeShard:
 
 
 @dataclass
+class ColocationConfig:
+    """config/v1alpha1 ColocationConfig — per-nodepool colocation/QoS
+    settings (cpu burst, memory qos, oversubscription ratio, network
+    bandwidth watermarks) keyed by a node label selector."""
+
+    meta: ObjectMeta = field(default_factory=ObjectMeta)
+    node_selector: Dict[str, str] = field(default_factory=dict)
+    cpu_burst_enable: bool = False
+    memory_qos_enable: bool = False
+    oversubscription_enable: bool = False
+    oversubscription_ratio: float = 0.6
+    network_qos_enable: bool = False
+    offline_bandwidth_share: float = 0.3
+
+
+@dataclass
 class NumaZone:
     id: int = 0
     cpu_milli: float = 0.0       # allocatable millicores in this NUMA node
@@ -444,5 +460,5 @@ KINDS = {
     "Queue": Queue, "Command": Command, "HyperNode": HyperNode,
     "NodeShard": NodeShard, "JobFlow": JobFlow, "JobTemplate": JobTemplate,
     "CronJob": CronJob, "PodDisruptionBudget": PodDisruptionBudget,
-    "Numatopology": Numatopology,
+    "Numatopology": Numatopology, "ColocationConfig": ColocationConfig,
 }

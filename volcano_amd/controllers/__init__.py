@@ -6,3 +6,5 @@ from .jobflow import JobFlowController
 from .cronjob import CronJobController
 from .garbagecollector import GarbageCollector
 from .hypernode import HyperNodeController
+from .sharding import ShardingController
+from .colocationconfig import ColocationConfigController

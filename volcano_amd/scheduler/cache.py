@@ -211,12 +211,7 @@ class SchedulerCache:
         if "default" not in self.queues:
             self.queues["default"] = QueueInfo(
                 Queue(meta=ObjectMeta(name="default")))
-        nodes = sorted(self.nodes.values(), key=lambda n: n.name)
-        if self._tensors_dirty or self.node_tensors.alloc_t is None:
-            self.node_tensors.pack(nodes)
-            self._tensors_dirty = False
-        for i, ni in enumerate(nodes):
-            ni.node_id = i
+        self.ensure_packed()
         ssn.jobs = self.jobs
         ssn.nodes = self.nodes
         ssn.queues = self.queues
@@ -242,6 +237,14 @@ class SchedulerCache:
                 job.update_task_status(t, TaskStatus.PENDING)
             if pipelined:
                 self._tensors_dirty = True
+
+    def ensure_packed(self) -> None:
+        nodes = sorted(self.nodes.values(), key=lambda n: n.name)
+        if self._tensors_dirty or self.node_tensors.alloc_t is None:
+            self.node_tensors.pack(nodes)
+            self._tensors_dirty = False
+        for i, ni in enumerate(nodes):
+            ni.node_id = i
 
     # -- commit pipeline ------------------------------------------------------
     def bind_tasks(self, tasks: List[TaskInfo]) -> None:

@@ -1,0 +1,115 @@
+"""Scheduler daemon entrypoint (the ``cmd/scheduler`` analog).
+
+Run: ``python -m volcano_amd.scheduler.daemon --state /path/state.json``
+or with ``--serve-api`` to also expose the apiserver for vcctl.
+
+Signals (reference cache/dumper.go + SIGUSR1/2): SIGUSR1 dumps the
+scheduler cache as JSON to --dump-dir.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import signal
+import sys
+import time
+
+
+def dump_cache(cache, path: str) -> None:
+    """Cache dumper (reference pkg/scheduler/cache/dumper.go:127-146)."""
+    state = {
+        "nodes": {
+            name: {
+                "allocatable": dict(ni.allocatable.q),
+                "used": dict(ni.used.q),
+                "releasing": dict(ni.releasing.q),
+                "tasks": sorted(ni.tasks),
+            } for name, ni in cache.nodes.items()
+        },
+        "jobs": {
+            key: {
+                "queue": job.queue,
+                "min_available": job.min_available,
+                "phase": job.phase,
+                "tasks": {t.key: {"status": t.status.name,
+                                  "node": t.node_name}
+                          for t in job.tasks.values()},
+            } for key, job in cache.jobs.items()
+        },
+        "queues": sorted(cache.queues),
+        "ts": time.time(),
+    }
+    with open(path, "w") as f:
+        json.dump(state, f, indent=1)
+
+
+def main(argv=None) -> int:
+    from ..store import ObjectStore
+    from ..utils.metrics import METRICS
+    from .cache import SchedulerCache
+    from .config import SchedulerConfiguration, default_config
+    from .engine import Scheduler
+
+    ap = argparse.ArgumentParser(prog="volcano-amd-scheduler")
+    ap.add_argument("--state", default="/tmp/volcano-amd-state.json")
+    ap.add_argument("--conf", default=None,
+                    help="scheduler YAML conf (actions + tiers)")
+    ap.add_argument("--period", type=float, default=1.0)
+    ap.add_argument("--device", default="auto",
+                    help="cuda|cpu|auto for the decision plane")
+    ap.add_argument("--serve-api", action="store_true")
+    ap.add_argument("--api-port", type=int, default=8343)
+    ap.add_argument("--dump-dir", default="/tmp")
+    ap.add_argument("--once", action="store_true")
+    args = ap.parse_args(argv)
+
+    import torch
+    device = args.device
+    if device == "auto":
+        device = "cuda" if torch.cuda.is_available() else "cpu"
+
+    try:
+        store = ObjectStore.load(args.state)
+    except FileNotFoundError:
+        store = ObjectStore()
+
+    if args.conf:
+        with open(args.conf) as f:
+            config = SchedulerConfiguration.from_yaml(f.read())
+    else:
+        config = default_config()
+    config.use_hip = device == "cuda"
+    config.device = device
+    config.schedule_period = args.period
+
+    cache = SchedulerCache(store=store, device=device)
+    sched = Scheduler(cache, config)
+
+    if args.serve_api:
+        from ..store.apiserver import serve
+        serve(store, port=args.api_port)
+        print(f"apiserver on :{args.api_port}", flush=True)
+
+    def on_usr1(sig, frame):
+        path = f"{args.dump_dir}/volcano-amd-cache-{int(time.time())}.json"
+        dump_cache(cache, path)
+        print(f"cache dumped to {path}", file=sys.stderr, flush=True)
+
+    signal.signal(signal.SIGUSR1, on_usr1)
+
+    if args.once:
+        sched.run_once()
+        store.save(args.state)
+        print(METRICS.export_text())
+        return 0
+    try:
+        sched.run(period=args.period)
+    except KeyboardInterrupt:
+        pass
+    store.save(args.state)
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())
