@@ -77,7 +77,7 @@ def reset_cluster(cache: SchedulerCache, jobs):
         ni.releasing = Resource()
         ni.pipelined = Resource()
         ni.tasks.clear()
-    cache._tensors_dirty = True
+    cache._used_dirty = True
     if isinstance(cache.binder, FakeBinder):
         cache.binder.binds.clear()
 

@@ -279,9 +279,11 @@ class JobInfo:
         return r
 
     def allocated_resource(self) -> Resource:
+        """O(allocated tasks) via the status index, not O(all tasks)."""
         r = Resource()
-        for t in self.tasks.values():
-            if t.status.occupies_node:
+        idx = self.task_status_index
+        for s in ALLOCATED_STATUSES:
+            for t in idx.get(s, {}).values():
                 r.add(t.request)
         return r
 

@@ -110,7 +110,7 @@ class AgentScheduler:
                 return False
             task.node_name = ni.name
             ni.add_allocated_bulk([task], task.request, 1)
-            self.cache._tensors_dirty = True
+            self.cache._used_dirty = True
             self.cache.bind_tasks([task])
         return True
 

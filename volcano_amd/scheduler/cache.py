@@ -92,7 +92,8 @@ class SchedulerCache:
         self._lock = threading.RLock()
         self._watch = store.watch("Pod", "Node", "PodGroup", "Queue") \
             if store else None
-        self._tensors_dirty = True
+        self._tensors_dirty = True      # full repack (node set/labels changed)
+        self._used_dirty = False        # dynamic planes only (usage changed)
 
     # -- event ingestion (reference cache/event_handlers.go) -----------------
     def sync(self) -> int:
@@ -110,7 +111,10 @@ class SchedulerCache:
                 if handler:
                     handler(ev)
             if evs:
-                self._tensors_dirty = True
+                if any(ev.kind == "Node" for ev in evs):
+                    self._tensors_dirty = True      # static planes changed
+                else:
+                    self._used_dirty = True         # usage-only delta
         return len(evs)
 
     def _job_for(self, pg_key: str) -> JobInfo:
@@ -199,7 +203,7 @@ class SchedulerCache:
         for t in job.tasks.values():
             if t.node_name and t.node_name in self.nodes:
                 self.nodes[t.node_name].add_task(t)
-        self._tensors_dirty = True
+        self._used_dirty = True
 
     def add_queue_info(self, qi: QueueInfo) -> None:
         self.queues[qi.name] = qi
@@ -236,15 +240,21 @@ class SchedulerCache:
                 t.node_name = ""
                 job.update_task_status(t, TaskStatus.PENDING)
             if pipelined:
-                self._tensors_dirty = True
+                self._used_dirty = True
 
     def ensure_packed(self) -> None:
         nodes = sorted(self.nodes.values(), key=lambda n: n.name)
         if self._tensors_dirty or self.node_tensors.alloc_t is None:
             self.node_tensors.pack(nodes)
-            self._tensors_dirty = False
-        for i, ni in enumerate(nodes):
-            ni.node_id = i
+            self._tensors_dirty = self._used_dirty = False
+            for i, ni in enumerate(nodes):
+                ni.node_id = i
+        elif self._used_dirty:
+            if not self.node_tensors.pack_dynamic(nodes):
+                self.node_tensors.pack(nodes)
+                for i, ni in enumerate(nodes):
+                    ni.node_id = i
+            self._tensors_dirty = self._used_dirty = False
 
     # -- commit pipeline ------------------------------------------------------
     def bind_tasks(self, tasks: List[TaskInfo]) -> None:
@@ -270,7 +280,7 @@ class SchedulerCache:
             # flip accounting used -> releasing
             node.remove_task(task)
             node.add_task(task)
-        self._tensors_dirty = True
+        self._used_dirty = True
 
     def update_podgroup(self, job: JobInfo) -> None:
         if self.store is not None and job.podgroup is not None:

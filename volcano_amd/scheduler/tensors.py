@@ -145,13 +145,46 @@ class NodeTensors:
         return require, forbid
 
     # -- packing -------------------------------------------------------------
+    def pack_dynamic(self, nodes: List[NodeInfo]) -> bool:
+        """Re-pack only the per-cycle-mutable planes (used/extra) when the
+        static planes (allocatable/labels/taints/ready) are still valid —
+        the common steady-state path: pod churn changes usage, node churn
+        is rare (SURVEY §7 'snapshot cost' hard-part).  Returns False if a
+        full pack is required."""
+        if self.alloc_t is None or len(nodes) != self.n:
+            return False
+        if [ni.name for ni in nodes] != self.names:
+            return False
+        N, R = self.n, self.r
+        used = np.zeros((N, R), dtype=np.float32)
+        extra = np.zeros((N, R), dtype=np.float32)
+        didx = self.dims.index
+        for i, ni in enumerate(nodes):
+            uq = ni.used.q
+            if uq:
+                for k, v in uq.items():
+                    j = didx.get(k)
+                    if j is not None:
+                        used[i, j] = v
+            for k, v in ni.releasing.q.items():
+                j = didx.get(k)
+                if j is not None:
+                    extra[i, j] += v
+            for k, v in ni.pipelined.q.items():
+                j = didx.get(k)
+                if j is not None:
+                    extra[i, j] -= v
+        dev = self.device
+        self.used_t.copy_(torch.from_numpy(used.T.copy()).to(dev))
+        self.extra_t.copy_(torch.from_numpy(extra.T.copy()).to(dev))
+        return True
+
     def pack(self, nodes: List[NodeInfo]) -> None:
         """Full (re)pack from host NodeInfos.
 
         Vectorized via numpy staging buffers then one H2D per plane; at
         N=50k this is ~a few MB — the per-cycle upload is microseconds of
-        PCIe time.  Incremental (dirty-node) patching is the planned upgrade
-        (SURVEY.md §7 hard-parts)."""
+        PCIe time."""
         names = [ni.name for ni in nodes]
         self.names = names
         self.index = {n: i for i, n in enumerate(names)}
