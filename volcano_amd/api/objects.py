@@ -331,6 +331,16 @@ class NodeShard:
 
 
 @dataclass
+class ResourceQuota:
+    """core/v1 ResourceQuota as the resourcequota plugin consumes it:
+    per-namespace hard limits checked at job enqueue."""
+
+    meta: ObjectMeta = field(default_factory=ObjectMeta)
+    hard: Resource = field(default_factory=Resource)
+    used: Resource = field(default_factory=Resource)
+
+
+@dataclass
 class ColocationConfig:
     """config/v1alpha1 ColocationConfig — per-nodepool colocation/QoS
     settings (cpu burst, memory qos, oversubscription ratio, network
@@ -461,4 +471,5 @@ KINDS = {
     "NodeShard": NodeShard, "JobFlow": JobFlow, "JobTemplate": JobTemplate,
     "CronJob": CronJob, "PodDisruptionBudget": PodDisruptionBudget,
     "Numatopology": Numatopology, "ColocationConfig": ColocationConfig,
+    "ResourceQuota": ResourceQuota,
 }

@@ -6,6 +6,7 @@ Registry mirrors ``actions/factory.go``.
 from .allocate import AllocateAction
 from .backfill import BackfillAction
 from .enqueue import EnqueueAction
+from .gangpreempt import GangPreemptAction, GangReclaimAction
 from .preempt import PreemptAction
 from .reclaim import ReclaimAction
 from .shuffle import ShuffleAction
@@ -16,6 +17,8 @@ ACTION_REGISTRY = {
     "backfill": BackfillAction,
     "preempt": PreemptAction,
     "reclaim": ReclaimAction,
+    "gangpreempt": GangPreemptAction,
+    "gangreclaim": GangReclaimAction,
     "shuffle": ShuffleAction,
 }
 

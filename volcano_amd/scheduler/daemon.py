@@ -103,6 +103,35 @@ def main(argv=None) -> int:
         store.save(args.state)
         print(METRICS.export_text())
         return 0
+
+    # conf hot-reload (reference scheduler.go:165-215 fsnotify watcher —
+    # here an mtime poll folded into the cycle loop)
+    if args.conf:
+        import os
+        conf_mtime = os.stat(args.conf).st_mtime
+        orig_run_once = sched.run_once
+
+        def run_once_with_reload():
+            nonlocal conf_mtime
+            try:
+                m = os.stat(args.conf).st_mtime
+                if m != conf_mtime:
+                    conf_mtime = m
+                    with open(args.conf) as f:
+                        new_conf = SchedulerConfiguration.from_yaml(f.read())
+                    new_conf.use_hip = config.use_hip
+                    new_conf.device = config.device
+                    sched.config = new_conf
+                    from . import actions as actions_mod
+                    sched._actions = [actions_mod.new_action(a)
+                                      for a in new_conf.actions]
+                    print("scheduler conf reloaded", file=sys.stderr,
+                          flush=True)
+            except FileNotFoundError:
+                pass
+            return orig_run_once()
+
+        sched.run_once = run_once_with_reload
     try:
         sched.run(period=args.period)
     except KeyboardInterrupt:

@@ -9,7 +9,8 @@ from .base import PLUGIN_REGISTRY, Plugin, register
 from . import (binpack, capacity, cdp, conformance, deviceshare, drf,
                extender, gang, network_topology_aware, nodegroup, nodeorder,
                numaaware, overcommit, pdb, predicates, priority, proportion,
-               rescheduling, resource_strategy_fit, sla, task_topology, tdm,
+               rescheduling, resource_strategy_fit, resourcequota, sla,
+               task_topology, tdm,
                usage)  # noqa: F401 (side-effect registration)
 
 

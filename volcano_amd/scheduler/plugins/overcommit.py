@@ -20,11 +20,12 @@ class OvercommitPlugin(Plugin):
                  else np.zeros(nt.r, dtype=np.float32)) * factor
         total = total.astype(np.float64)
         mask = total > 0
-        # resource already admitted: inqueue + running job requests
+        # admitted-but-not-yet-running work (reference overcommit counts
+        # Inqueue-phase podgroups; Running jobs hold real allocations and
+        # no longer occupy admission headroom)
         inqueue = np.zeros(nt.r, dtype=np.float64)
         for job in ssn.jobs.values():
-            if job.phase in (PodGroupPhase.INQUEUE.value,
-                             PodGroupPhase.RUNNING.value):
+            if job.phase == PodGroupPhase.INQUEUE.value:
                 inqueue += job.minres_vec(nt)
 
         def job_enqueueable(job) -> int:
