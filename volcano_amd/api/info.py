@@ -38,6 +38,7 @@ class TaskInfo:
     best_effort: bool = False
     preemptable: bool = False
     revocable_zone: str = ""
+    gated: bool = False            # has scheduling gates (k8s SchedulingGates)
     pod: Optional[Pod] = None
 
     @cached_property
@@ -84,6 +85,7 @@ class TaskInfo:
             priority=pod.priority,
             best_effort=pod.best_effort,
             preemptable=pod.meta.annotations.get(ANN_PREEMPTABLE, "") == "true",
+            gated=bool(pod.scheduling_gates),
             pod=pod,
         )
 
@@ -261,6 +263,8 @@ class JobInfo:
         insertion order inside a class (no per-task sort — hot path)."""
         groups: Dict[str, TaskClass] = {}
         for t in self.task_status_index.get(TaskStatus.PENDING, {}).values():
+            if t.gated:
+                continue    # scheduling gates hold the pod back (k8s gate)
             sig = t.class_signature()
             g = groups.get(sig)
             if g is None:

@@ -2,12 +2,17 @@
 
 Admission control: pending PodGroups become Inqueue when the enqueueable
 vote (proportion/overcommit/sla) passes — queues popped in queue order,
-jobs in job order.
+jobs in job order.  Also plays the gate manager (reference
+``gate/schedulinggate.go:43-113`` + feature gate
+SchedulingGatesQueueAdmission): the ``volcano.sh/queue-allocation-gate``
+scheduling gate is lifted from a job's pods once its queue admits it.
 """
 
 from __future__ import annotations
 
 from ...api.types import PodGroupPhase
+
+QUEUE_GATE = "volcano.sh/queue-allocation-gate"
 
 
 class EnqueueAction:
@@ -30,3 +35,20 @@ class EnqueueAction:
                     for fn in getattr(ssn, "job_enqueued_fns", []):
                         fn(job)
                     ssn.cache.update_podgroup(job)
+                    self._lift_queue_gates(ssn, job)
+
+    @staticmethod
+    def _lift_queue_gates(ssn, job) -> None:
+        store = getattr(ssn.cache, "store", None)
+        for t in job.tasks.values():
+            if not t.gated or t.pod is None:
+                continue
+            gates = t.pod.scheduling_gates
+            if QUEUE_GATE in gates:
+                gates.remove(QUEUE_GATE)
+                t.gated = bool(gates)
+                if store is not None:
+                    try:
+                        store.update("Pod", t.pod)
+                    except KeyError:
+                        pass
