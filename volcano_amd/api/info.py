@@ -59,6 +59,11 @@ class TaskInfo:
             "tol": [(t.key, t.operator, t.value, t.effect) for t in (p.tolerations if p else [])],
             "aff": p.affinity if p else None,
             "prio": self.priority,
+            "gates": sorted(p.scheduling_gates) if p else [],
+            "dev": sorted(k for k in (p.meta.annotations if p else {})
+                          if k.startswith("volcano.sh/gpu")
+                          or k.startswith("volcano.sh/vgpu")
+                          or k == "volcano.sh/numa-topology-policy"),
         }
         digest = hashlib.md5(
             json.dumps(sig, sort_keys=True, default=str).encode()).hexdigest()
@@ -119,6 +124,9 @@ class JobInfo:
         # dense-vector caches (invalidated on task/status mutation)
         self._alloc_vec = None               # (r, np.ndarray)
         self._total_vec = None
+        # plan atom: (sig, role, request, priority) when the job's task
+        # set is a single class; False = known multi-class; None = unknown
+        self._atom = None
 
     # -- basic accessors ----------------------------------------------------
     @property
@@ -159,14 +167,33 @@ class JobInfo:
     def add_task(self, task: TaskInfo) -> None:
         self.tasks[task.key] = task
         self.task_status_index.setdefault(task.status, {})[task.key] = task
-        self._alloc_vec = self._total_vec = None
+        self._alloc_vec = self._total_vec = self._atom = None
 
     def remove_task(self, task_key: str) -> Optional[TaskInfo]:
         t = self.tasks.pop(task_key, None)
         if t is not None:
             self.task_status_index.get(t.status, {}).pop(task_key, None)
-            self._alloc_vec = self._total_vec = None
+            self._alloc_vec = self._total_vec = self._atom = None
         return t
+
+    def plan_atom(self):
+        """(sig, role, request, priority) if every task of the job forms
+        ONE scheduling class — the steady-state fast path of the allocate
+        plan builder (uniform gangs).  False when known multi-class."""
+        if self._atom is not None:
+            return self._atom
+        it = iter(self.tasks.values())
+        first = next(it, None)
+        if first is None:
+            self._atom = False
+            return False
+        sig = first.class_signature()
+        for t in it:
+            if t.class_signature() != sig:
+                self._atom = False
+                return False
+        self._atom = (sig, first.role, first.request, first.priority)
+        return self._atom
 
     def update_task_status(self, task: TaskInfo, status: TaskStatus) -> None:
         self.task_status_index.get(task.status, {}).pop(task.key, None)
@@ -176,7 +203,24 @@ class JobInfo:
 
     def move_tasks_status(self, tasks: List[TaskInfo],
                           status: TaskStatus) -> None:
-        """Bulk status move (hot path: a whole class commits at once)."""
+        """Bulk status move (hot path: a whole class commits at once).
+        When the batch is an entire status bucket (the common gang case:
+        all pending → bound), the bucket dict moves wholesale."""
+        if tasks:
+            src_status = tasks[0].status
+            src = self.task_status_index.get(src_status)
+            if src is not None and len(src) == len(tasks) and \
+                    all(t.status == src_status for t in tasks):
+                for t in tasks:
+                    t.status = status
+                dst = self.task_status_index.get(status)
+                if dst:
+                    dst.update(src)
+                else:
+                    self.task_status_index[status] = src
+                self.task_status_index[src_status] = {}
+                self._alloc_vec = None
+                return
         dst = self.task_status_index.setdefault(status, {})
         for t in tasks:
             self.task_status_index.get(t.status, {}).pop(t.key, None)

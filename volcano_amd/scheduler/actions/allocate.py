@@ -96,8 +96,73 @@ class AllocateAction:
                 plan.classes.append(open_bundle)
             open_bundle, open_key = None, None
 
+        def bundle_in(cp: ClassPlan, job, qi: int, gang_min: int,
+                      sig: str) -> None:
+            nonlocal open_bundle, open_key
+            key = (qi, sig)
+            if open_bundle is not None and key == open_key:
+                b = open_bundle
+                b.bundle.append(BundleEntry(job.key, list(cp.tclass.tasks),
+                                            cp.tclass.count, gang_min))
+                b.tclass.tasks.extend(cp.tclass.tasks)
+                return
+            close_bundle()
+            merged = TaskClass(signature=cp.tclass.signature,
+                               role=cp.tclass.role,
+                               request=cp.tclass.request,
+                               tasks=list(cp.tclass.tasks),
+                               priority=cp.tclass.priority)
+            cp = ClassPlan(
+                tclass=merged, job_key=job.key, queue_idx=qi, req=cp.req,
+                tolerated=cp.tolerated, require=cp.require,
+                forbid=cp.forbid, min_needed=gang_min,
+                w_least=cp.w_least, w_most=cp.w_most, w_bal=cp.w_bal,
+                bundle=[BundleEntry(job.key, list(merged.tasks),
+                                    merged.count, gang_min)])
+            open_bundle, open_key = cp, key
+
+        _MISS = object()
+        req_memo = {}
+        from ...api.types import TaskStatus as _TS
+        w_least = w.get("least", 1.0)
+        w_most = w.get("most", 0.0)
+        w_bal = w.get("bal", 0.0)
+
         for q, job in ordered_jobs:
             qi = ssn.queue_index[q.name]
+
+            # -- steady-state fast path: the whole job is one pending class
+            atom = job.plan_atom()
+            pend = job.task_status_index.get(_TS.PENDING, {})
+            if atom and pend and len(pend) == len(job.tasks):
+                sig, role, request, priority = atom
+                first = next(iter(pend.values()))
+                if first.gated:
+                    continue
+                req = req_memo.get(sig, _MISS)
+                if req is _MISS:
+                    req = nt.req_vector(first)
+                    req_memo[sig] = req
+                if req is None:
+                    continue      # asks for a resource no node offers
+                tc = TaskClass(signature=sig, role=role, request=request,
+                               tasks=list(pend.values()), priority=priority)
+                if predicates is not None:
+                    tol, require, forbid = predicates.class_constraints(tc, job)
+                else:
+                    tol = -1
+                    W = max(nt.labels.words, 1)
+                    require = np.zeros(W, dtype=np.int64)
+                    forbid = np.zeros(W, dtype=np.int64)
+                gang_min = max(job.min_available,
+                               job.min_task_member.get(role, 0))
+                cp = ClassPlan(tclass=tc, job_key=job.key, queue_idx=qi,
+                               req=req, tolerated=tol, require=require,
+                               forbid=forbid, min_needed=gang_min,
+                               w_least=w_least, w_most=w_most, w_bal=w_bal)
+                bundle_in(cp, job, qi, gang_min, sig)
+                continue
+
             classes: List[ClassPlan] = []
             skipped = False
             for tc in job.pending_classes():
@@ -131,27 +196,7 @@ class AllocateAction:
             if bundleable:
                 cp = classes[0]
                 gang_min = max(job.min_available, cp.min_needed)
-                key = (qi, cp.tclass.signature)
-                if open_bundle is not None and key == open_key:
-                    b = open_bundle
-                    b.bundle.append(BundleEntry(job.key, list(cp.tclass.tasks),
-                                                cp.tclass.count, gang_min))
-                    b.tclass.tasks.extend(cp.tclass.tasks)
-                    continue
-                close_bundle()
-                merged = TaskClass(signature=cp.tclass.signature,
-                                   role=cp.tclass.role,
-                                   request=cp.tclass.request,
-                                   tasks=list(cp.tclass.tasks),
-                                   priority=cp.tclass.priority)
-                cp = ClassPlan(
-                    tclass=merged, job_key=job.key, queue_idx=qi, req=cp.req,
-                    tolerated=cp.tolerated, require=cp.require,
-                    forbid=cp.forbid, min_needed=gang_min,
-                    w_least=cp.w_least, w_most=cp.w_most, w_bal=cp.w_bal,
-                    bundle=[BundleEntry(job.key, list(merged.tasks),
-                                        merged.count, gang_min)])
-                open_bundle, open_key = cp, key
+                bundle_in(cp, job, qi, gang_min, cp.tclass.signature)
             else:
                 close_bundle()
                 plan.add_job(job, classes)

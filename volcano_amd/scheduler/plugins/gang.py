@@ -23,9 +23,8 @@ from .base import Plugin, register
 class GangPlugin(Plugin):
     def on_session_open(self, ssn) -> None:
         def job_valid(job) -> bool:
-            valid = sum(1 for t in job.tasks.values()
-                        if t.status != TaskStatus.FAILED)
-            return valid >= job.min_available
+            failed = len(job.task_status_index.get(TaskStatus.FAILED, ()))
+            return len(job.tasks) - failed >= job.min_available
 
         def job_order(a, b) -> int:
             ra, rb = a.is_ready(), b.is_ready()
