@@ -151,9 +151,15 @@ class JobInfo:
 
     @property
     def priority(self) -> int:
+        p = getattr(self, "_prio", None)
+        if p is not None:
+            return p
         if self.podgroup and self.podgroup.meta.annotations.get("priority"):
-            return int(self.podgroup.meta.annotations["priority"])
-        return max((t.priority for t in self.tasks.values()), default=0)
+            p = int(self.podgroup.meta.annotations["priority"])
+        else:
+            p = max((t.priority for t in self.tasks.values()), default=0)
+        self._prio = p
+        return p
 
     @property
     def creation_timestamp(self) -> float:
@@ -168,6 +174,7 @@ class JobInfo:
         self.tasks[task.key] = task
         self.task_status_index.setdefault(task.status, {})[task.key] = task
         self._alloc_vec = self._total_vec = self._atom = None
+        self._prio = None
 
     def remove_task(self, task_key: str) -> Optional[TaskInfo]:
         t = self.tasks.pop(task_key, None)
@@ -381,6 +388,9 @@ class NodeInfo:
         return self.node.ready and not self.node.unschedulable
 
     def add_task(self, task: TaskInfo) -> None:
+        prev = self.tasks.get(task.key)
+        if prev is not None and prev is not task:
+            self.remove_task(prev)     # never silently overwrite accounting
         self.tasks[task.key] = task
         if task.status.occupies_node:
             self.used.add(task.request)

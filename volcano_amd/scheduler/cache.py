@@ -89,6 +89,7 @@ class SchedulerCache:
         self.queues: Dict[str, QueueInfo] = {}
         self._task_node: Dict[str, str] = {}    # task key -> node name
         self._task_job: Dict[str, str] = {}     # task key -> job key
+        self._task_ref: Dict[str, TaskInfo] = {}  # task key -> live TaskInfo
         self._lock = threading.RLock()
         self._watch = store.watch("Pod", "Node", "PodGroup", "Queue") \
             if store else None
@@ -134,21 +135,27 @@ class SchedulerCache:
         key = f"{pod.meta.namespace}/{pg}"
         tkey = pod.meta.key
 
-        # remove any previous incarnation
+        # remove any previous incarnation — node accounting keys off the
+        # tracked TaskInfo object, NOT the job lookup: a PodGroup DELETE in
+        # the same batch may have dropped the job already, and the node's
+        # usage must still unwind (found by tests/test_churn_e2e.py)
         old_node = self._task_node.pop(tkey, None)
         old_job = self._task_job.pop(tkey, None)
+        old_task = self._task_ref.pop(tkey, None)
         if old_job is not None and old_job in self.jobs:
-            t = self.jobs[old_job].remove_task(tkey)
-            if t is not None and old_node and old_node in self.nodes:
-                self.nodes[old_node].remove_task(t)
-            if ev.type == EventType.DELETED and t is not None:
-                self._release_devices(t, old_node)
+            self.jobs[old_job].remove_task(tkey)
+        if old_task is not None:
+            if old_node and old_node in self.nodes:
+                self.nodes[old_node].remove_task(old_task)
+            if ev.type == EventType.DELETED:
+                self._release_devices(old_task, old_node)
 
         if ev.type == EventType.DELETED:
             return
         task = TaskInfo.from_pod(pod, key)
         self._job_for(key).add_task(task)
         self._task_job[tkey] = key
+        self._task_ref[tkey] = task
         if task.node_name and task.node_name in self.nodes:
             self.nodes[task.node_name].add_task(task)
             self._task_node[tkey] = task.node_name
