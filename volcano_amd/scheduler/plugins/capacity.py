@@ -38,12 +38,13 @@ class CapacityPlugin(Plugin):
         by_name = {q.name: q for q in queues}
         qi_of = ssn.queue_index
 
-        # request per queue (demand)
-        request = torch.zeros((Q, R), dtype=torch.float32)
+        # request per queue (demand) — f64 accumulation (see proportion.py)
+        req_np = np.zeros((Q, R), dtype=np.float64)
         for job in ssn.jobs.values():
             qi = qi_of.get(job.queue)
             if qi is not None:
-                request[qi] += torch.from_numpy(job.total_req_vec(nt))
+                req_np[qi] += job.total_req_vec(nt)
+        request = torch.from_numpy(req_np).to(torch.float32)
 
         # per-queue vectors
         guarantee = torch.zeros((Q, R), dtype=torch.float32)

@@ -38,10 +38,12 @@ class ProportionPlugin(Plugin):
             torch.zeros(R)
 
         weight = torch.tensor([q.weight for q in queues], dtype=torch.float32)
-        request = torch.zeros((Q, R), dtype=torch.float32)
+        # f64 accumulation: at 1M-pod scale the per-queue demand sum sits
+        # at ~1e9 where f32 ulp is 64 — f32 accumulation under-counts by
+        # enough to starve the tail gang's quota
+        req_np = np.zeros((Q, R), dtype=np.float64)
         guarantee = torch.zeros((Q, R), dtype=torch.float32)
         capability = torch.full((Q, R), BIG_LIMIT, dtype=torch.float32)
-        req_np = request.numpy()
         for i, q in enumerate(queues):
             guarantee[i] = torch.from_numpy(nt.resource_vector(q.guarantee))
             cap_vec = nt.resource_vector(q.capability)
@@ -52,6 +54,7 @@ class ProportionPlugin(Plugin):
             qi = ssn.queue_index.get(job.queue)
             if qi is not None:
                 req_np[qi] += job.total_req_vec(nt)
+        request = torch.from_numpy(req_np).to(torch.float32)
 
         deserved = ref.waterfill(weight, request, guarantee, capability, total)
         ssn.queue_deserved = deserved
