@@ -372,6 +372,9 @@ class NodeInfo:
         self.used = Resource()
         self.releasing = Resource()
         self.pipelined = Resource()
+        # usage placed by OTHER scheduler ranks (soft sharding: tracked as
+        # an aggregate, not per-task — parallel/softshard.py)
+        self.remote_used = Resource()
         self.tasks: Dict[str, TaskInfo] = {}
 
     @property

@@ -47,12 +47,27 @@ class DistributedScheduler:
 
     def __init__(self, cache: SchedulerCache,
                  config: Optional[SchedulerConfiguration] = None,
-                 policy: Optional[ShardingPolicy] = None):
+                 policy: Optional[ShardingPolicy] = None,
+                 mode: str = "hard"):
+        """mode="hard": nodes AND jobs sharded (conflict-free, the
+        reference's NodeShard semantics).  mode="soft": nodes replicated,
+        jobs sharded, per-cycle delta all-reduce with deterministic
+        conflict resolution (parallel/softshard.py) — gangs may span the
+        whole cluster."""
+        assert mode in ("hard", "soft")
+        self.mode = mode
         self.policy = policy or init_distributed()
         self.cache = cache
         self._install_shard_filter()
         self.scheduler = Scheduler(cache, config)
         self.last_stats: Optional[torch.Tensor] = None
+        if mode == "soft" and self.policy.world > 1:
+            from ..scheduler.actions.allocate import AllocateAction
+            from .softshard import SoftShardCoordinator
+            coord = SoftShardCoordinator(self.policy.rank, self.policy.world)
+            for a in self.scheduler._actions:
+                if isinstance(a, AllocateAction):
+                    a.coordinator = coord
 
     def _install_shard_filter(self) -> None:
         policy = self.policy
@@ -64,7 +79,7 @@ class DistributedScheduler:
         orig_pod = cache._on_pod
 
         def on_node(ev):
-            if policy.owns_node(ev.obj.meta.name):
+            if self.mode == "soft" or policy.owns_node(ev.obj.meta.name):
                 orig_node(ev)
 
         def on_podgroup(ev):

@@ -1,0 +1,77 @@
+"""Soft sharding — replicated nodes, sharded jobs, RCCL delta exchange.
+
+The second coordination mode of SURVEY §2.9 C2 (the first, hard node
+sharding, is `sharding.py`): every rank sees ALL nodes (so a gang may
+span the whole cluster), jobs are sharded by hash, and each cycle the
+ranks reconcile their staged placements by exchanging the per-node usage
+DELTA tensors over RCCL/xGMI:
+
+1. every rank runs its plan against its local copy of ``used``;
+2. all_gather of ``delta = used_after − used_before`` ([R, N] f32 —
+   ~320 KB at 10k nodes, latency-bound on xGMI);
+3. deterministic admission: rank order prefix — rank r's placements on
+   node n stand iff the cumulative delta of ranks 0..r fits the node's
+   allocatable in every dimension;
+4. a rank reverts EVERY job that placed on a node it lost (gang
+   atomicity; `AllocateAction._apply(bad_nodes=...)`), which only frees
+   capacity, so the admitted prefix stays feasible;
+5. all_reduce(SUM) of the post-revert deltas establishes the global
+   ``used``; each rank folds other ranks' usage into its host mirror as
+   per-node ``remote_used``.
+
+Losing jobs return to Pending and re-place next cycle against the now
+visible global usage.
+"""
+
+from __future__ import annotations
+
+from typing import FrozenSet
+
+import torch
+
+
+class SoftShardCoordinator:
+    def __init__(self, rank: int, world: int):
+        self.rank = rank
+        self.world = world
+
+    # -- step 2+3: gather deltas, find this rank's lost nodes ----------------
+    def find_conflicts(self, nt, used_before: torch.Tensor) -> FrozenSet[int]:
+        delta = nt.used_t - used_before
+        if self.world <= 1 or not torch.distributed.is_initialized():
+            return frozenset()
+        gathered = [torch.zeros_like(delta) for _ in range(self.world)]
+        torch.distributed.all_gather(gathered, delta.contiguous())
+        stack = torch.stack(gathered)                    # [world, R, N]
+        cum = torch.cumsum(stack, dim=0)
+        base = used_before.unsqueeze(0)                  # staged-before state
+        ok = ((base + cum) <= nt.alloc_t.unsqueeze(0) + 0.1).all(dim=1)  # [world, N]
+        mine_active = delta.abs().sum(dim=0) > 1e-6      # [N]
+        lost = (~ok[self.rank]) & mine_active
+        self._gathered_sum = stack.sum(dim=0)
+        ids = torch.nonzero(lost, as_tuple=False).flatten()
+        return frozenset(int(i) for i in ids.cpu())
+
+    # -- step 5: establish the global used state -----------------------------
+    def finalize(self, ssn, nt, used_before: torch.Tensor) -> None:
+        if self.world <= 1 or not torch.distributed.is_initialized():
+            return
+        my_delta = (nt.used_t - used_before).contiguous()
+        total = my_delta.clone()
+        torch.distributed.all_reduce(total, op=torch.distributed.ReduceOp.SUM)
+        remote = total - my_delta                        # other ranks' usage
+        nt.used_t.copy_(used_before + total)
+        # fold remote usage into the host mirror so future packs and the
+        # preempt path see the global state
+        remote_nr = remote.t().cpu().numpy()             # [N, R]
+        names = nt.dims.names
+        nodes_sorted = sorted(ssn.nodes.values(), key=lambda n: n.name)
+        import numpy as np
+        hot = np.nonzero(np.abs(remote_nr).sum(axis=1) > 1e-6)[0]
+        for i in hot:
+            ni = nodes_sorted[int(i)]
+            for r, name in enumerate(names):
+                v = float(remote_nr[i, r])
+                if v:
+                    ni.remote_used.q[name] = ni.remote_used.q.get(name, 0.0) + v
+                    ni.used.q[name] = ni.used.q.get(name, 0.0) + v

@@ -35,6 +35,9 @@ class AllocateAction:
 
     def __init__(self, use_hip: bool = None):
         self.use_hip = use_hip
+        # soft-sharding coordinator (parallel/softshard.py) — set by
+        # DistributedScheduler when nodes are replicated across ranks
+        self.coordinator = None
 
     def _runner(self, ssn):
         if self.use_hip is None:
@@ -202,20 +205,33 @@ class AllocateAction:
                 plan.add_job(job, classes)
         close_bundle()
 
-        if plan.n_classes == 0:
+        if plan.n_classes == 0 and self.coordinator is None:
             return
+        coord = self.coordinator
+        used_before = nt.used_t.clone() if coord is not None else None
         plan.finalize()
         t1 = time.perf_counter()
-        result = self._runner(ssn)(plan)
+        result = self._runner(ssn)(plan) if plan.n_classes else None
         t2 = time.perf_counter()
-        self._apply(ssn, plan, result)
+        bad = frozenset()
+        if coord is not None:
+            bad = coord.find_conflicts(nt, used_before)
+        if result is not None:
+            self._apply(ssn, plan, result, bad_nodes=bad)
+        if coord is not None:
+            coord.finalize(ssn, nt, used_before)
         t3 = time.perf_counter()
         METRICS.observe("allocate:plan_build", t1 - t0)
         METRICS.observe("allocate:plan_run", t2 - t1)
         METRICS.observe("allocate:apply", t3 - t2)
 
     # -- statement commit (host mirror of the device-side state) ------------
-    def _apply(self, ssn, plan: CyclePlan, result) -> None:
+    def _apply(self, ssn, plan: CyclePlan, result,
+               bad_nodes: frozenset = frozenset()) -> None:
+        """Apply the cycle result.  ``bad_nodes`` (soft-sharding conflict
+        resolution) lists node ids this rank LOST in the cross-rank
+        reconcile: any job with a placement there is reverted wholesale
+        (gang atomicity) and its device usage unwound."""
         nt = ssn.node_tensors
         nodes_sorted = sorted(ssn.nodes.values(), key=lambda n: n.name)
 
@@ -247,6 +263,11 @@ class AllocateAction:
                 tasks = iter(cp.tclass.tasks)
                 pieces = [(nid, cnt, [next(tasks) for _ in range(cnt)])
                           for nid, cnt in cres.placements]
+                if bad_nodes and any(nid in bad_nodes
+                                     for nid, _ in cres.placements):
+                    self._revert_pieces(plan, cp,
+                                        [(nid, cnt) for nid, cnt, _ in pieces])
+                    continue
                 commit_pieces(job, cp, pieces)
                 ssn.fire_allocate(cp.tclass,
                                   [nid for nid, _ in cres.placements],
@@ -254,10 +275,12 @@ class AllocateAction:
                                   [t for _, _, ts in pieces for t in ts])
                 continue
 
-            # -- bundle: walk jobs over the placement stream, trim the tail
+            # -- bundle: walk jobs over the placement stream; entries that
+            # miss their gang minimum or touch a lost node are reverted
             stream = list(cres.placements)
             ei, eoff = 0, 0
-            used_slots = 0
+            reverted: List[tuple] = []
+            stop = False
             for be in cp.bundle:
                 pieces = []
                 need = be.ntasks
@@ -275,16 +298,27 @@ class AllocateAction:
                         eoff = 0
                 got = be.ntasks - need
                 if got < be.min_needed:
-                    break   # gang unmet: trim this partial + stop the walk
+                    # gang unmet: revert this partial + the untouched rest
+                    reverted.extend((nid, cnt) for nid, cnt, _ in pieces)
+                    stop = True
+                    break
+                if bad_nodes and any(nid in bad_nodes
+                                     for nid, _, _ in pieces):
+                    reverted.extend((nid, cnt) for nid, cnt, _ in pieces)
+                    continue
                 job = ssn.jobs[be.job_key]
                 commit_pieces(job, cp, pieces)
-                used_slots += got
                 ssn.fire_allocate(cp.tclass, [p[0] for p in pieces],
                                   [p[1] for p in pieces],
                                   [t for _, _, ts in pieces for t in ts])
-            trim = cres.placed - used_slots
-            if trim > 0:
-                self._trim_tail(plan, cp, stream, used_slots, trim)
+            if stop and ei < len(stream):
+                # slots placed by the kernel beyond the walk (tail)
+                if eoff:
+                    reverted.append((stream[ei][0], stream[ei][1] - eoff))
+                    ei += 1
+                reverted.extend(stream[ei:])
+            if reverted:
+                self._revert_pieces(plan, cp, reverted)
 
         if to_bind:
             ssn.cache.bind_tasks(to_bind)
@@ -304,25 +338,20 @@ class AllocateAction:
                     ssn.cache.update_podgroup(job)
 
     @staticmethod
-    def _trim_tail(plan: CyclePlan, cp, stream, keep: int, trim: int) -> None:
-        """Undo the device-side staging for the trimmed tail of a bundle:
-        the kernel placed `keep + trim` instances; the last `trim` belong
-        to jobs whose gang minimum could not be met."""
+    def _revert_pieces(plan: CyclePlan, cp, pieces) -> None:
+        """Unwind device-side staging for rejected (node_id, count) pieces
+        of one class (gang-unmet tails, soft-shard conflict losers).
+        index_add_ accumulates, so duplicate node ids are fine."""
         import torch
+        if not pieces:
+            return
         nt = plan.nt
-        # walk the stream to find per-node trimmed counts
-        acc = 0
-        nids, cnts = [], []
-        for nid, cnt in stream:
-            lo = max(acc, keep)
-            hi = acc + cnt
-            if hi > lo:
-                nids.append(nid)
-                cnts.append(hi - lo)
-            acc = hi
-        req = torch.from_numpy(cp.req).to(nt.used_t.device)
-        idx = torch.tensor(nids, dtype=torch.long, device=nt.used_t.device)
-        cvec = torch.tensor(cnts, dtype=torch.float32,
-                            device=nt.used_t.device)
+        dev = nt.used_t.device
+        req = torch.from_numpy(cp.req).to(dev)
+        idx = torch.tensor([nid for nid, _ in pieces], dtype=torch.long,
+                           device=dev)
+        cvec = torch.tensor([float(c) for _, c in pieces],
+                            dtype=torch.float32, device=dev)
         nt.used_t.index_add_(1, idx, -req.unsqueeze(1) * cvec.unsqueeze(0))
-        plan.queue_alloc[cp.queue_idx] -= float(trim) * torch.from_numpy(cp.req)
+        total = sum(c for _, c in pieces)
+        plan.queue_alloc[cp.queue_idx] -= float(total) * torch.from_numpy(cp.req)

@@ -183,6 +183,8 @@ class SchedulerCache:
         if old is not None:
             for t in old.tasks.values():
                 ni.add_task(t)
+            ni.remote_used = old.remote_used.clone()
+            ni.used.add(ni.remote_used)
         self.nodes[name] = ni
 
     def _on_podgroup(self, ev) -> None:
