@@ -38,9 +38,13 @@ GI = 1024 ** 3
 
 
 def build_cluster(cache: SchedulerCache, n_nodes: int, n_jobs: int,
-                  pods_per_job: int, rank: int, world: int):
-    """Synthetic inventory, sharded round-robin by rank (hard sharding)."""
-    for i in range(rank, n_nodes, world):
+                  pods_per_job: int, rank: int, world: int,
+                  node_rank: int = None, node_world: int = None):
+    """Synthetic inventory: jobs sharded round-robin by rank; nodes by
+    (node_rank, node_world) — full replication in soft mode."""
+    node_rank = rank if node_rank is None else node_rank
+    node_world = world if node_world is None else node_world
+    for i in range(node_rank, n_nodes, node_world):
         node = synth.make_node(f"node-{i:06d}", cpu_milli=32000,
                                mem=256 * GI, pods=256)
         cache.add_node_info(NodeInfo(node))
@@ -94,6 +98,9 @@ def main():
                     help="force the CPU torch-oracle path (no GPU)")
     ap.add_argument("--timing", action="store_true",
                     help="print per-phase timing summaries to stderr")
+    ap.add_argument("--shard-mode", choices=["hard", "soft"], default="hard",
+                    help="hard: nodes+jobs sharded (conflict-free); "
+                         "soft: nodes replicated, delta all-reduce")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -120,13 +127,32 @@ def main():
 
     cache = SchedulerCache(store=None, binder=FakeBinder(), device=device)
     sched = Scheduler(cache, config)
+    if dist and args.shard_mode == "soft":
+        from volcano_amd.parallel.softshard import SoftShardCoordinator
+        from volcano_amd.scheduler.actions.allocate import AllocateAction
+        coord = SoftShardCoordinator(rank, world)
+        for a in sched._actions:
+            if isinstance(a, AllocateAction):
+                a.coordinator = coord
+    node_world = 1 if (dist and args.shard_mode == "soft") else world
+    node_rank = 0 if (dist and args.shard_mode == "soft") else rank
     jobs = build_cluster(cache, args.nodes, args.jobs, args.pods_per_job,
-                         rank, world)
+                         rank, world, node_rank, node_world)
     total_pods = args.jobs * args.pods_per_job   # whole-job, all ranks
+
+    soft = dist and args.shard_mode == "soft"
 
     def step() -> int:
         reset_cluster(cache, jobs)
         sched.run_once()
+        if soft:
+            # soft mode: conflict losers retry next cycle — a step is
+            # schedule-to-completion (convergence cost measured honestly)
+            my_pods = len(jobs) * args.pods_per_job
+            for _ in range(world + 3):
+                if len(cache.binder.binds) >= my_pods:
+                    break
+                sched.run_once()
         return len(cache.binder.binds)
 
     def sync():
@@ -203,7 +229,7 @@ def main():
                 "model": "gang-schedule-10kn-100kp",
                 "global_batch": total_pods,
                 "seq_len": args.nodes,
-                "parallelism": f"hardshard{world}",
+                "parallelism": f"{args.shard_mode}shard{world}",
                 "nodes": args.nodes,
                 "jobs": args.jobs,
                 "pods_per_job": args.pods_per_job,

@@ -34,6 +34,19 @@ class SoftShardCoordinator:
     def __init__(self, rank: int, world: int):
         self.rank = rank
         self.world = world
+        self._stagger = None
+
+    def stagger_bias(self, nt) -> torch.Tensor:
+        """Rank-staggered tie-break: an epsilon-scale score bias that makes
+        each rank prefer a different region of the node space, so equal-
+        score placements from different ranks don't collide on the same
+        low-index nodes (real score differences still dominate)."""
+        if self._stagger is None or self._stagger.shape[0] != nt.n:
+            N = nt.n
+            offset = (self.rank * N) // max(self.world, 1)
+            order = (torch.arange(N, dtype=torch.float32) - offset) % max(N, 1)
+            self._stagger = (-1e-6 * order).to(nt.alloc_t.device)
+        return self._stagger
 
     # -- step 2+3: gather deltas, find this rank's lost nodes ----------------
     def find_conflicts(self, nt, used_before: torch.Tensor) -> FrozenSet[int]:

@@ -209,6 +209,9 @@ class AllocateAction:
             return
         coord = self.coordinator
         used_before = nt.used_t.clone() if coord is not None else None
+        if coord is not None:
+            stagger = coord.stagger_bias(nt)
+            plan.bias = stagger if plan.bias is None else plan.bias + stagger
         plan.finalize()
         t1 = time.perf_counter()
         result = self._runner(ssn)(plan) if plan.n_classes else None
