@@ -246,3 +246,23 @@ def test_jobflow_cycle_detected():
                FlowStep(name="b", depends_on=["a"])]))
     cm.sync_until_quiet()
     assert store.get("JobFlow", "default", "bad").status["state"] == "Failed"
+
+
+def test_task_depends_on():
+    store, cm, sched, kubelet = mk_world()
+    store.create("Job", mk_job(
+        "dep", min_available=1,
+        tasks=[TaskSpec(name="ps", replicas=1,
+                        template={"resources": {"cpu": "1"}}),
+               TaskSpec(name="worker", replicas=2, depends_on=["ps"],
+                        template={"resources": {"cpu": "1"}})]))
+    cm.sync_until_quiet()
+    # only ps pods exist until ps is Running
+    assert {p.meta.name for p in store.list("Pod")} == {"dep-ps-0"}
+    sched.run_once()
+    kubelet.tick()        # ps starts Running
+    cm.sync_until_quiet()
+    names = {p.meta.name for p in store.list("Pod")}
+    assert names == {"dep-ps-0", "dep-worker-0", "dep-worker-1"}
+    sched.run_once()
+    assert all(p.node_name for p in store.list("Pod"))

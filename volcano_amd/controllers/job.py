@@ -212,8 +212,27 @@ class JobController(Controller):
         self._ensure_podgroup(job)
         pods = {p.meta.name: p for p in self._job_pods(job)}
 
+        def dep_satisfied(ts) -> bool:
+            """Task-level dependsOn (reference job.go TaskSpec.DependsOn):
+            a task's pods are created only after each dependency task has
+            its minimum running."""
+            for dep in ts.depends_on:
+                dep_ts = next((t for t in job.spec.tasks if t.name == dep),
+                              None)
+                if dep_ts is None:
+                    continue
+                running = sum(
+                    1 for p in pods.values()
+                    if p.meta.labels.get(LBL_TASK_SPEC) == dep
+                    and p.phase == "Running")
+                if running < dep_ts.min_needed:
+                    return False
+            return True
+
         # create missing / delete excess pods per task
         for ts in job.spec.tasks:
+            if ts.depends_on and not dep_satisfied(ts):
+                continue
             for i in range(ts.replicas):
                 name = self._pod_name(job, ts.name, i)
                 if name not in pods:
