@@ -175,6 +175,12 @@ class JobInfo:
         self.task_status_index.setdefault(task.status, {})[task.key] = task
         self._alloc_vec = self._total_vec = self._atom = None
         self._prio = None
+        if task.gated:
+            self._gated = getattr(self, "_gated", 0) + 1
+
+    @property
+    def has_gated_tasks(self) -> bool:
+        return getattr(self, "_gated", 0) > 0
 
     def remove_task(self, task_key: str) -> Optional[TaskInfo]:
         t = self.tasks.pop(task_key, None)

@@ -19,7 +19,7 @@ same-share queues goes first; across cycles shares converge identically.
 from __future__ import annotations
 
 import time
-from typing import List
+from typing import Dict, List
 
 import numpy as np
 
@@ -239,6 +239,7 @@ class AllocateAction:
         nodes_sorted = sorted(ssn.nodes.values(), key=lambda n: n.name)
 
         to_bind = []
+        bind_by_job: Dict[str, List] = {}
         committed_jobs = set()
 
         def commit_pieces(job, cp, pieces):
@@ -247,6 +248,7 @@ class AllocateAction:
             PENDING→BOUND, inside bind_tasks (the reference's
             Allocated→Binding→Bound pipeline compressed — nothing observes
             the intermediate states between plan apply and bind here)."""
+            jl = bind_by_job.setdefault(job.key, [])
             for nid, count, tasks in pieces:
                 ni = nodes_sorted[nid]
                 name = ni.name
@@ -254,6 +256,7 @@ class AllocateAction:
                     t.node_name = name
                 ni.add_allocated_bulk(tasks, cp.tclass.request, count)
                 to_bind.extend(tasks)
+                jl.extend(tasks)
             committed_jobs.add(job.key)
 
         for c, (cp, cres) in enumerate(zip(plan.classes, result.class_results)):
@@ -324,7 +327,7 @@ class AllocateAction:
                 self._revert_pieces(plan, cp, reverted)
 
         if to_bind:
-            ssn.cache.bind_tasks(to_bind)
+            ssn.cache.bind_tasks(to_bind, by_job=bind_by_job)
 
         # flip gang-ready podgroups to Running (job_updater analog)
         seen = set()

@@ -39,6 +39,8 @@ class EnqueueAction:
 
     @staticmethod
     def _lift_queue_gates(ssn, job) -> None:
+        if not job.has_gated_tasks:
+            return
         store = getattr(ssn.cache, "store", None)
         for t in job.tasks.values():
             if not t.gated or t.pod is None:

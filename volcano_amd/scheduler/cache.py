@@ -266,14 +266,19 @@ class SchedulerCache:
             self._tensors_dirty = self._used_dirty = False
 
     # -- commit pipeline ------------------------------------------------------
-    def bind_tasks(self, tasks: List[TaskInfo]) -> None:
+    def bind_tasks(self, tasks: List[TaskInfo],
+                   by_job: Optional[Dict[str, List[TaskInfo]]] = None) -> None:
         """Async in the reference (cache.go:1343 AddBindTask → 20 ms drain);
-        here a batched call — the binder itself may thread if it wants."""
-        self._task_node.update((t.key, t.node_name) for t in tasks)
+        here a batched call — the binder itself may thread if it wants.
+        Callers that already have the per-job grouping pass it in."""
+        if self._watch is not None:
+            # incarnation bookkeeping only matters when store events flow
+            self._task_node.update((t.key, t.node_name) for t in tasks)
         self.binder.bind(tasks)
-        by_job: Dict[str, List[TaskInfo]] = {}
-        for t in tasks:
-            by_job.setdefault(t.job_key, []).append(t)
+        if by_job is None:
+            by_job = {}
+            for t in tasks:
+                by_job.setdefault(t.job_key, []).append(t)
         for key, ts in by_job.items():
             job = self.jobs.get(key)
             if job is not None:
