@@ -49,6 +49,8 @@ class EnqueueAction:
             if QUEUE_GATE in gates:
                 gates.remove(QUEUE_GATE)
                 t.gated = bool(gates)
+                if not t.gated:
+                    job._gated = max(0, getattr(job, "_gated", 1) - 1)
                 if store is not None:
                     try:
                         store.update("Pod", t.pod)
