@@ -146,3 +146,51 @@ def test_extender_predicate_http():
         assert set(binder.binds.values()) == {"allowed"}
     finally:
         srv.shutdown()
+
+
+def test_interpod_anti_affinity_excludes_conode():
+    store, binder, cache, sched = mk(("interpodaffinity", {}))
+    for n in synth.make_nodes(3, cpu_milli=8000, mem=32 * GI):
+        store.create("Node", n)
+    store.create("Queue", synth.make_queue("default"))
+    pg = synth.make_podgroup("aa", min_member=3)
+    store.create("PodGroup", pg)
+    for i in range(3):
+        pod = synth.make_pod(f"aa-w-{i}", "aa", cpu_milli=500, mem=GI)
+        pod.affinity = {"podAntiAffinity": {"group": "replica-set"}}
+        store.create("Pod", pod)
+    sched.run_once()
+    assert len(binder.binds) == 3
+    # each member on its own node
+    assert len(set(binder.binds.values())) == 3
+    # a 4th member has no node left
+    pg2 = synth.make_podgroup("aa2", min_member=1)
+    store.create("PodGroup", pg2)
+    pod = synth.make_pod("aa2-w-0", "aa2", cpu_milli=500, mem=GI)
+    pod.affinity = {"podAntiAffinity": {"group": "replica-set"}}
+    store.create("Pod", pod)
+    sched.run_once()
+    assert "default/aa2-w-0" not in binder.binds
+
+
+def test_interpod_affinity_colocates():
+    store, binder, cache, sched = mk(("interpodaffinity", {}))
+    for n in synth.make_nodes(3, cpu_milli=8000, mem=32 * GI):
+        store.create("Node", n)
+    store.create("Queue", synth.make_queue("default"))
+    # anchor pod of group "svc"
+    pg = synth.make_podgroup("anchor", min_member=1)
+    store.create("PodGroup", pg)
+    anchor = synth.make_pod("anchor-w-0", "anchor", cpu_milli=500, mem=GI)
+    anchor.affinity = {"podAffinity": {"group": "svc"}}
+    store.create("Pod", anchor)
+    sched.run_once()
+    anchor_node = binder.binds["default/anchor-w-0"]
+    # follower must land on the anchor's node
+    pg2 = synth.make_podgroup("fol", min_member=1)
+    store.create("PodGroup", pg2)
+    fol = synth.make_pod("fol-w-0", "fol", cpu_milli=500, mem=GI)
+    fol.affinity = {"podAffinity": {"group": "svc"}}
+    store.create("Pod", fol)
+    sched.run_once()
+    assert binder.binds["default/fol-w-0"] == anchor_node

@@ -78,6 +78,15 @@ class TaskInfo:
             "Succeeded": TaskStatus.SUCCEEDED,
             "Failed": TaskStatus.FAILED,
         }.get(pod.phase, TaskStatus.UNKNOWN)
+        # pod anti-affinity groups ride the resource machinery: group G is
+        # a synthetic unit dim (every node offers 1 — tensors.py pack), so
+        # the capacity kernel enforces at-most-one-per-node with zero
+        # extra predicate work (plugins/interpodaffinity.py)
+        aff = pod.affinity if isinstance(pod.affinity, dict) else None
+        if aff:
+            anti = aff.get("podAntiAffinity")
+            if isinstance(anti, dict) and anti.get("group"):
+                pod.request.q.setdefault(f"paa:{anti['group']}", 1.0)
         return cls(
             uid=pod.meta.uid or pod.meta.key,
             name=pod.meta.name,

@@ -51,6 +51,19 @@ class AllocateAction:
         if nt is None or nt.n == 0:
             return
         t0 = time.perf_counter()
+        # a plugin may have grown the dim registry (synthetic dims, e.g.
+        # interpodaffinity) after the queue tensors were built — pad the
+        # queue rows: new dims are queue-unlimited
+        if ssn.queue_limit is not None and ssn.queue_limit.shape[1] < nt.r:
+            import torch
+            Q, old_r = ssn.queue_limit.shape
+            pad = nt.r - old_r
+            ssn.queue_limit = torch.cat(
+                [ssn.queue_limit,
+                 torch.full((Q, pad), 1.0e18, dtype=torch.float32)], dim=1)
+            ssn.queue_alloc = torch.cat(
+                [ssn.queue_alloc, torch.zeros((Q, pad), dtype=torch.float32)],
+                dim=1)
         plan = CyclePlan(nt, ssn.queue_limit, ssn.queue_alloc)
         plan.dim_w = ssn.dim_weight_vector()
         plan.bias = getattr(ssn, "score_bias", None)

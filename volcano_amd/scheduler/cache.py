@@ -153,6 +153,10 @@ class SchedulerCache:
         if ev.type == EventType.DELETED:
             return
         task = TaskInfo.from_pod(pod, key)
+        for k in task.request.q:
+            if k.startswith("paa:") and k not in self.dims:
+                self.dims.add(k)
+                self._tensors_dirty = True
         self._job_for(key).add_task(task)
         self._task_job[tkey] = key
         self._task_ref[tkey] = task

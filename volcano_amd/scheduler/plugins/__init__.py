@@ -7,10 +7,10 @@ mirrors the reference's ``plugins/factory.go``.
 
 from .base import PLUGIN_REGISTRY, Plugin, register
 from . import (binpack, capacity, cdp, conformance, deviceshare, drf,
-               extender, gang, network_topology_aware, nodegroup, nodeorder,
-               numaaware, overcommit, pdb, predicates, priority, proportion,
-               rescheduling, resource_strategy_fit, resourcequota, sla,
-               task_topology, tdm,
+               extender, gang, interpodaffinity, network_topology_aware,
+               nodegroup, nodeorder, numaaware, overcommit, pdb, predicates,
+               priority, proportion, rescheduling, resource_strategy_fit,
+               resourcequota, sla, task_topology, tdm,
                usage)  # noqa: F401 (side-effect registration)
 
 

@@ -153,6 +153,8 @@ class NodeTensors:
         full pack is required."""
         if self.alloc_t is None or len(nodes) != self.n:
             return False
+        if self.alloc_t.shape[0] != self.r:
+            return False    # dim registry grew since the static pack
         if [ni.name for ni in nodes] != self.names:
             return False
         N, R = self.n, self.r
@@ -211,6 +213,11 @@ class NodeTensors:
         planes = np.zeros((W, N), dtype=np.int64)
 
         didx = self.dims.index
+        # synthetic unit dims (pod anti-affinity groups): every node offers
+        # exactly 1 unless it opts out explicitly
+        for name, j in didx.items():
+            if name.startswith("paa:"):
+                alloc[:, j] = 1.0
         for i, ni in enumerate(nodes):
             for k, v in ni.allocatable.q.items():
                 alloc[i, didx[k]] = v
