@@ -1,0 +1,46 @@
+"""Daemon entrypoints (--once), metrics exposition."""
+
+import json
+
+from volcano_amd.store import ObjectStore
+from volcano_amd.utils import synth
+from volcano_amd.utils.metrics import METRICS
+
+
+def test_scheduler_and_controller_daemons_once(tmp_path, capsys):
+    from volcano_amd.api.objects import Job, JobSpec, ObjectMeta, TaskSpec
+    from volcano_amd.controllers.daemon import main as cm_main
+    from volcano_amd.scheduler.daemon import main as sched_main
+
+    state = str(tmp_path / "state.json")
+    store = ObjectStore()
+    for n in synth.make_nodes(2, cpu_milli=4000, mem=8 * 1024 ** 3):
+        store.create("Node", n)
+    store.create("Queue", synth.make_queue("default"))
+    store.create("Job", Job(meta=ObjectMeta(name="d1"), spec=JobSpec(
+        tasks=[TaskSpec(name="w", replicas=2,
+                        template={"resources": {"cpu": "1",
+                                                "memory": "1Gi"}})])))
+    store.save(state)
+
+    assert cm_main(["--state", state, "--once",
+                    "--controllers", "job,podgroup,queue"]) == 0
+    assert sched_main(["--state", state, "--once"]) == 0
+    out = capsys.readouterr().out
+    assert "e2e_scheduling_latency" in out     # metrics printed on --once
+
+    final = ObjectStore.load(state)
+    pods = final.list("Pod")
+    assert len(pods) == 2
+    assert all(p.node_name for p in pods)
+
+
+def test_metrics_export_reference_names():
+    METRICS.reset()
+    METRICS.observe("e2e_scheduling_latency", 0.01)
+    METRICS.observe("action_scheduling_latency:allocate", 0.002)
+    METRICS.inc("schedule_attempts_total")
+    text = METRICS.export_text()
+    assert "e2e_scheduling_latency_p99" in text
+    assert "action_scheduling_latency:allocate_mean" in text
+    assert "schedule_attempts_total 1.0" in text
