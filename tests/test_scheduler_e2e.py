@@ -212,3 +212,56 @@ def test_bound_pods_survive_restart():
     assert binder2.binds == {}
     used = sum(ni.used.milli_cpu for ni in cache2.nodes.values())
     assert used == 2000
+
+
+def test_fifo_dequeue_blocks_queue():
+    store, binder, cache, sched = mk()
+    for n in synth.make_nodes(1, cpu_milli=10000, mem=64 * GI):
+        store.create("Node", n)
+    q = synth.make_queue("fq")
+    q.spec.dequeue_strategy = "fifo"
+    q.spec.capability = __import__("volcano_amd.api.resource",
+                                   fromlist=["Resource"]).Resource(
+        {"cpu": 4000.0})
+    store.create("Queue", q)
+    # head job too big even for overcommitted capacity → enqueue rejects
+    # it, and fifo blocks the line behind it
+    synth.make_gang(store, "big-head", replicas=16, queue="fq",
+                    cpu_milli=1000, mem=GI)
+    import time as _t
+    _t.sleep(0.01)
+    synth.make_gang(store, "small", replicas=1, queue="fq", cpu_milli=1000,
+                    mem=GI)
+    sched.run_once()
+    assert binder.binds == {}       # fifo: small waits behind big-head
+
+    # traverse queue admits the small job past the stuck head
+    q.spec.dequeue_strategy = "traverse"
+    store.update("Queue", q)
+    sched.run_once()
+    assert any(k.startswith("default/small") for k in binder.binds)
+
+
+def test_bundle_slot_recycling_quota_starved_head():
+    """Regression: a quota-starved big gang bundled with a small identical
+    job must not drag the small job down — its slots recycle."""
+    store, binder, cache, sched = mk()
+    for n in synth.make_nodes(1, cpu_milli=10000, mem=64 * GI):
+        store.create("Node", n)
+    q = synth.make_queue("rq")
+    from volcano_amd.api.resource import Resource
+    q.spec.capability = Resource({"cpu": 4000.0})
+    store.create("Queue", q)
+    # head: 8-gang (min 8) > queue quota 4 → cannot place
+    synth.make_gang(store, "head", replicas=8, queue="rq", cpu_milli=1000,
+                    mem=GI)
+    import time
+    time.sleep(0.01)
+    # small identical-signature 1-gang bundles with it
+    synth.make_gang(store, "tiny", replicas=1, queue="rq", cpu_milli=1000,
+                    mem=GI)
+    sched.run_once()
+    assert "default/tiny-worker-0" in binder.binds
+    assert not any(k.startswith("default/head") for k in binder.binds)
+    # staged usage fully unwound for the failed gang
+    assert sum(ni.used.milli_cpu for ni in cache.nodes.values()) == 1000

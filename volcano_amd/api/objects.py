@@ -263,6 +263,9 @@ class QueueSpec:
     # {"affinity": {"required": [...], "preferred": [...]},
     #  "antiAffinity": {"required": [...], "preferred": [...]}}
     affinity: Optional[Dict[str, Any]] = None
+    # fifo: a job that cannot enqueue blocks the jobs behind it;
+    # traverse (default): keep trying the rest (reference dequeueStrategy)
+    dequeue_strategy: str = "traverse"
 
 
 @dataclass

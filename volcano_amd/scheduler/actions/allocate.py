@@ -103,12 +103,15 @@ class AllocateAction:
         def close_bundle():
             nonlocal open_bundle, open_key
             if open_bundle is not None:
+                # in-kernel revert only when even the EASIEST bundled gang
+                # cannot place (the apply walk settles per-job minimums)
+                easiest = min(be.min_needed for be in open_bundle.bundle)
                 plan.jobs.append(JobPlan(
                     job_key=open_bundle.job_key,
                     class_begin=len(plan.classes),
                     class_end=len(plan.classes) + 1,
                     occupied=0,
-                    min_available=open_bundle.bundle[0].min_needed))
+                    min_available=easiest))
                 plan.classes.append(open_bundle)
             open_bundle, open_key = None, None
 
@@ -294,13 +297,16 @@ class AllocateAction:
                                   [t for _, _, ts in pieces for t in ts])
                 continue
 
-            # -- bundle: walk jobs over the placement stream; entries that
-            # miss their gang minimum or touch a lost node are reverted
+            # -- bundle: walk jobs over the placement stream.  An entry
+            # that misses its gang minimum releases its slots back to the
+            # pool (cursor reset) so smaller jobs behind it still place;
+            # an entry touching a lost node (soft-shard conflict) reverts
+            # its slots outright.  Unclaimed slots revert at the end.
             stream = list(cres.placements)
             ei, eoff = 0, 0
             reverted: List[tuple] = []
-            stop = False
             for be in cp.bundle:
+                ei0, eoff0 = ei, eoff
                 pieces = []
                 need = be.ntasks
                 toff = 0
@@ -317,10 +323,8 @@ class AllocateAction:
                         eoff = 0
                 got = be.ntasks - need
                 if got < be.min_needed:
-                    # gang unmet: revert this partial + the untouched rest
-                    reverted.extend((nid, cnt) for nid, cnt, _ in pieces)
-                    stop = True
-                    break
+                    ei, eoff = ei0, eoff0      # recycle the slots
+                    continue
                 if bad_nodes and any(nid in bad_nodes
                                      for nid, _, _ in pieces):
                     reverted.extend((nid, cnt) for nid, cnt, _ in pieces)
@@ -330,8 +334,8 @@ class AllocateAction:
                 ssn.fire_allocate(cp.tclass, [p[0] for p in pieces],
                                   [p[1] for p in pieces],
                                   [t for _, _, ts in pieces for t in ts])
-            if stop and ei < len(stream):
-                # slots placed by the kernel beyond the walk (tail)
+            if ei < len(stream):
+                # slots the walk never claimed
                 if eoff:
                     reverted.append((stream[ei][0], stream[ei][1] - eoff))
                     ei += 1

@@ -28,6 +28,7 @@ class EnqueueAction:
         for q in ssn.sorted_queues(queues):
             if not q.is_open:
                 continue
+            fifo = q.queue.spec.dequeue_strategy == "fifo"
             for job in ssn.sorted_jobs(by_queue[q.name]):
                 if ssn.job_enqueueable(job):
                     if job.podgroup is not None:
@@ -36,6 +37,8 @@ class EnqueueAction:
                         fn(job)
                     ssn.cache.update_podgroup(job)
                     self._lift_queue_gates(ssn, job)
+                elif fifo:
+                    break   # head-of-line blocks the queue (dequeueStrategy)
 
     @staticmethod
     def _lift_queue_gates(ssn, job) -> None:

@@ -57,7 +57,7 @@ class CapacityPlugin(Plugin):
             cv = nt.resource_vector(q.capability)
             for r in range(R):
                 if cv[r] > 0:
-                    capability[i, r] = cv[r]
+                    capability[i, r] = float(cv[r])
             dv = nt.resource_vector(q.deserved_spec)
             if dv.any():
                 deserved_spec[i] = torch.from_numpy(dv)

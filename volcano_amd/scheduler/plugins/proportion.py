@@ -49,7 +49,7 @@ class ProportionPlugin(Plugin):
             cap_vec = nt.resource_vector(q.capability)
             for r in range(R):
                 if cap_vec[r] > 0:
-                    capability[i, r] = cap_vec[r]
+                    capability[i, r] = float(cap_vec[r])
         for job in ssn.jobs.values():
             qi = ssn.queue_index.get(job.queue)
             if qi is not None:
