@@ -106,6 +106,7 @@ class AllocateAction:
                 # in-kernel revert only when even the EASIEST bundled gang
                 # cannot place (the apply walk settles per-job minimums)
                 easiest = min(be.min_needed for be in open_bundle.bundle)
+                open_bundle.min_needed = easiest
                 plan.jobs.append(JobPlan(
                     job_key=open_bundle.job_key,
                     class_begin=len(plan.classes),
