@@ -120,25 +120,29 @@ class AllocateAction:
                       sig: str) -> None:
             nonlocal open_bundle, open_key
             key = (qi, sig)
+            tasks = cp.tclass.tasks        # already a fresh per-job list
             if open_bundle is not None and key == open_key:
                 b = open_bundle
-                b.bundle.append(BundleEntry(job.key, list(cp.tclass.tasks),
-                                            cp.tclass.count, gang_min))
-                b.tclass.tasks.extend(cp.tclass.tasks)
+                b.bundle.append(BundleEntry(job.key, tasks, len(tasks),
+                                            gang_min))
+                b.ntasks_override += len(tasks)
                 return
             close_bundle()
+            # the merged class keeps ONE representative task (handlers and
+            # constraint builders read tasks[0]); the instance count lives
+            # in ntasks_override — no 100k-element merged list
             merged = TaskClass(signature=cp.tclass.signature,
                                role=cp.tclass.role,
                                request=cp.tclass.request,
-                               tasks=list(cp.tclass.tasks),
+                               tasks=[tasks[0]],
                                priority=cp.tclass.priority)
             cp = ClassPlan(
                 tclass=merged, job_key=job.key, queue_idx=qi, req=cp.req,
                 tolerated=cp.tolerated, require=cp.require,
                 forbid=cp.forbid, min_needed=gang_min,
                 w_least=cp.w_least, w_most=cp.w_most, w_bal=cp.w_bal,
-                bundle=[BundleEntry(job.key, list(merged.tasks),
-                                    merged.count, gang_min)])
+                ntasks_override=len(tasks),
+                bundle=[BundleEntry(job.key, tasks, len(tasks), gang_min)])
             open_bundle, open_key = cp, key
 
         _MISS = object()

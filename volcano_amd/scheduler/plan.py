@@ -55,12 +55,21 @@ class ClassPlan:
     use_future: bool = False   # score against future-idle (pipelining)
     log_off: int = 0
     log_cap: int = 0
+    # instance count override (bundles track the sum here instead of
+    # materializing a merged 100k-task list; tclass keeps ONE
+    # representative task for constraint/handler access)
+    ntasks_override: Optional[int] = None
     # Gang bundling: consecutive jobs with identical single-class shape are
-    # fused into ONE kernel pass (ntasks = sum); the per-job gang boundaries
-    # live here and are enforced at apply time (tail-trim).  The kernel-side
-    # fuse_min is the FIRST job's minimum: if even that gang can't place,
-    # the whole bundle reverts in-kernel.
+    # fused into ONE kernel pass (ntasks = sum); the per-job gang
+    # boundaries live here and are enforced at apply time (slot walk with
+    # recycling).  The kernel-side fuse_min is the EASIEST entry's
+    # minimum: if even that gang can't place, the bundle reverts in-kernel.
     bundle: Optional[List[BundleEntry]] = None
+
+    @property
+    def ntasks(self) -> int:
+        return self.ntasks_override if self.ntasks_override is not None \
+            else self.tclass.count
 
 
 @dataclass
@@ -119,7 +128,7 @@ class CyclePlan:
         n = self.nt.n
         for cp in self.classes:
             cp.log_off = off
-            cp.log_cap = max(1, min(cp.tclass.count, n))
+            cp.log_cap = max(1, min(cp.ntasks, n))
             off += cp.log_cap
             if len(cp.require) < W:
                 cp.require = np.pad(cp.require, (0, W - len(cp.require)))
@@ -178,7 +187,7 @@ def run_plan_torch(plan: CyclePlan) -> CycleResult:
                           cp.w_least, cp.w_most, cp.w_bal, dim_w, plan.bias,
                           score, cap)
             sl = slice(cp.log_off, cp.log_off + cp.log_cap)
-            ref.select_commit(score, cap, req, cp.tclass.count, used,
+            ref.select_commit(score, cap, req, cp.ntasks, used,
                               plan.queue_alloc[cp.queue_idx],
                               plan.queue_limit[cp.queue_idx],
                               log_nodes[sl], log_counts[sl], log_len[c],
@@ -228,7 +237,7 @@ def run_plan_hip(plan: CyclePlan) -> CycleResult:
         d = cds[c]
         d.job_idx = job_index[cp.job_key]
         d.queue_idx = cp.queue_idx
-        d.ntasks = cp.tclass.count
+        d.ntasks = cp.ntasks
         d.min_needed = cp.min_needed
         d.log_off = cp.log_off
         d.log_cap = cp.log_cap
