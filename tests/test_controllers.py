@@ -266,3 +266,24 @@ def test_task_depends_on():
     assert names == {"dep-ps-0", "dep-worker-0", "dep-worker-1"}
     sched.run_once()
     assert all(p.node_name for p in store.list("Pod"))
+
+
+def test_hyperjob_expands_and_aggregates():
+    store, cm, sched, kubelet = mk_world()
+    from volcano_amd.api.objects import HyperJob, JobSpec, TaskSpec
+    from volcano_amd.controllers import ControllerManager
+    cm2 = ControllerManager(store, ["job", "podgroup", "queue", "hyperjob"])
+    store.create("HyperJob", HyperJob(
+        meta=ObjectMeta(name="hj"), replicas=2,
+        job_template=JobSpec(tasks=[TaskSpec(
+            name="w", replicas=1,
+            template={"resources": {"cpu": "1"}})])))
+    cm2.sync_until_quiet()
+    assert store.get("Job", "default", "hj-0") is not None
+    assert store.get("Job", "default", "hj-1") is not None
+    sched.run_once()
+    kubelet.tick()
+    kubelet.tick(complete=lambda p: "Succeeded")
+    cm2.sync_until_quiet()
+    hj = store.get("HyperJob", "default", "hj")
+    assert hj.status["state"] == "Completed"

@@ -410,6 +410,18 @@ class JobTemplate:
 
 
 @dataclass
+class HyperJob:
+    """training/v1alpha1 HyperJob (reference apis/training — incubating):
+    one logical job split into per-replica member Jobs (the multi-cluster
+    splitting shape, applied here to replica groups)."""
+
+    meta: ObjectMeta = field(default_factory=ObjectMeta)
+    replicas: int = 1                       # member jobs to create
+    job_template: JobSpec = field(default_factory=JobSpec)
+    status: Dict[str, Any] = field(default_factory=dict)
+
+
+@dataclass
 class CronJob:
     """batch/v1alpha1 CronJob (reference cronjob controller)."""
 
@@ -474,5 +486,5 @@ KINDS = {
     "NodeShard": NodeShard, "JobFlow": JobFlow, "JobTemplate": JobTemplate,
     "CronJob": CronJob, "PodDisruptionBudget": PodDisruptionBudget,
     "Numatopology": Numatopology, "ColocationConfig": ColocationConfig,
-    "ResourceQuota": ResourceQuota,
+    "ResourceQuota": ResourceQuota, "HyperJob": HyperJob,
 }

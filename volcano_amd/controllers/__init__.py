@@ -8,3 +8,4 @@ from .garbagecollector import GarbageCollector
 from .hypernode import HyperNodeController
 from .sharding import ShardingController
 from .colocationconfig import ColocationConfigController
+from .hyperjob import HyperJobController
