@@ -108,7 +108,8 @@ def main():
     dist = world > 1
     use_gpu = torch.cuda.is_available() and not args.cpu
     if dist:
-        backend = "nccl" if use_gpu else "gloo"
+        backend = os.environ.get("VAMD_DIST_BACKEND") or \
+            ("nccl" if use_gpu else "gloo")
         local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
         if use_gpu:
             torch.cuda.set_device(local_rank)
@@ -179,7 +180,7 @@ def main():
     if dist:
         te = torch.tensor([elapsed])
         tb = torch.tensor([float(bound)])
-        if use_gpu:
+        if use_gpu and torch.distributed.get_backend() == "nccl":
             te, tb = te.cuda(), tb.cuda()
         torch.distributed.all_reduce(te, op=torch.distributed.ReduceOp.MAX)
         torch.distributed.all_reduce(tb, op=torch.distributed.ReduceOp.SUM)

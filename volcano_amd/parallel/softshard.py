@@ -49,12 +49,22 @@ class SoftShardCoordinator:
         return self._stagger
 
     # -- step 2+3: gather deltas, find this rank's lost nodes ----------------
+    @staticmethod
+    def _comm_dev(t: torch.Tensor) -> torch.Tensor:
+        """Collectives run on the backend's native device: RCCL moves
+        device tensors over xGMI; gloo needs host staging."""
+        if torch.distributed.get_backend() == "gloo" and t.is_cuda:
+            return t.cpu()
+        return t.contiguous()
+
     def find_conflicts(self, nt, used_before: torch.Tensor) -> FrozenSet[int]:
         delta = nt.used_t - used_before
         if self.world <= 1 or not torch.distributed.is_initialized():
             return frozenset()
-        gathered = [torch.zeros_like(delta) for _ in range(self.world)]
-        torch.distributed.all_gather(gathered, delta.contiguous())
+        comm = self._comm_dev(delta)
+        gathered = [torch.zeros_like(comm) for _ in range(self.world)]
+        torch.distributed.all_gather(gathered, comm)
+        gathered = [g.to(delta.device) for g in gathered]
         stack = torch.stack(gathered)                    # [world, R, N]
         cum = torch.cumsum(stack, dim=0)
         base = used_before.unsqueeze(0)                  # staged-before state
@@ -70,8 +80,9 @@ class SoftShardCoordinator:
         if self.world <= 1 or not torch.distributed.is_initialized():
             return
         my_delta = (nt.used_t - used_before).contiguous()
-        total = my_delta.clone()
+        total = self._comm_dev(my_delta).clone()
         torch.distributed.all_reduce(total, op=torch.distributed.ReduceOp.SUM)
+        total = total.to(my_delta.device)
         remote = total - my_delta                        # other ranks' usage
         nt.used_t.copy_(used_before + total)
         # fold remote usage into the host mirror so future packs and the
