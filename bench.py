@@ -112,7 +112,8 @@ def main():
             ("nccl" if use_gpu else "gloo")
         local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
         if use_gpu:
-            torch.cuda.set_device(local_rank)
+            # modulo: lets N ranks share fewer GPUs (gloo-backend testing)
+            torch.cuda.set_device(local_rank % torch.cuda.device_count())
         torch.distributed.init_process_group(backend=backend)
     elif use_gpu:
         torch.cuda.set_device(0)
