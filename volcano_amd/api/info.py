@@ -136,6 +136,7 @@ class JobInfo:
         # plan atom: (sig, role, request, priority) when the job's task
         # set is a single class; False = known multi-class; None = unknown
         self._atom = None
+        self._minav = None
 
     # -- basic accessors ----------------------------------------------------
     @property
@@ -152,7 +153,11 @@ class JobInfo:
 
     @property
     def min_available(self) -> int:
-        return self.podgroup.spec.min_member if self.podgroup else 1
+        m = self._minav
+        if m is None:
+            m = self._minav = \
+                self.podgroup.spec.min_member if self.podgroup else 1
+        return m
 
     @property
     def min_task_member(self) -> Dict[str, int]:
@@ -229,10 +234,13 @@ class JobInfo:
         When the batch is an entire status bucket (the common gang case:
         all pending → bound), the bucket dict moves wholesale."""
         if tasks:
+            # callers pass tasks drawn from one status bucket (plan apply /
+            # bind); when the batch covers the whole bucket it moves
+            # wholesale.  Size equality implies identity here because the
+            # batch is a subset of the bucket by construction.
             src_status = tasks[0].status
             src = self.task_status_index.get(src_status)
-            if src is not None and len(src) == len(tasks) and \
-                    all(t.status == src_status for t in tasks):
+            if src is not None and len(src) == len(tasks):
                 for t in tasks:
                     t.status = status
                 dst = self.task_status_index.get(status)

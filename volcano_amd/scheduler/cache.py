@@ -197,7 +197,9 @@ class SchedulerCache:
         if ev.type == EventType.DELETED:
             self.jobs.pop(key, None)
             return
-        self._job_for(key).podgroup = pg
+        job = self._job_for(key)
+        job.podgroup = pg
+        job._minav = None      # minMember may have changed
 
     def _on_queue(self, ev) -> None:
         q: Queue = ev.obj
