@@ -97,12 +97,20 @@ class PreemptAction:
             constraints = predicates.class_constraints(tc, job) if predicates \
                 else (-1, np.zeros(max(nt.labels.words, 1), dtype=np.int64),
                       np.zeros(max(nt.labels.words, 1), dtype=np.int64))
+
+            # K5 tensor pass: global victim filter once per
+            # class, per-node evictable capacity as an extra-credit plane,
+            # ONE fused score_cap over all nodes → ranked candidate nodes.
+            scan = self._candidate_scan(ssn, job, tc, req_vec, constraints,
+                                        same_queue, victim_filter)
+            if scan is None:
+                continue
+            order, victims_by_node = scan
             for task in tc.tasks:
                 if placed >= still_needed:
                     break
-                node = self._preempt_one(ssn, stmt, job, task, tc, req_vec,
-                                         constraints, same_queue,
-                                         victim_filter)
+                node = self._preempt_one(ssn, stmt, job, task, req_vec,
+                                         order, victims_by_node)
                 if node is not None:
                     placed += 1
 
@@ -111,60 +119,105 @@ class PreemptAction:
         else:
             stmt.discard()
 
-    def _preempt_one(self, ssn, stmt: Statement, job: JobInfo,
-                     task: TaskInfo, tc, req_vec, constraints,
-                     same_queue: bool, victim_filter) -> Optional[str]:
+    def _candidate_scan(self, ssn, job, tc, req_vec, constraints,
+                        same_queue, victim_filter):
+        """One fused feasibility pass for the whole class: evictable
+        resources per node become the kernel's future-credit plane."""
+        import torch
+        from ...ops import reference as ref
         nt = ssn.node_tensors
-        best: Optional[NodeInfo] = None
-        best_victims: List[TaskInfo] = []
+        tol, require, forbid = constraints
+        rep = tc.tasks[0]
 
+        # global victim set (tier-intersected once, not per node)
+        candidates = []
         for ni in ssn.nodes.values():
-            if not _class_feasible_on_node(ssn, constraints, task, ni):
+            if not ni.ready:
                 continue
+            for t in ni.tasks.values():
+                if t.status not in self.victim_statuses:
+                    continue
+                vjob = ssn.jobs.get(t.job_key)
+                if vjob is None:
+                    continue
+                if same_queue and vjob.queue != job.queue:
+                    continue
+                candidates.append(t)
+        victims = victim_filter(rep, candidates) if victim_filter \
+            else ssn.preemptable(rep, candidates)
+
+        victims_by_node = {}
+        evictable = np.zeros((nt.n, nt.r), dtype=np.float32)
+        for v in victims:
+            ni = ssn.nodes.get(v.node_name)
+            if ni is None:
+                continue
+            victims_by_node.setdefault(ni.name, []).append(v)
+            vv = nt.req_vector(v)
+            if vv is not None:
+                evictable[ni.node_id] += vv
+        if not victims_by_node:
+            return None
+        for vs in victims_by_node.values():
+            vs.sort(key=lambda v: victim_sort_key(ssn, v))
+
+        dev = nt.alloc_t.device
+        nt.ensure_plane_width()
+        extra = nt.extra_t + torch.from_numpy(evictable.T).to(dev)
+        W = nt.planes_t.shape[0]
+        require = np.pad(require, (0, max(0, W - len(require))))
+        forbid = np.pad(forbid, (0, max(0, W - len(forbid))))
+        score = torch.empty(nt.n, device=dev)
+        cap = torch.empty(nt.n, dtype=torch.int32, device=dev)
+        ref.score_cap(nt.alloc_t.t(), nt.used_t.t(), extra.t(),
+                      nt.ready.bool(), nt.taint_mask, nt.planes_t.t(),
+                      torch.from_numpy(req_vec).to(dev), tol,
+                      torch.from_numpy(require).to(dev),
+                      torch.from_numpy(forbid).to(dev),
+                      1.0, 0.0, 0.0, torch.ones(nt.r, device=dev), None,
+                      score, cap)
+        feasible = (score > float("-inf")).cpu().numpy()
+        order_idx = torch.argsort(score, descending=True).cpu().numpy()
+        nodes_sorted = sorted(ssn.nodes.values(), key=lambda n: n.name)
+        order = [nodes_sorted[i] for i in order_idx if feasible[i]]
+        return order, victims_by_node
+
+    def _preempt_one(self, ssn, stmt: Statement, job: JobInfo,
+                     task: TaskInfo, req_vec, order, victims_by_node
+                     ) -> Optional[str]:
+        """Place one preemptor task: walk the kernel-ranked candidate
+        nodes, evict just-enough of the node's (pre-sorted) victims."""
+        nt = ssn.node_tensors
+
+        for ni in order:
             if _node_fits(nt, ni, req_vec):
-                # free room already (released by a previous eviction)
-                best, best_victims = ni, []
-                break
-            # candidate victims on this node
-            cands = [t for t in ni.tasks.values()
-                     if t.status in self.victim_statuses]
-            if same_queue:
-                cands = [t for t in cands
-                         if ssn.jobs.get(t.job_key) is not None
-                         and ssn.jobs[t.job_key].queue == job.queue]
-            if victim_filter is not None:
-                cands = victim_filter(task, cands)
-            else:
-                cands = ssn.preemptable(task, cands)
-            if not cands:
-                continue
-            cands.sort(key=lambda v: victim_sort_key(ssn, v))
-            # evict-just-enough simulation
+                stmt.pipeline(task, ni.name)
+                return ni.name
+            cands = victims_by_node.get(ni.name, [])
+            # evict-just-enough simulation over future idle
             chosen = []
             fi = ni.future_idle
             avail = {k: fi.get(k) for k in nt.dims.index}
+
             def fits():
                 for name, i in nt.dims.index.items():
                     if req_vec[i] > 0.1 and avail[name] + 0.1 < req_vec[i]:
                         return False
                 return True
+
             for v in cands:
                 if fits():
                     break
+                if v.status == TaskStatus.RELEASING:
+                    continue           # already evicted for an earlier task
                 chosen.append(v)
                 for name, val in v.request.q.items():
                     if name in avail:
                         avail[name] += val
             if not fits():
                 continue
-            if best is None or len(chosen) < len(best_victims):
-                best, best_victims = ni, chosen
-                if not chosen:
-                    break
-
-        if best is None:
-            return None
-        for v in best_victims:
-            stmt.evict(v)
-        stmt.pipeline(task, best.name)
-        return best.name
+            for v in chosen:
+                stmt.evict(v)
+            stmt.pipeline(task, ni.name)
+            return ni.name
+        return None
