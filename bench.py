@@ -39,27 +39,49 @@ GI = 1024 ** 3
 
 def build_cluster(cache: SchedulerCache, n_nodes: int, n_jobs: int,
                   pods_per_job: int, rank: int, world: int,
-                  node_rank: int = None, node_world: int = None):
+                  node_rank: int = None, node_world: int = None,
+                  mix: bool = False):
     """Synthetic inventory: jobs sharded round-robin by rank; nodes by
-    (node_rank, node_world) — full replication in soft mode."""
+    (node_rank, node_world) — full replication in soft mode.
+
+    ``mix``: heterogeneous inventory (varied gang sizes/requests/
+    priorities across 4 weighted queues, zoned nodes) — defeats the
+    uniform-gang fast paths on purpose."""
+    import random
+    rng = random.Random(12345)
     node_rank = rank if node_rank is None else node_rank
     node_world = world if node_world is None else node_world
     for i in range(node_rank, n_nodes, node_world):
+        labels = {"zone": f"z{i % 4}"} if mix else None
         node = synth.make_node(f"node-{i:06d}", cpu_milli=32000,
-                               mem=256 * GI, pods=256)
+                               mem=256 * GI, pods=256, labels=labels)
         cache.add_node_info(NodeInfo(node))
-    qi = QueueInfo(synth.make_queue("default", weight=1))
-    cache.add_queue_info(qi)
+    queues = ["default"] if not mix else ["default", "batch", "svc", "ml"]
+    for qn in queues:
+        cache.add_queue_info(QueueInfo(synth.make_queue(
+            qn, weight=1 if qn == "default" else rng.randint(1, 4))))
     jobs = []
     for j in range(rank, n_jobs, world):
         name = f"job-{j:06d}"
-        pg = synth.make_podgroup(name, min_member=pods_per_job,
-                                 min_resources={CPU: 1000.0 * pods_per_job,
-                                                MEMORY: float(GI) * pods_per_job})
+        if mix:
+            size = rng.choice([1, 2, 4, 8, 16])
+            cpu = float(rng.choice([250, 500, 1000, 2000, 4000]))
+            mem = float(rng.choice([1, 2, 4])) * GI
+            queue = rng.choice(queues)
+            prio = rng.randint(0, 10)
+            selector = {"zone": f"z{rng.randrange(4)}"} \
+                if rng.random() < 0.2 else None
+        else:
+            size, cpu, mem = pods_per_job, 1000.0, float(GI)
+            queue, prio, selector = "default", 0, None
+        pg = synth.make_podgroup(name, queue=queue, min_member=size,
+                                 min_resources={CPU: cpu * size,
+                                                MEMORY: mem * size})
         job = JobInfo(f"default/{name}", pg)
-        for p in range(pods_per_job):
-            pod = synth.make_pod(f"{name}-w-{p}", name, cpu_milli=1000,
-                                 mem=GI, role="worker")
+        for p in range(size):
+            pod = synth.make_pod(f"{name}-w-{p}", name, queue=queue,
+                                 cpu_milli=cpu, mem=mem, role="worker",
+                                 priority=prio, node_selector=selector)
             job.add_task(TaskInfo.from_pod(pod, job.key))
         cache.add_job_info(job)
         jobs.append(job)
@@ -101,6 +123,9 @@ def main():
     ap.add_argument("--shard-mode", choices=["hard", "soft"], default="hard",
                     help="hard: nodes+jobs sharded (conflict-free); "
                          "soft: nodes replicated, delta all-reduce")
+    ap.add_argument("--mix", action="store_true",
+                    help="heterogeneous inventory (varied gangs/queues/"
+                         "priorities/selectors)")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -139,8 +164,17 @@ def main():
     node_world = 1 if (dist and args.shard_mode == "soft") else world
     node_rank = 0 if (dist and args.shard_mode == "soft") else rank
     jobs = build_cluster(cache, args.nodes, args.jobs, args.pods_per_job,
-                         rank, world, node_rank, node_world)
-    total_pods = args.jobs * args.pods_per_job   # whole-job, all ranks
+                         rank, world, node_rank, node_world, mix=args.mix)
+    if args.mix:
+        my_pods = sum(len(j.tasks) for j in jobs)
+        if dist:
+            tp = torch.tensor([float(my_pods)])
+            torch.distributed.all_reduce(tp)
+            total_pods = int(tp.item())
+        else:
+            total_pods = my_pods
+    else:
+        total_pods = args.jobs * args.pods_per_job   # whole-job, all ranks
 
     soft = dist and args.shard_mode == "soft"
 
@@ -235,6 +269,7 @@ def main():
                 "nodes": args.nodes,
                 "jobs": args.jobs,
                 "pods_per_job": args.pods_per_job,
+                "mix": args.mix,
                 "decision_plane": "hip-gfx950" if use_gpu else "torch-cpu-oracle",
                 "p99_cycle_ms": round(p99 * 1000.0, 2),
             },

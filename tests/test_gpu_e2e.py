@@ -77,3 +77,33 @@ def test_gpu_full_stack_churn():
         assert abs(dev_used[ni.node_id] - ni.used.get(CPU)) < 1.0
     # work flowed through: most submitted gangs completed and were GC'd
     assert store.count("Job") < submitted
+
+
+def test_gpu_priority_preemption():
+    """Preempt path on device tensors: fused candidate scan + eviction."""
+    from volcano_amd.scheduler import (FakeBinder, Scheduler, SchedulerCache,
+                                       default_config)
+    from volcano_amd.store import ObjectStore
+    from volcano_amd.utils import synth
+
+    store = ObjectStore()
+    binder = FakeBinder()
+    for n in synth.make_nodes(4, cpu_milli=2000, mem=8 * GI):
+        store.create("Node", n)
+    store.create("Queue", synth.make_queue("default"))
+    config = default_config()
+    config.use_hip = True
+    config.device = "cuda"
+    config.actions = ["enqueue", "allocate", "preempt", "backfill"]
+    cache = SchedulerCache(store=store, binder=binder, device="cuda")
+    sched = Scheduler(cache, config)
+
+    synth.make_gang(store, "low", replicas=8, min_member=1, cpu_milli=1000,
+                    mem=GI, priority=1)
+    sched.run_once()
+    assert len(binder.binds) == 8
+    synth.make_gang(store, "high", replicas=3, cpu_milli=1000, mem=GI,
+                    priority=100)
+    sched.run_once()
+    assert len(binder.evictions) == 3
+    assert cache.jobs["default/high"].waiting_count == 3
