@@ -44,3 +44,37 @@ def test_metrics_export_reference_names():
     assert "e2e_scheduling_latency_p99" in text
     assert "action_scheduling_latency:allocate_mean" in text
     assert "schedule_attempts_total 1.0" in text
+
+
+def test_prometheus_metrics_source():
+    """PrometheusSource against a stub /api/v1/query endpoint."""
+    import json as _json
+    import threading
+    from http.server import BaseHTTPRequestHandler, HTTPServer
+
+    from volcano_amd.scheduler.metrics_source import PrometheusSource
+
+    class Handler(BaseHTTPRequestHandler):
+        def do_GET(self):
+            val = "77.5" if "cpu" in self.path else "33.0"
+            body = _json.dumps({"data": {"result": [
+                {"value": [0, val]}]}}).encode()
+            self.send_response(200)
+            self.send_header("Content-Length", str(len(body)))
+            self.end_headers()
+            self.wfile.write(body)
+
+        def log_message(self, *a):
+            pass
+
+    srv = HTTPServer(("127.0.0.1", 0), Handler)
+    t = threading.Thread(target=srv.serve_forever, daemon=True)
+    t.start()
+    try:
+        src = PrometheusSource(f"http://127.0.0.1:{srv.server_address[1]}")
+        nodes = [synth.make_node("n1")]
+        usage = src.node_usage(nodes)
+        assert usage["n1"]["cpu"] == 77.5
+        assert usage["n1"]["memory"] == 33.0
+    finally:
+        srv.shutdown()

@@ -34,13 +34,24 @@ class UsagePlugin(Plugin):
             else float(self.args.get("usage.memory-threshold", 80))
         weight = float(self.args.get("usage.weight", 5))
 
+        # pluggable usage source (reference metrics/source MetricsClient):
+        # annotations by default, prometheus via arguments
+        from ..metrics_source import new_source
+        kind = self.args.get("usage.metrics-source", "annotation")
+        src_kw = {}
+        if kind == "prometheus":
+            src_kw["address"] = self.args.get("usage.prometheus-address",
+                                              "http://127.0.0.1:9090")
+        source = new_source(kind, **src_kw)
+        usage = source.node_usage([ni.node for ni in ssn.nodes.values()])
+
         N = nt.n
         cpu = np.zeros(N, dtype=np.float32)
         mem = np.zeros(N, dtype=np.float32)
         for ni in ssn.nodes.values():
-            ann = ni.node.meta.annotations
-            cpu[ni.node_id] = float(ann.get(ANN_CPU_USAGE, 0.0))
-            mem[ni.node_id] = float(ann.get(ANN_MEM_USAGE, 0.0))
+            u = usage.get(ni.name, {})
+            cpu[ni.node_id] = u.get("cpu", 0.0)
+            mem[ni.node_id] = u.get("memory", 0.0)
 
         over = np.nonzero((cpu > cpu_thresh) | (mem > mem_thresh))[0]
         if len(over):
