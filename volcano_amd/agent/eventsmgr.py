@@ -10,7 +10,7 @@ the cadvisor/metriccollect analog), handlers react.
 from __future__ import annotations
 
 from dataclasses import dataclass
-from typing import Dict, List, Optional
+from typing import List, Optional
 
 from ..store import ObjectStore
 

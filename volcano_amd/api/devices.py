@@ -19,7 +19,7 @@ fit one 288 GiB card even though 320 GiB aggregate might.
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, List, Optional, Tuple
 
 ANN_GPU_NUMBER = "volcano.sh/gpu-number"

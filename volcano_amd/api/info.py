@@ -14,10 +14,10 @@ import hashlib
 import json
 from dataclasses import dataclass, field
 from functools import cached_property
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Optional
 
 from .objects import (ANN_PREEMPTABLE, LBL_TASK_SPEC, Pod, Node, PodGroup,
-                      Queue, Toleration)
+                      Queue)
 from .resource import Resource
 from .types import ALLOCATED_STATUSES, PodGroupPhase, TaskStatus
 

@@ -11,8 +11,6 @@ from __future__ import annotations
 
 import argparse
 import sys
-from typing import Optional
-
 import yaml
 
 from ..api.objects import (Command, Job, JobSpec, ObjectMeta, Queue,

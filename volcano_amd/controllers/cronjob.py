@@ -7,7 +7,7 @@ from __future__ import annotations
 import time
 from typing import List
 
-from ..api.objects import Job, JobSpec, ObjectMeta
+from ..api.objects import Job, ObjectMeta
 from ..api.types import JobPhase
 from ..utils.cron import CronSchedule
 from .framework import Controller, register

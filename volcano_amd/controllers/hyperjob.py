@@ -7,7 +7,6 @@ from __future__ import annotations
 from ..api.objects import HyperJob, Job, ObjectMeta
 from ..api.types import JobPhase
 from .framework import Controller, register
-from .garbagecollector import FINISHED
 
 
 @register("hyperjob")

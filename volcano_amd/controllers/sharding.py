@@ -7,7 +7,7 @@ inUse) so two schedulers never own one node simultaneously
 
 from __future__ import annotations
 
-from typing import Dict, List
+from typing import List
 
 from ..api.objects import NodeShard, ObjectMeta
 from ..parallel.sharding import ShardingPolicy

@@ -9,8 +9,7 @@ are the CPU oracle, not a GPU execution path).
 from __future__ import annotations
 
 import ctypes
-from ctypes import (POINTER, Structure, c_float, c_int, c_int32, c_int64,
-                    c_uint8, c_void_p)
+from ctypes import (Structure, c_float, c_int, c_int32, c_int64, c_void_p)
 from typing import Optional
 
 import torch

@@ -27,7 +27,7 @@ Decision determinism: score ties break toward the lower node index
 
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Optional
 
 import torch
 

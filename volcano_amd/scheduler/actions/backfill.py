@@ -12,7 +12,7 @@ from typing import List
 
 import numpy as np
 
-from ...api.types import PodGroupPhase, TaskStatus
+from ...api.types import PodGroupPhase
 from ..plan import ClassPlan, CyclePlan, run_plan_hip, run_plan_torch
 
 

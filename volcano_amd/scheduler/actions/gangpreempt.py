@@ -8,7 +8,7 @@ the most resource for the fewest evictions / lowest priority first).
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Tuple
+from typing import List, Optional
 
 import numpy as np
 

@@ -15,7 +15,7 @@ packed dense vectors, not per-pod Resource maps.
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional
+from typing import Optional
 
 import numpy as np
 

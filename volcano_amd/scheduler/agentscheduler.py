@@ -17,9 +17,8 @@ from __future__ import annotations
 
 import queue
 import threading
-from typing import List, Optional
+from typing import Optional
 
-import numpy as np
 import torch
 
 from ..api.info import NodeInfo, TaskInfo

@@ -16,8 +16,6 @@ from __future__ import annotations
 import threading
 from typing import Dict, List, Optional
 
-import torch
-
 from ..api.info import JobInfo, NodeInfo, QueueInfo, TaskInfo
 from ..api.objects import ObjectMeta, PodGroup, Queue
 from ..api.resource import ResourceDims

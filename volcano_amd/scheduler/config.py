@@ -4,7 +4,7 @@ actions/tiers format of ``util.go:38-51 DefaultSchedulerConf``)."""
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 import yaml
 

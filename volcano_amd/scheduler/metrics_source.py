@@ -10,7 +10,7 @@ matching the reference's query shape).
 
 from __future__ import annotations
 
-from typing import Dict, Optional
+from typing import Optional
 
 ANN_CPU_USAGE = "volcano.sh/cpu-usage"
 ANN_MEM_USAGE = "volcano.sh/memory-usage"

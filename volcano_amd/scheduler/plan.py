@@ -17,7 +17,7 @@ descriptors — and the whole cycle executes as one enqueue:
 from __future__ import annotations
 
 import ctypes
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, List, Optional, Tuple
 
 import numpy as np

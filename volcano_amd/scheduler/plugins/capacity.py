@@ -21,7 +21,7 @@ import torch
 from ...api.types import PodGroupPhase
 from ...ops import reference as ref
 from ..plan import BIG_LIMIT
-from ..session import ABSTAIN, PERMIT, REJECT
+from ..session import PERMIT, REJECT
 from .base import Plugin, register
 
 

@@ -15,7 +15,7 @@ order/valid/victim semantics.
 from __future__ import annotations
 
 from ...api.types import TaskStatus
-from ..session import ABSTAIN, PERMIT, REJECT
+from ..session import PERMIT, REJECT
 from .base import Plugin, register
 
 

@@ -23,7 +23,6 @@ from typing import Dict, Optional
 import numpy as np
 
 from ...api.hypernode import HyperNodeTree
-from ...api.types import TaskStatus
 from ..tensors import set_plane_bit
 from .base import Plugin, register
 

@@ -7,7 +7,7 @@ from __future__ import annotations
 import numpy as np
 
 from ...api.types import PodGroupPhase
-from ..session import ABSTAIN, PERMIT, REJECT
+from ..session import PERMIT, REJECT
 from .base import Plugin, register
 
 

@@ -22,7 +22,7 @@ import torch
 
 from ...ops import reference as ref
 from ..plan import BIG_LIMIT
-from ..session import ABSTAIN, PERMIT, REJECT
+from ..session import PERMIT, REJECT
 from .base import Plugin, register
 
 

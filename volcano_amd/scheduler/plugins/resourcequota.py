@@ -6,7 +6,7 @@ from __future__ import annotations
 
 from ...api.resource import Resource
 from ...api.types import PodGroupPhase
-from ..session import ABSTAIN, PERMIT, REJECT
+from ..session import ABSTAIN, REJECT
 from .base import Plugin, register
 
 

@@ -12,8 +12,6 @@ import time
 from typing import Dict, Set
 
 from ...api.objects import LBL_REVOCABLE_ZONE
-from ...api.types import TaskStatus
-from ..session import ABSTAIN, PERMIT, REJECT
 from ..tensors import set_plane_bit
 from .base import Plugin, register
 

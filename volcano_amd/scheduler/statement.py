@@ -13,9 +13,9 @@ and must roll back if the preemptor gang cannot reach its minimum.
 
 from __future__ import annotations
 
-from typing import List, Optional, Tuple
+from typing import List, Tuple
 
-from ..api.info import JobInfo, NodeInfo, TaskInfo
+from ..api.info import TaskInfo
 from ..api.types import TaskStatus
 
 

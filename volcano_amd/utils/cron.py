@@ -7,7 +7,7 @@ Supports: ``*``, lists ``1,2,3``, ranges ``1-5``, steps ``*/15`` and
 from __future__ import annotations
 
 import time
-from typing import List, Optional, Set
+from typing import Optional, Set
 
 _BOUNDS = [(0, 59), (0, 23), (1, 31), (1, 12), (0, 6)]
 

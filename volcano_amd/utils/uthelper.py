@@ -13,8 +13,7 @@ from typing import Dict, List, Optional
 
 from ..api.objects import HyperNode, Node, Pod, PodGroup, Queue
 from ..scheduler import FakeBinder, Scheduler, SchedulerCache
-from ..scheduler.config import (PluginOption, SchedulerConfiguration, Tier,
-                                default_config)
+from ..scheduler.config import PluginOption, Tier, default_config
 from ..store import ObjectStore
 
 

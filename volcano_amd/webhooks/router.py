@@ -11,7 +11,7 @@ same paths.
 from __future__ import annotations
 
 from dataclasses import dataclass
-from typing import Callable, Dict, List, Optional
+from typing import Callable, List, Optional
 
 from ..store import ObjectStore
 
