@@ -190,6 +190,33 @@ def jobflow_list(backend, args):
     print(_fmt_table(rows, ["NAME", "STATE", "STEPS"]))
 
 
+def jobtemplate_list(backend, args):
+    rows = []
+    for t in backend.list("JobTemplate"):
+        rows.append((t.meta.name, len(t.spec.tasks),
+                     t.spec.total_replicas))
+    print(_fmt_table(rows, ["NAME", "TASKS", "REPLICAS"]))
+
+
+def hypernode_list(backend, args):
+    rows = []
+    for h in backend.list("HyperNode"):
+        members = sum(len(m.selector.exact_match) for m in h.members)
+        rows.append((h.meta.name, h.tier, members))
+    print(_fmt_table(rows, ["NAME", "TIER", "MEMBERS"]))
+
+
+def node_list(backend, args):
+    rows = []
+    for n in backend.list("Node"):
+        rows.append((n.meta.name,
+                     f"{n.allocatable.milli_cpu / 1000:g}",
+                     f"{n.allocatable.memory / (1024 ** 3):g}Gi",
+                     "Ready" if n.ready else "NotReady",
+                     "unschedulable" if n.unschedulable else "-"))
+    print(_fmt_table(rows, ["NAME", "CPU", "MEMORY", "STATUS", "TAINT"]))
+
+
 def build_parser() -> argparse.ArgumentParser:
     ap = argparse.ArgumentParser(prog="vcctl",
                                  description="volcano_amd CLI")
@@ -247,6 +274,17 @@ def build_parser() -> argparse.ArgumentParser:
 
     jf = sub.add_parser("jobflow").add_subparsers(dest="cmd", required=True)
     jf.add_parser("list").set_defaults(fn=jobflow_list)
+
+    jt = sub.add_parser("jobtemplate").add_subparsers(dest="cmd",
+                                                      required=True)
+    jt.add_parser("list").set_defaults(fn=jobtemplate_list)
+
+    hn = sub.add_parser("hypernode").add_subparsers(dest="cmd",
+                                                    required=True)
+    hn.add_parser("list").set_defaults(fn=hypernode_list)
+
+    nd = sub.add_parser("node").add_subparsers(dest="cmd", required=True)
+    nd.add_parser("list").set_defaults(fn=node_list)
 
     return ap
 
