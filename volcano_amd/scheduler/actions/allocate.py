@@ -87,9 +87,13 @@ class AllocateAction:
         for q in ssn.sorted_queues(queues):
             if not q.is_open or ssn.queue_overused(q):
                 continue
-            for job in ssn.sorted_jobs(by_queue[q.name]):
-                if not ssn.allocatable(q, job):
-                    continue
+            # Allocatable is queue-scoped (quota gates; the exact per-task
+            # bound is enforced in-kernel) — hoist it out of the job loop
+            jobs_q = by_queue[q.name]
+            gate = ssn.allocatable(q, jobs_q[0]) if jobs_q else True
+            if not gate:
+                continue
+            for job in ssn.sorted_jobs(jobs_q):
                 ordered_jobs.append((q, job))
 
         # Gang bundling: a run of consecutive jobs whose single pending

@@ -30,7 +30,7 @@ from .tensors import NodeTensors
 BIG_LIMIT = 1.0e18     # "no queue limit" sentinel (finite: kernel does int cast)
 
 
-@dataclass
+@dataclass(slots=True)
 class BundleEntry:
     """One gang job inside a fused bundle class (see ClassPlan.bundle)."""
     job_key: str
@@ -72,7 +72,7 @@ class ClassPlan:
             else self.tclass.count
 
 
-@dataclass
+@dataclass(slots=True)
 class JobPlan:
     job_key: str
     class_begin: int
