@@ -203,3 +203,81 @@ def test_cycle_runner_matches_torch_plan(hip):
     assert len(cpu_binds) > 0
     assert cpu_binds == gpu_binds, (
         f"CPU and GPU cycles disagree: {len(cpu_binds)} vs {len(gpu_binds)} binds")
+
+
+def test_cycle_equivalence_feature_rich():
+    """CPU vs GPU whole-cycle decisions on an inventory exercising taints,
+    selectors, multi-role gangs, queues, priorities and anti-affinity."""
+    from volcano_amd.api.objects import Taint, Toleration
+    from volcano_amd.scheduler import FakeBinder, Scheduler, SchedulerCache, \
+        default_config
+    from volcano_amd.scheduler.config import PluginOption
+    from volcano_amd.store import ObjectStore
+    from volcano_amd.utils import synth
+
+    GI = 1024 ** 3
+
+    def build(device, use_hip):
+        store = ObjectStore()
+        rng = np.random.RandomState(99)
+        for i in range(300):
+            labels = {}
+            taints = []
+            if i % 3 == 0:
+                labels["zone"] = f"z{i % 2}"
+            if i % 17 == 0:
+                taints.append(Taint("dedicated", "infra", "NoSchedule"))
+            store.create("Node", synth.make_node(
+                f"n-{i:04d}", cpu_milli=float(rng.choice([8000, 16000])),
+                mem=float(rng.choice([16, 32])) * GI,
+                labels=labels, taints=taints))
+        store.create("Queue", synth.make_queue("qa", weight=3))
+        store.create("Queue", synth.make_queue("qb", weight=1))
+        rng2 = np.random.RandomState(5)
+        for j in range(40):
+            kind = j % 4
+            q = "qa" if j % 2 == 0 else "qb"
+            if kind == 0:       # plain gang
+                synth.make_gang(store, f"g{j:03d}", replicas=int(rng2.randint(1, 9)),
+                                queue=q, cpu_milli=1000, mem=GI,
+                                priority=int(rng2.randint(0, 5)))
+            elif kind == 1:     # selector gang
+                synth.make_gang(store, f"g{j:03d}", replicas=3, queue=q,
+                                cpu_milli=500, mem=GI,
+                                node_selector={"zone": f"z{j % 2}"})
+            elif kind == 2:     # tolerant gang
+                synth.make_gang(store, f"g{j:03d}", replicas=2, queue=q,
+                                cpu_milli=2000, mem=GI,
+                                tolerations=[Toleration(
+                                    key="dedicated", value="infra",
+                                    effect="NoSchedule")])
+            else:               # multi-role + anti-affinity
+                pg = synth.make_podgroup(f"g{j:03d}", queue=q, min_member=3,
+                                         min_task_member={"ps": 1,
+                                                          "worker": 2})
+                store.create("PodGroup", pg)
+                ps = synth.make_pod(f"g{j:03d}-ps-0", f"g{j:03d}", queue=q,
+                                    role="ps", cpu_milli=1000, mem=GI)
+                ps.affinity = {"podAntiAffinity": {"group": f"aa{j}"}}
+                store.create("Pod", ps)
+                for w in range(2):
+                    store.create("Pod", synth.make_pod(
+                        f"g{j:03d}-w-{w}", f"g{j:03d}", queue=q,
+                        role="worker", cpu_milli=500, mem=GI))
+        config = default_config()
+        config.use_hip = use_hip
+        config.device = device
+        config.tiers[1].plugins.append(PluginOption("interpodaffinity"))
+        binder = FakeBinder()
+        cache = SchedulerCache(store=store, binder=binder, device=device)
+        sched = Scheduler(cache, config)
+        sched.run_once()
+        sched.run_once()        # second cycle over mutated state
+        return binder.binds
+
+    cpu = build("cpu", False)
+    gpu = build("cuda", True)
+    assert len(cpu) > 50
+    assert cpu == gpu, (
+        f"decision divergence: {len(cpu)} vs {len(gpu)}; "
+        f"diff={set(cpu.items()) ^ set(gpu.items())}")
