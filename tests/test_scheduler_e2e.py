@@ -265,3 +265,21 @@ def test_bundle_slot_recycling_quota_starved_head():
     assert not any(k.startswith("default/head") for k in binder.binds)
     # staged usage fully unwound for the failed gang
     assert sum(ni.used.milli_cpu for ni in cache.nodes.values()) == 1000
+
+
+def test_multi_value_node_affinity_in():
+    store, binder, cache, sched = mk()
+    for name, zone in [("a", "z1"), ("b", "z2"), ("c", "z3")]:
+        store.create("Node", synth.make_node(name, cpu_milli=4000,
+                                             mem=8 * GI,
+                                             labels={"zone": zone}))
+    store.create("Queue", synth.make_queue("default"))
+    pg = synth.make_podgroup("mv", min_member=2)
+    store.create("PodGroup", pg)
+    for i in range(2):
+        pod = synth.make_pod(f"mv-w-{i}", "mv", cpu_milli=1000, mem=GI)
+        pod.affinity = {"in": {"zone": ["z1", "z3"]}}   # OR over values
+        store.create("Pod", pod)
+    sched.run_once()
+    assert len(binder.binds) == 2
+    assert set(binder.binds.values()) <= {"a", "c"}

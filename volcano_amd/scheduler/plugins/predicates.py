@@ -49,10 +49,41 @@ class PredicatesPlugin(Plugin):
         pod = t.pod
         tolerations = pod.tolerations if pod else []
         tolerated = nt.tolerated_mask(tolerations)
+        # register OR-set bits BEFORE the word arrays are sized so the
+        # widths line up
+        extra_bits = self._multi_value_in_bits(pod)
         require, forbid = nt.selector_bits(
             pod.node_selector if pod else {}, pod.affinity if pod else None)
+        if extra_bits:
+            from ..tensors import set_plane_bit
+            for bit in extra_bits:
+                set_plane_bit(require, bit)
         for hook in hooks:
             hook(tclass, job, require, forbid)
         out = (tolerated, require, forbid)
         self._memo[key] = out
         return out
+
+    def _multi_value_in_bits(self, pod):
+        """Multi-value nodeAffinity ``In`` lists are an OR over label
+        values — not expressible as an AND-mask directly, so each matching
+        node SET becomes one dynamic require bit (memoized)."""
+        aff = pod.affinity if pod is not None else None
+        if not isinstance(aff, dict):
+            return ()
+        nt = self._nt
+        bits = []
+        for k, vals in (aff.get("in") or {}).items():
+            if len(vals) <= 1:
+                continue       # single-value handled by selector_bits
+            memo_key = ("in", k, tuple(sorted(vals)))
+            bit = self._memo.get(memo_key)
+            if bit is None:
+                vset = set(vals)
+                ids = [ni.node_id for ni in self._ssn.nodes.values()
+                       if ni.node.meta.labels.get(k) in vset]
+                bit = nt.add_dynamic_bit(
+                    f"selin:{k}:{','.join(sorted(vals))}", ids)
+                self._memo[memo_key] = bit
+            bits.append(bit)
+        return bits
