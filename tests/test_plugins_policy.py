@@ -203,3 +203,33 @@ def test_resource_strategy_fit_directions():
     synth.make_gang(store, "probe", replicas=1, cpu_milli=100, mem=GI)
     sched.run_once()
     assert binder.binds["default/probe-worker-0"] == other
+
+
+def test_topology_spread_across_zones():
+    tiers = tiers_with(("topologyspread", {}))
+    store, binder, cache, sched = mk(tiers=tiers)
+    for i, zone in enumerate(["z1", "z1", "z2", "z2"]):
+        store.create("Node", synth.make_node(
+            f"n{i}", cpu_milli=8000, mem=32 * GI, labels={"zone": zone}))
+    store.create("Queue", synth.make_queue("default"))
+
+    def spread_pod(name, pg):
+        p = synth.make_pod(name, pg, cpu_milli=500, mem=GI)
+        p.affinity = {"topologySpread": {"group": "web", "topologyKey":
+                                         "zone", "maxSkew": 1}}
+        return p
+
+    zones = {"n0": "z1", "n1": "z1", "n2": "z2", "n3": "z2"}
+    placed_zones = []
+    # one pod per cycle (exact enforcement granularity)
+    for i in range(4):
+        pg = synth.make_podgroup(f"sp{i}", min_member=1)
+        store.create("PodGroup", pg)
+        store.create("Pod", spread_pod(f"sp{i}-w-0", f"sp{i}"))
+        sched.run_once()
+        node = binder.binds.get(f"default/sp{i}-w-0")
+        assert node is not None
+        placed_zones.append(zones[node])
+    # with maxSkew 1 the four pods alternate: 2 in each zone
+    assert placed_zones.count("z1") == 2
+    assert placed_zones.count("z2") == 2

@@ -10,7 +10,7 @@ from . import (binpack, capacity, cdp, conformance, deviceshare, drf,
                extender, gang, interpodaffinity, network_topology_aware,
                nodegroup, nodeorder, numaaware, overcommit, pdb, predicates,
                priority, proportion, rescheduling, resource_strategy_fit,
-               resourcequota, sla, task_topology, tdm,
+               resourcequota, sla, task_topology, tdm, topologyspread,
                usage)  # noqa: F401 (side-effect registration)
 
 
