@@ -259,6 +259,9 @@ def main():
             "higher_is_better": True,
             "scaling": "strong",
             "vs_baseline": round(value / 40.0, 2),
+            # second half of the BASELINE metric ("pods scheduled/sec +
+            # p99 scheduling-cycle latency")
+            "p99_cycle_ms": round(p99 * 1000.0, 2),
             "dtype": "fp32",
             "data": "synthetic",
             "config": {
