@@ -229,8 +229,12 @@ def main():
 
     value = bound / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
-    p99 = sorted(cycle_times)[max(0, int(len(cycle_times) * 0.99) - 1)] \
-        if cycle_times else 0.0
+    if cycle_times:
+        import math
+        n = len(cycle_times)
+        p99 = sorted(cycle_times)[min(n - 1, math.ceil(n * 0.99) - 1)]
+    else:
+        p99 = 0.0
 
     if args.timing and rank == 0:
         import sys
