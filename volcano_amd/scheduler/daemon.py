@@ -69,6 +69,12 @@ def main(argv=None) -> int:
     if device == "auto":
         device = "cuda" if torch.cuda.is_available() else "cpu"
 
+    # long-lived daemon with a large object graph: raise the cyclic-GC
+    # thresholds so gen-2 scans don't land inside scheduling cycles
+    # (docs/tuning.md; the bench freezes its permanent inventory outright)
+    import gc
+    gc.set_threshold(100000, 50, 50)
+
     try:
         store = ObjectStore.load(args.state)
     except FileNotFoundError:
