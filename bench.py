@@ -203,6 +203,8 @@ def main():
     import gc
     gc.collect()
     gc.freeze()
+    if os.environ.get("VAMD_GC_DISABLE"):
+        gc.disable()
 
     for _ in range(args.warmup):
         n = step()
