@@ -42,6 +42,20 @@ class Scheduler:
     def close_session(self, ssn: Session) -> None:
         for plugin in ssn.plugins:
             plugin.on_session_close(ssn)
+        # plugin closures reference the session and live in its registries
+        # — drop them so sessions are reclaimed by refcount alone (no
+        # reliance on the cyclic GC in the long-running daemon)
+        for name in ("job_order_fns", "queue_order_fns", "task_order_fns",
+                     "job_order_keys", "queue_order_keys", "job_valid_fns",
+                     "job_ready_fns", "job_pipelined_fns",
+                     "job_enqueueable_fns", "job_starving_fns",
+                     "overused_fns", "allocatable_fns", "preemptable_fns",
+                     "reclaimable_fns", "victim_tasks_fns",
+                     "victim_filter_fns", "event_handlers",
+                     "class_constraint_hooks", "plugins"):
+            getattr(ssn, name, []).clear()
+        if hasattr(ssn, "job_enqueued_fns"):
+            ssn.job_enqueued_fns.clear()
 
     # -- cycle ---------------------------------------------------------------
     def run_once(self) -> Session:
@@ -56,11 +70,25 @@ class Scheduler:
         METRICS.observe("e2e_scheduling_latency", time.perf_counter() - t0)
         return ssn
 
+    def _idle(self) -> bool:
+        """Nothing to do: no store events since last cycle and no pending
+        or pipelined work anywhere (daemon fast path; the reference runs
+        a full snapshot every second regardless)."""
+        if self.cache.sync() > 0:
+            return False
+        from ..api.types import TaskStatus
+        for job in self.cache.jobs.values():
+            idx = job.task_status_index
+            if idx.get(TaskStatus.PENDING) or idx.get(TaskStatus.PIPELINED):
+                return False
+        return True
+
     def run(self, period: Optional[float] = None) -> None:
         period = period if period is not None else self.config.schedule_period
         while not self._stop.is_set():
             start = time.perf_counter()
-            self.run_once()
+            if not self._idle():
+                self.run_once()
             elapsed = time.perf_counter() - start
             self._stop.wait(max(0.0, period - elapsed))
 
