@@ -287,3 +287,57 @@ def test_hyperjob_expands_and_aggregates():
     cm2.sync_until_quiet()
     hj = store.get("HyperJob", "default", "hj")
     assert hj.status["state"] == "Completed"
+
+
+def test_elastic_scale_up_down():
+    store, cm, sched, kubelet = mk_world()
+    store.create("Job", mk_job("el", replicas=2, min_available=1))
+    cm.sync_until_quiet()
+    sched.run_once()
+    kubelet.tick()
+    cm.sync_until_quiet()
+    assert store.count("Pod") == 2
+    # scale up
+    job = store.get("Job", "default", "el")
+    job.spec.tasks[0].replicas = 5
+    store.update("Job", job)
+    cm.sync_until_quiet()
+    assert store.count("Pod") == 5
+    sched.run_once()
+    assert sum(1 for p in store.list("Pod") if p.node_name) == 5
+    # scale down
+    job = store.get("Job", "default", "el")
+    job.spec.tasks[0].replicas = 3
+    store.update("Job", job)
+    cm.sync_until_quiet()
+    assert store.count("Pod") == 3
+
+
+def test_queue_close_drains_then_closes():
+    store, cm, sched, kubelet = mk_world()
+    store.create("Job", mk_job("qc", replicas=1,
+                               ttl_seconds_after_finished=0.0))
+    cm.sync_until_quiet()
+    store.create("Command", Command(meta=ObjectMeta(name="close-default"),
+                                    action="CloseQueue", target_kind="Queue",
+                                    target_name="default"))
+    cm.sync_until_quiet()
+    q = store.get("Queue", "default", "default")
+    assert q.status.state == "Closing"       # podgroup still exists
+    # closed queues admit nothing new
+    sched.run_once()
+    assert all(not p.node_name for p in store.list("Pod"))
+    # drain: job finishes and is GC'd → queue transitions Closed
+    job = store.get("Job", "default", "qc")
+    job.status.phase = "Completed"
+    store.update("Job", job)
+    cm.sync_until_quiet()
+    q = store.get("Queue", "default", "default")
+    assert q.status.state == "Closed"
+    # reopen works
+    store.create("Command", Command(meta=ObjectMeta(name="open-default"),
+                                    action="OpenQueue", target_kind="Queue",
+                                    target_name="default"))
+    cm.sync_until_quiet()
+    assert store.get("Queue", "default",
+                     "default").status.state == "Open"

@@ -32,7 +32,13 @@ class QueueController(Controller):
             q = self.store.get("Queue", "default", cmd.target_name)
             if q is not None:
                 if cmd.action == Action.CLOSE_QUEUE.value:
-                    q.status.state = QueueState.CLOSED.value
+                    # Closing while podgroups remain; Closed once drained
+                    # (reference queue/state machine factory.go:39-52)
+                    active = self.store.list(
+                        "PodGroup",
+                        selector=lambda g: g.spec.queue == q.meta.name)
+                    q.status.state = QueueState.CLOSING.value if active \
+                        else QueueState.CLOSED.value
                 elif cmd.action == Action.OPEN_QUEUE.value:
                     q.status.state = QueueState.OPEN.value
                 self.store.update("Queue", q)
@@ -53,5 +59,10 @@ class QueueController(Controller):
                 1 for g in pgs if g.status.phase == PodGroupPhase.RUNNING.value)
             q.status.inqueue = sum(
                 1 for g in pgs if g.status.phase == PodGroupPhase.INQUEUE.value)
-            if (q.status.pending, q.status.running, q.status.inqueue) != old:
+            changed = (q.status.pending, q.status.running,
+                       q.status.inqueue) != old
+            if q.status.state == QueueState.CLOSING.value and not pgs:
+                q.status.state = QueueState.CLOSED.value
+                changed = True
+            if changed:
                 self.store.update("Queue", q)
