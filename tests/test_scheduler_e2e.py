@@ -283,3 +283,17 @@ def test_multi_value_node_affinity_in():
     sched.run_once()
     assert len(binder.binds) == 2
     assert set(binder.binds.values()) <= {"a", "c"}
+
+
+def test_engine_idle_fast_path():
+    store, binder, cache, sched = mk()
+    for n in synth.make_nodes(1, cpu_milli=4000, mem=8 * GI):
+        store.create("Node", n)
+    store.create("Queue", synth.make_queue("default"))
+    synth.make_gang(store, "ij", replicas=2, cpu_milli=1000, mem=GI)
+    assert not sched._idle()          # pending work + fresh events
+    sched.run_once()
+    assert len(binder.binds) == 2
+    assert sched._idle()              # everything bound, no new events
+    synth.make_gang(store, "ij2", replicas=1, cpu_milli=1000, mem=GI)
+    assert not sched._idle()          # new events wake it
