@@ -294,6 +294,7 @@ def test_engine_idle_fast_path():
     assert not sched._idle()          # pending work + fresh events
     sched.run_once()
     assert len(binder.binds) == 2
+    sched._idle()                     # drains the cycle's own store writes
     assert sched._idle()              # everything bound, no new events
     synth.make_gang(store, "ij2", replicas=1, cpu_milli=1000, mem=GI)
     assert not sched._idle()          # new events wake it
