@@ -31,6 +31,9 @@ def main(argv=None) -> int:
     ap.add_argument("--period", type=float, default=0.5)
     ap.add_argument("--conf", default=None)
     ap.add_argument("--device", default="auto")
+    ap.add_argument("--agents", action="store_true",
+                    help="run node agents (oversubscription/QoS/eviction) "
+                         "for every node in the store")
     args = ap.parse_args(argv)
 
     import torch
@@ -57,6 +60,33 @@ def main(argv=None) -> int:
     cache = SchedulerCache(store=store, device=device)
     sched = Scheduler(cache, config)
 
+    agent_stop = None
+    if args.agents:
+        import threading
+
+        from .agent import (CpuQosHandler, EventsManager, EvictionHandler,
+                            MemoryQosHandler, OversubscriptionHandler)
+
+        agent_stop = threading.Event()
+
+        def agents_loop():
+            managers = {}
+            while not agent_stop.is_set():
+                for node in store.list("Node"):
+                    name = node.meta.name
+                    if name not in managers:
+                        m = EventsManager(store, name)
+                        m.register(OversubscriptionHandler())
+                        m.register(EvictionHandler())
+                        m.register(CpuQosHandler())
+                        m.register(MemoryQosHandler())
+                        managers[name] = m
+                for m in managers.values():
+                    m.tick()
+                agent_stop.wait(5.0)
+
+        threading.Thread(target=agents_loop, daemon=True).start()
+
     stopping = []
 
     def on_term(sig, frame):
@@ -70,6 +100,8 @@ def main(argv=None) -> int:
     try:
         sched.run(period=args.period)
     finally:
+        if agent_stop is not None:
+            agent_stop.set()
         cm.stop()
         store.save(args.state)
         print("state saved", flush=True)
