@@ -167,3 +167,22 @@ def test_cache_dumper(tmp_path):
     statuses = {t["status"] for t in
                 data["jobs"]["default/dj"]["tasks"].values()}
     assert statuses == {"BOUND"}
+
+
+def test_agent_scheduler_capacity_race():
+    """Two workers, one slot: the serialized commit section must admit
+    exactly one of the racing pods (optimistic conflict → unschedulable)."""
+    store = ObjectStore()
+    store.create("Node", synth.make_node("solo", cpu_milli=1000, mem=4 * GI))
+    binder = FakeBinder()
+    asched = AgentScheduler(store, binder=binder, workers=2)
+    for i in range(2):
+        p = synth.make_pod(f"race-{i}", podgroup="", cpu_milli=1000, mem=GI)
+        p.meta.annotations.pop("scheduling.volcano.sh/group-name", None)
+        store.create("Pod", p)
+    bound = asched.run_once()
+    assert bound == 1
+    assert len(binder.binds) == 1
+    # node not oversubscribed
+    ni = asched.cache.nodes["solo"]
+    assert ni.used.milli_cpu <= 1000.5
