@@ -194,3 +194,36 @@ def test_interpod_affinity_colocates():
     store.create("Pod", fol)
     sched.run_once()
     assert binder.binds["default/fol-w-0"] == anchor_node
+
+
+def test_dynamic_bits_not_stale_across_cycles():
+    """Regression: per-session node sets (device-unfit etc.) must not
+    linger once the condition clears (pod deletes don't repack planes)."""
+    store, binder, cache, sched = mk(("deviceshare", {}))
+    gpu = synth.make_node("gpu", cpu_milli=8000, mem=32 * GI)
+    gpu.meta.annotations["volcano.sh/gpu-count"] = "1"
+    gpu.meta.annotations["volcano.sh/gpu-memory-per-card"] = "1000"
+    store.create("Node", gpu)
+    store.create("Queue", synth.make_queue("default"))
+
+    def dev_pod(name, pg):
+        p = synth.make_pod(name, pg, cpu_milli=500, mem=GI)
+        p.meta.annotations["volcano.sh/gpu-number"] = "1"
+        p.meta.annotations["volcano.sh/gpu-memory"] = "700"
+        return p
+
+    pg = synth.make_podgroup("d1", min_member=1)
+    store.create("PodGroup", pg)
+    store.create("Pod", dev_pod("d1-w-0", "d1"))
+    sched.run_once()
+    assert binder.binds["default/d1-w-0"] == "gpu"
+
+    pg2 = synth.make_podgroup("d2", min_member=1)
+    store.create("PodGroup", pg2)
+    store.create("Pod", dev_pod("d2-w-0", "d2"))
+    sched.run_once()
+    assert "default/d2-w-0" not in binder.binds     # card full
+
+    store.delete("Pod", "default", "d1-w-0")        # releases the slices
+    sched.run_once()
+    assert binder.binds.get("default/d2-w-0") == "gpu"

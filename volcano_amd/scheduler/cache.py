@@ -229,6 +229,8 @@ class SchedulerCache:
             self.queues["default"] = QueueInfo(
                 Queue(meta=ObjectMeta(name="default")))
         self.ensure_packed()
+        # dynamic node-set bits are per-session; plugins re-project them
+        self.node_tensors.clear_dynamic_bits()
         ssn.jobs = self.jobs
         ssn.nodes = self.nodes
         ssn.queues = self.queues

@@ -262,6 +262,24 @@ class NodeTensors:
             out[i] = v
         return out
 
+    def clear_dynamic_bits(self) -> None:
+        """Zero every dynamic bit across all nodes.  Called at session
+        open: dynamic sets (device-unfit, usage-over, hypernode domains,
+        …) are session state — plugins re-project them each cycle, and a
+        stale membership from the previous cycle must not linger (a full
+        pack clears them implicitly; incremental cycles do not)."""
+        if self.planes_t is None or not getattr(self, "_dynamic_bits", None):
+            return
+        masks: Dict[int, int] = {}
+        for bit in self._dynamic_bits:
+            masks[bit // 64] = masks.get(bit // 64, 0) | (1 << (bit % 64))
+        for w, m in masks.items():
+            if w >= self.planes_np.shape[0]:
+                continue
+            inv = np.int64(_to_signed64(~m))
+            self.planes_np[w] &= inv
+            self.planes_t[w] &= int(inv)
+
     def add_dynamic_bit(self, name: str, node_ids) -> int:
         """Register (or reuse) a synthetic label bit and set it for the
         given node ids — plugins project arbitrary node sets (nodegroup
@@ -269,6 +287,9 @@ class NodeTensors:
         bit planes the predicate kernel already tests, so a new node-set
         predicate costs zero extra kernel work.  Returns the bit index."""
         bit = self.labels.bit(f"dyn:{name}")
+        if not hasattr(self, "_dynamic_bits"):
+            self._dynamic_bits = set()
+        self._dynamic_bits.add(bit)
         w, b = bit // 64, bit % 64
         val = np.int64(_to_signed64(1 << b))
         if self.planes_t is None:
