@@ -29,7 +29,26 @@ class EnqueueAction:
             if not q.is_open:
                 continue
             fifo = q.queue.spec.dequeue_strategy == "fifo"
-            for job in ssn.sorted_jobs(by_queue[q.name]):
+            jobs_q = by_queue[q.name]
+            # bulk fast path: when every enqueueable plugin can admit the
+            # whole batch (total demand fits), skip the per-job votes —
+            # identical outcome for monotone-sum admission
+            bulk = ssn.job_enqueueable_bulk_fns
+            if bulk and len(bulk) == len(ssn.job_enqueueable_fns):
+                # each fn returns a commit thunk iff the whole batch fits;
+                # commit only once EVERY plugin agreed (no partial updates)
+                commits = [fn(q.name, jobs_q) for fn in bulk]
+                if all(c is not None for c in commits):
+                    for c in commits:
+                        c()
+                    for job in jobs_q:
+                        if job.podgroup is not None:
+                            job.podgroup.status.phase = \
+                                PodGroupPhase.INQUEUE.value
+                        ssn.cache.update_podgroup(job)
+                        self._lift_queue_gates(ssn, job)
+                    continue
+            for job in ssn.sorted_jobs(jobs_q):
                 if ssn.job_enqueueable(job):
                     if job.podgroup is not None:
                         job.podgroup.status.phase = PodGroupPhase.INQUEUE.value

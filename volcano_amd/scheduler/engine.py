@@ -48,7 +48,8 @@ class Scheduler:
         for name in ("job_order_fns", "queue_order_fns", "task_order_fns",
                      "job_order_keys", "queue_order_keys", "job_valid_fns",
                      "job_ready_fns", "job_pipelined_fns",
-                     "job_enqueueable_fns", "job_starving_fns",
+                     "job_enqueueable_fns", "job_enqueueable_bulk_fns",
+                     "job_starving_fns",
                      "overused_fns", "allocatable_fns", "preemptable_fns",
                      "reclaimable_fns", "victim_tasks_fns",
                      "victim_filter_fns", "event_handlers",

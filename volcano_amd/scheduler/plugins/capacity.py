@@ -191,9 +191,24 @@ class CapacityPlugin(Plugin):
         ssn.add_queue_order_fn(
             queue_order,
             key=lambda q: (-q.priority, q_share(qi_of[q.name])))
+        def job_enqueueable_bulk(qname, jobs):
+            qi = qi_of.get(qname)
+            if qi is None:
+                return None
+            demand = np.zeros(R, dtype=np.float64)
+            for j in jobs:
+                demand += j.minres_vec(nt)
+            head = (alloc_np[qi][mask].astype(np.float64)
+                    + inqueue_np[qi][mask] + demand[mask])
+            des = des_np[qi][mask].astype(np.float64)
+            if bool((head <= des + 0.1 + 1e-6 * des).all()):
+                return lambda: inqueue_np[qi].__iadd__(demand)
+            return None
+
         ssn.overused_fns.append(overused)
         ssn.allocatable_fns.append(allocatable)
         ssn.job_enqueueable_fns.append(job_enqueueable)
+        ssn.job_enqueueable_bulk_fns.append(job_enqueueable_bulk)
         ssn.job_enqueued_fns = getattr(ssn, "job_enqueued_fns", [])
         ssn.job_enqueued_fns.append(job_enqueued)
         ssn.reclaimable_fns.append(reclaimable)

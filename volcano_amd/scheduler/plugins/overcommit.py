@@ -36,6 +36,19 @@ class OvercommitPlugin(Plugin):
         def job_enqueued(job) -> None:
             inqueue[:] += job.minres_vec(nt)
 
+        def job_enqueueable_bulk(qname, jobs):
+            """Whole-batch admission: total demand fits <=> every prefix
+            fits (monotone sum), so admitting all == per-job votes.
+            Returns a commit thunk (run only when all plugins agree)."""
+            demand = np.zeros(nt.r, dtype=np.float64)
+            for j in jobs:
+                demand += j.minres_vec(nt)
+            head = inqueue[mask] + demand[mask]
+            if bool((head <= total[mask] + 0.1 + 1e-6 * total[mask]).all()):
+                return lambda: inqueue.__iadd__(demand)
+            return None
+
         ssn.job_enqueueable_fns.append(job_enqueueable)
+        ssn.job_enqueueable_bulk_fns.append(job_enqueueable_bulk)
         ssn.job_enqueued_fns = getattr(ssn, "job_enqueued_fns", [])
         ssn.job_enqueued_fns.append(job_enqueued)

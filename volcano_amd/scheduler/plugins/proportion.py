@@ -134,11 +134,29 @@ class ProportionPlugin(Plugin):
                     out.append(v)
             return out
 
+        def job_enqueueable_bulk(qname, jobs):
+            """Whole-queue batch vote: total pending demand under deserved
+            <=> every sequential per-job vote permits (monotone inqueue
+            sum).  Thunk commits the accounting once all plugins agree."""
+            qi = ssn.queue_index.get(qname)
+            if qi is None:
+                return None
+            demand = np.zeros(R, dtype=np.float64)
+            for j in jobs:
+                demand += j.minres_vec(nt)
+            head = (alloc_np[qi][mask].astype(np.float64)
+                    + inqueue_np[qi][mask] + demand[mask])
+            des = des_np[qi][mask].astype(np.float64)
+            if bool((head <= des + 0.1 + 1e-6 * des).all()):
+                return lambda: inqueue_np[qi].__iadd__(demand)
+            return None
+
         ssn.add_queue_order_fn(
             queue_order, key=lambda q: q_share(ssn.queue_index[q.name]))
         ssn.overused_fns.append(overused)
         ssn.allocatable_fns.append(allocatable)
         ssn.job_enqueueable_fns.append(job_enqueueable)
+        ssn.job_enqueueable_bulk_fns.append(job_enqueueable_bulk)
         ssn.job_enqueued_fns = getattr(ssn, "job_enqueued_fns", [])
         ssn.job_enqueued_fns.append(job_enqueued)
         ssn.reclaimable_fns.append(reclaimable)

@@ -60,6 +60,11 @@ class Session:
         self.job_ready_fns: List[Callable[[JobInfo], bool]] = []
         self.job_pipelined_fns: List[Callable[[JobInfo], int]] = []
         self.job_enqueueable_fns: List[Callable[[JobInfo], int]] = []
+        # optional bulk counterparts: fn(queue_name, jobs) -> True when the
+        # WHOLE batch can be admitted (and internal accounting updated),
+        # None to fall back to per-job votes.  Exact when admission
+        # depends on monotone accumulated sums (proportion/overcommit).
+        self.job_enqueueable_bulk_fns: List[Callable] = []
         self.job_starving_fns: List[Callable[[JobInfo], bool]] = []
         self.overused_fns: List[Callable[[QueueInfo], bool]] = []
         self.allocatable_fns: List[Callable[[QueueInfo, JobInfo], bool]] = []
