@@ -10,8 +10,6 @@ util/predicate_helper.go:45).
 
 from __future__ import annotations
 
-import hashlib
-import json
 from dataclasses import dataclass, field
 from functools import cached_property
 from typing import Dict, List, Optional
@@ -45,30 +43,34 @@ class TaskInfo:
     def key(self) -> str:
         return f"{self.namespace}/{self.name}"
 
-    def class_signature(self) -> str:
+    def class_signature(self):
         """Tasks with equal signatures are scheduled as one batch.
-        Memoized — constraints are immutable for a task's lifetime."""
+        Memoized — constraints are immutable for a task's lifetime.
+        A plain hashable tuple (not a digest: hashing+json per task was
+        the dominant cold-snapshot cost at 100k pods).  Distinct
+        constraints always yield distinct tuples; `aff` uses repr so two
+        equal-but-differently-ordered dicts may split a batch (safe:
+        batching is an optimization, never a correctness input)."""
         sig = getattr(self, "_sig", None)
         if sig is not None:
             return sig
         p = self.pod
-        sig = {
-            "role": self.role,
-            "req": sorted(self.request.q.items()),
-            "sel": sorted((p.node_selector or {}).items()) if p else [],
-            "tol": [(t.key, t.operator, t.value, t.effect) for t in (p.tolerations if p else [])],
-            "aff": p.affinity if p else None,
-            "prio": self.priority,
-            "gates": sorted(p.scheduling_gates) if p else [],
-            "dev": sorted(k for k in (p.meta.annotations if p else {})
-                          if k.startswith("volcano.sh/gpu")
-                          or k.startswith("volcano.sh/vgpu")
-                          or k == "volcano.sh/numa-topology-policy"),
-        }
-        digest = hashlib.md5(
-            json.dumps(sig, sort_keys=True, default=str).encode()).hexdigest()
-        self._sig = digest
-        return digest
+        sig = (
+            self.role,
+            tuple(sorted(self.request.q.items())),
+            tuple(sorted((p.node_selector or {}).items())) if p else (),
+            tuple((t.key, t.operator, t.value, t.effect)
+                  for t in (p.tolerations if p else [])),
+            repr(p.affinity) if p is not None and p.affinity else "",
+            self.priority,
+            tuple(sorted(p.scheduling_gates)) if p else (),
+            tuple(sorted(k for k in (p.meta.annotations if p else {})
+                         if k.startswith("volcano.sh/gpu")
+                         or k.startswith("volcano.sh/vgpu")
+                         or k == "volcano.sh/numa-topology-policy")),
+        )
+        self._sig = sig
+        return sig
 
     @classmethod
     def from_pod(cls, pod: Pod, job_key: str) -> "TaskInfo":
