@@ -51,39 +51,81 @@ class PredicatesPlugin(Plugin):
         tolerated = nt.tolerated_mask(tolerations)
         # register OR-set bits BEFORE the word arrays are sized so the
         # widths line up
-        extra_bits = self._multi_value_in_bits(pod)
+        extra_req, extra_forbid = self._affinity_bits(pod)
         require, forbid = nt.selector_bits(
             pod.node_selector if pod else {}, pod.affinity if pod else None)
-        if extra_bits:
+        if extra_req or extra_forbid:
             from ..tensors import set_plane_bit
-            for bit in extra_bits:
+            for bit in extra_req:
                 set_plane_bit(require, bit)
+            for bit in extra_forbid:
+                set_plane_bit(forbid, bit)
         for hook in hooks:
             hook(tclass, job, require, forbid)
         out = (tolerated, require, forbid)
         self._memo[key] = out
         return out
 
-    def _multi_value_in_bits(self, pod):
-        """Multi-value nodeAffinity ``In`` lists are an OR over label
-        values — not expressible as an AND-mask directly, so each matching
-        node SET becomes one dynamic require bit (memoized)."""
+    def _node_set_bit(self, memo_key, name, match):
+        """Memoized dynamic require/forbid bit for the node set where
+        ``match(labels)`` holds."""
+        bit = self._memo.get(memo_key)
+        if bit is None:
+            ids = [ni.node_id for ni in self._ssn.nodes.values()
+                   if match(ni.node.meta.labels)]
+            bit = self._nt.add_dynamic_bit(name, ids)
+            self._memo[memo_key] = bit
+        return bit
+
+    def _affinity_bits(self, pod):
+        """Node-affinity operators beyond exact-match AND: each compiles
+        to a dynamic bit over its matching node SET (memoized per
+        expression).  Covers the k8s NodeSelectorOperator surface
+        (k8s.io/api/core/v1 In/NotIn/Exists/DoesNotExist/Gt/Lt wrapped by
+        reference plugins/predicates/predicates.go nodeaffinity):
+
+        * ``in`` with >1 value — OR over values → one require bit;
+        * ``exists: [key, ...]`` — label present → require bit per key;
+        * ``notExists: [key, ...]`` — label present → forbid bit;
+        * ``gt/lt: {key: num}`` — numeric label compare → require bit.
+        """
         aff = pod.affinity if pod is not None else None
         if not isinstance(aff, dict):
-            return ()
-        nt = self._nt
-        bits = []
+            return (), ()
+        req, forbid = [], []
+
+        def num(x):
+            try:
+                return float(x)
+            except (TypeError, ValueError):
+                return None
+
         for k, vals in (aff.get("in") or {}).items():
             if len(vals) <= 1:
                 continue       # single-value handled by selector_bits
-            memo_key = ("in", k, tuple(sorted(vals)))
-            bit = self._memo.get(memo_key)
-            if bit is None:
-                vset = set(vals)
-                ids = [ni.node_id for ni in self._ssn.nodes.values()
-                       if ni.node.meta.labels.get(k) in vset]
-                bit = nt.add_dynamic_bit(
-                    f"selin:{k}:{','.join(sorted(vals))}", ids)
-                self._memo[memo_key] = bit
-            bits.append(bit)
-        return bits
+            vset = set(vals)
+            req.append(self._node_set_bit(
+                ("in", k, tuple(sorted(vals))),
+                f"selin:{k}:{','.join(sorted(vals))}",
+                lambda lbl, k=k, vset=vset: lbl.get(k) in vset))
+        for k in (aff.get("exists") or []):
+            req.append(self._node_set_bit(
+                ("exists", k), f"selex:{k}",
+                lambda lbl, k=k: k in lbl))
+        for k in (aff.get("notExists") or []):
+            forbid.append(self._node_set_bit(
+                ("exists", k), f"selex:{k}",
+                lambda lbl, k=k: k in lbl))
+        for k, v in (aff.get("gt") or {}).items():
+            vf = float(v)
+            req.append(self._node_set_bit(
+                ("gt", k, vf), f"selgt:{k}:{v}",
+                lambda lbl, k=k, vf=vf: (lambda n: n is not None and n > vf)(
+                    num(lbl.get(k)))))
+        for k, v in (aff.get("lt") or {}).items():
+            vf = float(v)
+            req.append(self._node_set_bit(
+                ("lt", k, vf), f"sellt:{k}:{v}",
+                lambda lbl, k=k, vf=vf: (lambda n: n is not None and n < vf)(
+                    num(lbl.get(k)))))
+        return req, forbid

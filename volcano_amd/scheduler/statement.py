@@ -80,6 +80,57 @@ class Statement:
                     task.status = TaskStatus.PENDING
         self.ops.clear()
 
+    # -- composition / replay (statement.go:433 Merge, :440 SaveOperations,
+    #    :457 RecoverOperations) ----------------------------------------------
+    def merge(self, *stmts: "Statement") -> None:
+        """Absorb other statements' pending ops so they commit or discard
+        with this one; sources are cleared (no double-commit)."""
+        for st in stmts:
+            self.ops.extend(st.ops)
+            st.ops.clear()
+
+    def save_operations(self) -> List[Tuple]:
+        """Detached record of the pending ops (task identity + target),
+        usable to replay the same decisions in a later session via
+        :meth:`recover_operations` — the cross-action replay primitive."""
+        saved = []
+        for op in self.ops:
+            if op[0] == "evict":
+                _, victim, _, reason = op
+                saved.append(("evict", victim.key, reason))
+            elif op[0] == "pipeline":
+                _, task, node_name = op
+                saved.append(("pipeline", task.key, node_name))
+        return saved
+
+    def recover_operations(self, saved: List[Tuple]) -> bool:
+        """Re-apply a saved op list against THIS session's live state.
+        Tasks are looked up by key (a later session rebuilds its infos);
+        missing tasks abort the replay and roll back (all-or-nothing,
+        mirroring the reference's error-out)."""
+        by_key = {}
+        for job in self.ssn.jobs.values():
+            for t in job.tasks.values():
+                by_key[t.key] = t
+        start = len(self.ops)
+        for op in saved:
+            t = by_key.get(op[1])
+            if t is None:
+                # roll back only what this replay added
+                tail = Statement(self.ssn)
+                tail.ops = self.ops[start:]
+                tail.discard()
+                del self.ops[start:]
+                return False
+            if op[0] == "evict":
+                self.evict(t, op[2])
+            elif op[0] == "pipeline":
+                self.pipeline(t, op[2])
+        return True
+
+    def has_evictions(self) -> bool:
+        return any(op[0] == "evict" for op in self.ops)
+
     def commit(self) -> None:
         """Flush evictions to the cache/binder (statement.go:402); pipelined
         tasks hold their reservation for this cycle (see engine notes)."""
