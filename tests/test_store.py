@@ -47,3 +47,32 @@ def test_save_load_roundtrip(tmp_path):
     pod = s2.get("Pod", "default", "p1")
     assert pod.request.milli_cpu == 250
     assert pod.podgroup_name == "pg1"
+
+
+def test_watch_overflow_drops_and_flags():
+    from volcano_amd.utils import synth
+    store = ObjectStore()
+    w = store.watch("Node", replay=False, maxsize=4)
+    for i in range(10):
+        store.create("Node", synth.make_node(f"ovf-{i}"))
+    assert w.overflowed                    # writers never blocked
+    assert len(w.drain()) == 4             # bounded queue kept the head
+    w.stop()
+
+
+def test_cache_resyncs_after_watch_overflow():
+    from volcano_amd.scheduler import FakeBinder, SchedulerCache
+    from volcano_amd.utils import synth
+    store = ObjectStore()
+    cache = SchedulerCache(store=store, binder=FakeBinder())
+    cache._watch.stop()
+    cache._watch = store.watch("Pod", "Node", "PodGroup", "Queue", maxsize=8)
+    for i in range(30):
+        store.create("Node", synth.make_node(f"rs-{i}"))
+    synth.make_gang(store, "g", replicas=2, cpu_milli=100)
+    assert cache._watch.overflowed
+    cache.sync()
+    # the relist rebuilt the complete world despite the dropped events
+    assert len(cache.nodes) == 30
+    assert "default/g" in cache.jobs
+    assert not cache._watch.overflowed

@@ -99,6 +99,8 @@ class SchedulerCache:
         """Drain pending store events into the infos; returns event count."""
         if self._watch is None:
             return 0
+        if self._watch.overflowed:
+            return self._resync()
         evs = self._watch.drain()
         # within a batch, structure before pods (a replayed Pod may refer to
         # a Node/PodGroup whose ADDED event is later in the same drain)
@@ -115,6 +117,22 @@ class SchedulerCache:
                 else:
                     self._used_dirty = True         # usage-only delta
         return len(evs)
+
+    def _resync(self) -> int:
+        """Watch overflowed (events dropped under backpressure): discard
+        the stale stream and rebuild every info from the store's current
+        state — the informer relist path (k8s 410 Gone → re-list)."""
+        with self._lock:
+            self._watch.stop()
+            self.jobs.clear()
+            self.nodes.clear()
+            self.queues.clear()
+            self._task_node.clear()
+            self._task_job.clear()
+            self._task_ref.clear()
+            self._watch = self.store.watch("Pod", "Node", "PodGroup", "Queue")
+            self._tensors_dirty = True
+        return self.sync()
 
     def _job_for(self, pg_key: str) -> JobInfo:
         job = self.jobs.get(pg_key)

@@ -50,10 +50,18 @@ class Watch:
         self.kinds = kinds
         self._q: "queue.Queue[Optional[Event]]" = queue.Queue(maxsize=maxsize)
         self._closed = False
+        # backpressure: a slow consumer must never block writers — when
+        # the queue is full the event is DROPPED and the stream marked
+        # overflowed; the consumer detects it and relists from the store
+        # (k8s "watch too old" → informer resync semantics)
+        self.overflowed = False
 
     def _push(self, ev: Event) -> None:
         if not self._closed and (not self.kinds or ev.kind in self.kinds):
-            self._q.put(ev)
+            try:
+                self._q.put_nowait(ev)
+            except queue.Full:
+                self.overflowed = True
 
     def next(self, timeout: Optional[float] = None) -> Optional[Event]:
         try:
@@ -74,7 +82,10 @@ class Watch:
     def stop(self) -> None:
         self._closed = True
         self._store._remove_watch(self)
-        self._q.put(None)
+        try:
+            self._q.put_nowait(None)    # wake a blocked next(); best-effort
+        except queue.Full:
+            pass                        # consumer has pending events anyway
 
 
 class ObjectStore:
