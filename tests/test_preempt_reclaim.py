@@ -153,3 +153,39 @@ def test_eviction_converges_to_bind():
     high_pod = [p for p in store.list("Pod")
                 if p.meta.name.startswith("high") and p.node_name]
     assert len(high_pod) == 1
+
+
+def test_nominated_node_fast_path_after_preempt():
+    """Preempt pipelines the gang; once victims terminate, the next cycle
+    commits on the nominated nodes WITHOUT re-scoring (allocate fast
+    path, reference allocate.go:797 NominatedNodeName)."""
+    from volcano_amd.utils.metrics import METRICS
+    store = ObjectStore()
+    cache = SchedulerCache(store=store)          # StoreBinder: evict deletes
+    config = default_config()
+    config.actions = PREEMPT
+    sched = Scheduler(cache, config)
+    for n in synth.make_nodes(2, cpu_milli=2000, mem=8 * GI):
+        store.create("Node", n)
+    store.create("Queue", synth.make_queue("default"))
+    synth.make_gang(store, "low", replicas=4, min_member=1, cpu_milli=1000,
+                    mem=GI, priority=1)
+    sched.run_once()
+    synth.make_gang(store, "high", replicas=2, cpu_milli=1000, mem=GI,
+                    priority=100)
+    sched.run_once()                             # evict + pipeline
+    high = cache.jobs["default/high"]
+    assert high.waiting_count == 2
+    nominated = sorted(t.node_name
+                       for t in high.tasks.values() if t.node_name)
+    # evicted pods: StoreBinder marked them Failed; drop them fully so
+    # the capacity is free (kubelet termination analog)
+    for p in list(store.list("Pod")):
+        if p.phase == "Failed":
+            store.delete("Pod", p.meta.namespace, p.meta.name)
+    before = METRICS.counter("allocate:nominated_fastpath")
+    sched.run_once()                             # fast-path commit
+    assert METRICS.counter("allocate:nominated_fastpath") == before + 1
+    placed = sorted(t.node_name for t in high.tasks.values())
+    assert placed == nominated                   # stayed on freed nodes
+    assert high.occupied_count == 2

@@ -37,6 +37,10 @@ class TaskInfo:
     preemptable: bool = False
     revocable_zone: str = ""
     gated: bool = False            # has scheduling gates (k8s SchedulingGates)
+    # node a preemption pipelined this task onto last cycle — the allocate
+    # fast path re-checks fit there before full scoring (reference
+    # NominatedNodeName, allocate.go:797)
+    nominated_node: str = ""
     pod: Optional[Pod] = None
 
     @cached_property

@@ -159,6 +159,15 @@ class AllocateAction:
         for q, job in ordered_jobs:
             qi = ssn.queue_index[q.name]
 
+            # -- nomination fast path (reference allocate.go:797
+            # NominatedNodeName): a gang pipelined by last cycle's preempt
+            # re-checks fit on its nominated nodes and commits host-side,
+            # skipping scoring entirely.  Disabled under soft-sharding
+            # (commits must flow through the conflict reconcile).
+            if self.coordinator is None and \
+                    self._try_nominated(ssn, job, qi, predicates):
+                continue
+
             # -- steady-state fast path: the whole job is one pending class
             atom = job.plan_atom()
             pend = job.task_status_index.get(_TS.PENDING, {})
@@ -252,6 +261,113 @@ class AllocateAction:
         METRICS.observe("allocate:plan_build", t1 - t0)
         METRICS.observe("allocate:plan_run", t2 - t1)
         METRICS.observe("allocate:apply", t3 - t2)
+
+    def _try_nominated(self, ssn, job, qi: int, predicates) -> bool:
+        """Commit a fully-nominated gang onto its nominated nodes if they
+        still fit (capacity + ready + taints + label planes re-checked on
+        the packed mirrors).  Any miss clears the nominations and returns
+        False — the job takes the normal scored path."""
+        import torch
+        from ...api.resource import Resource
+        pend = job.pending_tasks
+        if not pend or not pend[0].nominated_node:
+            return False
+        nt = ssn.node_tensors
+        by_node: Dict[str, List] = {}
+        for t in pend:
+            ni = ssn.nodes.get(t.nominated_node)
+            if t.gated or ni is None or ni.node_id < 0:
+                self._clear_nominations(pend)
+                return False
+            by_node.setdefault(t.nominated_node, []).append(t)
+
+        # constraint planes per class signature (memoized by predicates)
+        sig_cons = {}
+        total_vec = np.zeros(nt.r, dtype=np.float64)
+        for nn, ts in by_node.items():
+            ni = ssn.nodes[nn]
+            nid = ni.node_id
+            if not nt.ready_np[nid]:
+                self._clear_nominations(pend)
+                return False
+            need = Resource()
+            for t in ts:
+                need.add(t.request)
+                if predicates is not None:
+                    sig = t.class_signature()
+                    cons = sig_cons.get(sig)
+                    if cons is None:
+                        tc = TaskClass(signature=sig, role=t.role,
+                                       request=t.request, tasks=[t],
+                                       priority=t.priority)
+                        cons = sig_cons[sig] = \
+                            predicates.class_constraints(tc, job)
+                    tol, require, forbid = cons
+                    if (nt.taints_np[nid] & ~np.int64(tol)) != 0:
+                        self._clear_nominations(pend)
+                        return False
+                    col = nt.planes_np[:, nid]
+                    W = min(len(require), len(col))
+                    if (require[:W] & ~col[:W]).any() or require[W:].any() \
+                            or (forbid[:W] & col[:W]).any():
+                        self._clear_nominations(pend)
+                        return False
+            if not need.less_equal(ni.idle):
+                self._clear_nominations(pend)
+                return False
+            total_vec += nt.resource_vector(need)
+
+        # queue bound (the in-kernel clamp's host equivalent)
+        if ssn.queue_limit is not None:
+            head = ssn.queue_alloc[qi].numpy().astype(np.float64) + total_vec
+            lim = ssn.queue_limit[qi].numpy()
+            if not bool((head <= lim + 1e-3).all()):
+                self._clear_nominations(pend)
+                return False
+
+        # -- commit (host infos + device planes + queue rows) ---------------
+        dev = nt.used_t.device
+        ids, vecs = [], []
+        to_bind = []
+        for nn, ts in by_node.items():
+            ni = ssn.nodes[nn]
+            by_sig: Dict[object, List] = {}
+            for t in ts:
+                t.node_name = nn
+                by_sig.setdefault(t.class_signature(), []).append(t)
+            node_need = Resource()
+            for group in by_sig.values():
+                ni.add_allocated_bulk(group, group[0].request, len(group))
+                g0 = group[0]
+                tc = TaskClass(signature=g0.class_signature(), role=g0.role,
+                               request=g0.request, tasks=group,
+                               priority=g0.priority)
+                ssn.fire_allocate(tc, [ni.node_id], [len(group)], group)
+                for t in group:
+                    node_need.add(t.request)
+            ids.append(ni.node_id)
+            vecs.append(nt.resource_vector(node_need))
+            to_bind.extend(ts)
+        delta = torch.from_numpy(
+            np.stack(vecs).astype(np.float32).T).to(dev)      # [R, k]
+        idx = torch.tensor(ids, dtype=torch.long, device=dev)
+        nt.used_t.index_add_(1, idx, delta)
+        if ssn.queue_alloc is not None:
+            ssn.queue_alloc[qi] += torch.from_numpy(
+                total_vec.astype(np.float32))
+        self._clear_nominations(pend)
+        ssn.cache.bind_tasks(to_bind, by_job={job.key: to_bind})
+        if ssn.job_ready(job) and job.podgroup is not None and \
+                job.phase != PodGroupPhase.RUNNING.value:
+            job.podgroup.status.phase = PodGroupPhase.RUNNING.value
+            ssn.cache.update_podgroup(job)
+        METRICS.inc("allocate:nominated_fastpath")
+        return True
+
+    @staticmethod
+    def _clear_nominations(tasks) -> None:
+        for t in tasks:
+            t.nominated_node = ""
 
     # -- statement commit (host mirror of the device-side state) ------------
     def _apply(self, ssn, plan: CyclePlan, result,

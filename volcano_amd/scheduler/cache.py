@@ -270,6 +270,7 @@ class SchedulerCache:
                 node = self.nodes.get(t.node_name)
                 if node is not None:
                     node.remove_task(t)
+                t.nominated_node = t.node_name   # allocate fast-path hint
                 t.node_name = ""
                 job.update_task_status(t, TaskStatus.PENDING)
             if pipelined:
