@@ -123,3 +123,30 @@ def test_hypernode_controller_label_discovery():
     assert hns["spine-s0"].tier == 2
     tree = HyperNodeTree(list(hns.values()), [f"n{i}" for i in range(4)])
     assert tree.members["spine-s0"] == {"n0", "n1", "n2", "n3"}
+
+
+def test_hypernode_controller_fabric_discovery(tmp_path):
+    """UFM-dump provider: leaf/spine HyperNodes from interface records
+    (reference discovery/ufm/ufm.go buildHyperNodes)."""
+    import json
+    dump = [
+        {"system_name": "lsw1", "tier": 1, "node_description": "n0"},
+        {"system_name": "lsw1", "tier": 1, "node_description": "n1"},
+        {"system_name": "lsw2", "tier": 1, "node_description": "n2"},
+        {"system_name": "sp0", "tier": 2, "peer_node_name": "lsw1"},
+        {"system_name": "sp0", "tier": 2, "peer_node_name": "lsw2"},
+    ]
+    path = tmp_path / "fabric.json"
+    path.write_text(json.dumps(dump))
+    store = ObjectStore()
+    cm = ControllerManager(store, ["hypernode"])
+    for c in cm.controllers:
+        if c.name == "hypernode":
+            c.fabric_file = str(path)
+    cm.sync_until_quiet()
+    hns = {h.meta.name: h for h in store.list("HyperNode")}
+    assert set(hns) == {"leaf-lsw1", "leaf-lsw2", "spine-sp0"}
+    tree = HyperNodeTree(list(hns.values()), ["n0", "n1", "n2"])
+    assert tree.members["spine-sp0"] == {"n0", "n1", "n2"}
+    assert tree.lca_tier("n0", "n1") == 1
+    assert tree.lca_tier("n0", "n2") == 2
