@@ -18,6 +18,7 @@ extern "C" void vamd_run_cycle(
     const float* alloc, float* used, const float* extra,
     const uint8_t* ready, const int64_t* taints, const int64_t* planes,
     const float* bias,
+    const float* bias_rows,
     const float* class_req, const int64_t* class_tol,
     const int64_t* class_require, const int64_t* class_forbid,
     const int32_t* class_min, const float* dim_w,
@@ -36,12 +37,16 @@ extern "C" void vamd_run_cycle(
         for (int c = job.class_begin; c < job.class_end; ++c) {
             const VamdClassDesc& cd = classes[c];
             const float* ext = (cd.flags & 1) ? extra : nullptr;
+            // per-class bias row (task-topology bucket packing) overrides
+            // the plan-wide bias plane
+            const float* b = (cd.bias_row >= 0 && bias_rows)
+                ? bias_rows + (size_t)cd.bias_row * N : bias;
 
             vamd_score_cap(alloc, used, ext, ready, taints, planes,
                            class_req + (size_t)c * R, class_tol[c],
                            class_require + (size_t)c * W,
                            class_forbid + (size_t)c * W,
-                           cd.w_least, cd.w_most, cd.w_bal, dim_w, bias,
+                           cd.w_least, cd.w_most, cd.w_bal, dim_w, b,
                            score_scratch, cap_scratch, N, R, W, stream);
 
             // single-class jobs: gang check fused into the commit.  The

@@ -41,7 +41,7 @@ struct VamdClassDesc {
     int32_t log_off;      // offset into the log arrays
     int32_t log_cap;      // slot capacity (>= min(ntasks, N))
     int32_t flags;        // bit0: use extra (future-idle) credit
-    int32_t _pad;
+    int32_t bias_row;     // row of bias_rows for this class, or -1 → plan bias
     float w_least, w_most, w_bal;  // score weights for this class's queue tier
     float _padf;
 };
@@ -66,6 +66,7 @@ void vamd_run_cycle(
     const float* alloc, float* used, const float* extra,
     const uint8_t* ready, const int64_t* taints, const int64_t* planes,
     const float* bias,
+    const float* bias_rows,        // [B, N] per-class rows (nullable)
     // per-class constraint rows
     const float* class_req,        // [C, R]
     const int64_t* class_tol,      // [C]

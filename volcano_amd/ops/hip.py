@@ -23,7 +23,7 @@ class VamdClassDesc(Structure):
     _fields_ = [
         ("job_idx", c_int32), ("queue_idx", c_int32), ("ntasks", c_int32),
         ("min_needed", c_int32), ("log_off", c_int32), ("log_cap", c_int32),
-        ("flags", c_int32), ("_pad", c_int32),
+        ("flags", c_int32), ("bias_row", c_int32),
         ("w_least", c_float), ("w_most", c_float), ("w_bal", c_float),
         ("_padf", c_float),
     ]
@@ -128,10 +128,10 @@ def cond_revert(flag, log_nodes, log_counts, log_len, req, used_t,
 
 def run_cycle(class_descs: bytes, n_classes: int, job_descs: bytes,
               n_jobs: int, alloc_t, used_t, extra_t, ready, taints, planes_t,
-              bias, class_req, class_tol, class_require, class_forbid,
-              class_min, dim_w, queue_alloc, queue_limit, score_scratch,
-              cap_scratch, log_nodes, log_counts, log_len, class_placed,
-              job_placed, job_flag):
+              bias, bias_rows, class_req, class_tol, class_require,
+              class_forbid, class_min, dim_w, queue_alloc, queue_limit,
+              score_scratch, cap_scratch, log_nodes, log_counts, log_len,
+              class_placed, job_placed, job_flag):
     """One library call = one whole allocate cycle (plan built host-side)."""
     lib = _load()
     R, N = alloc_t.shape
@@ -140,7 +140,7 @@ def run_cycle(class_descs: bytes, n_classes: int, job_descs: bytes,
         class_descs, c_int(n_classes),
         job_descs, c_int(n_jobs),
         _p(alloc_t), _p(used_t), _p(extra_t), _p(ready), _p(taints),
-        _p(planes_t), _p(bias), _p(class_req), _p(class_tol),
+        _p(planes_t), _p(bias), _p(bias_rows), _p(class_req), _p(class_tol),
         _p(class_require), _p(class_forbid), _p(class_min), _p(dim_w),
         _p(queue_alloc), _p(queue_limit), _p(score_scratch), _p(cap_scratch),
         _p(log_nodes), _p(log_counts), _p(log_len), _p(class_placed),

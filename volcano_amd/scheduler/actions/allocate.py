@@ -155,6 +155,15 @@ class AllocateAction:
         w_least = w.get("least", 1.0)
         w_most = w.get("most", 0.0)
         w_bal = w.get("bal", 0.0)
+        bias_fns = ssn.class_bias_fns
+
+        def class_bias(tc, job):
+            out = None
+            for fn in bias_fns:
+                b = fn(tc, job)
+                if b is not None:
+                    out = b if out is None else out + b
+            return out
 
         for q, job in ordered_jobs:
             qi = ssn.queue_index[q.name]
@@ -197,7 +206,13 @@ class AllocateAction:
                                req=req, tolerated=tol, require=require,
                                forbid=forbid, min_needed=gang_min,
                                w_least=w_least, w_most=w_most, w_bal=w_bal)
-                bundle_in(cp, job, qi, gang_min, sig)
+                if bias_fns:
+                    cp.bias = class_bias(tc, job)
+                if cp.bias is not None:     # biased classes never fuse
+                    close_bundle()
+                    plan.add_job(job, [cp])
+                else:
+                    bundle_in(cp, job, qi, gang_min, sig)
                 continue
 
             classes: List[ClassPlan] = []
@@ -216,11 +231,14 @@ class AllocateAction:
                     forbid = np.zeros(W, dtype=np.int64)
                 need = job.min_task_member.get(tc.role, 0)
                 min_needed = max(0, need - job.role_occupied(tc.role))
-                classes.append(ClassPlan(
+                cp = ClassPlan(
                     tclass=tc, job_key=job.key, queue_idx=qi, req=req,
                     tolerated=tol, require=require, forbid=forbid,
                     min_needed=min_needed, w_least=w.get("least", 1.0),
-                    w_most=w.get("most", 0.0), w_bal=w.get("bal", 0.0)))
+                    w_most=w.get("most", 0.0), w_bal=w.get("bal", 0.0))
+                if bias_fns:
+                    cp.bias = class_bias(tc, job)
+                classes.append(cp)
             if skipped and classes and job.min_available > sum(
                     c.tclass.count for c in classes) + job.occupied_count:
                 # gang can never be satisfied this cycle
@@ -229,7 +247,8 @@ class AllocateAction:
                 continue
 
             bundleable = (len(classes) == 1 and job.occupied_count == 0
-                          and classes[0].tclass.count == len(job.tasks))
+                          and classes[0].tclass.count == len(job.tasks)
+                          and classes[0].bias is None)
             if bundleable:
                 cp = classes[0]
                 gang_min = max(job.min_available, cp.min_needed)

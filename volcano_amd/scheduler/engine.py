@@ -53,7 +53,7 @@ class Scheduler:
                      "overused_fns", "allocatable_fns", "preemptable_fns",
                      "reclaimable_fns", "victim_tasks_fns",
                      "victim_filter_fns", "event_handlers",
-                     "class_constraint_hooks", "plugins"):
+                     "class_constraint_hooks", "class_bias_fns", "plugins"):
             getattr(ssn, name, []).clear()
         if hasattr(ssn, "job_enqueued_fns"):
             ssn.job_enqueued_fns.clear()

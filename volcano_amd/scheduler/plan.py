@@ -53,6 +53,10 @@ class ClassPlan:
     w_most: float = 0.0
     w_bal: float = 0.0
     use_future: bool = False   # score against future-idle (pipelining)
+    # optional per-class additive score plane [N] f32 (task-topology
+    # bucket packing); overrides the plan-wide bias for this class —
+    # builders that want both must pre-add plan.bias into the row
+    bias: Optional[np.ndarray] = None
     log_off: int = 0
     log_cap: int = 0
     # instance count override (bundles track the sum here instead of
@@ -121,6 +125,22 @@ class CyclePlan:
         """Assign undo-log slots (one contiguous region per class) and
         normalize constraint widths (the label-bit registry may have grown
         while classes were built)."""
+        # per-class bias rows REPLACE the plan bias in the kernel — fold
+        # the plan-wide plane (score_bias / soft-shard stagger) into them
+        # so biased classes don't lose it.  Shared rows fold once.
+        if self.bias is not None:
+            pb = self.bias.cpu().numpy().astype(np.float32)
+            originals = [cp.bias for cp in self.classes
+                         if cp.bias is not None]     # pin ids during fold
+            folded: Dict[int, np.ndarray] = {}
+            for cp in self.classes:
+                if cp.bias is not None:
+                    nb = folded.get(id(cp.bias))
+                    if nb is None:
+                        nb = folded[id(cp.bias)] = \
+                            (cp.bias + pb).astype(np.float32)
+                    cp.bias = nb
+            del originals
         self.nt.ensure_plane_width()
         W = self.nt.planes_t.shape[0] if self.nt.planes_t is not None \
             else max(self.nt.labels.words, 1)
@@ -182,9 +202,11 @@ def run_plan_torch(plan: CyclePlan) -> CycleResult:
             require = torch.from_numpy(cp.require).to(dev)
             forbid = torch.from_numpy(cp.forbid).to(dev)
             extra = extra_full if cp.use_future else zeros_extra
+            b = plan.bias if cp.bias is None \
+                else torch.from_numpy(cp.bias).to(dev)
             ref.score_cap(alloc, used, extra, nt.ready.bool(), nt.taint_mask,
                           planes, req, cp.tolerated, require, forbid,
-                          cp.w_least, cp.w_most, cp.w_bal, dim_w, plan.bias,
+                          cp.w_least, cp.w_most, cp.w_bal, dim_w, b,
                           score, cap)
             sl = slice(cp.log_off, cp.log_off + cp.log_cap)
             ref.select_commit(score, cap, req, cp.ntasks, used,
@@ -233,6 +255,8 @@ def run_plan_hip(plan: CyclePlan) -> CycleResult:
 
     cds = (hip.VamdClassDesc * C)()
     job_index = {jp.job_key: j for j, jp in enumerate(plan.jobs)}
+    bias_rows_list: List[np.ndarray] = []
+    bias_row_of: Dict[int, int] = {}      # id(array) -> row index
     for c, cp in enumerate(plan.classes):
         d = cds[c]
         d.job_idx = job_index[cp.job_key]
@@ -243,6 +267,14 @@ def run_plan_hip(plan: CyclePlan) -> CycleResult:
         d.log_cap = cp.log_cap
         d.flags = _flag_of(cp)
         d.w_least, d.w_most, d.w_bal = cp.w_least, cp.w_most, cp.w_bal
+        if cp.bias is None:
+            d.bias_row = -1
+        else:
+            row = bias_row_of.get(id(cp.bias))
+            if row is None:
+                row = bias_row_of[id(cp.bias)] = len(bias_rows_list)
+                bias_rows_list.append(cp.bias)
+            d.bias_row = row
     jds = (hip.VamdJobDesc * len(plan.jobs))()
     for j, jp in enumerate(plan.jobs):
         d = jds[j]
@@ -265,6 +297,9 @@ def run_plan_hip(plan: CyclePlan) -> CycleResult:
     q_alloc = plan.queue_alloc.to(dev).contiguous()
     q_limit = plan.queue_limit.to(dev).contiguous()
     bias = plan.bias.to(dev) if plan.bias is not None else None
+    bias_rows = torch.from_numpy(
+        np.stack(bias_rows_list).astype(np.float32)).to(dev).contiguous() \
+        if bias_rows_list else None                                # [B,N]
 
     score = torch.empty(N, dtype=torch.float32, device=dev)
     cap = torch.empty(N, dtype=torch.int32, device=dev)
@@ -279,9 +314,9 @@ def run_plan_hip(plan: CyclePlan) -> CycleResult:
         ctypes.cast(cds, ctypes.c_void_p), C,
         ctypes.cast(jds, ctypes.c_void_p), len(plan.jobs),
         nt.alloc_t, nt.used_t, nt.extra_t, nt.ready, nt.taint_mask,
-        nt.planes_t, bias, class_req, class_tol, class_require, class_forbid,
-        class_min, dim_w, q_alloc, q_limit, score, cap, log_nodes, log_counts,
-        log_len, class_placed, job_placed, job_flag)
+        nt.planes_t, bias, bias_rows, class_req, class_tol, class_require,
+        class_forbid, class_min, dim_w, q_alloc, q_limit, score, cap,
+        log_nodes, log_counts, log_len, class_placed, job_placed, job_flag)
 
     res = _collect(plan, log_nodes.cpu(), log_counts.cpu(), log_len.cpu(),
                    class_placed.cpu(), job_placed.cpu(), job_flag.cpu())

@@ -55,3 +55,73 @@ class TaskTopologyPlugin(Plugin):
         if anti_roles:
             ssn.score_weights["least"] = max(
                 ssn.score_weights.get("least", 0.0), 1.0)
+
+        # -- bucket scoring (reference topology.go:138 calcBucketScore +
+        # :193 NodeOrderFn): classes of affine roles get a per-class score
+        # bias toward nodes already hosting the job's bucket members, and
+        # away from nodes hosting anti-affine peers.  The bias plane rides
+        # the kernel's additive bias input (ClassPlan.bias), so bucket
+        # preference is evaluated for ALL nodes in the same fused pass as
+        # feasibility — no host (task, node) callbacks.
+        import numpy as np
+        from ...api.types import TaskStatus
+        weight = float(self.args.get("task-topology.weight", 10.0))
+        nt = ssn.node_tensors
+        group_memo: Dict[str, tuple] = {}
+
+        def job_groups(job):
+            """(affinity groups, anti groups) for a job: plugin args plus
+            the reference's podgroup annotations
+            (util.go:36 volcano.sh/task-topology-affinity)."""
+            got = group_memo.get(job.key)
+            if got is not None:
+                return got
+            aff_g = [list(g) for g in affinity]
+            anti_g = [list(g) for g in anti]
+            ann = job.podgroup.meta.annotations if job.podgroup else {}
+            a = ann.get("volcano.sh/task-topology-affinity")
+            if a:
+                aff_g += [s.split(",") for s in a.split(";") if s]
+            b = ann.get("volcano.sh/task-topology-anti-affinity")
+            if b:
+                anti_g += [s.split(",") for s in b.split(";") if s]
+            got = group_memo[job.key] = (aff_g, anti_g)
+            return got
+
+        def class_bias(tclass, job):
+            if job is None or nt is None or nt.n == 0:
+                return None
+            aff_g, anti_g = job_groups(job)
+            if not aff_g and not anti_g:
+                return None
+            role = tclass.role
+            partners = set()
+            for g in aff_g:
+                if role in g:
+                    partners.update(g)
+            repel = set()
+            for g in anti_g:
+                if role in g:
+                    repel.update(g)
+            if not partners and not repel:
+                return None
+            bias = None
+            for t in job.tasks.values():
+                if not t.node_name:
+                    continue
+                if not (t.status.occupies_node
+                        or t.status == TaskStatus.PIPELINED):
+                    continue
+                w = (weight if t.role in partners else 0.0) \
+                    - (weight if t.role in repel else 0.0)
+                if w == 0.0:
+                    continue
+                ni = ssn.nodes.get(t.node_name)
+                if ni is None or ni.node_id < 0:
+                    continue
+                if bias is None:
+                    bias = np.zeros(nt.n, dtype=np.float32)
+                bias[ni.node_id] += w
+            return bias
+
+        ssn.class_bias_fns.append(class_bias)

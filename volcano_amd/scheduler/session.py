@@ -80,6 +80,9 @@ class Session:
         self.event_handlers: List[object] = []   # objects w/ allocate/evict hooks
         # fns(tclass, job, require, forbid) mutating per-class plane bits
         self.class_constraint_hooks: List[Callable] = []
+        # fns(tclass, job) -> Optional[np.ndarray [N] f32]: additive
+        # per-class node score bias (task-topology bucket packing)
+        self.class_bias_fns: List[Callable] = []
 
         # tensor-plane configuration contributed by plugins
         self.score_weights = {"least": 1.0, "most": 0.0, "bal": 0.0}
