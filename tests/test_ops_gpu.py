@@ -268,10 +268,19 @@ def test_cycle_equivalence_feature_rich():
         config.use_hip = use_hip
         config.device = device
         config.tiers[1].plugins.append(PluginOption("interpodaffinity"))
+        config.tiers[1].plugins.append(PluginOption(
+            "task-topology", arguments={"affinity": [["ps", "worker"]]}))
         binder = FakeBinder()
         cache = SchedulerCache(store=store, binder=binder, device=device)
         sched = Scheduler(cache, config)
         sched.run_once()
+        # late workers join the multi-role jobs: their cycle-2 classes get
+        # task-topology bucket bias rows (per-class kernel bias path)
+        for j in range(3, 40, 4):
+            store.create("Pod", synth.make_pod(
+                f"g{j:03d}-w-late", f"g{j:03d}",
+                queue="qa" if j % 2 == 0 else "qb",
+                role="worker", cpu_milli=500, mem=GI))
         sched.run_once()        # second cycle over mutated state
         return binder.binds
 
