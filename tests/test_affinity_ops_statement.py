@@ -110,3 +110,37 @@ def test_statement_recover_missing_task_rolls_back():
     # first op must have been rolled back with the failure
     assert task.status != TaskStatus.RELEASING and s.ops == []
     sched.close_session(ssn)
+
+
+# -- volume zone filter ------------------------------------------------------
+
+def test_volume_zone_confines_pod():
+    from volcano_amd.api.objects import (ObjectMeta, PersistentVolume,
+                                         PersistentVolumeClaim, ZONE_LABEL)
+    store, binder, cache, sched = world({
+        "a": {ZONE_LABEL: "z1"}, "b": {ZONE_LABEL: "z2"}, "c": {}})
+    store.create("PersistentVolume", PersistentVolume(
+        meta=ObjectMeta(name="pv1", labels={ZONE_LABEL: "z2"}),
+        capacity=10 * GI))
+    store.create("PersistentVolumeClaim", PersistentVolumeClaim(
+        meta=ObjectMeta(name="data", namespace="default"),
+        volume_name="pv1"))
+    synth.make_gang(store, "vj", replicas=1, cpu_milli=500, mem=GI)
+    pod = store.get("Pod", "default", "vj-worker-0")
+    pod.volumes = ["data"]
+    store.update("Pod", pod)
+    sched.run_once()
+    assert binder.binds.get("default/vj-worker-0") == "b"   # PV's zone
+
+
+def test_unbound_pvc_constrains_nothing():
+    from volcano_amd.api.objects import (ObjectMeta, PersistentVolumeClaim)
+    store, binder, cache, sched = world({"a": {}, "b": {}})
+    store.create("PersistentVolumeClaim", PersistentVolumeClaim(
+        meta=ObjectMeta(name="later", namespace="default")))  # unbound
+    synth.make_gang(store, "uj", replicas=1, cpu_milli=500, mem=GI)
+    pod = store.get("Pod", "default", "uj-worker-0")
+    pod.volumes = ["later"]
+    store.update("Pod", pod)
+    sched.run_once()
+    assert binder.binds.get("default/uj-worker-0") in ("a", "b")

@@ -72,6 +72,7 @@ class TaskInfo:
                          if k.startswith("volcano.sh/gpu")
                          or k.startswith("volcano.sh/vgpu")
                          or k == "volcano.sh/numa-topology-policy")),
+            tuple(sorted(p.volumes)) if p is not None and p.volumes else (),
         )
         self._sig = sig
         return sig

@@ -113,6 +113,7 @@ class Pod:
     affinity: Optional[dict] = None     # simplified nodeAffinity expression tree
     best_effort: bool = field(default=False)
     scheduling_gates: List[str] = field(default_factory=list)
+    volumes: List[str] = field(default_factory=list)   # PVC names (same ns)
 
     def __post_init__(self):
         if self.request.is_empty():
@@ -333,6 +334,37 @@ class NodeShard:
     nodes_to_remove: List[str] = field(default_factory=list)
 
 
+ZONE_LABEL = "topology.kubernetes.io/zone"
+
+
+@dataclass
+class PersistentVolume:
+    """core/v1 PV reduced to the volume-zone filter's needs: a bound PV
+    whose zone label confines its consumers' pods to that zone (k8s
+    volumezone filter, wrapped by reference plugins/predicates)."""
+
+    meta: ObjectMeta = field(default_factory=ObjectMeta)
+    capacity: float = 0.0               # bytes
+    storage_class: str = ""
+    # zone read from meta.labels[ZONE_LABEL] ("" = zone-free volume)
+
+    @property
+    def zone(self) -> str:
+        return self.meta.labels.get(ZONE_LABEL, "")
+
+
+@dataclass
+class PersistentVolumeClaim:
+    """core/v1 PVC: binds a pod's volume to a PV.  Unbound claims
+    (volume_name == "") constrain nothing — WaitForFirstConsumer
+    semantics: the volume binds where the pod lands."""
+
+    meta: ObjectMeta = field(default_factory=ObjectMeta)
+    volume_name: str = ""               # bound PV ("" = unbound)
+    storage_class: str = ""
+    request: float = 0.0                # bytes
+
+
 @dataclass
 class ResourceQuota:
     """core/v1 ResourceQuota as the resourcequota plugin consumes it:
@@ -487,4 +519,6 @@ KINDS = {
     "CronJob": CronJob, "PodDisruptionBudget": PodDisruptionBudget,
     "Numatopology": Numatopology, "ColocationConfig": ColocationConfig,
     "ResourceQuota": ResourceQuota, "HyperJob": HyperJob,
+    "PersistentVolume": PersistentVolume,
+    "PersistentVolumeClaim": PersistentVolumeClaim,
 }

@@ -52,6 +52,7 @@ class PredicatesPlugin(Plugin):
         # register OR-set bits BEFORE the word arrays are sized so the
         # widths line up
         extra_req, extra_forbid = self._affinity_bits(pod)
+        extra_req = tuple(extra_req) + self._volume_zone_bits(pod)
         require, forbid = nt.selector_bits(
             pod.node_selector if pod else {}, pod.affinity if pod else None)
         if extra_req or extra_forbid:
@@ -65,6 +66,39 @@ class PredicatesPlugin(Plugin):
         out = (tolerated, require, forbid)
         self._memo[key] = out
         return out
+
+    def _volume_zone_bits(self, pod) -> tuple:
+        """Volume-zone filter (k8s volumezone, wrapped by reference
+        plugins/predicates): a pod consuming a bound PVC whose PV carries
+        a zone label must land on nodes of that zone.  Unbound claims
+        constrain nothing (WaitForFirstConsumer).  Zones ride the static
+        label bit planes — zero kernel cost."""
+        if pod is None or not pod.volumes:
+            return ()
+        store = getattr(self._ssn.cache, "store", None)
+        if store is None:
+            return ()
+        from ...api.objects import ZONE_LABEL
+        key = ("volzone", pod.meta.namespace, tuple(sorted(pod.volumes)))
+        got = self._memo.get(key)
+        if got is not None:
+            return got
+        pv_index = self._memo.get("_pv_index")
+        if pv_index is None:
+            pv_index = self._memo["_pv_index"] = {
+                pv.meta.name: pv
+                for pv in store.list("PersistentVolume")}
+        bits = []
+        for vname in pod.volumes:
+            pvc = store.get("PersistentVolumeClaim", pod.meta.namespace,
+                            vname)
+            if pvc is None or not pvc.volume_name:
+                continue
+            pv = pv_index.get(pvc.volume_name)
+            if pv is not None and pv.zone:
+                bits.append(self._nt.label_bit(ZONE_LABEL, pv.zone))
+        got = self._memo[key] = tuple(bits)
+        return got
 
     def _node_set_bit(self, memo_key, name, match):
         """Memoized dynamic require/forbid bit for the node set where
