@@ -128,6 +128,9 @@ class TaskClass:
         return len(self.tasks)
 
 
+_OCCUPIED_STATUSES = tuple(ALLOCATED_STATUSES) + (TaskStatus.SUCCEEDED,)
+
+
 class JobInfo:
     """A PodGroup plus its tasks (reference api/job_info.go:399-445)."""
 
@@ -308,10 +311,16 @@ class JobInfo:
     @property
     def occupied_count(self) -> int:
         """Tasks holding or promised resources (reference ReadyTaskNum:
-        Bound+Binding+Running+Allocated+Succeeded)."""
+        Bound+Binding+Running+Allocated+Succeeded).  Unrolled — this is
+        the hottest job property (gang checks, ordering keys)."""
         idx = self.task_status_index
-        n = sum(len(idx.get(s, ())) for s in ALLOCATED_STATUSES)
-        return n + len(idx.get(TaskStatus.SUCCEEDED, ()))
+        g = idx.get
+        n = 0
+        for s in _OCCUPIED_STATUSES:
+            b = g(s)
+            if b:
+                n += len(b)
+        return n
 
     @property
     def waiting_count(self) -> int:

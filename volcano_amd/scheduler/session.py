@@ -241,6 +241,8 @@ class Session:
         self.queue_index = {name: i for i, name in enumerate(sorted(self.queues))}
         alloc = np.zeros((Q, R), dtype=np.float32)
         for job in self.jobs.values():
+            if job.occupied_count == 0:
+                continue      # nothing allocated — skip the vector build
             qi = self.queue_index.get(job.queue)
             if qi is not None:
                 alloc[qi] += job.alloc_vec(nt)
