@@ -11,7 +11,6 @@ util/predicate_helper.go:45).
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from functools import cached_property
 from typing import Dict, List, Optional
 
 from .objects import (ANN_PREEMPTABLE, LBL_TASK_SPEC, Pod, Node, PodGroup,
@@ -20,9 +19,12 @@ from .resource import Resource
 from .types import ALLOCATED_STATUSES, PodGroupPhase, TaskStatus
 
 
-@dataclass
+@dataclass(slots=True)
 class TaskInfo:
-    """One schedulable pod (reference api/job_info.go:118-175)."""
+    """One schedulable pod (reference api/job_info.go:118-175).
+    Slots: at 100k+ live tasks the per-instance dict is the dominant
+    host-memory cost and attribute access is the hottest interpreter op
+    in the commit path."""
 
     uid: str
     name: str
@@ -42,10 +44,18 @@ class TaskInfo:
     # NominatedNodeName, allocate.go:797)
     nominated_node: str = ""
     pod: Optional[Pod] = None
+    # lazy caches (slots-compatible; excluded from init/repr/compare)
+    _key: Optional[str] = field(default=None, init=False, repr=False,
+                                compare=False)
+    _sig: Optional[tuple] = field(default=None, init=False, repr=False,
+                                  compare=False)
 
-    @cached_property
+    @property
     def key(self) -> str:
-        return f"{self.namespace}/{self.name}"
+        k = self._key
+        if k is None:
+            k = self._key = f"{self.namespace}/{self.name}"
+        return k
 
     def class_signature(self):
         """Tasks with equal signatures are scheduled as one batch.
@@ -55,7 +65,7 @@ class TaskInfo:
         constraints always yield distinct tuples; `aff` uses repr so two
         equal-but-differently-ordered dicts may split a batch (safe:
         batching is an optimization, never a correctness input)."""
-        sig = getattr(self, "_sig", None)
+        sig = self._sig
         if sig is not None:
             return sig
         p = self.pod
