@@ -286,11 +286,15 @@ class AllocateAction:
         still fit (capacity + ready + taints + label planes re-checked on
         the packed mirrors).  Any miss clears the nominations and returns
         False — the job takes the normal scored path."""
+        from ...api.types import TaskStatus as _TS
+        pend_idx = job.task_status_index.get(_TS.PENDING)
+        if not pend_idx:
+            return False
+        if not next(iter(pend_idx.values())).nominated_node:
+            return False    # common case: no nomination — zero-alloc exit
         import torch
         from ...api.resource import Resource
-        pend = job.pending_tasks
-        if not pend or not pend[0].nominated_node:
-            return False
+        pend = list(pend_idx.values())
         nt = ssn.node_tensors
         by_node: Dict[str, List] = {}
         for t in pend:
