@@ -26,11 +26,21 @@ def test_full_pipeline_churn_invariants():
     config.tiers[1].plugins.append(PluginOption("pdb"))
     config.tiers[1].plugins.append(
         PluginOption("cdp", arguments={"cdp.cooldown-time": "0s"}))
+    config.tiers[1].plugins.append(PluginOption(
+        "task-topology", arguments={"affinity": [["ps", "worker"]]}))
     sched = Scheduler(cache, config)
 
+    from volcano_amd.api.objects import (ObjectMeta, PersistentVolume,
+                                         PersistentVolumeClaim, ZONE_LABEL)
     for i in range(12):
         store.create("Node", synth.make_node(
-            f"n-{i:03d}", cpu_milli=8000, mem=32 * GI))
+            f"n-{i:03d}", cpu_milli=8000, mem=32 * GI,
+            labels={ZONE_LABEL: f"z{i % 3}"}))
+    store.create("PersistentVolume", PersistentVolume(
+        meta=ObjectMeta(name="pv-z1", labels={ZONE_LABEL: "z1"})))
+    store.create("PersistentVolumeClaim", PersistentVolumeClaim(
+        meta=ObjectMeta(name="claim-z1", namespace="default"),
+        volume_name="pv-z1"))
     store.create("Queue", synth.make_queue("qa", weight=3))
     store.create("Queue", synth.make_queue("qb", weight=1))
 
@@ -42,12 +52,25 @@ def test_full_pipeline_churn_invariants():
             kw = {}
             if rng.random() < 0.4:
                 kw["phase"] = "Pending"
+            roll = rng.random()
             synth.make_gang(store, f"c{jid:04d}",
                             replicas=rng.randint(1, 6),
                             min_member=None if rng.random() < 0.5 else 1,
                             queue=rng.choice(["qa", "qb"]),
                             cpu_milli=rng.choice([500, 1000, 2000]),
-                            mem=GI, priority=rng.randint(0, 20))
+                            mem=GI, priority=rng.randint(0, 20),
+                            role=rng.choice(["ps", "worker", "etl"]))
+            if roll < 0.2:
+                # sprinkle the new predicate surfaces into the churn
+                for p in store.list("Pod"):
+                    if p.meta.name.startswith(f"c{jid:04d}-"):
+                        if roll < 0.07:
+                            p.volumes = ["claim-z1"]
+                        elif roll < 0.14:
+                            p.affinity = {"exists": [ZONE_LABEL]}
+                        else:
+                            p.affinity = {"gt": {"missing-numeric": 5}}
+                        store.update("Pod", p)
         # random completions: delete a random bound pod (simulates finish)
         bound_pods = [p for p in store.list("Pod") if p.node_name]
         for p in rng.sample(bound_pods, min(3, len(bound_pods))):
