@@ -83,3 +83,24 @@ def test_soft_shard_conflict_resolution():
         assert r[6], f"rank {r[0]} over-allocated: {r[5]}"
         assert r[5]["node-00000"] == 2000.0
         assert r[5]["node-00001"] == 2000.0
+
+
+def test_conflict_repulsion_decays():
+    """Lost nodes get a transient stagger penalty that decays (soft-shard
+    anti-storm measure)."""
+    import torch
+    from volcano_amd.parallel.softshard import SoftShardCoordinator
+
+    class NT:
+        n = 8
+        alloc_t = torch.zeros(2, 8)
+    coord = SoftShardCoordinator(rank=0, world=2)
+    base = coord.stagger_bias(NT).clone()
+    coord._note_losses({3, 5})
+    b1 = coord.stagger_bias(NT)
+    assert b1[3] < base[3] and b1[5] < base[5]
+    assert torch.equal(b1[[0, 1, 2, 4, 6, 7]], base[[0, 1, 2, 4, 6, 7]])
+    # decay: after several loss-free reconciles the penalty vanishes
+    for _ in range(40):
+        coord._note_losses(())
+    assert torch.equal(coord.stagger_bias(NT), base)

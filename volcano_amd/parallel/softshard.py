@@ -31,10 +31,20 @@ import torch
 
 
 class SoftShardCoordinator:
+    # conflict-loser repulsion: a node this rank LOST in the reconcile
+    # gets a transient extra penalty (decaying ×0.5/cycle) so adversarial
+    # same-score workloads stop re-colliding on the same nodes cycle
+    # after cycle.  Amplitude stays tie-break scale — real score
+    # differences (and feasibility) always dominate.
+    REPEL = 1e-4
+    REPEL_DECAY = 0.5
+    REPEL_FLOOR = 1e-7
+
     def __init__(self, rank: int, world: int):
         self.rank = rank
         self.world = world
         self._stagger = None
+        self._repel = {}               # node_id -> penalty strength
 
     def stagger_bias(self, nt) -> torch.Tensor:
         """Rank-staggered tie-break: an epsilon-scale score bias that makes
@@ -46,7 +56,26 @@ class SoftShardCoordinator:
             offset = (self.rank * N) // max(self.world, 1)
             order = (torch.arange(N, dtype=torch.float32) - offset) % max(N, 1)
             self._stagger = (-1e-6 * order).to(nt.alloc_t.device)
-        return self._stagger
+        if not self._repel:
+            return self._stagger
+        bias = self._stagger.clone()
+        ids = torch.tensor(list(self._repel), dtype=torch.long,
+                           device=bias.device)
+        pen = torch.tensor(list(self._repel.values()), dtype=torch.float32,
+                           device=bias.device)
+        bias.index_add_(0, ids, -pen)
+        return bias
+
+    def _note_losses(self, lost_ids) -> None:
+        """Decay old repulsion, strengthen freshly-lost nodes."""
+        decayed = {}
+        for nid, s in self._repel.items():
+            s *= self.REPEL_DECAY
+            if s > self.REPEL_FLOOR:
+                decayed[nid] = s
+        for nid in lost_ids:
+            decayed[nid] = decayed.get(nid, 0.0) + self.REPEL
+        self._repel = decayed
 
     # -- step 2+3: gather deltas, find this rank's lost nodes ----------------
     @staticmethod
@@ -73,7 +102,9 @@ class SoftShardCoordinator:
         lost = (~ok[self.rank]) & mine_active
         self._gathered_sum = stack.sum(dim=0)
         ids = torch.nonzero(lost, as_tuple=False).flatten()
-        return frozenset(int(i) for i in ids.cpu())
+        out = frozenset(int(i) for i in ids.cpu())
+        self._note_losses(out)
+        return out
 
     # -- step 5: establish the global used state -----------------------------
     def finalize(self, ssn, nt, used_before: torch.Tensor) -> None:
