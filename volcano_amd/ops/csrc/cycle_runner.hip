@@ -26,6 +26,7 @@ extern "C" void vamd_run_cycle(
     float* score_scratch, int* cap_scratch,
     int* log_nodes, int* log_counts, int* log_len,
     int* class_placed, int* job_placed, uint8_t* job_flag,
+    unsigned* sort_scratch,
     int N, int R, int W, hipStream_t stream)
 {
     (void)n_classes;
@@ -66,7 +67,8 @@ extern "C" void vamd_run_cycle(
                                queue_limit + (size_t)cd.queue_idx * R,
                                log_nodes + cd.log_off, log_counts + cd.log_off,
                                log_len + c, class_placed + c, job_placed + j,
-                               fuse_min, N, R, cd.log_cap, stream);
+                               fuse_min, sort_scratch, N, R, cd.log_cap,
+                               stream);
         }
 
         if (!single) {

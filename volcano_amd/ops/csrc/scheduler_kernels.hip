@@ -318,6 +318,247 @@ __global__ void cond_revert_kernel(
     for (int e = threadIdx.x; e < len; e += blockDim.x) log_counts[e] = 0;
 }
 
+// ---------------------------------------------------------------------------
+// K2/K4 bulk select for mega-classes (gang bundles): the serial argmax
+// loop above costs one block-reduce + owner rescan PER CONSUMED NODE
+// (~10k sequential steps for a 100k-task bundle).  Scores are static
+// within a class, so the fill equals "stable-sort nodes by score desc,
+// take caps until the budget" — exactly the torch oracle
+// (reference.py select_commit: argsort + cumsum + clamp).  This kernel
+// computes that directly: a workgroup LSD radix sort (8 passes of 4-bit
+// digits, rank-stable) followed by a chunked block scan that emits the
+// undo log in sorted order.  One workgroup; ~16N global accesses total
+// instead of ~N per consumed node.
+// ---------------------------------------------------------------------------
+#define BS_THREADS 1024
+#define BS_WAVES (BS_THREADS / WAVE)   // 16
+#define BS_DIGITS 16                   // 4-bit LSD
+#define BS_PASSES 8
+
+// monotonic float->uint map, then invert so ASCENDING key == score DESC
+DEVINL unsigned score_key(float s) {
+    unsigned u = __float_as_uint(s);
+    unsigned m = (u >> 31) ? 0xFFFFFFFFu : 0x80000000u;
+    return ~(u ^ m);
+}
+
+__global__ void __launch_bounds__(BS_THREADS)
+bulk_select_kernel(
+    const float* __restrict__ score,       // [N]
+    const int* __restrict__ cap,           // [N]
+    const float* __restrict__ req,         // [R]
+    int ntasks,
+    float* __restrict__ used,              // [R, N] in place
+    float* __restrict__ queue_alloc,       // [R] in place
+    const float* __restrict__ queue_limit, // [R]
+    int* __restrict__ log_nodes,           // [K]
+    int* __restrict__ log_counts,          // [K]
+    int* __restrict__ log_len,             // [1]
+    int* __restrict__ placed,              // [1]
+    int* __restrict__ job_placed,          // [1]
+    int fuse_min,
+    unsigned* __restrict__ scratch,        // [4N]: keys/ids ping-pong
+    int N, int R, int K)
+{
+    __shared__ unsigned s_hist[BS_DIGITS * BS_THREADS];     // 64 KB
+    __shared__ unsigned s_digit_base[BS_DIGITS];
+    __shared__ long long s_budget;
+    __shared__ long long s_carry_cap;      // running csum across chunks
+    __shared__ int s_carry_log;            // running log-entry count
+    __shared__ long long s_wave_ll[BS_WAVES];
+    __shared__ int s_wave_i[BS_WAVES];
+
+    const int tid = threadIdx.x;
+    const int lane = tid & (WAVE - 1);
+    const int wid = tid / WAVE;
+
+    unsigned* keys_a = scratch;
+    unsigned* ids_a = scratch + (size_t)N;
+    unsigned* keys_b = scratch + 2 * (size_t)N;
+    unsigned* ids_b = scratch + 3 * (size_t)N;
+
+    if (tid == 0) {
+        long long quota = BIG_CAP;
+        for (int r = 0; r < R; ++r) {
+            float rq = req[r];
+            if (rq > EPS) {
+                float head = queue_limit[r] - queue_alloc[r];
+                float qq = (head + EPS) / rq;
+                if (qq > (float)BIG_CAP) qq = (float)BIG_CAP;
+                long long q = (long long)floorf(qq);
+                quota = min(quota, max(q, 0ll));
+            }
+        }
+        s_budget = min((long long)ntasks, quota);
+        s_carry_cap = 0;
+        s_carry_log = 0;
+    }
+
+    // -- build (key, id) pairs; array position i = node i (stable base) --
+    for (int i = tid; i < N; i += BS_THREADS) {
+        keys_a[i] = score_key(score[i]);
+        ids_a[i] = (unsigned)i;
+    }
+    __syncthreads();
+
+    // -- 8-pass LSD radix (each thread owns a CONTIGUOUS chunk: rank
+    //    order inside a digit bin is (thread, position) = input order,
+    //    which is what makes the sort stable) -------------------------------
+    const int chunk = (N + BS_THREADS - 1) / BS_THREADS;
+    const int lo = min(tid * chunk, N);
+    const int hi = min(lo + chunk, N);
+
+    for (int pass = 0; pass < BS_PASSES; ++pass) {
+        unsigned* in_k = (pass & 1) ? keys_b : keys_a;
+        unsigned* in_i = (pass & 1) ? ids_b : ids_a;
+        unsigned* out_k = (pass & 1) ? keys_a : keys_b;
+        unsigned* out_i = (pass & 1) ? ids_a : ids_b;
+        const int shift = pass * 4;
+
+        unsigned cnt[BS_DIGITS];
+        for (int d = 0; d < BS_DIGITS; ++d) cnt[d] = 0;
+        for (int i = lo; i < hi; ++i)
+            cnt[(in_k[i] >> shift) & 15u]++;
+        for (int d = 0; d < BS_DIGITS; ++d)
+            s_hist[d * BS_THREADS + tid] = cnt[d];
+        __syncthreads();
+
+        // wave `wid` scans digit row `wid` (inclusive, 16 chunks of 64)
+        {
+            unsigned carry = 0;
+            unsigned* row = s_hist + wid * BS_THREADS;
+            for (int c = 0; c < BS_THREADS / WAVE; ++c) {
+                unsigned v = row[c * WAVE + lane];
+                for (int off = 1; off < WAVE; off <<= 1) {
+                    unsigned o = __shfl_up(v, off);
+                    if (lane >= off) v += o;
+                }
+                v += carry;
+                row[c * WAVE + lane] = v;
+                carry = __shfl(v, WAVE - 1);
+            }
+            if (lane == 0) s_digit_base[wid] = carry;   // digit total
+        }
+        __syncthreads();
+        if (tid == 0) {
+            unsigned run = 0;
+            for (int d = 0; d < BS_DIGITS; ++d) {
+                unsigned t = s_digit_base[d];
+                s_digit_base[d] = run;
+                run += t;
+            }
+        }
+        __syncthreads();
+
+        unsigned off[BS_DIGITS];
+        for (int d = 0; d < BS_DIGITS; ++d)
+            off[d] = s_digit_base[d] + s_hist[d * BS_THREADS + tid] - cnt[d];
+        for (int i = lo; i < hi; ++i) {
+            unsigned k = in_k[i];
+            unsigned d = (k >> shift) & 15u;
+            unsigned p = off[d]++;
+            out_k[p] = k;
+            out_i[p] = in_i[i];
+        }
+        __syncthreads();
+    }
+    // even pass count -> sorted result is back in keys_a / ids_a
+
+    // -- chunked block scan over the sorted order: cumulative caps ->
+    //    per-position take, log compaction, usage commit ---------------------
+    const long long budget = s_budget;
+    const int nchunks = (N + BS_THREADS - 1) / BS_THREADS;
+    long long total_ll = 0;
+
+    for (int c = 0; c < nchunks; ++c) {
+        int j = c * BS_THREADS + tid;
+        long long cj = 0;
+        int node = -1;
+        if (j < N) {
+            node = (int)ids_a[j];
+            float s = score[node];
+            if (s != NEG_INF) cj = (long long)cap[node];
+        }
+        // block inclusive scan of cj (wave scan + cross-wave offsets)
+        long long v = cj;
+        for (int off = 1; off < WAVE; off <<= 1) {
+            long long o = __shfl_up(v, off);
+            if (lane >= off) v += o;
+        }
+        if (lane == WAVE - 1) s_wave_ll[wid] = v;
+        __syncthreads();
+        if (wid == 0) {
+            long long w = (lane < BS_WAVES) ? s_wave_ll[lane] : 0;
+            for (int off = 1; off < BS_WAVES; off <<= 1) {
+                long long o = __shfl_up(w, off);
+                if (lane >= off) w += o;
+            }
+            if (lane < BS_WAVES) s_wave_ll[lane] = w;
+        }
+        __syncthreads();
+        long long csum = v + (wid > 0 ? s_wave_ll[wid - 1] : 0) + s_carry_cap;
+
+        long long over = csum - budget; if (over < 0) over = 0;
+        long long take_ll = cj - over; if (take_ll < 0) take_ll = 0;
+        int take = (int)take_ll;
+        int flag = take > 0 ? 1 : 0;
+
+        // block inclusive scan of flags -> log slot
+        int f = flag;
+        for (int off = 1; off < WAVE; off <<= 1) {
+            int o = __shfl_up(f, off);
+            if (lane >= off) f += o;
+        }
+        if (lane == WAVE - 1) s_wave_i[wid] = f;
+        __syncthreads();
+        if (wid == 0) {
+            int w = (lane < BS_WAVES) ? s_wave_i[lane] : 0;
+            for (int off = 1; off < BS_WAVES; off <<= 1) {
+                int o = __shfl_up(w, off);
+                if (lane >= off) w += o;
+            }
+            if (lane < BS_WAVES) s_wave_i[lane] = w;
+        }
+        __syncthreads();
+        int fsum = f + (wid > 0 ? s_wave_i[wid - 1] : 0) + s_carry_log;
+
+        if (flag && fsum - 1 < K) {
+            log_nodes[fsum - 1] = node;
+            log_counts[fsum - 1] = take;
+        }
+        __syncthreads();
+        if (tid == BS_THREADS - 1) {
+            s_carry_cap = csum;
+            s_carry_log = fsum;
+        }
+        __syncthreads();
+    }
+    total_ll = min(s_carry_cap, budget);
+    int total = (int)total_ll;
+    int m = min(s_carry_log, K);
+    bool revert = (fuse_min >= 0) && (total < fuse_min);
+
+    if (revert) {
+        // gang can't reach its minimum: nothing is applied; the log keeps
+        // zeroed counts (same end state as the serial kernel's discard)
+        for (int e = tid; e < m; e += BS_THREADS) log_counts[e] = 0;
+        if (tid == 0) { *log_len = m; *placed = 0; }
+        return;
+    }
+    // commit: apply the logged takes to the staged usage
+    for (int e = wid; e < m; e += BS_WAVES) {
+        int node = log_nodes[e];
+        int cnt = log_counts[e];
+        if (lane < R) used[(size_t)lane * N + node] += (float)cnt * req[lane];
+    }
+    if (tid < R) queue_alloc[tid] += (float)total * req[tid];
+    if (tid == 0) {
+        *log_len = m;
+        *placed = total;
+        *job_placed += total;
+    }
+}
+
 }  // namespace vamd
 
 // ---------------------------------------------------------------------------
@@ -346,8 +587,19 @@ void vamd_select_commit(
     float* score, const int* cap, const float* req, int ntasks, float* used,
     float* queue_alloc, const float* queue_limit, int* log_nodes,
     int* log_counts, int* log_len, int* placed, int* job_placed, int fuse_min,
-    int N, int R, int K, hipStream_t stream)
+    unsigned* sort_scratch, int N, int R, int K, hipStream_t stream)
 {
+    // mega-classes (gang bundles): sort-based bulk fill — identical
+    // decisions, ~16N total accesses instead of ~N per consumed node.
+    // Small classes: the iterative argmax loop wins (few consumptions).
+    if (sort_scratch != nullptr && ntasks >= 512 && N >= 512) {
+        hipLaunchKernelGGL(vamd::bulk_select_kernel, dim3(1),
+                           dim3(BS_THREADS), 0, stream, score, cap, req,
+                           ntasks, used, queue_alloc, queue_limit, log_nodes,
+                           log_counts, log_len, placed, job_placed, fuse_min,
+                           sort_scratch, N, R, K);
+        return;
+    }
     hipLaunchKernelGGL(vamd::select_commit_kernel, dim3(1), dim3(SC_THREADS),
                        0, stream, score, cap, req, ntasks, used, queue_alloc,
                        queue_limit, log_nodes, log_counts, log_len, placed,

@@ -18,6 +18,7 @@ void vamd_select_commit(
     float* score, const int* cap, const float* req, int ntasks, float* used,
     float* queue_alloc, const float* queue_limit, int* log_nodes,
     int* log_counts, int* log_len, int* placed, int* job_placed, int fuse_min,
+    unsigned* sort_scratch,  // [4N] bulk-fill radix scratch (nullable)
     int N, int R, int K, hipStream_t stream);
 
 void vamd_finalize_job(
@@ -84,6 +85,7 @@ void vamd_run_cycle(
     int* class_placed,             // [C]
     int* job_placed,               // [J]
     uint8_t* job_flag,             // [J]
+    unsigned* sort_scratch,        // [4N] bulk-select radix scratch (nullable)
     int N, int R, int W,
     hipStream_t stream);
 

@@ -93,7 +93,7 @@ def score_cap(alloc_t, used_t, extra_t, ready, taints, planes_t, req,
 
 def select_commit(score, cap, req, ntasks, used_t, queue_alloc_row,
                   queue_limit_row, log_nodes, log_counts, log_len, placed,
-                  job_placed, fuse_min):
+                  job_placed, fuse_min, sort_scratch=None):
     lib = _load()
     N = score.shape[0]
     R = req.shape[0]
@@ -102,7 +102,8 @@ def select_commit(score, cap, req, ntasks, used_t, queue_alloc_row,
         _p(score), _p(cap), _p(req), c_int(int(ntasks)), _p(used_t),
         _p(queue_alloc_row), _p(queue_limit_row), _p(log_nodes),
         _p(log_counts), _p(log_len), _p(placed), _p(job_placed),
-        c_int(int(fuse_min)), c_int(N), c_int(R), c_int(K), _stream())
+        c_int(int(fuse_min)), _p(sort_scratch),
+        c_int(N), c_int(R), c_int(K), _stream())
 
 
 def finalize_job(job_placed, occupied, min_available, class_placed,
@@ -131,7 +132,7 @@ def run_cycle(class_descs: bytes, n_classes: int, job_descs: bytes,
               bias, bias_rows, class_req, class_tol, class_require,
               class_forbid, class_min, dim_w, queue_alloc, queue_limit,
               score_scratch, cap_scratch, log_nodes, log_counts, log_len,
-              class_placed, job_placed, job_flag):
+              class_placed, job_placed, job_flag, sort_scratch=None):
     """One library call = one whole allocate cycle (plan built host-side)."""
     lib = _load()
     R, N = alloc_t.shape
@@ -144,4 +145,5 @@ def run_cycle(class_descs: bytes, n_classes: int, job_descs: bytes,
         _p(class_require), _p(class_forbid), _p(class_min), _p(dim_w),
         _p(queue_alloc), _p(queue_limit), _p(score_scratch), _p(cap_scratch),
         _p(log_nodes), _p(log_counts), _p(log_len), _p(class_placed),
-        _p(job_placed), _p(job_flag), c_int(N), c_int(R), c_int(W), _stream())
+        _p(job_placed), _p(job_flag), _p(sort_scratch),
+        c_int(N), c_int(R), c_int(W), _stream())

@@ -309,6 +309,9 @@ def run_plan_hip(plan: CyclePlan) -> CycleResult:
     class_placed = torch.zeros(C, dtype=torch.int32, device=dev)
     job_placed = torch.zeros(len(plan.jobs), dtype=torch.int32, device=dev)
     job_flag = torch.ones(len(plan.jobs), dtype=torch.uint8, device=dev)
+    # radix scratch for the bulk select path (keys/ids ping-pong)
+    sort_scratch = torch.empty(4 * N, dtype=torch.int32, device=dev) \
+        if any(cp.ntasks >= 512 for cp in plan.classes) and N >= 512 else None
 
     hip.run_cycle(
         ctypes.cast(cds, ctypes.c_void_p), C,
@@ -316,7 +319,8 @@ def run_plan_hip(plan: CyclePlan) -> CycleResult:
         nt.alloc_t, nt.used_t, nt.extra_t, nt.ready, nt.taint_mask,
         nt.planes_t, bias, bias_rows, class_req, class_tol, class_require,
         class_forbid, class_min, dim_w, q_alloc, q_limit, score, cap,
-        log_nodes, log_counts, log_len, class_placed, job_placed, job_flag)
+        log_nodes, log_counts, log_len, class_placed, job_placed, job_flag,
+        sort_scratch)
 
     res = _collect(plan, log_nodes.cpu(), log_counts.cpu(), log_len.cpu(),
                    class_placed.cpu(), job_placed.cpu(), job_flag.cpu())
