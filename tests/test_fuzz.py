@@ -94,3 +94,43 @@ def test_job_dict_roundtrip(job):
     d = to_dict(job)
     back = from_dict(Job, d)
     assert to_dict(back) == d
+
+
+@given(st.integers(1, 25), st.integers(1, 40), st.integers(100, 6000),
+       st.integers(0, 3))
+@settings(max_examples=25, deadline=None)
+def test_bulk_enqueue_equals_per_job_votes(n_nodes, n_jobs, cpu_each, seed):
+    """Property: the whole-batch enqueue fast path admits EXACTLY the jobs
+    the per-job vote loop admits (monotone-sum argument,
+    actions/enqueue.py)."""
+    from volcano_amd.scheduler import (FakeBinder, Scheduler, SchedulerCache,
+                                       default_config)
+
+    def run(bulk):
+        store = ObjectStore()
+        cache = SchedulerCache(store=store, binder=FakeBinder())
+        config = default_config()
+        config.actions = ["enqueue"]
+        sched = Scheduler(cache, config)
+        if not bulk:
+            orig = sched.open_session
+
+            def patched():
+                ssn = orig()
+                ssn.job_enqueueable_bulk_fns.clear()
+                return ssn
+            sched.open_session = patched
+        for i in range(n_nodes):
+            store.create("Node", synth.make_node(f"n{i}", cpu_milli=4000))
+        store.create("Queue", synth.make_queue("qa", weight=1))
+        import random
+        rng = random.Random(seed)
+        for j in range(n_jobs):
+            synth.make_gang(store, f"g{j:03d}",
+                            replicas=rng.randint(1, 4), queue="qa",
+                            cpu_milli=cpu_each, mem=1, phase="Pending")
+        sched.run_once()
+        return sorted(pg.meta.name for pg in store.list("PodGroup")
+                      if pg.status.phase == "Inqueue")
+
+    assert run(True) == run(False)
