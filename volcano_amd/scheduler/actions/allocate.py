@@ -25,6 +25,8 @@ import numpy as np
 
 from ...api.info import TaskClass
 from ...api.types import PodGroupPhase, TaskStatus
+
+PENDING_S = TaskStatus.PENDING
 from ...utils.metrics import METRICS
 from ..plan import (BundleEntry, ClassPlan, CyclePlan, JobPlan, run_plan_hip,
                     run_plan_torch)
@@ -76,8 +78,8 @@ class AllocateAction:
             if job.phase not in (PodGroupPhase.INQUEUE.value,
                                  PodGroupPhase.RUNNING.value):
                 continue
-            if not job.pending_tasks:
-                continue
+            if not job.task_status_index.get(PENDING_S):
+                continue      # no list build — emptiness check only
             if not ssn.job_valid(job):
                 continue
             by_queue.setdefault(job.queue, []).append(job)

@@ -28,7 +28,10 @@ class BackfillAction:
         plan.bias = getattr(ssn, "score_bias", None)
         predicates = getattr(ssn, "predicates", None)
 
+        from ...api.types import TaskStatus
         for job in ssn.jobs.values():
+            if not job.task_status_index.get(TaskStatus.PENDING):
+                continue      # emptiness check before any property work
             if job.phase not in (PodGroupPhase.INQUEUE.value,
                                  PodGroupPhase.RUNNING.value):
                 continue

@@ -264,8 +264,10 @@ class SchedulerCache:
         session in which its evictions were committed, and converges to
         Bound the following cycle.)"""
         for job in self.jobs.values():
-            pipelined = list(job.task_status_index.get(
-                TaskStatus.PIPELINED, {}).values())
+            bucket = job.task_status_index.get(TaskStatus.PIPELINED)
+            if not bucket:
+                continue
+            pipelined = list(bucket.values())
             for t in pipelined:
                 node = self.nodes.get(t.node_name)
                 if node is not None:
