@@ -120,7 +120,9 @@ class SoftShardCoordinator:
         # preempt path see the global state
         remote_nr = remote.t().cpu().numpy()             # [N, R]
         names = nt.dims.names
-        nodes_sorted = sorted(ssn.nodes.values(), key=lambda n: n.name)
+        nodes_sorted = getattr(ssn.cache, "nodes_sorted", None)
+        if nodes_sorted is None or len(nodes_sorted) != len(ssn.nodes):
+            nodes_sorted = sorted(ssn.nodes.values(), key=lambda n: n.name)
         import numpy as np
         hot = np.nonzero(np.abs(remote_nr).sum(axis=1) > 1e-6)[0]
         for i in hot:

@@ -402,7 +402,9 @@ class AllocateAction:
         reconcile: any job with a placement there is reverted wholesale
         (gang atomicity) and its device usage unwound."""
         nt = ssn.node_tensors
-        nodes_sorted = sorted(ssn.nodes.values(), key=lambda n: n.name)
+        nodes_sorted = getattr(ssn.cache, "nodes_sorted", None)
+        if nodes_sorted is None or len(nodes_sorted) != len(ssn.nodes):
+            nodes_sorted = sorted(ssn.nodes.values(), key=lambda n: n.name)
 
         to_bind = []
         bind_by_job: Dict[str, List] = {}

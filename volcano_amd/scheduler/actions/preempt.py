@@ -178,7 +178,9 @@ class PreemptAction:
                       score, cap)
         feasible = (score > float("-inf")).cpu().numpy()
         order_idx = torch.argsort(score, descending=True).cpu().numpy()
-        nodes_sorted = sorted(ssn.nodes.values(), key=lambda n: n.name)
+        nodes_sorted = getattr(ssn.cache, "nodes_sorted", None)
+        if nodes_sorted is None or len(nodes_sorted) != len(ssn.nodes):
+            nodes_sorted = sorted(ssn.nodes.values(), key=lambda n: n.name)
         order = [nodes_sorted[i] for i in order_idx if feasible[i]]
         return order, victims_by_node
 

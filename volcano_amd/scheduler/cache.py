@@ -280,6 +280,9 @@ class SchedulerCache:
 
     def ensure_packed(self) -> None:
         nodes = sorted(self.nodes.values(), key=lambda n: n.name)
+        # node_id-ordered list for apply/preempt paths (avoids re-sorting
+        # 10k+ NodeInfos once per action)
+        self.nodes_sorted = nodes
         if self._tensors_dirty or self.node_tensors.alloc_t is None:
             self.node_tensors.pack(nodes)
             self._tensors_dirty = self._used_dirty = False
