@@ -104,6 +104,13 @@ class TaskInfo:
             anti = aff.get("podAntiAffinity")
             if isinstance(anti, dict) and anti.get("group"):
                 pod.request.q.setdefault(f"paa:{anti['group']}", 1.0)
+        # host ports and attachable volumes ride the same machinery: each
+        # port is a per-node unit dim (k8s nodeports filter), the volume
+        # count draws from the node's attach budget (nodevolumelimits)
+        for p in pod.host_ports:
+            pod.request.q.setdefault(f"hp:{int(p)}", 1.0)
+        if pod.volumes:
+            pod.request.q.setdefault("nvl:attach", float(len(pod.volumes)))
         return cls(
             uid=pod.meta.uid or pod.meta.key,
             name=pod.meta.name,

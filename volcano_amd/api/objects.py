@@ -95,6 +95,7 @@ class Node:
     ready: bool = True
     # colocation / oversubscription (reference node_info.go:83-89)
     oversubscription: Resource = field(default_factory=Resource)
+    images: List[str] = field(default_factory=list)  # cached container images
 
 
 @dataclass
@@ -114,6 +115,8 @@ class Pod:
     best_effort: bool = field(default=False)
     scheduling_gates: List[str] = field(default_factory=list)
     volumes: List[str] = field(default_factory=list)   # PVC names (same ns)
+    host_ports: List[int] = field(default_factory=list)  # requested hostPorts
+    image: str = ""                                    # container image
 
     def __post_init__(self):
         if self.request.is_empty():

@@ -170,7 +170,7 @@ class SchedulerCache:
             return
         task = TaskInfo.from_pod(pod, key)
         for k in task.request.q:
-            if k.startswith("paa:") and k not in self.dims:
+            if k.startswith(("paa:", "hp:", "nvl:")) and k not in self.dims:
                 self.dims.add(k)
                 self._tensors_dirty = True
         self._job_for(key).add_task(task)

@@ -7,10 +7,11 @@ mirrors the reference's ``plugins/factory.go``.
 
 from .base import PLUGIN_REGISTRY, Plugin, register
 from . import (binpack, capacity, cdp, conformance, deviceshare, drf,
-               extender, gang, interpodaffinity, network_topology_aware,
-               nodegroup, nodeorder, numaaware, overcommit, pdb, predicates,
-               priority, proportion, rescheduling, resource_strategy_fit,
-               resourcequota, sla, task_topology, tdm, topologyspread,
+               extender, gang, imagelocality, interpodaffinity,
+               network_topology_aware, nodegroup, nodeorder, numaaware,
+               overcommit, pdb, predicates, priority, proportion,
+               rescheduling, resource_strategy_fit, resourcequota, sla,
+               task_topology, tdm, topologyspread,
                usage)  # noqa: F401 (side-effect registration)
 
 

@@ -213,12 +213,24 @@ class NodeTensors:
         planes = np.zeros((W, N), dtype=np.int64)
 
         didx = self.dims.index
-        # synthetic unit dims (pod anti-affinity groups): every node offers
-        # exactly 1 unless it opts out explicitly
+        # synthetic unit dims: anti-affinity groups ("paa:") and host
+        # ports ("hp:<port>") — every node offers exactly 1, so the
+        # capacity kernel enforces at-most-one-per-node (k8s nodeports
+        # filter) with zero predicate work.  "nvl:attach" models the
+        # node's attachable-volume limit (k8s nodevolumelimits): default
+        # 256, overridden per node by the volcano.sh/max-volumes
+        # annotation.
+        nvl_j = didx.get("nvl:attach")
         for name, j in didx.items():
-            if name.startswith("paa:"):
+            if name.startswith("paa:") or name.startswith("hp:"):
                 alloc[:, j] = 1.0
+        if nvl_j is not None:
+            alloc[:, nvl_j] = 256.0
         for i, ni in enumerate(nodes):
+            if nvl_j is not None:
+                lim = ni.node.meta.annotations.get("volcano.sh/max-volumes")
+                if lim:
+                    alloc[i, nvl_j] = float(lim)
             for k, v in ni.allocatable.q.items():
                 alloc[i, didx[k]] = v
             for k, v in ni.used.q.items():
