@@ -98,3 +98,34 @@ def test_image_locality_no_cache_no_pin():
     key = mk_pod(store, "img2", image="registry/other:v1")
     sched.run_once()
     assert key in binder.binds                    # schedules anywhere
+
+
+def test_volume_binding_assume_on_bind():
+    """Unbound PVC binds to a fitting free PV in the landing node's zone
+    when the pod schedules (k8s volumebinding WaitForFirstConsumer)."""
+    from volcano_amd.api.objects import (ObjectMeta, PersistentVolume,
+                                         PersistentVolumeClaim, ZONE_LABEL)
+    store = ObjectStore()
+    cache = SchedulerCache(store=store)               # StoreBinder
+    sched = Scheduler(cache, default_config())
+    store.create("Node", synth.make_node(
+        "z1n", cpu_milli=8000, mem=32 * GI, labels={ZONE_LABEL: "z1"}))
+    store.create("Queue", synth.make_queue("default"))
+    # two free PVs: wrong zone (big) and right zone (fits)
+    store.create("PersistentVolume", PersistentVolume(
+        meta=ObjectMeta(name="pv-z2", labels={ZONE_LABEL: "z2"}),
+        capacity=100 * GI))
+    store.create("PersistentVolume", PersistentVolume(
+        meta=ObjectMeta(name="pv-z1", labels={ZONE_LABEL: "z1"}),
+        capacity=50 * GI))
+    store.create("PersistentVolumeClaim", PersistentVolumeClaim(
+        meta=ObjectMeta(name="want", namespace="default"),
+        request=10 * GI))
+    key = mk_pod(store, "wf", volumes=["want"])
+    sched.run_once()
+    pod = store.get("Pod", "default", "wf-worker-0")
+    assert pod.node_name == "z1n"
+    pvc = store.get("PersistentVolumeClaim", "default", "want")
+    assert pvc.volume_name == "pv-z1"                 # zone-matched PV
+    pv = store.get("PersistentVolume", "default", "pv-z1")
+    assert pv.claim_ref == "default/want"

@@ -349,6 +349,7 @@ class PersistentVolume:
     meta: ObjectMeta = field(default_factory=ObjectMeta)
     capacity: float = 0.0               # bytes
     storage_class: str = ""
+    claim_ref: str = ""                 # "ns/name" of the bound PVC ("" = free)
     # zone read from meta.labels[ZONE_LABEL] ("" = zone-free volume)
 
     @property

@@ -61,7 +61,44 @@ class StoreBinder(Binder):
             pod = self.store.get("Pod", t.namespace, t.name)
             if pod is not None:
                 pod.node_name = t.node_name
+                if pod.volumes:
+                    self._bind_volumes(pod)
                 self.store.update("Pod", pod)
+
+    def _bind_volumes(self, pod) -> None:
+        """Schedule-time volume binding (k8s volumebinding's assume step,
+        WaitForFirstConsumer): each of the pod's still-unbound claims gets
+        a free PV that fits — capacity, storage class, and the zone of the
+        node the pod just landed on."""
+        from ..api.objects import ZONE_LABEL
+        node = self.store.get("Node", "default", pod.node_name) or \
+            next(iter(self.store.list(
+                "Node", selector=lambda n: n.meta.name == pod.node_name)),
+                None)
+        node_zone = node.meta.labels.get(ZONE_LABEL, "") if node else ""
+        free = None
+        for vname in pod.volumes:
+            pvc = self.store.get("PersistentVolumeClaim",
+                                 pod.meta.namespace, vname)
+            if pvc is None or pvc.volume_name:
+                continue
+            if free is None:
+                free = sorted((pv for pv in self.store.list("PersistentVolume")
+                               if not pv.claim_ref),
+                              key=lambda pv: (pv.capacity, pv.meta.name))
+            for i, pv in enumerate(free):
+                if pv.capacity < pvc.request:
+                    continue
+                if pvc.storage_class and pv.storage_class != pvc.storage_class:
+                    continue
+                if pv.zone and node_zone and pv.zone != node_zone:
+                    continue
+                pv.claim_ref = pvc.meta.key
+                pvc.volume_name = pv.meta.name
+                self.store.update("PersistentVolume", pv)
+                self.store.update("PersistentVolumeClaim", pvc)
+                free.pop(i)
+                break
 
     def evict(self, task: TaskInfo, reason: str = "") -> None:
         pod = self.store.get("Pod", task.namespace, task.name)
