@@ -301,12 +301,16 @@ def test_cycle_equivalence_feature_rich():
                 synth.make_gang(store, f"g{j:03d}", replicas=3, queue=q,
                                 cpu_milli=500, mem=GI,
                                 node_selector={"zone": f"z{j % 2}"})
-            elif kind == 2:     # tolerant gang
+            elif kind == 2:     # tolerant gang with a shared host port
                 synth.make_gang(store, f"g{j:03d}", replicas=2, queue=q,
                                 cpu_milli=2000, mem=GI,
                                 tolerations=[Toleration(
                                     key="dedicated", value="infra",
                                     effect="NoSchedule")])
+                for pod in store.list("Pod"):
+                    if pod.meta.name.startswith(f"g{j:03d}-"):
+                        pod.host_ports = [7000 + (j % 3)]
+                        store.update("Pod", pod)
             else:               # multi-role + anti-affinity
                 pg = synth.make_podgroup(f"g{j:03d}", queue=q, min_member=3,
                                          min_task_member={"ps": 1,
