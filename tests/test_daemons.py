@@ -78,3 +78,26 @@ def test_prometheus_metrics_source():
         assert usage["n1"]["memory"] == 33.0
     finally:
         srv.shutdown()
+
+
+def test_cycle_recorder_writes_jsonl(tmp_path, monkeypatch):
+    import json
+    from volcano_amd.scheduler import (FakeBinder, Scheduler, SchedulerCache,
+                                       default_config)
+    from volcano_amd.store import ObjectStore
+    from volcano_amd.utils import synth
+    log = tmp_path / "cycles.jsonl"
+    monkeypatch.setenv("VAMD_CYCLE_LOG", str(log))
+    store = ObjectStore()
+    cache = SchedulerCache(store=store, binder=FakeBinder())
+    sched = Scheduler(cache, default_config())
+    store.create("Node", synth.make_node("n0"))
+    synth.make_gang(store, "g", replicas=2, cpu_milli=100)
+    sched.run_once()
+    sched.run_once()
+    lines = [json.loads(x) for x in log.read_text().splitlines()]
+    assert len(lines) == 2
+    assert lines[0]["cycle"] == 1 and lines[1]["cycle"] == 2
+    assert "allocate" in lines[0]["actions_ms"]
+    assert lines[0]["nodes"] == 1 and lines[0]["jobs"] >= 1
+    assert lines[1]["pending_after"] == 0        # gang placed

@@ -62,14 +62,48 @@ class Scheduler:
     def run_once(self) -> Session:
         t0 = time.perf_counter()
         ssn = self.open_session()
+        timings = {}
         for action in self._actions:
             ta = time.perf_counter()
             action.execute(ssn)
-            METRICS.observe(f"action_scheduling_latency:{action.name}",
-                            time.perf_counter() - ta)
+            dt = time.perf_counter() - ta
+            timings[action.name] = dt
+            METRICS.observe(f"action_scheduling_latency:{action.name}", dt)
         self.close_session(ssn)
         METRICS.observe("e2e_scheduling_latency", time.perf_counter() - t0)
+        self._record_cycle(ssn, t0, timings)
         return ssn
+
+    _cycle_idx = 0
+
+    def _record_cycle(self, ssn, t0, timings) -> None:
+        """Per-cycle decision recorder (reference actions/allocate
+        recorder.go): one JSON line per cycle to $VAMD_CYCLE_LOG —
+        enough to reconstruct what the cycle saw and decided."""
+        import os
+        path = os.environ.get("VAMD_CYCLE_LOG")
+        if not path:
+            return
+        import json
+        from ..api.types import TaskStatus
+        pending = sum(len(j.task_status_index.get(TaskStatus.PENDING, ()))
+                      for j in ssn.jobs.values())
+        self._cycle_idx += 1
+        rec = {
+            "cycle": self._cycle_idx,
+            "ts": time.time(),
+            "e2e_ms": round((time.perf_counter() - t0) * 1000, 3),
+            "actions_ms": {k: round(v * 1000, 3) for k, v in timings.items()},
+            "nodes": len(ssn.nodes),
+            "jobs": len(ssn.jobs),
+            "queues": len(ssn.queues),
+            "pending_after": pending,
+        }
+        try:
+            with open(path, "a") as f:
+                f.write(json.dumps(rec) + "\n")
+        except OSError:
+            pass
 
     def _idle(self) -> bool:
         """Nothing to do: no store events since last cycle and no pending
