@@ -128,7 +128,7 @@ class TaskInfo:
         )
 
 
-@dataclass
+@dataclass(slots=True)
 class TaskClass:
     """A batch of identical pending tasks of one job — the kernel launch unit."""
 

@@ -39,7 +39,7 @@ class BundleEntry:
     min_needed: int            # gang minimum for THIS job
 
 
-@dataclass
+@dataclass(slots=True)
 class ClassPlan:
     tclass: TaskClass
     job_key: str
