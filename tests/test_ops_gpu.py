@@ -190,6 +190,35 @@ def test_bulk_select_matches_oracle(hip, seed, N, ntasks):
     assert torch.allclose(qa, qa_g.cpu(), atol=1e-2)
 
 
+def test_bulk_select_fused_gang_revert(hip):
+    """Bulk path with fuse_min > achievable placements: in-kernel
+    discard — usage and queue untouched, placed 0, log counts zeroed."""
+    N, R = 2000, 4
+    g = torch.Generator().manual_seed(9)
+    score = torch.rand(N, generator=g)
+    cap = torch.ones(N, dtype=torch.int32)          # 2000 instances max
+    req = torch.ones(R)
+    dev = "cuda"
+    used_t = torch.zeros(R, N, device=dev)
+    qa = torch.zeros(R, device=dev)
+    qlimit = torch.full((R,), 1.0e18)
+    K = N
+    ln = torch.zeros(K, dtype=torch.int32, device=dev)
+    lc = torch.zeros(K, dtype=torch.int32, device=dev)
+    ll = torch.zeros(1, dtype=torch.int32, device=dev)
+    pl = torch.zeros(1, dtype=torch.int32, device=dev)
+    jp = torch.zeros(1, dtype=torch.int32, device=dev)
+    scratch = torch.empty(4 * N, dtype=torch.int32, device=dev)
+    hip.select_commit(score.to(dev), cap.to(dev), req.to(dev), 3000,
+                      used_t, qa, qlimit.to(dev), ln, lc, ll, pl, jp,
+                      2500, sort_scratch=scratch)   # needs 2500, only 2000
+    torch.cuda.synchronize()
+    assert int(pl.cpu()) == 0
+    assert float(used_t.abs().sum().cpu()) == 0.0
+    assert float(qa.abs().sum().cpu()) == 0.0
+    assert int(lc.cpu().sum()) == 0                 # discard zeroed the log
+
+
 def test_select_commit_gang_fused_revert(hip):
     """fuse_min above achievable → in-kernel revert leaves zero state."""
     N, R = 1000, 4
