@@ -296,9 +296,9 @@ __global__ void cond_revert_kernel(
     if (*flag) return;
     int len = *log_len;
     int tid = blockIdx.x * blockDim.x + threadIdx.x;
-    int lane_r = tid % 16;           // up to 16 dims handled per entry
-    int entry = tid / 16;
-    int stride = (gridDim.x * blockDim.x) / 16;
+    int lane_r = tid % WAVE;         // one wave per entry: R <= 64 dims
+    int entry = tid / WAVE;
+    int stride = (gridDim.x * blockDim.x) / WAVE;
     for (int e = entry; e < len; e += stride) {
         int cnt = log_counts[e];
         if (cnt == 0) continue;

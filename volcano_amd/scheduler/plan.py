@@ -252,6 +252,13 @@ def run_plan_hip(plan: CyclePlan) -> CycleResult:
     C = plan.n_classes
     if C == 0:
         return CycleResult([], {}, {})
+    if R > 64:
+        # revert/commit kernels map one wavefront (64 lanes) per log
+        # entry — more resource dims than lanes would silently skip dims
+        raise RuntimeError(
+            f"HIP decision plane supports up to 64 resource dims, got {R} "
+            "(synthetic paa:/hp: dims count); split the inventory or use "
+            "the torch oracle path")
 
     cds = (hip.VamdClassDesc * C)()
     job_index = {jp.job_key: j for j, jp in enumerate(plan.jobs)}
