@@ -67,7 +67,6 @@ __global__ void score_cap_kernel(
 
         long long cap = BIG_CAP;
         float least = 0.f, most = 0.f, mean = 0.f;
-        float frac[16];
         for (int r = 0; r < R; ++r) {
             size_t off = (size_t)r * N + i;
             float a = alloc[off];
@@ -83,19 +82,29 @@ __global__ void score_cap_kernel(
                 cap = min(cap, max(c, 0ll));
             }
             float f = fminf((u + rq) / fmaxf(a, EPS), 1.0f);
-            frac[r] = f;
             float dw = dim_w[r];
             least += (1.0f - f) * dw;
             most += f * dw;
             mean += f;
         }
         mean /= (float)R;
-        float var = 0.f;
-        for (int r = 0; r < R; ++r) {
-            float d = frac[r] - mean;
-            var += d * d;
+        // balanced-allocation variance: second pass RECOMPUTES f instead
+        // of staging a per-thread frac[] array (a fixed-size array caps R
+        // and spills VGPRs; inputs are L2-resident so the re-read is
+        // cheap, and the float op sequence is identical — bit-equal to
+        // the torch oracle).  Skipped entirely when the weight is 0.
+        float bal = 0.f;
+        if (w_bal != 0.f) {
+            float var = 0.f;
+            for (int r = 0; r < R; ++r) {
+                size_t off = (size_t)r * N + i;
+                float f = fminf((used[off] + req[r]) /
+                                fmaxf(alloc[off], EPS), 1.0f);
+                float d = f - mean;
+                var += d * d;
+            }
+            bal = 1.0f - sqrtf(var / (float)R);
         }
-        float bal = 1.0f - sqrtf(var / (float)R);
         float s = w_least * least / wsum + w_most * most / wsum + w_bal * bal;
         if (bias) s += bias[i];
 
