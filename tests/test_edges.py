@@ -82,3 +82,60 @@ def test_agent_scheduler_skips_gang_pods():
     # a gang pod (has podgroup annotation) is NOT fast-path eligible
     store.create("Pod", synth.make_pod("gangpod", podgroup="pg1"))
     assert asched.run_once() == 0
+
+
+def test_hip_plan_rejects_too_many_dims():
+    """The HIP path's revert/commit kernels map one 64-lane wave per log
+    entry; >64 resource dims must fail loudly, not silently skip dims."""
+    import numpy as np
+    import pytest
+    import torch
+    from volcano_amd.api.resource import ResourceDims
+    from volcano_amd.scheduler.plan import CyclePlan, ClassPlan, run_plan_hip
+    from volcano_amd.scheduler.tensors import NodeTensors
+    from volcano_amd.api.info import NodeInfo, TaskClass
+    from volcano_amd.utils import synth
+
+    dims = ResourceDims()
+    for i in range(70):
+        dims.add(f"paa:g{i}")
+    nt = NodeTensors(dims)
+    nt.pack([NodeInfo(synth.make_node("n0"))])
+    assert nt.r > 64
+    plan = CyclePlan(nt, torch.full((1, nt.r), 1e18), torch.zeros(1, nt.r))
+    tc = TaskClass(signature="s", role="", request=None, tasks=[object()])
+    plan.classes.append(ClassPlan(
+        tclass=tc, job_key="j", queue_idx=0,
+        req=np.zeros(nt.r, dtype=np.float32), tolerated=-1,
+        require=np.zeros(1, dtype=np.int64),
+        forbid=np.zeros(1, dtype=np.int64), min_needed=0))
+    with pytest.raises(RuntimeError, match="64 resource dims"):
+        run_plan_hip(plan)
+
+
+def test_many_synthetic_dims_cpu_cycle():
+    """A cycle with >16 dims (anti-affinity + host ports) through the
+    torch oracle — the dim count that overflowed the GPU kernel's
+    per-thread array before the fix."""
+    from volcano_amd.scheduler import (FakeBinder, Scheduler, SchedulerCache,
+                                       default_config)
+    from volcano_amd.store import ObjectStore
+    from volcano_amd.utils import synth
+    GI = 1024 ** 3
+    store = ObjectStore()
+    binder = FakeBinder()
+    cache = SchedulerCache(store=store, binder=binder)
+    sched = Scheduler(cache, default_config())
+    for i in range(6):
+        store.create("Node", synth.make_node(f"n{i}", cpu_milli=8000,
+                                             mem=32 * GI))
+    store.create("Queue", synth.make_queue("default"))
+    for j in range(12):
+        synth.make_gang(store, f"g{j}", replicas=1, cpu_milli=500, mem=GI)
+        pod = store.get("Pod", "default", f"g{j}-worker-0")
+        pod.affinity = {"podAntiAffinity": {"group": f"grp{j}"}}
+        pod.host_ports = [8000 + j]
+        store.update("Pod", pod)
+    sched.run_once()
+    assert cache.dims and len(cache.dims.names) > 16
+    assert len(binder.binds) == 12
