@@ -27,7 +27,12 @@ class DrfPlugin(Plugin):
         jobs = list(ssn.jobs.values())
         if not jobs:
             return
-        alloc = np.stack([job.alloc_vec(nt) for job in jobs]).astype(np.float64)
+        # zero-alloc jobs (the common start-of-cycle state) skip the
+        # vector build entirely — share is exactly 0
+        alloc = np.zeros((len(jobs), nt.r), dtype=np.float64)
+        for i, job in enumerate(jobs):
+            if job.occupied_count:
+                alloc[i] = job.alloc_vec(nt)
         total = ssn.total_resource.numpy().astype(np.float64) \
             if ssn.total_resource is not None else np.ones(nt.r)
         shares = (alloc / np.maximum(total, 1.0)).max(axis=1)     # [J]
