@@ -104,3 +104,32 @@ def test_two_rank_gloo_sharded_cycle():
     assert len(nodes0) + len(nodes1) == 8
     assert not (set(binds0) & set(binds1))
     assert len(binds0) + len(binds1) == 20
+
+
+def test_four_rank_gloo_sharded_cycle():
+    """World 4 — the driver's SCALE shape (minus GPUs): shards stay
+    disjoint and the global count converges."""
+    import socket
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_rank_main, args=(r, 4, port, q))
+             for r in range(4)]
+    for p in procs:
+        p.start()
+    results = []
+    for _ in range(4):
+        results.append(q.get(timeout=180))
+    for p in procs:
+        p.join(timeout=30)
+    errors = [r for r in results if r[1] == "ERROR"]
+    assert not errors, errors
+    results.sort()
+    totals = {r[1] for r in results}
+    assert totals == {20}                   # every rank agrees globally
+    all_nodes = [n for r in results for n in r[2]]
+    assert len(all_nodes) == len(set(all_nodes)) == 8
+    all_binds = [b for r in results for b in r[3]]
+    assert len(all_binds) == len(set(all_binds)) == 20
