@@ -137,9 +137,16 @@ def main():
             ("nccl" if use_gpu else "gloo")
         local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
         if use_gpu:
-            # modulo: lets N ranks share fewer GPUs (gloo-backend testing)
-            torch.cuda.set_device(local_rank % torch.cuda.device_count())
-        torch.distributed.init_process_group(backend=backend)
+            # modulo: lets N ranks share fewer GPUs (rehearsal on 1 GPU)
+            dev_idx = local_rank % torch.cuda.device_count()
+            torch.cuda.set_device(dev_idx)
+        if use_gpu and backend == "nccl":
+            # pin the communicator device: barrier() otherwise guesses
+            # cuda:<global rank>, which breaks when ranks share GPUs
+            torch.distributed.init_process_group(
+                backend=backend, device_id=torch.device("cuda", dev_idx))
+        else:
+            torch.distributed.init_process_group(backend=backend)
     elif use_gpu:
         torch.cuda.set_device(0)
 
