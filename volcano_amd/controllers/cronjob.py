@@ -31,6 +31,10 @@ class CronJobController(Controller):
                                selector=lambda j: j.meta.owner == owner)
 
     def resync(self) -> None:
+        from ..utils.features import enabled
+        if not enabled("CronVolcanoJobSupport"):
+            return
+
         for cj in self.store.list("CronJob"):
             self._sync(cj)
 

@@ -40,6 +40,9 @@ class JobController(Controller):
 
     # -- event routing --------------------------------------------------------
     def handle(self, ev) -> None:
+        from ..utils.features import enabled
+        if not enabled("VolcanoJobSupport"):
+            return
         if ev.kind == "Job":
             if ev.type != EventType.DELETED:
                 self._dirty.add(ev.obj.meta.key)

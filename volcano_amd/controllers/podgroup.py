@@ -16,6 +16,9 @@ class PodGroupController(Controller):
     watch_kinds = ("Pod",)
 
     def handle(self, ev) -> None:
+        from ..utils.features import enabled
+        if not enabled("WorkLoadSupport"):
+            return          # plain-pod wrapping is the workload gate
         if ev.type == EventType.DELETED:
             return
         pod = ev.obj

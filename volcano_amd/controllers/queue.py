@@ -26,6 +26,9 @@ class QueueController(Controller):
         elif ev.kind == "PodGroup":
             self._dirty.add(ev.obj.spec.queue)
         elif ev.kind == "Command" and ev.type != EventType.DELETED:
+            from ..utils.features import enabled
+            if not enabled("QueueCommandSync"):
+                return
             cmd = ev.obj
             if cmd.target_kind != "Queue":
                 return

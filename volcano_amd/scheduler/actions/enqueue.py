@@ -63,6 +63,9 @@ class EnqueueAction:
     def _lift_queue_gates(ssn, job) -> None:
         if not job.has_gated_tasks:
             return
+        from ...utils.features import enabled
+        if not enabled("SchedulingGatesQueueAdmission"):
+            return
         store = getattr(ssn.cache, "store", None)
         for t in job.tasks.values():
             if not t.gated or t.pod is None:

@@ -21,6 +21,7 @@ from ..api.objects import ObjectMeta, PodGroup, Queue
 from ..api.resource import ResourceDims
 from ..api.types import TaskStatus
 from ..store import EventType, ObjectStore
+from ..utils.features import enabled as features_enabled
 from .tensors import NodeTensors
 
 
@@ -61,7 +62,7 @@ class StoreBinder(Binder):
             pod = self.store.get("Pod", t.namespace, t.name)
             if pod is not None:
                 pod.node_name = t.node_name
-                if pod.volumes:
+                if pod.volumes and features_enabled("CSIStorage"):
                     self._bind_volumes(pod)
                 self.store.update("Pod", pod)
 

@@ -28,6 +28,7 @@ class SchedulerConfiguration:
     schedule_period: float = 1.0
     use_hip: bool = False          # decision plane on GPU (HIP kernels)
     device: str = "cpu"
+    feature_gates: Dict[str, bool] = field(default_factory=dict)
 
     @classmethod
     def from_yaml(cls, text: str) -> "SchedulerConfiguration":
@@ -55,6 +56,9 @@ class SchedulerConfiguration:
             conf.tiers = tiers
         for c in data.get("configurations", []) or []:
             conf.configurations[c.get("name", "")] = c.get("arguments", {}) or {}
+        fg = data.get("feature_gates") or data.get("featureGates") or {}
+        if isinstance(fg, dict):
+            conf.feature_gates = {str(k): bool(v) for k, v in fg.items()}
         return conf
 
 

@@ -21,6 +21,9 @@ class Scheduler:
                  config: Optional[SchedulerConfiguration] = None):
         self.cache = cache
         self.config = config or default_config()
+        if self.config.feature_gates:
+            from ..utils import features
+            features.set_gates(self.config.feature_gates)
         self._stop = threading.Event()
         self._actions = [actions_mod.new_action(a) for a in self.config.actions]
 

@@ -45,6 +45,9 @@ class NumaState:
 @register("numaaware")
 class NumaAwarePlugin(Plugin):
     def on_session_open(self, ssn) -> None:
+        from ...utils.features import enabled
+        if not enabled("ResourceTopology"):
+            return
         store = getattr(ssn.cache, "store", None)
         topos = store.list("Numatopology") if store is not None else []
         if not topos:

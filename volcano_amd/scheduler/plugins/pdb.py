@@ -13,6 +13,9 @@ from .base import Plugin, register
 @register("pdb")
 class PdbPlugin(Plugin):
     def on_session_open(self, ssn) -> None:
+        from ...utils.features import enabled
+        if not enabled("PodDisruptionBudgetsSupport"):
+            return
         store = getattr(ssn.cache, "store", None)
         pdbs = store.list("PodDisruptionBudget") if store is not None else []
         if not pdbs:

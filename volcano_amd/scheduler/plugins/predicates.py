@@ -75,6 +75,9 @@ class PredicatesPlugin(Plugin):
         label bit planes — zero kernel cost."""
         if pod is None or not pod.volumes:
             return ()
+        from ...utils.features import enabled
+        if not enabled("CSIStorage"):
+            return ()
         store = getattr(self._ssn.cache, "store", None)
         if store is None:
             return ()

@@ -9,6 +9,9 @@ from .base import Plugin, register
 @register("priority")
 class PriorityPlugin(Plugin):
     def on_session_open(self, ssn) -> None:
+        from ...utils.features import enabled
+        if not enabled("PriorityClass"):
+            return
         def job_order(a, b) -> int:
             if a.priority != b.priority:
                 return -1 if a.priority > b.priority else 1
