@@ -139,6 +139,10 @@ class TaskClass:
     priority: int = 0
     # filled by the snapshot packer:
     class_id: int = -1
+    # per-class network-topology override (SubGroupPolicy): the
+    # network-topology-aware plugin confines THIS class to one domain
+    # instead of using the job-level spec
+    topology: Optional[dict] = None
 
     @property
     def count(self) -> int:

@@ -235,6 +235,13 @@ class PodGroupSpec:
     priority_class: str = ""
     min_resources: Resource = field(default_factory=Resource)
     network_topology: Optional[dict] = None
+    # SubGroupPolicy (reference scheduling/v1beta1 types.go:218-267):
+    # secondary grouping — each entry {"name", "subGroupSize",
+    # "minSubGroups", "networkTopology", "labelSelector",
+    # "matchLabelKeys"} partitions matching pods into gang-atomic
+    # subgroups of subGroupSize; the job commits only when at least
+    # minSubGroups of them place.
+    sub_group_policy: List[dict] = field(default_factory=list)
 
 
 @dataclass

@@ -170,6 +170,18 @@ class AllocateAction:
         for q, job in ordered_jobs:
             qi = ssn.queue_index[q.name]
 
+            # -- SubGroupPolicy (reference types.go:218 SubGroupPolicySpec
+            # + allocate.go allocateForSubJob): matching pods partition
+            # into gang-atomic subgroups of subGroupSize, each planned as
+            # its own single-class gang (in-kernel revert); the job-level
+            # minSubGroups gate is applied after the run (_apply gating).
+            if job.podgroup is not None and \
+                    job.podgroup.spec.sub_group_policy:
+                close_bundle()
+                self._plan_subgroups(ssn, plan, job, qi, predicates,
+                                     w_least, w_most, w_bal)
+                continue
+
             # -- nomination fast path (reference allocate.go:797
             # NominatedNodeName): a gang pipelined by last cycle's preempt
             # re-checks fit on its nominated nodes and commits host-side,
@@ -275,6 +287,7 @@ class AllocateAction:
         if coord is not None:
             bad = coord.find_conflicts(nt, used_before)
         if result is not None:
+            self._gate_subgroup_families(plan, result)
             self._apply(ssn, plan, result, bad_nodes=bad)
         if coord is not None:
             coord.finalize(ssn, nt, used_before)
@@ -282,6 +295,95 @@ class AllocateAction:
         METRICS.observe("allocate:plan_build", t1 - t0)
         METRICS.observe("allocate:plan_run", t2 - t1)
         METRICS.observe("allocate:apply", t3 - t2)
+
+    def _plan_subgroups(self, ssn, plan, job, qi: int, predicates,
+                        w_least, w_most, w_bal) -> None:
+        """Partition the job's pending pods per SubGroupPolicy entry and
+        emit one single-class gang JobPlan per complete subgroup.  Pods
+        left unmatched (or in an incomplete tail chunk) stay pending
+        this cycle.  Synthetic job keys alias back to the real job for
+        the apply walk; the minSubGroups family gate lives in plan
+        metadata and is enforced post-run."""
+        from ...api.types import TaskStatus as _TS
+        pend = [t for t in job.task_status_index.get(
+            _TS.PENDING, {}).values() if not t.gated]
+        claimed = set()
+        if not hasattr(plan, "job_alias"):
+            plan.job_alias = {}
+            plan.subgroup_families = []
+        for pi, pol in enumerate(job.podgroup.spec.sub_group_policy):
+            size = max(1, int(pol.get("subGroupSize", 1)))
+            min_subs = int(pol.get("minSubGroups", 0))
+            sel = pol.get("labelSelector") or {}
+            keys = pol.get("matchLabelKeys") or []
+            topo = pol.get("networkTopology")
+            matched = []
+            for t in pend:
+                if id(t) in claimed or t.pod is None:
+                    continue
+                lbl = t.pod.meta.labels
+                if all(lbl.get(k) == v for k, v in sel.items()):
+                    matched.append(t)
+            groups: Dict[tuple, List] = {}
+            for t in matched:
+                gk = tuple(t.pod.meta.labels.get(k) for k in keys)
+                groups.setdefault(gk, []).append(t)
+            members = []
+            sg = 0
+            for gk in sorted(groups, key=lambda g: tuple(map(str, g))):
+                ts = groups[gk]
+                for off in range(0, len(ts) - size + 1, size):
+                    chunk = ts[off:off + size]
+                    for t in chunk:
+                        claimed.add(id(t))
+                    first = chunk[0]
+                    req = ssn.node_tensors.req_vector(first)
+                    if req is None:
+                        continue
+                    tc = TaskClass(signature=first.class_signature(),
+                                   role=first.role, request=first.request,
+                                   tasks=chunk, priority=first.priority,
+                                   topology=topo)
+                    if predicates is not None:
+                        tol, require, forbid = \
+                            predicates.class_constraints(tc, job)
+                    else:
+                        W = max(ssn.node_tensors.labels.words, 1)
+                        tol = -1
+                        require = np.zeros(W, dtype=np.int64)
+                        forbid = np.zeros(W, dtype=np.int64)
+                    skey = f"{job.key}#sg{pi}:{sg}"
+                    sg += 1
+                    cp = ClassPlan(tclass=tc, job_key=skey, queue_idx=qi,
+                                   req=req, tolerated=tol, require=require,
+                                   forbid=forbid, min_needed=size,
+                                   w_least=w_least, w_most=w_most,
+                                   w_bal=w_bal)
+                    plan.jobs.append(JobPlan(
+                        job_key=skey, class_begin=len(plan.classes),
+                        class_end=len(plan.classes) + 1, occupied=0,
+                        min_available=size))
+                    plan.classes.append(cp)
+                    plan.job_alias[skey] = job.key
+                    members.append(skey)
+            if members:
+                plan.subgroup_families.append((job.key, min_subs, members))
+
+    def _gate_subgroup_families(self, plan, result) -> None:
+        """minSubGroups: if fewer than the required number of a policy's
+        subgroups fully placed, revert the ones that did (whole-family
+        atomicity above the per-subgroup gangs)."""
+        for real_key, min_subs, members in getattr(
+                plan, "subgroup_families", []):
+            placed = [k for k in members if result.job_committed.get(k)]
+            if len(placed) >= min_subs:
+                continue
+            for c, cp in enumerate(plan.classes):
+                if cp.job_key in placed:
+                    cres = result.class_results[c]
+                    self._revert_pieces(plan, cp, cres.placements)
+                    cres.placements = []
+                    result.job_committed[cp.job_key] = False
 
     def _try_nominated(self, ssn, job, qi: int, predicates) -> bool:
         """Commit a fully-nominated gang onto its nominated nodes if they
@@ -433,7 +535,8 @@ class AllocateAction:
             if not cres.placements:
                 continue
             if cp.bundle is None:
-                job = ssn.jobs[cp.job_key]
+                job = ssn.jobs.get(cp.job_key) or \
+                    ssn.jobs[getattr(plan, "job_alias", {})[cp.job_key]]
                 tasks = iter(cp.tclass.tasks)
                 pieces = [(nid, cnt, [next(tasks) for _ in range(cnt)])
                           for nid, cnt in cres.placements]
@@ -500,9 +603,11 @@ class AllocateAction:
 
         # flip gang-ready podgroups to Running (job_updater analog)
         seen = set()
+        alias = getattr(plan, "job_alias", {})
         for cp in plan.classes:
             keys = [be.job_key for be in cp.bundle] if cp.bundle else [cp.job_key]
             for key in keys:
+                key = alias.get(key, key)
                 if key in seen or key not in committed_jobs:
                     continue
                 seen.add(key)

@@ -85,7 +85,35 @@ class NetworkTopologyAwarePlugin(Plugin):
             choice[job.key] = best
             return best
 
+        # per-SUBGROUP domains (SubGroupPolicy networkTopology): classes
+        # carrying a topology override pick their own domain against a
+        # build-time reservation ledger, so several subgroups of one job
+        # spread across racks within ONE cycle (a failed subgroup's
+        # reservation is simply re-chosen next cycle)
+        free_resv = {h: f.copy() for h, f in free.items()}
+
+        def choose_domain_for_class(tclass):
+            spec = tclass.topology
+            max_tier = spec.get("highestTierAllowed")
+            v = nt.req_vector(tclass.tasks[0])
+            if v is None:
+                return None
+            need = v.astype(np.float64) * len(tclass.tasks)
+            for hname in tree.domains_by_tier(max_tier):
+                if (free_resv[hname] + 0.1 >= need).all():
+                    free_resv[hname] = free_resv[hname] - need
+                    return hname
+            return None
+
         def hook(tclass, job, require, forbid):
+            spec2 = getattr(tclass, "topology", None)
+            if spec2:
+                best = choose_domain_for_class(tclass)
+                if best is not None:
+                    set_plane_bit(require, self.domain_bit[best])
+                elif spec2.get("mode", "hard") == "hard":
+                    set_plane_bit(require, impossible_bit)
+                return
             if job is None or job.podgroup is None:
                 return
             spec = job.podgroup.spec.network_topology

@@ -40,10 +40,14 @@ class PredicatesPlugin(Plugin):
         (nodegroup/tdm/usage) extend the bits through
         ``ssn.class_constraint_hooks``."""
         hooks = getattr(self._ssn, "class_constraint_hooks", [])
+        # per-class topology overrides (SubGroupPolicy) choose a domain
+        # PER SUBGROUP — never share a memo entry across subgroups
+        memoize = getattr(tclass, "topology", None) is None
         key = (tclass.signature, job.queue if (job and hooks) else None)
-        got = self._memo.get(key)
-        if got is not None:
-            return got
+        if memoize:
+            got = self._memo.get(key)
+            if got is not None:
+                return got
         nt = self._nt
         t = tclass.tasks[0]
         pod = t.pod
@@ -64,7 +68,8 @@ class PredicatesPlugin(Plugin):
         for hook in hooks:
             hook(tclass, job, require, forbid)
         out = (tolerated, require, forbid)
-        self._memo[key] = out
+        if memoize:
+            self._memo[key] = out
         return out
 
     def _volume_zone_bits(self, pod) -> tuple:
