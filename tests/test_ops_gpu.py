@@ -383,3 +383,41 @@ def test_cycle_equivalence_feature_rich():
     assert cpu == gpu, (
         f"decision divergence: {len(cpu)} vs {len(gpu)}; "
         f"diff={set(cpu.items()) ^ set(gpu.items())}")
+
+
+def test_cycle_equivalence_subgroups(hip):
+    """CPU vs GPU decisions for SubGroupPolicy jobs (per-subgroup gangs
+    ride the fused in-kernel revert; minSubGroups gates post-run)."""
+    from volcano_amd.scheduler import FakeBinder, Scheduler, SchedulerCache, \
+        default_config
+    from volcano_amd.store import ObjectStore
+    from volcano_amd.utils import synth
+
+    GI = 1024 ** 3
+
+    def build(device, use_hip):
+        store = ObjectStore()
+        for i in range(40):
+            store.create("Node", synth.make_node(
+                f"n-{i:03d}", cpu_milli=4000, mem=16 * GI))
+        store.create("Queue", synth.make_queue("default"))
+        for j in range(6):
+            pg = synth.make_podgroup(f"sg{j}", min_member=1)
+            pg.spec.sub_group_policy = [
+                {"subGroupSize": 3, "minSubGroups": 2}]
+            store.create("PodGroup", pg)
+            for i in range(7):      # 2 complete triples + 1 tail
+                store.create("Pod", synth.make_pod(
+                    f"sg{j}-w-{i}", f"sg{j}", cpu_milli=1000, mem=GI))
+        config = default_config()
+        config.use_hip = use_hip
+        config.device = device
+        binder = FakeBinder()
+        cache = SchedulerCache(store=store, binder=binder, device=device)
+        Scheduler(cache, config).run_once()
+        return binder.binds
+
+    cpu = build("cpu", False)
+    gpu = build("cuda", True)
+    assert len(cpu) == 36               # 6 jobs × 2 triples
+    assert cpu == gpu
