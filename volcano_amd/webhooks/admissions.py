@@ -105,6 +105,18 @@ def validate_podgroup(store, pg, op) -> None:
     for role, n in pg.spec.min_task_member.items():
         if n < 0:
             raise AdmissionError(f"minTaskMember[{role}] must be >= 0")
+    for i, pol in enumerate(pg.spec.sub_group_policy or []):
+        if int(pol.get("subGroupSize", 1)) < 1:
+            raise AdmissionError(
+                f"subGroupPolicy[{i}].subGroupSize must be >= 1")
+        if int(pol.get("minSubGroups", 0)) < 0:
+            raise AdmissionError(
+                f"subGroupPolicy[{i}].minSubGroups must be >= 0")
+        nt = pol.get("networkTopology")
+        if nt is not None and nt.get("mode", "hard") not in ("hard", "soft"):
+            raise AdmissionError(
+                f"subGroupPolicy[{i}].networkTopology.mode must be "
+                "hard|soft")
 
 
 # -- pods ---------------------------------------------------------------------
