@@ -12,6 +12,22 @@
 
 #include "vamd_api.h"
 
+#include <cstdlib>
+#include <cstring>
+
+// one launch for the whole plan when every class is small (the
+// heterogeneous shape: thousands of per-job classes would otherwise
+// cost ~2 enqueues each).  VAMD_MEGACYCLE=0 falls back to per-class
+// launches.
+static bool megacycle_enabled() {
+    static int v = -1;
+    if (v < 0) {
+        const char* e = getenv("VAMD_MEGACYCLE");
+        v = (e == nullptr || atoi(e) != 0) ? 1 : 0;
+    }
+    return v == 1;
+}
+
 extern "C" void vamd_run_cycle(
     const VamdClassDesc* classes, int n_classes,
     const VamdJobDesc* jobs, int n_jobs,
@@ -29,7 +45,39 @@ extern "C" void vamd_run_cycle(
     unsigned* sort_scratch,
     int N, int R, int W, hipStream_t stream)
 {
-    (void)n_classes;
+    bool any_bulk = false;
+    for (int c = 0; c < n_classes; ++c)
+        if (classes[c].ntasks >= 512) { any_bulk = true; break; }
+    if (megacycle_enabled() && !any_bulk && n_classes >= 32) {
+        // stage descriptors + taint masks on device (stream-ordered)
+        size_t sz_c = (size_t)n_classes * sizeof(VamdClassDesc);
+        size_t sz_j = (size_t)n_jobs * sizeof(VamdJobDesc);
+        size_t sz_t = (size_t)n_classes * sizeof(int64_t);
+        void* dev_blob = nullptr;
+        if (hipMallocAsync(&dev_blob, sz_c + sz_j + sz_t, stream)
+                == hipSuccess && dev_blob != nullptr) {
+            char* p = (char*)dev_blob;
+            (void)hipMemcpyAsync(p, classes, sz_c, hipMemcpyHostToDevice,
+                                 stream);
+            (void)hipMemcpyAsync(p + sz_c, jobs, sz_j,
+                                 hipMemcpyHostToDevice, stream);
+            (void)hipMemcpyAsync(p + sz_c + sz_j, class_tol, sz_t,
+                                 hipMemcpyHostToDevice, stream);
+            vamd_megacycle((const VamdClassDesc*)p,
+                           (const VamdJobDesc*)(p + sz_c), n_jobs,
+                           alloc, used, extra, ready, taints, planes,
+                           bias, bias_rows, class_req,
+                           (const int64_t*)(p + sz_c + sz_j),
+                           class_require, class_forbid, class_min,
+                           dim_w, queue_alloc, queue_limit,
+                           score_scratch, cap_scratch, log_nodes,
+                           log_counts, log_len, class_placed,
+                           job_placed, job_flag, N, R, W, stream);
+            (void)hipFreeAsync(dev_blob, stream);
+            return;
+        }
+        // allocation failed: fall through to per-class launches
+    }
     for (int j = 0; j < n_jobs; ++j) {
         const VamdJobDesc& job = jobs[j];
         int nc = job.class_end - job.class_begin;

@@ -19,6 +19,8 @@
 #include <cfloat>
 #include <cstdint>
 
+#include "vamd_api.h"   // VamdClassDesc/VamdJobDesc for the megacycle
+
 #define DEVINL __device__ __forceinline__
 
 namespace vamd {
@@ -32,22 +34,25 @@ constexpr long long BIG_CAP = 1 << 30;
 // K1+K2 fused: per-node feasibility + capacity + score for one task class.
 //   grid-stride over N; one thread = one node; all loads coalesced ([R,N]).
 // ---------------------------------------------------------------------------
-__global__ void score_cap_kernel(
-    const float* __restrict__ alloc,   // [R, N]
-    const float* __restrict__ used,    // [R, N]
-    const float* __restrict__ extra,   // [R, N] (releasing - pipelined) or nullptr
-    const uint8_t* __restrict__ ready, // [N]
-    const int64_t* __restrict__ taints,// [N]
-    const int64_t* __restrict__ planes,// [W, N]
-    const float* __restrict__ req,     // [R]
+// core of K1+K2: score/cap for nodes {start, start+step, ...} — shared
+// by the grid-stride kernel and the fused megacycle (block stride)
+__device__ void score_pass(
+    int start, int step,
+    const float* __restrict__ alloc,
+    const float* __restrict__ used,
+    const float* __restrict__ extra,
+    const uint8_t* __restrict__ ready,
+    const int64_t* __restrict__ taints,
+    const int64_t* __restrict__ planes,
+    const float* __restrict__ req,
     int64_t tolerated,
-    const int64_t* __restrict__ require, // [W]
-    const int64_t* __restrict__ forbid,  // [W]
+    const int64_t* __restrict__ require,
+    const int64_t* __restrict__ forbid,
     float w_least, float w_most, float w_bal,
-    const float* __restrict__ dim_w,   // [R]
-    const float* __restrict__ bias,    // [N] or nullptr
-    float* __restrict__ score_out,     // [N]
-    int* __restrict__ cap_out,         // [N]
+    const float* __restrict__ dim_w,
+    const float* __restrict__ bias,
+    float* __restrict__ score_out,
+    int* __restrict__ cap_out,
     int N, int R, int W)
 {
     // req/dim_w are tiny and wave-uniform: scalar-cached by L1 automatically.
@@ -55,8 +60,7 @@ __global__ void score_cap_kernel(
     for (int r = 0; r < R; ++r) wsum += dim_w[r];
     wsum = fmaxf(wsum, EPS);
 
-    for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < N;
-         i += gridDim.x * blockDim.x) {
+    for (int i = start; i < N; i += step) {
         bool feasible = ready[i] != 0;
         feasible = feasible && ((taints[i] & ~tolerated) == 0);
         for (int w = 0; w < W; ++w) {
@@ -114,6 +118,21 @@ __global__ void score_cap_kernel(
     }
 }
 
+__global__ void score_cap_kernel(
+    const float* alloc, const float* used, const float* extra,
+    const uint8_t* ready, const int64_t* taints, const int64_t* planes,
+    const float* req, int64_t tolerated, const int64_t* require,
+    const int64_t* forbid, float w_least, float w_most, float w_bal,
+    const float* dim_w, const float* bias, float* score_out, int* cap_out,
+    int N, int R, int W)
+{
+    score_pass(blockIdx.x * blockDim.x + threadIdx.x,
+               gridDim.x * blockDim.x,
+               alloc, used, extra, ready, taints, planes, req, tolerated,
+               require, forbid, w_least, w_most, w_bal, dim_w, bias,
+               score_out, cap_out, N, R, W);
+}
+
 // ---------------------------------------------------------------------------
 // (value, index) argmax with ties -> lowest index, across a wave then a block.
 // ---------------------------------------------------------------------------
@@ -146,8 +165,7 @@ DEVINL ValIdx wave_reduce(ValIdx x) {
 #define SC_THREADS 1024
 #define SC_WAVES (SC_THREADS / WAVE)
 
-__global__ void __launch_bounds__(SC_THREADS)
-select_commit_kernel(
+__device__ void dev_select_commit(
     float* __restrict__ score,        // [N] (consumed: selected entries -> -inf)
     const int* __restrict__ cap,      // [N]
     const float* __restrict__ req,    // [R]
@@ -267,17 +285,25 @@ select_commit_kernel(
     }
 }
 
+__global__ void __launch_bounds__(SC_THREADS)
+select_commit_kernel(
+    float* score, const int* cap, const float* req, int ntasks, float* used,
+    float* queue_alloc, const float* queue_limit, int* log_nodes,
+    int* log_counts, int* log_len, int* placed, int* job_placed,
+    int fuse_min, int N, int R, int K)
+{
+    dev_select_commit(score, cap, req, ntasks, used, queue_alloc,
+                      queue_limit, log_nodes, log_counts, log_len, placed,
+                      job_placed, fuse_min, N, R, K);
+}
+
 // ---------------------------------------------------------------------------
 // K4 gang readiness for multi-class jobs (JobReady AND-aggregation,
 // session_plugins.go:483).  One tiny block.
 // ---------------------------------------------------------------------------
-__global__ void finalize_job_kernel(
-    const int* __restrict__ job_placed,   // [1]
-    int occupied, int min_available,
-    const int* __restrict__ class_placed, // [nc] (slice)
-    const int* __restrict__ class_min,    // [nc]
-    uint8_t* __restrict__ flag,           // [1]
-    int nc)
+DEVINL void dev_finalize_job(
+    const int* job_placed, int occupied, int min_available,
+    const int* class_placed, const int* class_min, uint8_t* flag, int nc)
 {
     if (threadIdx.x == 0) {
         bool ok = (*job_placed + occupied) >= min_available;
@@ -287,10 +313,23 @@ __global__ void finalize_job_kernel(
     }
 }
 
+__global__ void finalize_job_kernel(
+    const int* __restrict__ job_placed,   // [1]
+    int occupied, int min_available,
+    const int* __restrict__ class_placed, // [nc] (slice)
+    const int* __restrict__ class_min,    // [nc]
+    uint8_t* __restrict__ flag,           // [1]
+    int nc)
+{
+    dev_finalize_job(job_placed, occupied, min_available, class_placed,
+                     class_min, flag, nc);
+}
+
 // ---------------------------------------------------------------------------
 // K4 conditional rollback of one class, gated on the job's device flag.
+// Block-scope (callers launch ONE block / run inside the megacycle).
 // ---------------------------------------------------------------------------
-__global__ void cond_revert_kernel(
+__device__ void dev_cond_revert(
     const uint8_t* __restrict__ flag,  // [1] 1 = keep
     int* __restrict__ log_nodes,       // [K]
     int* __restrict__ log_counts,      // [K]
@@ -304,10 +343,10 @@ __global__ void cond_revert_kernel(
 {
     if (*flag) return;
     int len = *log_len;
-    int tid = blockIdx.x * blockDim.x + threadIdx.x;
+    int tid = threadIdx.x;
     int lane_r = tid % WAVE;         // one wave per entry: R <= 64 dims
     int entry = tid / WAVE;
-    int stride = (gridDim.x * blockDim.x) / WAVE;
+    int stride = blockDim.x / WAVE;
     for (int e = entry; e < len; e += stride) {
         int cnt = log_counts[e];
         if (cnt == 0) continue;
@@ -321,10 +360,18 @@ __global__ void cond_revert_kernel(
         *job_placed -= p;
         *placed = 0;
     }
-    // zero the counts after the subtraction pass completes (same block
-    // ordering is not guaranteed across blocks — run with ONE block)
+    // zero the counts after the subtraction pass completes
     __syncthreads();
     for (int e = threadIdx.x; e < len; e += blockDim.x) log_counts[e] = 0;
+}
+
+__global__ void cond_revert_kernel(
+    const uint8_t* flag, int* log_nodes, int* log_counts,
+    const int* log_len, const float* req, float* used, float* queue_alloc,
+    int* placed, int* job_placed, int N, int R)
+{
+    dev_cond_revert(flag, log_nodes, log_counts, log_len, req, used,
+                    queue_alloc, placed, job_placed, N, R);
 }
 
 // ---------------------------------------------------------------------------
@@ -568,6 +615,93 @@ bulk_select_kernel(
     }
 }
 
+// ---------------------------------------------------------------------------
+// Megacycle: a WHOLE small-class plan in ONE launch.  Heterogeneous
+// inventories produce thousands of per-job classes; at ~2 enqueues per
+// class the cycle becomes launch-bound.  This kernel sequences the
+// per-job kernel bodies device-side (score -> select -> gang
+// finalize/revert), preserving the exact sequential semantics: one
+// workgroup, every class sees the previous classes' staged usage.
+// Dispatched only when no class takes the bulk path (cycle_runner).
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(SC_THREADS)
+megacycle_kernel(
+    const VamdClassDesc* __restrict__ classes,   // [C] (device copy)
+    const VamdJobDesc* __restrict__ jobs,        // [J] (device copy)
+    int n_jobs,
+    const float* __restrict__ alloc, float* __restrict__ used,
+    const float* __restrict__ extra,
+    const uint8_t* __restrict__ ready, const int64_t* __restrict__ taints,
+    const int64_t* __restrict__ planes,
+    const float* __restrict__ bias, const float* __restrict__ bias_rows,
+    const float* __restrict__ class_req,        // [C, R]
+    const int64_t* __restrict__ class_tol,      // [C] (device copy)
+    const int64_t* __restrict__ class_require,  // [C, W]
+    const int64_t* __restrict__ class_forbid,   // [C, W]
+    const int32_t* __restrict__ class_min,      // [C]
+    const float* __restrict__ dim_w,
+    float* __restrict__ queue_alloc, const float* __restrict__ queue_limit,
+    float* __restrict__ score, int* __restrict__ cap,
+    int* __restrict__ log_nodes, int* __restrict__ log_counts,
+    int* __restrict__ log_len, int* __restrict__ class_placed,
+    int* __restrict__ job_placed, uint8_t* __restrict__ job_flag,
+    int N, int R, int W)
+{
+    const int tid = threadIdx.x;
+    for (int j = 0; j < n_jobs; ++j) {
+        const VamdJobDesc jd = jobs[j];
+        const int nc = jd.class_end - jd.class_begin;
+        const bool single = (nc == 1);
+
+        for (int c = jd.class_begin; c < jd.class_end; ++c) {
+            const VamdClassDesc cd = classes[c];
+            const float* ext = (cd.flags & 1) ? extra : nullptr;
+            const float* b = (cd.bias_row >= 0 && bias_rows)
+                ? bias_rows + (size_t)cd.bias_row * N : bias;
+
+            score_pass(tid, SC_THREADS, alloc, used, ext, ready, taints,
+                       planes, class_req + (size_t)c * R, class_tol[c],
+                       class_require + (size_t)c * W,
+                       class_forbid + (size_t)c * W,
+                       cd.w_least, cd.w_most, cd.w_bal, dim_w, b,
+                       score, cap, N, R, W);
+            __syncthreads();
+
+            int fuse_min = -1;
+            if (single) {
+                int need = jd.min_available - jd.occupied;
+                if (cd.min_needed > need) need = cd.min_needed;
+                fuse_min = need > 0 ? need : 0;
+            }
+            dev_select_commit(score, cap, class_req + (size_t)c * R,
+                              cd.ntasks, used,
+                              queue_alloc + (size_t)cd.queue_idx * R,
+                              queue_limit + (size_t)cd.queue_idx * R,
+                              log_nodes + cd.log_off,
+                              log_counts + cd.log_off,
+                              log_len + c, class_placed + c, job_placed + j,
+                              fuse_min, N, R, cd.log_cap);
+            __syncthreads();
+        }
+
+        if (!single) {
+            dev_finalize_job(job_placed + j, jd.occupied, jd.min_available,
+                             class_placed + jd.class_begin,
+                             class_min + jd.class_begin, job_flag + j, nc);
+            __syncthreads();
+            for (int c = jd.class_begin; c < jd.class_end; ++c) {
+                const VamdClassDesc cd = classes[c];
+                dev_cond_revert(job_flag + j, log_nodes + cd.log_off,
+                                log_counts + cd.log_off, log_len + c,
+                                class_req + (size_t)c * R, used,
+                                queue_alloc + (size_t)cd.queue_idx * R,
+                                class_placed + c, job_placed + j, N, R);
+                __syncthreads();
+            }
+        }
+    }
+}
+
 }  // namespace vamd
 
 // ---------------------------------------------------------------------------
@@ -633,6 +767,30 @@ void vamd_cond_revert(
     hipLaunchKernelGGL(vamd::cond_revert_kernel, dim3(1), dim3(1024), 0,
                        stream, flag, log_nodes, log_counts, log_len, req,
                        used, queue_alloc, placed, job_placed, N, R);
+}
+
+void vamd_megacycle(
+    const VamdClassDesc* classes_dev, const VamdJobDesc* jobs_dev,
+    int n_jobs,
+    const float* alloc, float* used, const float* extra,
+    const uint8_t* ready, const int64_t* taints, const int64_t* planes,
+    const float* bias, const float* bias_rows,
+    const float* class_req, const int64_t* class_tol_dev,
+    const int64_t* class_require, const int64_t* class_forbid,
+    const int32_t* class_min, const float* dim_w,
+    float* queue_alloc, const float* queue_limit,
+    float* score_scratch, int* cap_scratch,
+    int* log_nodes, int* log_counts, int* log_len,
+    int* class_placed, int* job_placed, uint8_t* job_flag,
+    int N, int R, int W, hipStream_t stream)
+{
+    hipLaunchKernelGGL(vamd::megacycle_kernel, dim3(1), dim3(SC_THREADS), 0,
+                       stream, classes_dev, jobs_dev, n_jobs, alloc, used,
+                       extra, ready, taints, planes, bias, bias_rows,
+                       class_req, class_tol_dev, class_require, class_forbid,
+                       class_min, dim_w, queue_alloc, queue_limit,
+                       score_scratch, cap_scratch, log_nodes, log_counts,
+                       log_len, class_placed, job_placed, job_flag, N, R, W);
 }
 
 }  // extern "C"

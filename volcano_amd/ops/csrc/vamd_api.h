@@ -60,6 +60,23 @@ struct VamdJobDesc {
 // finalize/revert.  Single-class jobs fuse the gang check into
 // select_commit (2 launches per job).  Everything stays on `stream`;
 // NO host synchronisation happens here.
+// One launch for a whole small-class plan (heterogeneous shapes):
+// descriptor/taint arrays must be DEVICE pointers here.
+void vamd_megacycle(
+    const VamdClassDesc* classes_dev, const VamdJobDesc* jobs_dev,
+    int n_jobs,
+    const float* alloc, float* used, const float* extra,
+    const uint8_t* ready, const int64_t* taints, const int64_t* planes,
+    const float* bias, const float* bias_rows,
+    const float* class_req, const int64_t* class_tol_dev,
+    const int64_t* class_require, const int64_t* class_forbid,
+    const int32_t* class_min, const float* dim_w,
+    float* queue_alloc, const float* queue_limit,
+    float* score_scratch, int* cap_scratch,
+    int* log_nodes, int* log_counts, int* log_len,
+    int* class_placed, int* job_placed, uint8_t* job_flag,
+    int N, int R, int W, hipStream_t stream);
+
 void vamd_run_cycle(
     const VamdClassDesc* classes, int n_classes,
     const VamdJobDesc* jobs, int n_jobs,
