@@ -421,3 +421,46 @@ def test_cycle_equivalence_subgroups(hip):
     gpu = build("cuda", True)
     assert len(cpu) == 36               # 6 jobs × 2 triples
     assert cpu == gpu
+
+
+def test_cycle_equivalence_megacycle(hip, monkeypatch):
+    """The opt-in single-launch megacycle path (VAMD_MEGACYCLE=1) makes
+    the same decisions as the per-class launch sequence and the CPU
+    oracle."""
+    monkeypatch.setenv("VAMD_MEGACYCLE", "1")
+    from volcano_amd.scheduler import FakeBinder, Scheduler, SchedulerCache, \
+        default_config
+    from volcano_amd.store import ObjectStore
+    from volcano_amd.utils import synth
+
+    GI = 1024 ** 3
+
+    def build(device, use_hip):
+        store = ObjectStore()
+        for i in range(500):
+            store.create("Node", synth.make_node(
+                f"n-{i:03d}", cpu_milli=8000, mem=32 * GI))
+        store.create("Queue", synth.make_queue("qa", weight=2))
+        store.create("Queue", synth.make_queue("qb", weight=1))
+        import numpy as np
+        rng = np.random.RandomState(11)
+        for j in range(60):       # >= 32 classes → megacycle dispatch
+            synth.make_gang(store, f"m{j:03d}",
+                            replicas=int(rng.randint(1, 5)),
+                            queue="qa" if j % 2 else "qb",
+                            cpu_milli=float(rng.choice([500, 1000, 1500])),
+                            mem=GI, priority=int(rng.randint(0, 3)))
+        config = default_config()
+        config.use_hip = use_hip
+        config.device = device
+        binder = FakeBinder()
+        cache = SchedulerCache(store=store, binder=binder, device=device)
+        sched = Scheduler(cache, config)
+        sched.run_once()
+        sched.run_once()
+        return binder.binds
+
+    cpu = build("cpu", False)
+    gpu = build("cuda", True)
+    assert len(cpu) > 100
+    assert cpu == gpu

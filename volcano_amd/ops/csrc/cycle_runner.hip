@@ -15,17 +15,17 @@
 #include <cstdlib>
 #include <cstring>
 
-// one launch for the whole plan when every class is small (the
-// heterogeneous shape: thousands of per-job classes would otherwise
-// cost ~2 enqueues each).  VAMD_MEGACYCLE=0 falls back to per-class
-// launches.
+// one launch for the whole plan when every class is small.  Measured
+// OFF by default: at 10k nodes the async per-class enqueue pipeline
+// overlaps launches with host plan iteration, while the single-
+// workgroup megacycle serializes the per-class score passes (mix bench
+// 559 ms vs 778 ms/step — gpurun_out/g28_mix*.log).  Kept as an
+// opt-in (VAMD_MEGACYCLE=1) for launch-latency-bound deployments
+// (small N, many classes); a cooperative multi-block variant is the
+// roadmap item.  getenv per cycle so tests can flip it.
 static bool megacycle_enabled() {
-    static int v = -1;
-    if (v < 0) {
-        const char* e = getenv("VAMD_MEGACYCLE");
-        v = (e == nullptr || atoi(e) != 0) ? 1 : 0;
-    }
-    return v == 1;
+    const char* e = getenv("VAMD_MEGACYCLE");
+    return e != nullptr && atoi(e) != 0;
 }
 
 extern "C" void vamd_run_cycle(
