@@ -32,12 +32,16 @@ class Controller:
         """Periodic full pass (cron triggers, TTL sweeps)."""
 
     def sync_once(self) -> int:
+        from ..utils.metrics import METRICS
         n = 0
         if self._watch is not None:
             for ev in self._watch.drain():
                 self.handle(ev)
                 n += 1
         self.resync()
+        if n:
+            # reference controllers/metrics: per-controller event counters
+            METRICS.inc(f"controller_handled_events:{self.name}", n)
         return n
 
 
