@@ -123,6 +123,9 @@ def main():
     ap.add_argument("--shard-mode", choices=["hard", "soft"], default="hard",
                     help="hard: nodes+jobs sharded (conflict-free); "
                          "soft: nodes replicated, delta all-reduce")
+    ap.add_argument("--extra-plugins", default="",
+                    help="comma list of extra plugins for the last tier "
+                         "(e.g. task-topology,numaaware — BASELINE config 5)")
     ap.add_argument("--mix", action="store_true",
                     help="heterogeneous inventory (varied gangs/queues/"
                          "priorities/selectors)")
@@ -152,6 +155,11 @@ def main():
 
     device = "cuda" if use_gpu else "cpu"
     config = default_config()
+    if args.extra_plugins:
+        from volcano_amd.scheduler.config import PluginOption
+        for name in args.extra_plugins.split(","):
+            if name.strip():
+                config.tiers[-1].plugins.append(PluginOption(name.strip()))
     config.use_hip = use_gpu
     config.device = device
     if use_gpu:

@@ -31,9 +31,11 @@ CONFIGS = {
     3: {"desc": "10k nodes / 100k pods, queue proportion, sharded N GPUs",
         "bench": ["--nodes", "10000", "--jobs", "10000",
                   "--pods-per-job", "10"]},
-    4: {"desc": "50k nodes / 1M pods, 8 GPUs (also meaningful at 1)",
+    4: {"desc": "50k nodes / 1M pods, task-topology + numaaware, "
+             "8 GPUs (also meaningful at 1)",
         "bench": ["--nodes", "50000", "--jobs", "50000",
-                  "--pods-per-job", "20", "--steps", "1"]},
+                  "--pods-per-job", "20", "--steps", "1",
+                  "--extra-plugins", "task-topology,numaaware"]},
 }
 
 
