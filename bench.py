@@ -97,6 +97,7 @@ def reset_cluster(cache: SchedulerCache, jobs):
             t.status = TaskStatus.PENDING
             t.node_name = ""
         job.task_status_index = {TaskStatus.PENDING: dict(job.tasks)}
+        job._occ = 0                       # counter matches the fresh index
         job._alloc_vec = None
     for ni in cache.nodes.values():
         ni.used = Resource()

@@ -90,6 +90,12 @@ def test_full_pipeline_churn_invariants():
             assert ni.used.get(CPU) <= ni.allocatable.get(CPU) + 1.0
         for job in cache.jobs.values():
             occ = job.occupied_count
+            # the incremental counter must equal a from-scratch recount
+            from volcano_amd.api.info import _OCC_SET
+            recount = sum(1 for t in job.tasks.values()
+                          if t.status in _OCC_SET)
+            assert occ == recount, \
+                f"cycle {cycle}: {job.key} occ counter drift {occ} vs {recount}"
             # gang atomicity (pipelined reservations count toward min)
             assert occ == 0 or occ + job.waiting_count >= \
                 min(job.min_available, len(job.tasks)), \

@@ -150,6 +150,7 @@ class TaskClass:
 
 
 _OCCUPIED_STATUSES = tuple(ALLOCATED_STATUSES) + (TaskStatus.SUCCEEDED,)
+_OCC_SET = frozenset(_OCCUPIED_STATUSES)
 
 
 class JobInfo:
@@ -168,6 +169,7 @@ class JobInfo:
         # set is a single class; False = known multi-class; None = unknown
         self._atom = None
         self._minav = None
+        self._occ = 0                        # occupied-count (incremental)
 
     # -- basic accessors ----------------------------------------------------
     @property
@@ -218,6 +220,8 @@ class JobInfo:
     def add_task(self, task: TaskInfo) -> None:
         self.tasks[task.key] = task
         self.task_status_index.setdefault(task.status, {})[task.key] = task
+        if task.status in _OCC_SET:
+            self._occ += 1
         self._alloc_vec = self._total_vec = self._atom = None
         self._prio = None
         if task.gated:
@@ -231,6 +235,8 @@ class JobInfo:
         t = self.tasks.pop(task_key, None)
         if t is not None:
             self.task_status_index.get(t.status, {}).pop(task_key, None)
+            if t.status in _OCC_SET:
+                self._occ -= 1
             self._alloc_vec = self._total_vec = self._atom = None
         return t
 
@@ -255,6 +261,7 @@ class JobInfo:
 
     def update_task_status(self, task: TaskInfo, status: TaskStatus) -> None:
         self.task_status_index.get(task.status, {}).pop(task.key, None)
+        self._occ += (status in _OCC_SET) - (task.status in _OCC_SET)
         task.status = status
         self.task_status_index.setdefault(status, {})[task.key] = task
         self._alloc_vec = None
@@ -270,6 +277,8 @@ class JobInfo:
             # wholesale.  Size equality implies identity here because the
             # batch is a subset of the bucket by construction.
             src_status = tasks[0].status
+            self._occ += ((status in _OCC_SET) -
+                          (src_status in _OCC_SET)) * len(tasks)
             src = self.task_status_index.get(src_status)
             if src is not None and len(src) == len(tasks):
                 for t in tasks:
@@ -283,8 +292,10 @@ class JobInfo:
                 self._alloc_vec = None
                 return
         dst = self.task_status_index.setdefault(status, {})
+        in_occ = status in _OCC_SET
         for t in tasks:
             self.task_status_index.get(t.status, {}).pop(t.key, None)
+            self._occ += in_occ - (t.status in _OCC_SET)
             t.status = status
             dst[t.key] = t
         self._alloc_vec = None
@@ -332,16 +343,10 @@ class JobInfo:
     @property
     def occupied_count(self) -> int:
         """Tasks holding or promised resources (reference ReadyTaskNum:
-        Bound+Binding+Running+Allocated+Succeeded).  Unrolled — this is
-        the hottest job property (gang checks, ordering keys)."""
-        idx = self.task_status_index
-        g = idx.get
-        n = 0
-        for s in _OCCUPIED_STATUSES:
-            b = g(s)
-            if b:
-                n += len(b)
-        return n
+        Bound+Binding+Running+Allocated+Succeeded).  Maintained
+        incrementally by the four mutation methods — this is the hottest
+        job property (gang checks, ordering keys, queue tensors)."""
+        return self._occ
 
     @property
     def waiting_count(self) -> int:
