@@ -277,10 +277,10 @@ class JobInfo:
             # wholesale.  Size equality implies identity here because the
             # batch is a subset of the bucket by construction.
             src_status = tasks[0].status
-            self._occ += ((status in _OCC_SET) -
-                          (src_status in _OCC_SET)) * len(tasks)
             src = self.task_status_index.get(src_status)
             if src is not None and len(src) == len(tasks):
+                self._occ += ((status in _OCC_SET) -
+                              (src_status in _OCC_SET)) * len(tasks)
                 for t in tasks:
                     t.status = status
                 dst = self.task_status_index.get(status)
