@@ -131,3 +131,51 @@ def test_match_label_keys_grouping():
     sched.run_once()
     # s0 pair + one s1 pair place; s1's odd pod stays pending
     assert len(binder.binds) == 4
+
+
+def test_subgroups_with_full_pipeline_churn():
+    """Subgroup jobs under the full corrective pipeline across cycles:
+    gang atomicity of subgroups holds, accounting stays consistent."""
+    import random
+    from volcano_amd.api.resource import CPU
+    from volcano_amd.api.info import _OCC_SET
+    from volcano_amd.scheduler.config import PluginOption as PO
+    rng = random.Random(11)
+    store = ObjectStore()
+    binder = FakeBinder()
+    cache = SchedulerCache(store=store, binder=binder)
+    config = default_config()
+    config.actions = ["enqueue", "allocate", "preempt", "backfill"]
+    sched = Scheduler(cache, config)
+    for i in range(8):
+        store.create("Node", synth.make_node(f"n{i}", cpu_milli=4000,
+                                             mem=32 * GI))
+    store.create("Queue", synth.make_queue("default"))
+    jid = 0
+    for cycle in range(12):
+        for _ in range(rng.randint(1, 3)):
+            jid += 1
+            size = rng.choice([2, 3])
+            mk_subgroup_job(store, f"s{jid:03d}",
+                            replicas=size * rng.randint(1, 3) +
+                            rng.randint(0, 1),
+                            policy=[{"subGroupSize": size,
+                                     "minSubGroups": 1}],
+                            cpu_milli=rng.choice([500, 1000]))
+        bound = [p for p in store.list("Pod") if p.node_name]
+        for p in rng.sample(bound, min(2, len(bound))):
+            store.delete("Pod", p.meta.namespace, p.meta.name)
+        sched.run_once()
+        # invariants: node accounting consistent; subgroup-size multiples
+        for ni in cache.nodes.values():
+            rec = sum(t.request.get(CPU) for t in ni.tasks.values()
+                      if t.status in _OCC_SET)
+            assert abs(rec - ni.used.get(CPU)) < 1.0
+        for job in cache.jobs.values():
+            if not (job.podgroup and job.podgroup.spec.sub_group_policy):
+                continue
+            size = job.podgroup.spec.sub_group_policy[0]["subGroupSize"]
+            assert job.occupied_count % size == 0, \
+                f"cycle {cycle}: {job.key} occupied {job.occupied_count} " \
+                f"not a multiple of subgroup size {size}"
+    assert len(binder.binds) > 10

@@ -303,7 +303,13 @@ class AllocateAction:
         left unmatched (or in an incomplete tail chunk) stay pending
         this cycle.  Synthetic job keys alias back to the real job for
         the apply walk; the minSubGroups family gate lives in plan
-        metadata and is enforced post-run."""
+        metadata and is enforced post-run.
+
+        Semantics notes (reference types.go:224: SubGroupPolicy is the
+        evolution of minTaskMember): when present it supersedes the
+        job-level minMember for admission-to-placement — minSubGroups is
+        the job gate; corrective actions (preempt/reclaim) treat the job
+        conventionally through its pending tasks."""
         from ...api.types import TaskStatus as _TS
         pend = [t for t in job.task_status_index.get(
             _TS.PENDING, {}).values() if not t.gated]
