@@ -139,3 +139,32 @@ def test_many_synthetic_dims_cpu_cycle():
     sched.run_once()
     assert cache.dims and len(cache.dims.names) > 16
     assert len(binder.binds) == 12
+
+
+def test_sorted_jobs_lexsort_equals_tuple_sort():
+    """The columnar (lexsort) job ordering must equal the tuple sort
+    exactly, including creation-time and key tie-breaks."""
+    import random
+    from volcano_amd.scheduler import (FakeBinder, Scheduler, SchedulerCache,
+                                       default_config)
+    from volcano_amd.store import ObjectStore
+    from volcano_amd.utils import synth
+    rng = random.Random(4)
+    store = ObjectStore()
+    cache = SchedulerCache(store=store, binder=FakeBinder())
+    sched = Scheduler(cache, default_config())
+    store.create("Node", synth.make_node("n0"))
+    store.create("Queue", synth.make_queue("default"))
+    for j in range(400):      # >= 256 triggers the columnar path
+        synth.make_gang(store, f"ls{j:04d}", replicas=1, cpu_milli=10,
+                        priority=rng.randint(0, 3))
+    ssn = sched.open_session()
+    jobs = list(ssn.jobs.values())
+    rng.shuffle(jobs)
+    fast = [j.key for j in ssn.sorted_jobs(jobs)]
+    keys = [k for tier in ssn.job_order_keys for k in tier]
+    slow = [j.key for j in sorted(
+        jobs, key=lambda j: tuple(k(j) for k in keys)
+        + (j.creation_timestamp, j.key))]
+    assert fast == slow
+    sched.close_session(ssn)

@@ -227,6 +227,16 @@ class Session:
         keys = [k for tier in self.job_order_keys for k in tier]
         fns = [f for tier in self.job_order_fns for f in tier]
         if len(keys) == len(fns) and all(k is not None for k in keys):
+            if len(jobs) >= 256:
+                # columnar sort: same ordering as the tuple sort (keys
+                # ascending, creation/key tie-break) without building a
+                # tuple per job — np.lexsort's LAST column is primary
+                import numpy as np
+                cols = [np.array([k(j) for j in jobs]) for k in keys]
+                ts = np.array([j.creation_timestamp for j in jobs])
+                jk = np.array([j.key for j in jobs])
+                order = np.lexsort(tuple([jk, ts] + cols[::-1]))
+                return [jobs[i] for i in order]
             return sorted(jobs, key=lambda j: tuple(k(j) for k in keys)
                           + (j.creation_timestamp, j.key))
         return sorted(jobs, key=functools.cmp_to_key(self.job_order))
