@@ -199,6 +199,20 @@ class AllocateAction:
                 first = next(iter(pend.values()))
                 if first.gated:
                     continue
+                # bundle continuation: identical (queue, signature) run —
+                # append the gang entry directly, skipping the
+                # TaskClass/ClassPlan/constraint work bundle_in would
+                # discard anyway (dominant shape: thousands of identical
+                # gangs back to back)
+                if not bias_fns and open_bundle is not None \
+                        and open_key == (qi, sig):
+                    tasks = list(pend.values())
+                    gang_min = max(job.min_available,
+                                   job.min_task_member.get(role, 0))
+                    open_bundle.bundle.append(
+                        BundleEntry(job.key, tasks, len(tasks), gang_min))
+                    open_bundle.ntasks_override += len(tasks)
+                    continue
                 req = req_memo.get(sig, _MISS)
                 if req is _MISS:
                     req = nt.req_vector(first)
