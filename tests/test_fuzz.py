@@ -134,3 +134,33 @@ def test_bulk_enqueue_equals_per_job_votes(n_nodes, n_jobs, cpu_each, seed):
                       if pg.status.phase == "Inqueue")
 
     assert run(True) == run(False)
+
+
+@given(st.integers(1, 4), st.integers(0, 14), st.integers(0, 3),
+       st.integers(0, 5))
+@settings(max_examples=30, deadline=None)
+def test_subgroup_partition_properties(size, replicas, min_subs, seed):
+    """SubGroupPolicy partition: every planned subgroup is complete
+    (exactly subGroupSize tasks), subgroups are disjoint, and leftover
+    pods stay pending."""
+    from volcano_amd.scheduler import (FakeBinder, Scheduler, SchedulerCache,
+                                       default_config)
+    store = ObjectStore()
+    binder = FakeBinder()
+    cache = SchedulerCache(store=store, binder=binder)
+    sched = Scheduler(cache, default_config())
+    for i in range(6):
+        store.create("Node", synth.make_node(f"n{i}", cpu_milli=32000))
+    store.create("Queue", synth.make_queue("default"))
+    pg = synth.make_podgroup("fz", min_member=1)
+    pg.spec.sub_group_policy = [{"subGroupSize": size,
+                                 "minSubGroups": min_subs}]
+    store.create("PodGroup", pg)
+    for i in range(replicas):
+        store.create("Pod", synth.make_pod(f"fz-w-{i}", "fz",
+                                           cpu_milli=100, mem=1))
+    sched.run_once()
+    complete = replicas // size
+    expected = complete * size if complete >= min_subs else 0
+    assert len(binder.binds) == expected, \
+        (size, replicas, min_subs, len(binder.binds))
