@@ -168,3 +168,37 @@ def test_sorted_jobs_lexsort_equals_tuple_sort():
         + (j.creation_timestamp, j.key))]
     assert fast == slow
     sched.close_session(ssn)
+
+
+def test_scheduler_determinism():
+    """Two scheduler instances over identical inventories make identical
+    decisions — the decision plane has no hidden randomness (ties break
+    structurally: lower node index, stable job order)."""
+    from volcano_amd.scheduler import (FakeBinder, Scheduler, SchedulerCache,
+                                       default_config)
+    from volcano_amd.store import ObjectStore
+    from volcano_amd.utils import synth
+    import random
+
+    def run():
+        rng = random.Random(77)
+        store = ObjectStore()
+        binder = FakeBinder()
+        cache = SchedulerCache(store=store, binder=binder)
+        sched = Scheduler(cache, default_config())
+        for i in range(20):
+            store.create("Node", synth.make_node(f"n{i:02d}",
+                                                 cpu_milli=8000))
+        store.create("Queue", synth.make_queue("qa", weight=2))
+        store.create("Queue", synth.make_queue("qb", weight=1))
+        for j in range(40):
+            synth.make_gang(store, f"d{j:03d}",
+                            replicas=rng.randint(1, 4),
+                            queue=rng.choice(["qa", "qb"]),
+                            cpu_milli=rng.choice([500, 1000]),
+                            priority=rng.randint(0, 3))
+        sched.run_once()
+        sched.run_once()
+        return dict(binder.binds)
+
+    assert run() == run()
