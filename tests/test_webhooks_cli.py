@@ -133,3 +133,27 @@ def test_vcctl_flow(tmp_path, capsys):
     assert main(["--state", state, "job", "delete", "-N", "train"]) == 0
     store = ObjectStore.load(state)
     assert store.get("Job", "default", "train") is None
+
+
+def test_hypernode_webhook_validation():
+    from volcano_amd.api.objects import (HyperNode, HyperNodeMember,
+                                         MemberSelector, ObjectMeta)
+    from volcano_amd.store import ObjectStore
+    from volcano_amd.webhooks import AdmissionError, default_chain
+    import pytest
+    store = ObjectStore()
+    guarded = default_chain(store).guard(store)
+    good = HyperNode(meta=ObjectMeta(name="ok"), tier=1, members=[
+        HyperNodeMember(type="Node",
+                        selector=MemberSelector(regex_match=r"gpu-\d+"))])
+    guarded.create("HyperNode", good)
+    bad_re = HyperNode(meta=ObjectMeta(name="badre"), tier=1, members=[
+        HyperNodeMember(type="Node",
+                        selector=MemberSelector(regex_match="gpu-("))])
+    with pytest.raises(AdmissionError, match="regexMatch"):
+        guarded.create("HyperNode", bad_re)
+    bad_tier = HyperNode(meta=ObjectMeta(name="badtier"), tier=0, members=[
+        HyperNodeMember(type="Node",
+                        selector=MemberSelector(exact_match=["n1"]))])
+    with pytest.raises(AdmissionError, match="tier"):
+        guarded.create("HyperNode", bad_tier)

@@ -158,6 +158,24 @@ def validate_jobflow(store, flow, op) -> None:
 
 # -- cronjobs -----------------------------------------------------------------
 
+def validate_hypernode(store, hn, op) -> None:
+    """reference admission/hypernodes/validate: regex selectors must
+    compile; tier >= 1; every member needs a selector."""
+    import re
+    if hn.tier < 1:
+        raise AdmissionError("hypernode tier must be >= 1")
+    for i, m in enumerate(hn.members or []):
+        sel = m.selector
+        if not (sel.exact_match or sel.regex_match or sel.label_match):
+            raise AdmissionError(f"members[{i}]: empty selector")
+        if sel.regex_match:
+            try:
+                re.compile(sel.regex_match)
+            except re.error as e:
+                raise AdmissionError(
+                    f"members[{i}]: bad regexMatch pattern: {e}")
+
+
 def validate_cronjob(store, cj, op) -> None:
     try:
         CronSchedule(cj.schedule)
@@ -179,5 +197,7 @@ def register_all(chain: AdmissionChain) -> None:
         AdmissionService("/pods/mutate", "Pod", mutate_pod),
         AdmissionService("/jobflows/validate", "JobFlow", validate_jobflow),
         AdmissionService("/cronjobs/validate", "CronJob", validate_cronjob),
+        AdmissionService("/hypernodes/validate", "HyperNode",
+                         validate_hypernode),
     ]:
         chain.register(svc)
