@@ -92,3 +92,15 @@ def test_gate_via_scheduler_config():
     Scheduler(SchedulerCache(store=ObjectStore(), binder=FakeBinder()),
               config)
     assert not features.enabled("PriorityClass")
+
+
+def test_env_parsing(monkeypatch):
+    monkeypatch.setenv("VAMD_FEATURE_GATES",
+                       "CSIStorage=false, PriorityClass=true,Bad")
+    features.reset()        # re-reads the env
+    assert not features.enabled("CSIStorage")
+    assert features.enabled("PriorityClass")
+    assert features.enabled("WorkLoadSupport")   # untouched default
+    monkeypatch.delenv("VAMD_FEATURE_GATES")
+    features.reset()
+    assert features.enabled("CSIStorage")
