@@ -77,3 +77,80 @@ def test_resourcequota_blocks_enqueue():
     assert sum(1 for k in binder.binds if k.startswith("default/ok")) == 2
     assert not any(k.startswith("default/toobig") for k in binder.binds)
     assert cache.jobs["default/toobig"].phase == "Pending"
+
+
+def test_gangpreempt_honors_pdb_veto():
+    """plugin vetoes are authoritative for bundles —
+    a PDB covering the low-prio gang's pods blocks gangpreempt even
+    though the victims are strictly lower priority."""
+    from volcano_amd.api.objects import ObjectMeta, PodDisruptionBudget
+    store, binder, cache, sched = mk(
+        actions=["enqueue", "allocate", "gangpreempt", "backfill"],
+        extra_plugins=("pdb",))
+    for n in synth.make_nodes(2, cpu_milli=2000, mem=8 * GI):
+        store.create("Node", n)
+    store.create("Queue", synth.make_queue("default"))
+    store.create("PodDisruptionBudget", PodDisruptionBudget(
+        meta=ObjectMeta(name="protect-low", namespace="default"),
+        selector={"app": "low"}, min_available=4))
+    pg = synth.make_gang(store, "low", replicas=4, cpu_milli=1000, mem=GI,
+                         priority=1)
+    for p in store.list("Pod"):
+        if p.meta.name.startswith("low"):
+            p.meta.labels["app"] = "low"
+            store.update("Pod", p)
+    sched.run_once()
+    # mark bound pods running so the PDB sees them healthy
+    for p in store.list("Pod"):
+        if p.meta.name.startswith("low"):
+            p.phase = "Running"
+            store.update("Pod", p)
+    sched.run_once()
+    synth.make_gang(store, "high", replicas=3, cpu_milli=1000, mem=GI,
+                    priority=100)
+    sched.run_once()
+    assert binder.evictions == []      # PDB veto holds for the bundle
+
+
+def test_gangpreempt_requires_lower_job_priority():
+    """Reference gangpreempt.go:210 — equal-priority jobs are not bundle
+    victims (the old OR fallback would still have evicted via the
+    plugin intersection)."""
+    store, binder, cache, sched = mk(
+        actions=["enqueue", "allocate", "gangpreempt", "backfill"])
+    for n in synth.make_nodes(2, cpu_milli=2000, mem=8 * GI):
+        store.create("Node", n)
+    store.create("Queue", synth.make_queue("default"))
+    synth.make_gang(store, "low", replicas=4, cpu_milli=1000, mem=GI,
+                    priority=50)
+    sched.run_once()
+    synth.make_gang(store, "high", replicas=3, cpu_milli=1000, mem=GI,
+                    priority=50)
+    sched.run_once()
+    assert binder.evictions == []
+
+
+def test_evict_task_credits_releasing():
+    """cache.evict_task must flip used→releasing so
+    future_idle sees the evicted capacity."""
+    from volcano_amd.api.info import NodeInfo, JobInfo, TaskInfo
+    from volcano_amd.api.resource import CPU
+    from volcano_amd.api.types import TaskStatus
+    cache = SchedulerCache(store=None, binder=FakeBinder())
+    node = NodeInfo(synth.make_node("n1", cpu_milli=4000, mem=8 * GI))
+    cache.add_node_info(node)
+    pg = synth.make_podgroup("j1")
+    job = JobInfo("default/j1", pg)
+    pod = synth.make_pod("j1-w-0", "j1", cpu_milli=1000, mem=GI,
+                         node_name="n1", phase="Running")
+    t = TaskInfo.from_pod(pod, job.key)
+    job.add_task(t)
+    cache.add_job_info(job)
+    used_before = node.used.get(CPU)
+    assert node.releasing.get(CPU) == 0
+    cache.evict_task(t, "test")
+    assert t.status == TaskStatus.RELEASING
+    assert node.releasing.get(CPU) == 1000      # credited
+    assert node.used.get(CPU) == used_before    # still held until released
+    fi = node.future_idle
+    assert fi.get(CPU) == 4000 - used_before + 1000

@@ -90,9 +90,12 @@ class AllocateAction:
             if not q.is_open or ssn.queue_overused(q):
                 continue
             # Allocatable is queue-scoped (quota gates; the exact per-task
-            # bound is enforced in-kernel) — hoist it out of the job loop
+            # bound is enforced in-kernel) — hoisted out of the job loop.
+            # CONTRACT: allocatable_fns must be job-independent (job=None
+            # is passed); a job-dependent gate belongs in the per-class
+            # plan constraints, not here.
             jobs_q = by_queue[q.name]
-            gate = ssn.allocatable(q, jobs_q[0]) if jobs_q else True
+            gate = ssn.allocatable(q, None) if jobs_q else True
             if not gate:
                 continue
             for job in ssn.sorted_jobs(jobs_q):

@@ -37,31 +37,47 @@ class GangPreemptAction:
             for job in ssn.sorted_jobs(starving):
                 self._gang_preempt(ssn, job)
 
-    def _victim_ok(self, ssn, preemptor_task, victim_job: JobInfo) -> bool:
-        """Whole-bundle evictability: EVERY occupied task of the victim
-        job must pass the victim callbacks (evicting all of them empties
-        the gang, so the gang-min protection is bypassed by design —
-        reference bundle semantics: the unit is the whole sub-job)."""
+    def _victim_ok(self, ssn, preemptor_task, preemptor_job: JobInfo,
+                   victim_job: JobInfo) -> bool:
+        """Whole-bundle evictability (reference gangpreempt.go:189-246):
+        EVERY occupied task of the victim job must pass the victim
+        plugin intersection — the plugin vetoes (PDB budget, conformance
+        kube-system protection, cooldown) are authoritative, with ONE
+        exception: the gang plugin's keep-above-minAvailable filter is
+        bypassed, because the unit of eviction is the whole bundle
+        (evicting all of a gang's tasks never strands it below min).
+        Additionally the reference requires (a) the victim job strictly
+        below the preemptor's priority, (b) the victim job not explicitly
+        marked non-preemptable (unset defaults to allowed —
+        utils.IsJobPreemptableForGangEviction, actions/utils/util.go:114),
+        and (c) no victim pod explicitly annotated non-preemptable
+        (GetPodPreemptable defaults true, api/pod_info.go:176)."""
         victims = victim_job.tasks_with_status(*VICTIM_STATUSES)
         if not victims:
             return False
-        if self.same_queue:
-            fns = [f for f in ssn.preemptable_fns]
-        else:
-            fns = [f for f in ssn.reclaimable_fns]
-        # priority/share/conformance checks apply per task; the gang-min
-        # filter is intentionally skipped (whole-bundle eviction)
-        for v in victims:
-            ok = False
-            allowed = {t.uid for t in ssn.preemptable(preemptor_task, [v])} \
-                if self.same_queue else \
-                {t.uid for t in ssn.reclaimable(preemptor_task, [v])}
-            # fall back: a task the tiered intersection admits OR whose
-            # job is entirely below the preemptor's priority
-            if v.uid in allowed or v.priority < preemptor_task.priority:
-                ok = True
-            if not ok:
+        if victim_job.priority >= preemptor_job.priority:
+            return False
+        pg = victim_job.podgroup
+        if pg is not None:
+            from ...api.objects import ANN_PREEMPTABLE
+            ann = pg.meta.annotations.get(ANN_PREEMPTABLE) or \
+                pg.meta.labels.get(ANN_PREEMPTABLE)
+            if ann is not None and ann != "true":
                 return False
+        for v in victims:
+            p = v.pod
+            if p is not None and (p.meta.annotations.get(
+                    "volcano.sh/preemptable") == "false"):
+                return False
+        fns = ssn.preemptable_fns if self.same_queue else ssn.reclaimable_fns
+        allowed = victims
+        for fn in fns:
+            if getattr(fn, "bundle_exempt", False):
+                continue    # gang-min protection — whole-bundle semantics
+            keep = {t.uid for t in fn(preemptor_task, allowed)}
+            allowed = [t for t in allowed if t.uid in keep]
+            if len(allowed) < len(victims):
+                return False    # some task vetoed → the bundle is not free
         return True
 
     def _gang_preempt(self, ssn, job: JobInfo) -> None:
@@ -96,7 +112,7 @@ class GangPreemptAction:
                 continue
             if not self.same_queue and vj.queue == job.queue:
                 continue
-            if self._victim_ok(ssn, rep, vj):
+            if self._victim_ok(ssn, rep, job, vj):
                 cands.append(vj)
         cands.sort(key=lambda j: (j.priority, len(j.tasks)))
 

@@ -355,12 +355,17 @@ class SchedulerCache:
     def evict_task(self, task: TaskInfo, reason: str = "") -> None:
         self.binder.evict(task, reason)
         job = self.jobs.get(task.job_key)
+        node = self.nodes.get(task.node_name)
+        # remove BEFORE the status flip (remove subtracts by the task's
+        # CURRENT status buckets), then re-add as RELEASING so the node's
+        # releasing total is credited — same order as Statement.evict
+        if node is not None:
+            node.remove_task(task)
         if job is not None:
             job.update_task_status(task, TaskStatus.RELEASING)
-        node = self.nodes.get(task.node_name)
+        else:
+            task.status = TaskStatus.RELEASING
         if node is not None:
-            # flip accounting used -> releasing
-            node.remove_task(task)
             node.add_task(task)
         self._used_dirty = True
 

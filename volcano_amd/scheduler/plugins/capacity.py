@@ -132,10 +132,20 @@ class CapacityPlugin(Plugin):
         des_np = deserved.numpy()
         total_np = total.numpy()
         mask = total_np > 0
-        tot64 = np.maximum(total_np.astype(np.float64), 1.0)
+        des64 = des_np.astype(np.float64)
 
         def q_share(qi: int) -> float:
-            return float((alloc_np[qi] / tot64).max())
+            # allocated/deserved per dim (reference capacity.go:1812-1824;
+            # no-deserved best-effort queues pin to share 1)
+            d = des64[qi][mask]
+            pos = d > 0
+            a = alloc_np[qi][mask].astype(np.float64)
+            if not pos.any():
+                return 1.0
+            s = float((a[pos] / d[pos]).max())
+            if (a[~pos] > 0.1).any():
+                s = max(s, 1.0)
+            return s
 
         def queue_order(a, b) -> int:
             # priority first (capacity honors queue priority), then share

@@ -59,6 +59,11 @@ class GangPlugin(Plugin):
                     taken[v.job_key] = taken.get(v.job_key, 0) + 1
             return out
 
+        # bundle actions (gangpreempt/gangreclaim) evict whole gangs, so
+        # the keep-above-min filter does not apply to them — they skip
+        # fns carrying this marker (actions/gangpreempt.py _victim_ok)
+        preemptable.bundle_exempt = True
+
         ssn.job_valid_fns.append(job_valid)
         ssn.add_job_order_fn(job_order, key=lambda j: j.is_ready())
         ssn.job_ready_fns.append(job_ready)

@@ -72,10 +72,23 @@ class ProportionPlugin(Plugin):
         des_np = deserved.numpy()
         total_np = total.numpy()
         mask = total_np > 0
-        tot64 = np.maximum(total_np.astype(np.float64), 1.0)
+
+        des64 = des_np.astype(np.float64)
 
         def q_share(qi: int) -> float:
-            return float((alloc_np[qi] / tot64).max())
+            # share_r = allocated_r / deserved_r (reference proportion.go
+            # updateQueueAttrShare); deserved==0 dims: share 1 if anything
+            # is allocated there, and a queue with NO deserved at all is
+            # share 1 (best-effort sorts last, capacity.go:1820)
+            d = des64[qi][mask]
+            pos = d > 0
+            a = alloc_np[qi][mask].astype(np.float64)
+            if not pos.any():
+                return 1.0
+            s = float((a[pos] / d[pos]).max())
+            if (a[~pos] > 0.1).any():
+                s = max(s, 1.0)
+            return s
 
         def queue_order(a, b) -> int:
             sa = q_share(ssn.queue_index[a.name])

@@ -139,4 +139,8 @@ class Statement:
             if op[0] == "evict":
                 _, victim, _, reason = op
                 cache.binder.evict(victim, reason)
+                # release hooks (deviceshare/numaaware pools) — the
+                # counterpart of fire_allocate (reference statement.go:402
+                # fires EventHandler.DeallocateFunc on commit)
+                self.ssn.fire_evict(victim)
         self.ops.clear()
