@@ -99,14 +99,9 @@ def reset_cluster(cache: SchedulerCache, jobs):
         job.task_status_index = {TaskStatus.PENDING: dict(job.tasks)}
         job._occ = 0                       # counter matches the fresh index
         job._alloc_vec = None
-    for ni in cache.nodes.values():
-        ni.used = Resource()
-        ni.releasing = Resource()
-        ni.pipelined = Resource()
-        ni.tasks.clear()
-    cache._used_dirty = True
+    cache.reset_usage()
     if isinstance(cache.binder, FakeBinder):
-        cache.binder.binds.clear()
+        cache.binder.clear()
 
 
 def main():
@@ -202,10 +197,10 @@ def main():
             # schedule-to-completion (convergence cost measured honestly)
             my_pods = len(jobs) * args.pods_per_job
             for _ in range(world + 3):
-                if len(cache.binder.binds) >= my_pods:
+                if cache.binder.bound_count >= my_pods:
                     break
                 sched.run_once()
-        return len(cache.binder.binds)
+        return cache.binder.bound_count
 
     def sync():
         if use_gpu:

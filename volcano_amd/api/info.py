@@ -414,7 +414,13 @@ class JobInfo:
 
 
 class NodeInfo:
-    """Per-node state (reference api/node_info.go:52-97)."""
+    """Per-node state (reference api/node_info.go:52-97).
+
+    Usage accounting (used/releasing/pipelined/remote_used) lives in the
+    cache-level :class:`~volcano_amd.api.ledger.NodeLedger` once the node
+    is adopted into the packed order; the properties below materialize
+    read-only ``Resource`` snapshots from the ledger row.  Mutations go
+    through the node methods (or ledger bulk ops) only."""
 
     def __init__(self, node: Node):
         self.node = node
@@ -434,59 +440,174 @@ class NodeInfo:
             self.oversubscription.q[MEMORY] = float(
                 ann["volcano.sh/oversubscription-memory"])
         self.allocatable.add(self.oversubscription)
-        self.used = Resource()
-        self.releasing = Resource()
-        self.pipelined = Resource()
-        # usage placed by OTHER scheduler ranks (soft sharding: tracked as
-        # an aggregate, not per-task — parallel/softshard.py)
-        self.remote_used = Resource()
-        self.tasks: Dict[str, TaskInfo] = {}
+        # pre-adoption local truth (ledger adoption copies these in)
+        self._used = Resource()
+        self._releasing = Resource()
+        self._pipelined = Resource()
+        self._remote_used = Resource()
+        self._ledger = None
+        self._row = -1
+        self._tasks: Dict[str, TaskInfo] = {}
+        # deferred membership batches (allocate commit appends task lists;
+        # the `tasks` property folds them on first read — cold paths only)
+        self._batches: List[List[TaskInfo]] = []
+
+    # -- usage views ---------------------------------------------------------
+    @property
+    def tasks(self) -> Dict[str, TaskInfo]:
+        b = self._batches
+        if b:
+            t = self._tasks
+            for ts in b:
+                for x in ts:
+                    t[x.key] = x
+            b.clear()
+        return self._tasks
+
+    def _plane(self, plane: int, local: Resource) -> Resource:
+        led = self._ledger
+        if led is None or self._row < 0:
+            return local
+        return led.resource(self._row, plane)
+
+    @property
+    def used(self) -> Resource:
+        from .ledger import USED
+        return self._plane(USED, self._used)
+
+    @used.setter
+    def used(self, r: Resource) -> None:
+        from .ledger import USED
+        if self._ledger is None or self._row < 0:
+            self._used = r
+        else:
+            self._ledger.set_row(self._row, USED, r)
+
+    @property
+    def releasing(self) -> Resource:
+        from .ledger import RELEASING
+        return self._plane(RELEASING, self._releasing)
+
+    @releasing.setter
+    def releasing(self, r: Resource) -> None:
+        from .ledger import RELEASING
+        if self._ledger is None or self._row < 0:
+            self._releasing = r
+        else:
+            self._ledger.set_row(self._row, RELEASING, r)
+
+    @property
+    def pipelined(self) -> Resource:
+        from .ledger import PIPELINED
+        return self._plane(PIPELINED, self._pipelined)
+
+    @pipelined.setter
+    def pipelined(self, r: Resource) -> None:
+        from .ledger import PIPELINED
+        if self._ledger is None or self._row < 0:
+            self._pipelined = r
+        else:
+            self._ledger.set_row(self._row, PIPELINED, r)
+
+    @property
+    def remote_used(self) -> Resource:
+        from .ledger import REMOTE
+        return self._plane(REMOTE, self._remote_used)
+
+    @remote_used.setter
+    def remote_used(self, r: Resource) -> None:
+        from .ledger import REMOTE
+        if self._ledger is None or self._row < 0:
+            self._remote_used = r
+        else:
+            self._ledger.set_row(self._row, REMOTE, r)
 
     @property
     def idle(self) -> Resource:
-        return self.allocatable.clone().sub(self.used)
+        led = self._ledger
+        if led is None or self._row < 0:
+            return self.allocatable.clone().sub(self._used)
+        from .ledger import USED
+        row = led.alloc[self._row] - led.planes[USED, self._row]
+        names = led.dims.names
+        out = Resource({names[i]: float(v) for i, v in enumerate(row)
+                        if v > 0.0 and i < len(names)})
+        return out
 
     @property
     def future_idle(self) -> Resource:
         """idle + releasing − pipelined (reference FutureIdle)."""
-        return self.allocatable.clone().sub(self.used).add(self.releasing).sub(self.pipelined)
+        led = self._ledger
+        if led is None or self._row < 0:
+            return self.allocatable.clone().sub(self._used) \
+                .add(self._releasing).sub(self._pipelined)
+        from .ledger import PIPELINED, RELEASING, USED
+        p = led.planes
+        row = (led.alloc[self._row] - p[USED, self._row]
+               + p[RELEASING, self._row] - p[PIPELINED, self._row])
+        names = led.dims.names
+        return Resource({names[i]: float(v) for i, v in enumerate(row)
+                         if v > 0.0 and i < len(names)})
 
     @property
     def ready(self) -> bool:
         return self.node.ready and not self.node.unschedulable
 
+    # -- accounting mutations -------------------------------------------------
+    def _acct(self, request: Resource, du: float, dr: float,
+              dp: float) -> None:
+        led = self._ledger
+        if led is not None and self._row >= 0:
+            led.apply(self._row, request.q, du, dr, dp)
+            return
+        if du > 0:
+            self._used.add(request)
+        elif du < 0:
+            self._used.sub(request)
+        if dr > 0:
+            self._releasing.add(request)
+        elif dr < 0:
+            self._releasing.sub(request)
+        if dp > 0:
+            self._pipelined.add(request)
+        elif dp < 0:
+            self._pipelined.sub(request)
+
     def add_task(self, task: TaskInfo) -> None:
-        prev = self.tasks.get(task.key)
+        tasks = self.tasks
+        prev = tasks.get(task.key)
         if prev is not None and prev is not task:
             self.remove_task(prev)     # never silently overwrite accounting
-        self.tasks[task.key] = task
+        tasks[task.key] = task
         if task.status.occupies_node:
-            self.used.add(task.request)
+            self._acct(task.request, 1, 0, 0)
         elif task.status == TaskStatus.RELEASING:
-            self.used.add(task.request)
-            self.releasing.add(task.request)
+            self._acct(task.request, 1, 1, 0)
         elif task.status == TaskStatus.PIPELINED:
-            self.pipelined.add(task.request)
+            self._acct(task.request, 0, 0, 1)
 
     def add_allocated_bulk(self, tasks: List[TaskInfo], request: Resource,
                            count: int) -> None:
-        """Bulk add for freshly-allocated identical tasks: ONE Resource op
+        """Bulk add for freshly-allocated identical tasks: ONE accounting op
         for the whole batch (hot path of the allocate commit)."""
-        for t in tasks:
-            self.tasks[t.key] = t
-        self.used.add(request.clone().multi(float(count)))
+        self._batches.append(tasks)
+        led = self._ledger
+        if led is not None and self._row >= 0:
+            led.apply(self._row, request.q, float(count), 0, 0)
+        else:
+            self._used.add(request.clone().multi(float(count)))
 
     def remove_task(self, task: TaskInfo) -> None:
-        if task.key not in self.tasks:
+        tasks = self.tasks
+        if task.key not in tasks:
             return
-        del self.tasks[task.key]
+        del tasks[task.key]
         if task.status.occupies_node:
-            self.used.sub(task.request)
+            self._acct(task.request, -1, 0, 0)
         elif task.status == TaskStatus.RELEASING:
-            self.used.sub(task.request)
-            self.releasing.sub(task.request)
+            self._acct(task.request, -1, -1, 0)
         elif task.status == TaskStatus.PIPELINED:
-            self.pipelined.sub(task.request)
+            self._acct(task.request, 0, 0, -1)
 
 
 class QueueInfo:

@@ -119,16 +119,30 @@ class SoftShardCoordinator:
         # fold remote usage into the host mirror so future packs and the
         # preempt path see the global state
         remote_nr = remote.t().cpu().numpy()             # [N, R]
-        names = nt.dims.names
-        nodes_sorted = getattr(ssn.cache, "nodes_sorted", None)
-        if nodes_sorted is None or len(nodes_sorted) != len(ssn.nodes):
-            nodes_sorted = sorted(ssn.nodes.values(), key=lambda n: n.name)
         import numpy as np
-        hot = np.nonzero(np.abs(remote_nr).sum(axis=1) > 1e-6)[0]
-        for i in hot:
-            ni = nodes_sorted[int(i)]
-            for r, name in enumerate(names):
-                v = float(remote_nr[i, r])
-                if v:
-                    ni.remote_used.q[name] = ni.remote_used.q.get(name, 0.0) + v
-                    ni.used.q[name] = ni.used.q.get(name, 0.0) + v
+        ledger = getattr(ssn.cache, "ledger", None)
+        if ledger is not None and ledger.n == remote_nr.shape[0]:
+            # vectorized fold into the columnar accounting (both the used
+            # truth and the remote-attribution plane)
+            from ..api.ledger import REMOTE, USED
+            R = remote_nr.shape[1]
+            if ledger.width < R:
+                ledger._widen(R)
+            ledger.planes[USED, :, :R] += remote_nr
+            ledger.planes[REMOTE, :, :R] += remote_nr
+            ledger.version += 1
+        else:
+            from ..api.resource import Resource
+            names = nt.dims.names
+            nodes_sorted = getattr(ssn.cache, "nodes_sorted", None)
+            if nodes_sorted is None or len(nodes_sorted) != len(ssn.nodes):
+                nodes_sorted = sorted(ssn.nodes.values(),
+                                      key=lambda n: n.name)
+            hot = np.nonzero(np.abs(remote_nr).sum(axis=1) > 1e-6)[0]
+            for i in hot:
+                ni = nodes_sorted[int(i)]
+                delta = Resource({names[r]: float(remote_nr[i, r])
+                                  for r in range(len(names))
+                                  if remote_nr[i, r]})
+                ni._acct(delta, 1, 0, 0)
+                ni._remote_used.add(delta)

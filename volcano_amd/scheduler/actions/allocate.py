@@ -530,24 +530,33 @@ class AllocateAction:
         nodes_sorted = getattr(ssn.cache, "nodes_sorted", None)
         if nodes_sorted is None or len(nodes_sorted) != len(ssn.nodes):
             nodes_sorted = sorted(ssn.nodes.values(), key=lambda n: n.name)
+        ledger = getattr(ssn.cache, "ledger", None)
 
         to_bind = []
         bind_by_job: Dict[str, List] = {}
         committed_jobs = set()
+        # fire event handlers only when someone registered one — building
+        # the per-piece argument lists for nobody was measurable at 10k jobs
+        fire = ssn.fire_allocate if ssn.event_handlers else None
 
-        def commit_pieces(job, cp, pieces):
-            """Assign (node_id, count) pieces to the next tasks of `job`;
-            bulk accounting: one Resource op per piece.  Status moves once,
-            PENDING→BOUND, inside bind_tasks (the reference's
-            Allocated→Binding→Bound pipeline compressed — nothing observes
-            the intermediate states between plan apply and bind here)."""
+        def commit_pieces(job, cp, pieces, led_rows, led_counts):
+            """Assign (node_id, count) pieces to the next tasks of `job`.
+            Node accounting is ledger-vectorized per CLASS (led_rows/
+            led_counts accumulate here, one np.add.at after the walk);
+            node→task membership defers into NodeInfo._batches (folded
+            lazily by cold readers).  Status moves once, PENDING→BOUND,
+            inside bind_tasks (the reference's Allocated→Binding→Bound
+            pipeline compressed — nothing observes the intermediate
+            states between plan apply and bind here)."""
             jl = bind_by_job.setdefault(job.key, [])
             for nid, count, tasks in pieces:
                 ni = nodes_sorted[nid]
                 name = ni.name
                 for t in tasks:
                     t.node_name = name
-                ni.add_allocated_bulk(tasks, cp.tclass.request, count)
+                ni._batches.append(tasks)
+                led_rows.append(nid)
+                led_counts.append(count)
                 to_bind.extend(tasks)
                 jl.extend(tasks)
             committed_jobs.add(job.key)
@@ -557,6 +566,8 @@ class AllocateAction:
                 continue
             if not cres.placements:
                 continue
+            led_rows: List[int] = []
+            led_counts: List[int] = []
             if cp.bundle is None:
                 job = ssn.jobs.get(cp.job_key) or \
                     ssn.jobs[getattr(plan, "job_alias", {})[cp.job_key]]
@@ -568,58 +579,71 @@ class AllocateAction:
                     self._revert_pieces(plan, cp,
                                         [(nid, cnt) for nid, cnt, _ in pieces])
                     continue
-                commit_pieces(job, cp, pieces)
-                ssn.fire_allocate(cp.tclass,
-                                  [nid for nid, _ in cres.placements],
-                                  [cnt for _, cnt in cres.placements],
-                                  [t for _, _, ts in pieces for t in ts])
-                continue
-
-            # -- bundle: walk jobs over the placement stream.  An entry
-            # that misses its gang minimum releases its slots back to the
-            # pool (cursor reset) so smaller jobs behind it still place;
-            # an entry touching a lost node (soft-shard conflict) reverts
-            # its slots outright.  Unclaimed slots revert at the end.
-            stream = list(cres.placements)
-            ei, eoff = 0, 0
-            reverted: List[tuple] = []
-            for be in cp.bundle:
-                ei0, eoff0 = ei, eoff
-                pieces = []
-                need = be.ntasks
-                toff = 0
-                while need > 0 and ei < len(stream):
-                    nid, cnt = stream[ei]
-                    avail = cnt - eoff
-                    take = min(avail, need)
-                    pieces.append((nid, take, be.tasks[toff:toff + take]))
-                    toff += take
-                    need -= take
-                    eoff += take
-                    if eoff == cnt:
+                commit_pieces(job, cp, pieces, led_rows, led_counts)
+                if fire is not None:
+                    fire(cp.tclass,
+                         [nid for nid, _ in cres.placements],
+                         [cnt for _, cnt in cres.placements],
+                         [t for _, _, ts in pieces for t in ts])
+            else:
+                # -- bundle: walk jobs over the placement stream.  An entry
+                # that misses its gang minimum releases its slots back to
+                # the pool (cursor reset) so smaller jobs behind it still
+                # place; an entry touching a lost node (soft-shard
+                # conflict) reverts its slots outright.  Unclaimed slots
+                # revert at the end.
+                stream = cres.placements
+                ei, eoff = 0, 0
+                nstream = len(stream)
+                reverted: List[tuple] = []
+                for be in cp.bundle:
+                    ei0, eoff0 = ei, eoff
+                    pieces = []
+                    need = be.ntasks
+                    toff = 0
+                    while need > 0 and ei < nstream:
+                        nid, cnt = stream[ei]
+                        avail = cnt - eoff
+                        take = avail if avail < need else need
+                        pieces.append((nid, take, be.tasks[toff:toff + take]))
+                        toff += take
+                        need -= take
+                        eoff += take
+                        if eoff == cnt:
+                            ei += 1
+                            eoff = 0
+                    if be.ntasks - need < be.min_needed:
+                        ei, eoff = ei0, eoff0      # recycle the slots
+                        continue
+                    if bad_nodes and any(nid in bad_nodes
+                                         for nid, _, _ in pieces):
+                        reverted.extend((nid, cnt) for nid, cnt, _ in pieces)
+                        continue
+                    job = ssn.jobs[be.job_key]
+                    commit_pieces(job, cp, pieces, led_rows, led_counts)
+                    if fire is not None:
+                        fire(cp.tclass, [p[0] for p in pieces],
+                             [p[1] for p in pieces],
+                             [t for _, _, ts in pieces for t in ts])
+                if ei < nstream:
+                    # slots the walk never claimed
+                    if eoff:
+                        reverted.append((stream[ei][0], stream[ei][1] - eoff))
                         ei += 1
-                        eoff = 0
-                got = be.ntasks - need
-                if got < be.min_needed:
-                    ei, eoff = ei0, eoff0      # recycle the slots
-                    continue
-                if bad_nodes and any(nid in bad_nodes
-                                     for nid, _, _ in pieces):
-                    reverted.extend((nid, cnt) for nid, cnt, _ in pieces)
-                    continue
-                job = ssn.jobs[be.job_key]
-                commit_pieces(job, cp, pieces)
-                ssn.fire_allocate(cp.tclass, [p[0] for p in pieces],
-                                  [p[1] for p in pieces],
-                                  [t for _, _, ts in pieces for t in ts])
-            if ei < len(stream):
-                # slots the walk never claimed
-                if eoff:
-                    reverted.append((stream[ei][0], stream[ei][1] - eoff))
-                    ei += 1
-                reverted.extend(stream[ei:])
-            if reverted:
-                self._revert_pieces(plan, cp, reverted)
+                    reverted.extend(stream[ei:])
+                if reverted:
+                    self._revert_pieces(plan, cp, reverted)
+            if led_rows:
+                if ledger is not None:
+                    ledger.add_used_bulk(
+                        np.asarray(led_rows, dtype=np.int64),
+                        np.asarray(led_counts, dtype=np.int64),
+                        cp.req)
+                else:   # no ledger (bare-cache tests): per-piece Resource math
+                    for nid, cnt in zip(led_rows, led_counts):
+                        nodes_sorted[nid]._acct(
+                            cp.tclass.request.clone().multi(float(cnt)),
+                            1, 0, 0)
 
         if to_bind:
             ssn.cache.bind_tasks(to_bind, by_job=bind_by_job)

@@ -38,14 +38,36 @@ class Binder:
 class FakeBinder(Binder):
     """Records binds/evictions (reference util/test_utils.go:536 FakeBinder)
     — also the bench-mode sink so the GPU loop, not a cluster API, is
-    measured (SURVEY.md §7 'binding throughput')."""
+    measured (SURVEY.md §7 'binding throughput').
+
+    Intake is batched: ``bind`` stores the task list; the ``binds`` dict
+    materializes lazily on read (tests) so the hot path pays no per-task
+    dict stores.  ``bound_count`` is the O(1) bench-side counter."""
 
     def __init__(self):
-        self.binds: Dict[str, str] = {}
+        self._binds: Dict[str, str] = {}
+        self._spans: List[List[TaskInfo]] = []
+        self.bound_count = 0
         self.evictions: List[str] = []
 
+    @property
+    def binds(self) -> Dict[str, str]:
+        if self._spans:
+            d = self._binds
+            for ts in self._spans:
+                for t in ts:
+                    d[t.key] = t.node_name
+            self._spans.clear()
+        return self._binds
+
     def bind(self, tasks: List[TaskInfo]) -> None:
-        self.binds.update((t.key, t.node_name) for t in tasks)
+        self._spans.append(tasks)
+        self.bound_count += len(tasks)
+
+    def clear(self) -> None:
+        self._binds.clear()
+        self._spans.clear()
+        self.bound_count = 0
 
     def evict(self, task: TaskInfo, reason: str = "") -> None:
         self.evictions.append(task.key)
@@ -322,16 +344,42 @@ class SchedulerCache:
         # 10k+ NodeInfos once per action)
         self.nodes_sorted = nodes
         if self._tensors_dirty or self.node_tensors.alloc_t is None:
-            self.node_tensors.pack(nodes)
+            # (re)adopt the node set into the columnar usage ledger first
+            # so the pack reads usage as one vectorized plane copy
+            from ..api.ledger import NodeLedger
+            self.ledger = NodeLedger.build(self.dims, nodes)
+            self.node_tensors.pack(nodes, self.ledger)
             self._tensors_dirty = self._used_dirty = False
             for i, ni in enumerate(nodes):
                 ni.node_id = i
         elif self._used_dirty:
-            if not self.node_tensors.pack_dynamic(nodes):
-                self.node_tensors.pack(nodes)
+            if not self.node_tensors.pack_dynamic(
+                    nodes, getattr(self, "ledger", None)):
+                from ..api.ledger import NodeLedger
+                self.ledger = NodeLedger.build(self.dims, nodes)
+                self.node_tensors.pack(nodes, self.ledger)
                 for i, ni in enumerate(nodes):
                     ni.node_id = i
             self._tensors_dirty = self._used_dirty = False
+
+    def reset_usage(self) -> None:
+        """Zero every node's usage accounting and task membership in one
+        vectorized pass (bench/soak step reset; also the crash-resume
+        rebuild path).  The ledger planes zero wholesale; unadopted nodes
+        fall back to fresh local Resources."""
+        from ..api.resource import Resource
+        led = getattr(self, "ledger", None)
+        if led is not None:
+            led.zero_usage()
+        for ni in self.nodes.values():
+            if ni._ledger is None or ni._row < 0:
+                ni._used = Resource()
+                ni._releasing = Resource()
+                ni._pipelined = Resource()
+                ni._remote_used = Resource()
+            ni._tasks.clear()
+            ni._batches.clear()
+        self._used_dirty = True
 
     # -- commit pipeline ------------------------------------------------------
     def bind_tasks(self, tasks: List[TaskInfo],

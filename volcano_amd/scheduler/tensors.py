@@ -145,12 +145,16 @@ class NodeTensors:
         return require, forbid
 
     # -- packing -------------------------------------------------------------
-    def pack_dynamic(self, nodes: List[NodeInfo]) -> bool:
+    def pack_dynamic(self, nodes: List[NodeInfo], ledger=None) -> bool:
         """Re-pack only the per-cycle-mutable planes (used/extra) when the
         static planes (allocatable/labels/taints/ready) are still valid —
         the common steady-state path: pod churn changes usage, node churn
         is rare (SURVEY §7 'snapshot cost' hard-part).  Returns False if a
-        full pack is required."""
+        full pack is required.
+
+        With a :class:`NodeLedger` (the steady state) this is a pure
+        vectorized astype/transpose of the ledger planes — no per-node
+        work at all."""
         if self.alloc_t is None or len(nodes) != self.n:
             return False
         if self.alloc_t.shape[0] != self.r:
@@ -158,6 +162,16 @@ class NodeTensors:
         if [ni.name for ni in nodes] != self.names:
             return False
         N, R = self.n, self.r
+        if ledger is not None and ledger.n == N and ledger.width >= R:
+            from ..api.ledger import PIPELINED, RELEASING, USED
+            p = ledger.planes
+            used = p[USED, :, :R].T.astype(np.float32)
+            extra = (p[RELEASING, :, :R] - p[PIPELINED, :, :R]) \
+                .T.astype(np.float32)
+            dev = self.device
+            self.used_t.copy_(torch.from_numpy(np.ascontiguousarray(used)).to(dev))
+            self.extra_t.copy_(torch.from_numpy(np.ascontiguousarray(extra)).to(dev))
+            return True
         used = np.zeros((N, R), dtype=np.float32)
         extra = np.zeros((N, R), dtype=np.float32)
         didx = self.dims.index
@@ -181,7 +195,7 @@ class NodeTensors:
         self.extra_t.copy_(torch.from_numpy(extra.T.copy()).to(dev))
         return True
 
-    def pack(self, nodes: List[NodeInfo]) -> None:
+    def pack(self, nodes: List[NodeInfo], ledger=None) -> None:
         """Full (re)pack from host NodeInfos.
 
         Vectorized via numpy staging buffers then one H2D per plane; at
@@ -226,6 +240,14 @@ class NodeTensors:
                 alloc[:, j] = 1.0
         if nvl_j is not None:
             alloc[:, nvl_j] = 256.0
+        use_ledger = ledger is not None and ledger.n == N \
+            and ledger.width >= R
+        if use_ledger:
+            from ..api.ledger import PIPELINED, RELEASING, USED
+            p = ledger.planes
+            used[:] = p[USED, :, :R].astype(np.float32)
+            extra[:] = (p[RELEASING, :, :R] - p[PIPELINED, :, :R]) \
+                .astype(np.float32)
         for i, ni in enumerate(nodes):
             if nvl_j is not None:
                 lim = ni.node.meta.annotations.get("volcano.sh/max-volumes")
@@ -233,12 +255,13 @@ class NodeTensors:
                     alloc[i, nvl_j] = float(lim)
             for k, v in ni.allocatable.q.items():
                 alloc[i, didx[k]] = v
-            for k, v in ni.used.q.items():
-                used[i, didx[k]] = v
-            for k, v in ni.releasing.q.items():
-                extra[i, didx[k]] += v
-            for k, v in ni.pipelined.q.items():
-                extra[i, didx[k]] -= v
+            if not use_ledger:
+                for k, v in ni.used.q.items():
+                    used[i, didx[k]] = v
+                for k, v in ni.releasing.q.items():
+                    extra[i, didx[k]] += v
+                for k, v in ni.pipelined.q.items():
+                    extra[i, didx[k]] -= v
             ready[i] = 1 if ni.ready else 0
             for t in ni.node.taints:
                 if t.effect in _BLOCKING_EFFECTS:
