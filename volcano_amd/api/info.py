@@ -170,6 +170,8 @@ class JobInfo:
         self._atom = None
         self._minav = None
         self._occ = 0                        # occupied-count (incremental)
+        self._tver = 0                       # static-shape version (JobTable)
+        self._jrow = -1                      # row in the cycle JobTable
 
     # -- basic accessors ----------------------------------------------------
     @property
@@ -224,6 +226,7 @@ class JobInfo:
             self._occ += 1
         self._alloc_vec = self._total_vec = self._atom = None
         self._prio = None
+        self._tver += 1
         if task.gated:
             self._gated = getattr(self, "_gated", 0) + 1
 
@@ -238,6 +241,7 @@ class JobInfo:
             if t.status in _OCC_SET:
                 self._occ -= 1
             self._alloc_vec = self._total_vec = self._atom = None
+            self._tver += 1
         return t
 
     def plan_atom(self):

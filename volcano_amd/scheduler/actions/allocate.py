@@ -73,33 +73,70 @@ class AllocateAction:
         predicates = getattr(ssn, "predicates", None)
 
         # -- build the worksheet (buildAllocateContext analog) --------------
-        by_queue = {}
-        for job in ssn.jobs.values():
-            if job.phase not in (PodGroupPhase.INQUEUE.value,
-                                 PodGroupPhase.RUNNING.value):
-                continue
-            if not job.task_status_index.get(PENDING_S):
-                continue      # no list build — emptiness check only
-            if not ssn.job_valid(job):
-                continue
-            by_queue.setdefault(job.queue, []).append(job)
-
-        queues = [ssn.queues[q] for q in by_queue if q in ssn.queues]
+        # Columnar path: selection (phase/pending/valid), queue grouping
+        # and job ordering all run as array ops over the JobTable; the
+        # legacy per-job path remains for unaligned tables (mid-cycle
+        # direct mutations in tests).
+        jt = getattr(ssn, "job_table", None)
+        if jt is not None and len(jt.jobs) != len(ssn.jobs):
+            jt = None
         ordered_jobs = []
-        for q in ssn.sorted_queues(queues):
-            if not q.is_open or ssn.queue_overused(q):
-                continue
-            # Allocatable is queue-scoped (quota gates; the exact per-task
-            # bound is enforced in-kernel) — hoisted out of the job loop.
-            # CONTRACT: allocatable_fns must be job-independent (job=None
-            # is passed); a job-dependent gate belongs in the per-class
-            # plan constraints, not here.
-            jobs_q = by_queue[q.name]
-            gate = ssn.allocatable(q, None) if jobs_q else True
-            if not gate:
-                continue
-            for job in ssn.sorted_jobs(jobs_q):
-                ordered_jobs.append((q, job))
+        if jt is not None:
+            from ..jobtable import PH_INQUEUE, PH_RUNNING
+            mask = ((jt.phase == PH_INQUEUE) | (jt.phase == PH_RUNNING)) \
+                & (jt.npend > 0)
+            vcols = ssn.job_valid_cols
+            if len(vcols) == len(ssn.job_valid_fns):
+                for fn in vcols:
+                    mask &= fn(jt)
+                sel = np.nonzero(mask)[0]
+            else:
+                sel = np.nonzero(mask)[0]
+                if len(sel):
+                    keep = np.fromiter(
+                        (ssn.job_valid(jt.jobs[int(k)]) for k in sel),
+                        dtype=bool, count=len(sel))
+                    sel = sel[keep]
+            if len(sel):
+                qis = jt.qi[sel]
+                qname_by_qi = {i: n for n, i in ssn.queue_index.items()}
+                present = np.unique(qis[qis >= 0])
+                queues = [ssn.queues[qname_by_qi[int(i)]] for i in present
+                          if qname_by_qi.get(int(i)) in ssn.queues]
+                jjobs = jt.jobs
+                for q in ssn.sorted_queues(queues):
+                    if not q.is_open or ssn.queue_overused(q):
+                        continue
+                    # Allocatable is queue-scoped (quota gates; the exact
+                    # per-task bound is enforced in-kernel).  CONTRACT:
+                    # allocatable_fns must be job-independent (job=None).
+                    if not ssn.allocatable(q, None):
+                        continue
+                    qi = ssn.queue_index[q.name]
+                    rows = ssn.ordered_job_rows(jt, sel[qis == qi])
+                    ordered_jobs.extend((q, jjobs[int(k)]) for k in rows)
+        else:
+            by_queue = {}
+            for job in ssn.jobs.values():
+                if job.phase not in (PodGroupPhase.INQUEUE.value,
+                                     PodGroupPhase.RUNNING.value):
+                    continue
+                if not job.task_status_index.get(PENDING_S):
+                    continue      # no list build — emptiness check only
+                if not ssn.job_valid(job):
+                    continue
+                by_queue.setdefault(job.queue, []).append(job)
+
+            queues = [ssn.queues[q] for q in by_queue if q in ssn.queues]
+            for q in ssn.sorted_queues(queues):
+                if not q.is_open or ssn.queue_overused(q):
+                    continue
+                jobs_q = by_queue[q.name]
+                gate = ssn.allocatable(q, None) if jobs_q else True
+                if not gate:
+                    continue
+                for job in ssn.sorted_jobs(jobs_q):
+                    ordered_jobs.append((q, job))
 
         # Gang bundling: a run of consecutive jobs whose single pending
         # class has identical (queue, signature) is fused into ONE kernel

@@ -19,6 +19,10 @@ class EnqueueAction:
     name = "enqueue"
 
     def execute(self, ssn) -> None:
+        jt = getattr(ssn, "job_table", None)
+        if jt is not None and len(jt.jobs) == len(ssn.jobs):
+            self._execute_columnar(ssn, jt)
+            return
         by_queue = {}
         for job in ssn.jobs.values():
             if job.phase != PodGroupPhase.PENDING.value:
@@ -28,36 +32,78 @@ class EnqueueAction:
         for q in ssn.sorted_queues(queues):
             if not q.is_open:
                 continue
-            fifo = q.queue.spec.dequeue_strategy == "fifo"
             jobs_q = by_queue[q.name]
-            # bulk fast path: when every enqueueable plugin can admit the
-            # whole batch (total demand fits), skip the per-job votes —
-            # identical outcome for monotone-sum admission
-            bulk = ssn.job_enqueueable_bulk_fns
-            if bulk and len(bulk) == len(ssn.job_enqueueable_fns):
-                # each fn returns a commit thunk iff the whole batch fits;
-                # commit only once EVERY plugin agreed (no partial updates)
-                commits = [fn(q.name, jobs_q) for fn in bulk]
-                if all(c is not None for c in commits):
-                    for c in commits:
-                        c()
-                    for job in jobs_q:
-                        if job.podgroup is not None:
-                            job.podgroup.status.phase = \
-                                PodGroupPhase.INQUEUE.value
-                        ssn.cache.update_podgroup(job)
-                        self._lift_queue_gates(ssn, job)
-                    continue
-            for job in ssn.sorted_jobs(jobs_q):
-                if ssn.job_enqueueable(job):
+            self._admit_queue(ssn, q, jobs_q, rows=None, jt=None)
+
+    def _execute_columnar(self, ssn, jt) -> None:
+        """Selection + grouping over the JobTable columns; admission per
+        queue in queue order (same semantics as the per-job path)."""
+        import numpy as np
+        from ..jobtable import PH_PENDING
+        sel = np.nonzero(jt.phase == PH_PENDING)[0]
+        if not len(sel):
+            return
+        qis = jt.qi[sel]
+        qname_by_qi = {i: n for n, i in ssn.queue_index.items()}
+        present = np.unique(qis[qis >= 0])
+        queues = [ssn.queues[qname_by_qi[int(i)]] for i in present
+                  if qname_by_qi.get(int(i)) in ssn.queues]
+        for q in ssn.sorted_queues(queues):
+            if not q.is_open:
+                continue
+            qi = ssn.queue_index[q.name]
+            rows = sel[qis == qi]
+            self._admit_queue(ssn, q, None, rows=rows, jt=jt)
+
+    def _admit_queue(self, ssn, q, jobs_q, rows=None, jt=None) -> None:
+        if jobs_q is None:
+            jobs_q = [jt.jobs[int(k)] for k in rows]
+        if not jobs_q:
+            return
+        fifo = q.queue.spec.dequeue_strategy == "fifo"
+        # bulk fast path: when every enqueueable plugin can admit the
+        # whole batch (total demand fits), skip the per-job votes —
+        # identical outcome for monotone-sum admission
+        bulk = ssn.job_enqueueable_bulk_fns
+        store = getattr(ssn.cache, "store", None)
+        if bulk and len(bulk) == len(ssn.job_enqueueable_fns):
+            # each fn returns a commit thunk iff the whole batch fits;
+            # commit only once EVERY plugin agreed (no partial updates)
+            commits = [fn(q.name, jobs_q, rows=rows, table=jt)
+                       for fn in bulk]
+            if all(c is not None for c in commits):
+                for c in commits:
+                    c()
+                inq = PodGroupPhase.INQUEUE.value
+                from ..jobtable import PH_INQUEUE
+                for job in jobs_q:
                     if job.podgroup is not None:
-                        job.podgroup.status.phase = PodGroupPhase.INQUEUE.value
-                    for fn in getattr(ssn, "job_enqueued_fns", []):
-                        fn(job)
-                    ssn.cache.update_podgroup(job)
-                    self._lift_queue_gates(ssn, job)
-                elif fifo:
-                    break   # head-of-line blocks the queue (dequeueStrategy)
+                        job.podgroup.status.phase = inq
+                    if jt is not None:
+                        jt.phase[job._jrow] = PH_INQUEUE
+                    if store is not None:
+                        ssn.cache.update_podgroup(job)
+                    if getattr(job, "_gated", 0):
+                        self._lift_queue_gates(ssn, job)
+                return
+        if rows is not None:
+            rows = ssn.ordered_job_rows(jt, rows)
+            ordered = (jt.jobs[int(k)] for k in rows)
+        else:
+            ordered = ssn.sorted_jobs(jobs_q)
+        from ..jobtable import PH_INQUEUE
+        for job in ordered:
+            if ssn.job_enqueueable(job):
+                if job.podgroup is not None:
+                    job.podgroup.status.phase = PodGroupPhase.INQUEUE.value
+                if jt is not None:
+                    jt.phase[job._jrow] = PH_INQUEUE
+                for fn in getattr(ssn, "job_enqueued_fns", []):
+                    fn(job)
+                ssn.cache.update_podgroup(job)
+                self._lift_queue_gates(ssn, job)
+            elif fifo:
+                break   # head-of-line blocks the queue (dequeueStrategy)
 
     @staticmethod
     def _lift_queue_gates(ssn, job) -> None:

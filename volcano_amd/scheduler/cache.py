@@ -143,6 +143,9 @@ class SchedulerCache:
         self.node_tensors = NodeTensors(self.dims, device=device)
 
         self.jobs: Dict[str, JobInfo] = {}
+        self.jobs_epoch = 0                     # bumped on job add/remove
+        from .jobtable import JobTable
+        self.job_table = JobTable()
         self.nodes: Dict[str, NodeInfo] = {}
         self.queues: Dict[str, QueueInfo] = {}
         self._task_node: Dict[str, str] = {}    # task key -> node name
@@ -192,12 +195,14 @@ class SchedulerCache:
             self._task_ref.clear()
             self._watch = self.store.watch("Pod", "Node", "PodGroup", "Queue")
             self._tensors_dirty = True
+            self.jobs_epoch += 1
         return self.sync()
 
     def _job_for(self, pg_key: str) -> JobInfo:
         job = self.jobs.get(pg_key)
         if job is None:
             job = self.jobs[pg_key] = JobInfo(pg_key)
+            self.jobs_epoch += 1
         return job
 
     def _on_pod(self, ev) -> None:
@@ -271,11 +276,13 @@ class SchedulerCache:
         pg: PodGroup = ev.obj
         key = pg.meta.key
         if ev.type == EventType.DELETED:
-            self.jobs.pop(key, None)
+            if self.jobs.pop(key, None) is not None:
+                self.jobs_epoch += 1
             return
         job = self._job_for(key)
         job.podgroup = pg
         job._minav = None      # minMember may have changed
+        job._tver += 1
 
     def _on_queue(self, ev) -> None:
         q: Queue = ev.obj
@@ -291,6 +298,7 @@ class SchedulerCache:
 
     def add_job_info(self, job: JobInfo) -> None:
         self.jobs[job.key] = job
+        self.jobs_epoch += 1
         for t in job.tasks.values():
             if t.node_name and t.node_name in self.nodes:
                 self.nodes[t.node_name].add_task(t)
@@ -314,6 +322,10 @@ class SchedulerCache:
         ssn.queues = self.queues
         ssn.node_tensors = self.node_tensors
         ssn.total_resource = self.node_tensors.alloc_t.sum(dim=1).to("cpu")
+        ssn.queue_index = {name: i
+                           for i, name in enumerate(sorted(self.queues))}
+        self.job_table.refresh(self, self.node_tensors, ssn.queue_index)
+        ssn.job_table = self.job_table
 
     def _demote_pipelined(self) -> None:
         """Pipelined reservations live one cycle: at the next snapshot they

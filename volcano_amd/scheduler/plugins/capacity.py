@@ -201,13 +201,17 @@ class CapacityPlugin(Plugin):
         ssn.add_queue_order_fn(
             queue_order,
             key=lambda q: (-q.priority, q_share(qi_of[q.name])))
-        def job_enqueueable_bulk(qname, jobs):
+        def job_enqueueable_bulk(qname, jobs, rows=None, table=None):
             qi = qi_of.get(qname)
             if qi is None:
                 return None
-            demand = np.zeros(R, dtype=np.float64)
-            for j in jobs:
-                demand += j.minres_vec(nt)
+            if rows is not None and table is not None \
+                    and table.minres.shape[1] == R:
+                demand = table.minres[rows].sum(axis=0)
+            else:
+                demand = np.zeros(R, dtype=np.float64)
+                for j in jobs:
+                    demand += j.minres_vec(nt)
             head = (alloc_np[qi][mask].astype(np.float64)
                     + inqueue_np[qi][mask] + demand[mask])
             des = des_np[qi][mask].astype(np.float64)

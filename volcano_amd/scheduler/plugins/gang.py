@@ -65,7 +65,11 @@ class GangPlugin(Plugin):
         preemptable.bundle_exempt = True
 
         ssn.job_valid_fns.append(job_valid)
-        ssn.add_job_order_fn(job_order, key=lambda j: j.is_ready())
+        ssn.job_valid_cols.append(
+            lambda jt: (jt.ntasks - jt.nfailed) >= jt.minav)
+        ssn.add_job_order_fn(job_order, key=lambda j: j.is_ready(),
+                             col=lambda jt, rows:
+                                 jt.occ[rows] >= jt.minav[rows])
         ssn.job_ready_fns.append(job_ready)
         ssn.job_pipelined_fns.append(job_pipelined)
         ssn.job_starving_fns.append(job_starving)

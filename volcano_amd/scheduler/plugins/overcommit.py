@@ -36,13 +36,17 @@ class OvercommitPlugin(Plugin):
         def job_enqueued(job) -> None:
             inqueue[:] += job.minres_vec(nt)
 
-        def job_enqueueable_bulk(qname, jobs):
+        def job_enqueueable_bulk(qname, jobs, rows=None, table=None):
             """Whole-batch admission: total demand fits <=> every prefix
             fits (monotone sum), so admitting all == per-job votes.
             Returns a commit thunk (run only when all plugins agree)."""
-            demand = np.zeros(nt.r, dtype=np.float64)
-            for j in jobs:
-                demand += j.minres_vec(nt)
+            if rows is not None and table is not None \
+                    and table.minres.shape[1] == nt.r:
+                demand = table.minres[rows].sum(axis=0)
+            else:
+                demand = np.zeros(nt.r, dtype=np.float64)
+                for j in jobs:
+                    demand += j.minres_vec(nt)
             head = inqueue[mask] + demand[mask]
             if bool((head <= total[mask] + 0.1 + 1e-6 * total[mask]).all()):
                 return lambda: inqueue.__iadd__(demand)

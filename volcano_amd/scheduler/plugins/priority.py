@@ -28,7 +28,8 @@ class PriorityPlugin(Plugin):
         def job_starving(job) -> bool:
             return job.is_starving()
 
-        ssn.add_job_order_fn(job_order, key=lambda j: -j.priority)
+        ssn.add_job_order_fn(job_order, key=lambda j: -j.priority,
+                             col=lambda jt, rows: -jt.prio[rows])
         ssn.add_task_order_fn(task_order)
         ssn.preemptable_fns.append(preemptable)
         ssn.job_starving_fns.append(job_starving)

@@ -50,10 +50,18 @@ class ProportionPlugin(Plugin):
             for r in range(R):
                 if cap_vec[r] > 0:
                     capability[i, r] = float(cap_vec[r])
-        for job in ssn.jobs.values():
-            qi = ssn.queue_index.get(job.queue)
-            if qi is not None:
-                req_np[qi] += job.total_req_vec(nt)
+        jt = getattr(ssn, "job_table", None)
+        if jt is None or len(jt.jobs) != len(ssn.jobs) \
+                or jt.totreq.shape[1] != R:
+            jt = None
+        if jt is not None:
+            ok = jt.qi >= 0
+            np.add.at(req_np, jt.qi[ok], jt.totreq[ok])
+        else:
+            for job in ssn.jobs.values():
+                qi = ssn.queue_index.get(job.queue)
+                if qi is not None:
+                    req_np[qi] += job.total_req_vec(nt)
         request = torch.from_numpy(req_np).to(torch.float32)
 
         deserved = ref.waterfill(weight, request, guarantee, capability, total)
@@ -109,11 +117,16 @@ class ProportionPlugin(Plugin):
         # allocated + already-admitted-but-unscheduled against deserved)
         from ...api.types import PodGroupPhase
         inqueue_np = np.zeros((Q, R), dtype=np.float64)
-        for job in ssn.jobs.values():
-            if job.phase == PodGroupPhase.INQUEUE.value:
-                qi = ssn.queue_index.get(job.queue)
-                if qi is not None:
-                    inqueue_np[qi] += job.minres_vec(nt)
+        if jt is not None:
+            from ..jobtable import PH_INQUEUE
+            sel = (jt.qi >= 0) & (jt.phase == PH_INQUEUE)
+            np.add.at(inqueue_np, jt.qi[sel], jt.minres[sel])
+        else:
+            for job in ssn.jobs.values():
+                if job.phase == PodGroupPhase.INQUEUE.value:
+                    qi = ssn.queue_index.get(job.queue)
+                    if qi is not None:
+                        inqueue_np[qi] += job.minres_vec(nt)
 
         def job_enqueueable(job) -> int:
             qi = ssn.queue_index.get(job.queue)
@@ -147,16 +160,21 @@ class ProportionPlugin(Plugin):
                     out.append(v)
             return out
 
-        def job_enqueueable_bulk(qname, jobs):
+        def job_enqueueable_bulk(qname, jobs, rows=None, table=None):
             """Whole-queue batch vote: total pending demand under deserved
             <=> every sequential per-job vote permits (monotone inqueue
-            sum).  Thunk commits the accounting once all plugins agree."""
+            sum).  Thunk commits the accounting once all plugins agree.
+            With JobTable rows the demand sum is one vectorized reduce."""
             qi = ssn.queue_index.get(qname)
             if qi is None:
                 return None
-            demand = np.zeros(R, dtype=np.float64)
-            for j in jobs:
-                demand += j.minres_vec(nt)
+            if rows is not None and table is not None \
+                    and table.minres.shape[1] == R:
+                demand = table.minres[rows].sum(axis=0)
+            else:
+                demand = np.zeros(R, dtype=np.float64)
+                for j in jobs:
+                    demand += j.minres_vec(nt)
             head = (alloc_np[qi][mask].astype(np.float64)
                     + inqueue_np[qi][mask] + demand[mask])
             des = des_np[qi][mask].astype(np.float64)

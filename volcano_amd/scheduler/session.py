@@ -55,8 +55,13 @@ class Session:
         self.queue_order_fns: List[List[CmpFn]] = []
         self.task_order_fns: List[List[CmpFn]] = []
         self.job_order_keys: List[List] = []
+        self.job_order_cols: List[List] = []   # vectorized key providers
         self.queue_order_keys: List[List] = []
         self.job_valid_fns: List[Callable[[JobInfo], bool]] = []
+        # vectorized job_valid counterparts: fn(job_table) -> bool mask [J].
+        # The columnar worksheet path applies them only when EVERY
+        # job_valid fn has one (1:1), else it falls back per job.
+        self.job_valid_cols: List[Callable] = []
         self.job_ready_fns: List[Callable[[JobInfo], bool]] = []
         self.job_pipelined_fns: List[Callable[[JobInfo], int]] = []
         self.job_enqueueable_fns: List[Callable[[JobInfo], int]] = []
@@ -100,15 +105,19 @@ class Session:
         self.queue_order_fns.append([])
         self.task_order_fns.append([])
         self.job_order_keys.append([])
+        self.job_order_cols.append([])
         self.queue_order_keys.append([])
 
-    def add_job_order_fn(self, fn: CmpFn, key=None) -> None:
+    def add_job_order_fn(self, fn: CmpFn, key=None, col=None) -> None:
         """Register a job compare fn; `key` is an optional *sort key*
         equivalent (ascending) — when every registered order fn provides
         one, ordering runs as a single tuple-key sort instead of
-        O(n log n) cmp callbacks (hot at 10k+ jobs)."""
+        O(n log n) cmp callbacks (hot at 10k+ jobs).  `col` is the
+        vectorized form: fn(job_table, rows) -> np.ndarray of ascending
+        keys for those table rows (the columnar worksheet path)."""
         self.job_order_fns[-1].append(fn)
         self.job_order_keys[-1].append(key)
+        self.job_order_cols[-1].append(col)
 
     def add_queue_order_fn(self, fn: CmpFn, key=None) -> None:
         self.queue_order_fns[-1].append(fn)
@@ -241,6 +250,31 @@ class Session:
                           + (j.creation_timestamp, j.key))
         return sorted(jobs, key=functools.cmp_to_key(self.job_order))
 
+    def ordered_job_rows(self, jt, rows):
+        """Vectorized job ordering over JobTable rows: same ordering as
+        ``sorted_jobs`` (tier keys ascending, creation/key tie-break) but
+        computed by np.lexsort over column providers."""
+        import numpy as np
+        if len(rows) <= 1:
+            return rows
+        keys = [k for tier in self.job_order_keys for k in tier]
+        fns = [f for tier in self.job_order_fns for f in tier]
+        cols = [c for tier in self.job_order_cols for c in tier]
+        if len(keys) != len(fns) or any(k is None for k in keys):
+            jobs = [jt.jobs[int(i)] for i in rows]
+            pos = {id(j): r for j, r in zip(jobs, rows)}
+            return np.array([pos[id(j)] for j in self.sorted_jobs(jobs)],
+                            dtype=rows.dtype)
+        arrs = []
+        for k, c in zip(keys, cols):
+            if c is not None:
+                arrs.append(np.asarray(c(jt, rows)))
+            else:
+                arrs.append(np.array([k(jt.jobs[int(i)]) for i in rows]))
+        order = np.lexsort(tuple([jt.keys[rows], jt.ctime[rows]]
+                                 + arrs[::-1]))
+        return rows[order]
+
     # -- queue tensor rows ----------------------------------------------------
     def build_queue_tensors(self) -> None:
         """queue_alloc from currently-allocated tasks; limit defaults open.
@@ -248,14 +282,25 @@ class Session:
         import numpy as np
         nt = self.node_tensors
         Q, R = len(self.queues), nt.r
-        self.queue_index = {name: i for i, name in enumerate(sorted(self.queues))}
+        if not self.queue_index or len(self.queue_index) != Q:
+            self.queue_index = {name: i
+                                for i, name in enumerate(sorted(self.queues))}
         alloc = np.zeros((Q, R), dtype=np.float32)
-        for job in self.jobs.values():
-            if job.occupied_count == 0:
-                continue      # nothing allocated — skip the vector build
-            qi = self.queue_index.get(job.queue)
-            if qi is not None:
-                alloc[qi] += job.alloc_vec(nt)
+        jt = getattr(self, "job_table", None)
+        if jt is not None and len(jt.jobs) == len(self.jobs):
+            rows = np.nonzero(jt.occ > 0)[0]
+            for k in rows:
+                job = jt.jobs[k]
+                qi = jt.qi[k]
+                if qi >= 0:
+                    alloc[qi] += job.alloc_vec(nt)
+        else:
+            for job in self.jobs.values():
+                if job.occupied_count == 0:
+                    continue      # nothing allocated — skip the vector build
+                qi = self.queue_index.get(job.queue)
+                if qi is not None:
+                    alloc[qi] += job.alloc_vec(nt)
         self.queue_alloc = torch.from_numpy(alloc)
         self.queue_limit = torch.full((Q, R), BIG_LIMIT, dtype=torch.float32)
 
