@@ -84,6 +84,11 @@ class TaskInfo:
                          or k == "volcano.sh/numa-topology-policy")),
             tuple(sorted(p.volumes)) if p is not None and p.volumes else (),
         )
+        # intern: equal signatures share ONE tuple object, so downstream
+        # equality checks (bundle-continuation keys, memo dicts) hit the
+        # identity fast path instead of deep tuple compares — measurable
+        # at 10k identical gangs per cycle.  Bounded by distinct classes.
+        sig = _SIG_INTERN.setdefault(sig, sig)
         self._sig = sig
         return sig
 
@@ -151,6 +156,7 @@ class TaskClass:
 
 _OCCUPIED_STATUSES = tuple(ALLOCATED_STATUSES) + (TaskStatus.SUCCEEDED,)
 _OCC_SET = frozenset(_OCCUPIED_STATUSES)
+_SIG_INTERN: Dict[tuple, tuple] = {}
 
 
 class JobInfo:
@@ -172,6 +178,8 @@ class JobInfo:
         self._occ = 0                        # occupied-count (incremental)
         self._tver = 0                       # static-shape version (JobTable)
         self._jrow = -1                      # row in the cycle JobTable
+        self._mtm = None                     # cached minTaskMember map
+        self._nom = False                    # any task carries a nomination
 
     # -- basic accessors ----------------------------------------------------
     @property
@@ -196,7 +204,11 @@ class JobInfo:
 
     @property
     def min_task_member(self) -> Dict[str, int]:
-        return self.podgroup.spec.min_task_member if self.podgroup else {}
+        m = self._mtm
+        if m is None:
+            m = self._mtm = \
+                self.podgroup.spec.min_task_member if self.podgroup else {}
+        return m
 
     @property
     def priority(self) -> int:
@@ -371,7 +383,10 @@ class JobInfo:
 
     def roles_ready(self) -> bool:
         """Per-role minimums (minTaskMember) — reference CheckTaskReady."""
-        for role, need in self.min_task_member.items():
+        mtm = self.min_task_member
+        if not mtm:
+            return True
+        for role, need in mtm.items():
             if self.role_occupied(role) < need:
                 return False
         return True

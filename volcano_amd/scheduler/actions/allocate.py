@@ -150,9 +150,9 @@ class AllocateAction:
             nonlocal open_bundle, open_key
             if open_bundle is not None:
                 # in-kernel revert only when even the EASIEST bundled gang
-                # cannot place (the apply walk settles per-job minimums)
-                easiest = min(be.min_needed for be in open_bundle.bundle)
-                open_bundle.min_needed = easiest
+                # cannot place (the apply walk settles per-job minimums);
+                # min_needed is min-tracked as entries append
+                easiest = open_bundle.min_needed
                 plan.jobs.append(JobPlan(
                     job_key=open_bundle.job_key,
                     class_begin=len(plan.classes),
@@ -172,6 +172,8 @@ class AllocateAction:
                 b.bundle.append(BundleEntry(job.key, tasks, len(tasks),
                                             gang_min))
                 b.ntasks_override += len(tasks)
+                if gang_min < b.min_needed:
+                    b.min_needed = gang_min
                 return
             close_bundle()
             # the merged class keeps ONE representative task (handlers and
@@ -227,7 +229,7 @@ class AllocateAction:
             # re-checks fit on its nominated nodes and commits host-side,
             # skipping scoring entirely.  Disabled under soft-sharding
             # (commits must flow through the conflict reconcile).
-            if self.coordinator is None and \
+            if self.coordinator is None and job._nom and \
                     self._try_nominated(ssn, job, qi, predicates):
                 continue
 
@@ -247,11 +249,17 @@ class AllocateAction:
                 if not bias_fns and open_bundle is not None \
                         and open_key == (qi, sig):
                     tasks = list(pend.values())
-                    gang_min = max(job.min_available,
-                                   job.min_task_member.get(role, 0))
-                    open_bundle.bundle.append(
+                    mtm = job._mtm
+                    if mtm is None:
+                        mtm = job.min_task_member
+                    gang_min = job.min_available if not mtm else \
+                        max(job.min_available, mtm.get(role, 0))
+                    b = open_bundle
+                    b.bundle.append(
                         BundleEntry(job.key, tasks, len(tasks), gang_min))
-                    open_bundle.ntasks_override += len(tasks)
+                    b.ntasks_override += len(tasks)
+                    if gang_min < b.min_needed:
+                        b.min_needed = gang_min
                     continue
                 req = req_memo.get(sig, _MISS)
                 if req is _MISS:
@@ -451,6 +459,7 @@ class AllocateAction:
         the packed mirrors).  Any miss clears the nominations and returns
         False — the job takes the normal scored path."""
         from ...api.types import TaskStatus as _TS
+        job._nom = False      # hint is single-shot per cycle
         pend_idx = job.task_status_index.get(_TS.PENDING)
         if not pend_idx:
             return False
@@ -688,6 +697,10 @@ class AllocateAction:
         # flip gang-ready podgroups to Running (job_updater analog)
         seen = set()
         alias = getattr(plan, "job_alias", {})
+        jr = ssn.job_ready_fns
+        fast_ready = len(jr) == 1 and getattr(jr[0], "is_gang", False)
+        running = PodGroupPhase.RUNNING.value
+        store = getattr(ssn.cache, "store", None)
         for cp in plan.classes:
             keys = [be.job_key for be in cp.bundle] if cp.bundle else [cp.job_key]
             for key in keys:
@@ -696,10 +709,13 @@ class AllocateAction:
                     continue
                 seen.add(key)
                 job = ssn.jobs[key]
-                if ssn.job_ready(job) and job.podgroup is not None and \
-                        job.phase != PodGroupPhase.RUNNING.value:
-                    job.podgroup.status.phase = PodGroupPhase.RUNNING.value
-                    ssn.cache.update_podgroup(job)
+                ready = (job.is_ready() and job.roles_ready()) if fast_ready \
+                    else ssn.job_ready(job)
+                pg = job.podgroup
+                if ready and pg is not None and pg.status.phase != running:
+                    pg.status.phase = running
+                    if store is not None:
+                        ssn.cache.update_podgroup(job)
 
     @staticmethod
     def _revert_pieces(plan: CyclePlan, cp, pieces) -> None:

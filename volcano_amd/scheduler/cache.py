@@ -282,6 +282,7 @@ class SchedulerCache:
         job = self._job_for(key)
         job.podgroup = pg
         job._minav = None      # minMember may have changed
+        job._mtm = None
         job._tver += 1
 
     def _on_queue(self, ev) -> None:
@@ -348,6 +349,7 @@ class SchedulerCache:
                 t.node_name = ""
                 job.update_task_status(t, TaskStatus.PENDING)
             if pipelined:
+                job._nom = True       # allocate checks this before the
                 self._used_dirty = True
 
     def ensure_packed(self) -> None:

@@ -34,6 +34,7 @@ class GangPlugin(Plugin):
 
         def job_ready(job) -> bool:
             return job.is_ready() and job.roles_ready()
+        job_ready.is_gang = True      # allocate's ready-flip fast path
 
         def job_pipelined(job) -> int:
             if job.is_pipelined():
