@@ -464,3 +464,62 @@ def test_cycle_equivalence_megacycle(hip, monkeypatch):
     gpu = build("cuda", True)
     assert len(cpu) > 100
     assert cpu == gpu
+
+
+def test_cycle_equivalence_chain_path(hip):
+    """Chain path (batched score + select-chain with lazy exact re-score)
+    vs the torch oracle: a heterogeneous many-small-class inventory
+    (hundreds of consecutive single-class jobs — the shape that takes
+    the chain path in vamd_run_cycle) must produce IDENTICAL binds.
+    Also re-runs with VAMD_NO_CHAIN=1 to confirm the chain changes
+    nothing vs the per-class kernel path."""
+    import os
+    from volcano_amd.scheduler import FakeBinder, Scheduler, SchedulerCache, \
+        default_config
+    from volcano_amd.store import ObjectStore
+    from volcano_amd.utils import synth
+
+    GI = 1024 ** 3
+
+    def build(device, use_hip):
+        store = ObjectStore()
+        rng = np.random.RandomState(21)
+        for i in range(400):
+            labels = {"zone": f"z{i % 4}"}
+            store.create("Node", synth.make_node(
+                f"n-{i:04d}", cpu_milli=float(rng.choice([8000, 16000, 32000])),
+                mem=float(rng.choice([16, 32, 64])) * GI, labels=labels))
+        for qn, w in (("qa", 3), ("qb", 1), ("qc", 2)):
+            store.create("Queue", synth.make_queue(qn, weight=w))
+        rng2 = np.random.RandomState(11)
+        for j in range(300):
+            q = ("qa", "qb", "qc")[j % 3]
+            sel = {"zone": f"z{int(rng2.randint(0, 4))}"} \
+                if rng2.rand() < 0.25 else None
+            synth.make_gang(
+                store, f"m{j:03d}", replicas=int(rng2.randint(1, 9)),
+                queue=q,
+                cpu_milli=float(rng2.choice([250, 500, 1000, 2000])),
+                mem=float(rng2.choice([1, 2, 4])) * GI,
+                priority=int(rng2.randint(0, 8)),
+                node_selector=sel)
+        config = default_config()
+        config.use_hip = use_hip
+        config.device = device
+        binder = FakeBinder()
+        cache = SchedulerCache(store=store, binder=binder, device=device)
+        Scheduler(cache, config).run_once()
+        return dict(binder.binds)
+
+    cpu = build("cpu", False)
+    gpu_chain = build("cuda", True)
+    assert len(cpu) > 500
+    assert cpu == gpu_chain, (
+        f"chain path diverges from oracle: {len(cpu)} vs {len(gpu_chain)}; "
+        f"sample diff={list(set(cpu.items()) ^ set(gpu_chain.items()))[:6]}")
+    os.environ["VAMD_NO_CHAIN"] = "1"
+    try:
+        gpu_legacy = build("cuda", True)
+    finally:
+        del os.environ["VAMD_NO_CHAIN"]
+    assert gpu_legacy == gpu_chain, "chain vs per-class kernel divergence"

@@ -78,7 +78,103 @@ extern "C" void vamd_run_cycle(
         }
         // allocation failed: fall through to per-class launches
     }
-    for (int j = 0; j < n_jobs; ++j) {
+
+    // ---- chain path: runs of consecutive single-class small jobs become
+    // 2 launches per chunk (batch score + select chain) instead of 2 per
+    // class.  Decisions match the per-class path exactly (lazy re-score
+    // of touched nodes — scheduler_kernels.hip).  VAMD_NO_CHAIN=1 kills it.
+    const int CHAIN_MIN = 32;
+    int CHUNK_MAX = 512;
+    auto chainable = [&](int j) {
+        const VamdJobDesc& jb = jobs[j];
+        if (jb.class_end - jb.class_begin != 1) return false;
+        return classes[jb.class_begin].ntasks < 512;
+    };
+    bool chain_on = false;
+    {
+        const char* e = getenv("VAMD_NO_CHAIN");
+        bool disabled = e != nullptr && atoi(e) != 0;
+        if (!disabled) {
+            int run = 0;
+            for (int j = 0; j < n_jobs; ++j) {
+                bool ok = chainable(j) &&
+                    (run == 0 || jobs[j].class_begin == jobs[j - 1].class_end);
+                run = ok ? run + 1 : (chainable(j) ? 1 : 0);
+                if (run >= CHAIN_MIN) { chain_on = true; break; }
+            }
+        }
+    }
+    char* chain_blob = nullptr;
+    const VamdClassDesc* cdev = nullptr;
+    const VamdJobDesc* jdev = nullptr;
+    const int64_t* tdev = nullptr;
+    float* score_buf = nullptr;
+    int* tlist = nullptr;
+    uint8_t* touched = nullptr;
+    if (chain_on) {
+        size_t per_class = (size_t)N * sizeof(float);
+        size_t mem_cap = (size_t)256 << 20;
+        if ((size_t)CHUNK_MAX * per_class > mem_cap)
+            CHUNK_MAX = (int)(mem_cap / per_class);
+        if (CHUNK_MAX < 8) CHUNK_MAX = 8;
+        size_t sz_c = (size_t)n_classes * sizeof(VamdClassDesc);
+        size_t sz_j = (size_t)n_jobs * sizeof(VamdJobDesc);
+        size_t sz_t = (size_t)n_classes * sizeof(int64_t);
+        size_t sz_s = (size_t)CHUNK_MAX * per_class;
+        size_t sz_l = (size_t)N * sizeof(int);
+        size_t sz_b = (size_t)N;
+        void* p = nullptr;
+        if (hipMallocAsync(&p, sz_c + sz_j + sz_t + sz_s + sz_l + sz_b,
+                           stream) == hipSuccess && p != nullptr) {
+            chain_blob = (char*)p;
+            char* q = chain_blob;
+            (void)hipMemcpyAsync(q, classes, sz_c, hipMemcpyHostToDevice,
+                                 stream);
+            cdev = (const VamdClassDesc*)q; q += sz_c;
+            (void)hipMemcpyAsync(q, jobs, sz_j, hipMemcpyHostToDevice,
+                                 stream);
+            jdev = (const VamdJobDesc*)q; q += sz_j;
+            (void)hipMemcpyAsync(q, class_tol, sz_t, hipMemcpyHostToDevice,
+                                 stream);
+            tdev = (const int64_t*)q; q += sz_t;
+            score_buf = (float*)q; q += sz_s;
+            tlist = (int*)q; q += sz_l;
+            touched = (uint8_t*)q;
+        } else {
+            chain_on = false;
+        }
+    }
+
+    for (int j = 0; j < n_jobs; ) {
+        if (chain_on && chainable(j)) {
+            // maximal contiguous chain run from j
+            int e = j + 1;
+            while (e < n_jobs && chainable(e)
+                   && jobs[e].class_begin == jobs[e - 1].class_end)
+                ++e;
+            if (e - j >= CHAIN_MIN) {
+                int cb = jobs[j].class_begin;
+                int ce = jobs[e - 1].class_end;
+                for (int cc = cb; cc < ce; cc += CHUNK_MAX) {
+                    int cend = cc + CHUNK_MAX < ce ? cc + CHUNK_MAX : ce;
+                    vamd_batch_score(cdev, cc, cend - cc, alloc, used, extra,
+                                     ready, taints, planes, bias, bias_rows,
+                                     class_req, tdev, class_require,
+                                     class_forbid, dim_w, score_buf,
+                                     N, R, W, stream);
+                    vamd_select_chain(cdev, jdev, cc, cend, alloc, used,
+                                      extra, ready, taints, planes, bias,
+                                      bias_rows, class_req, tdev,
+                                      class_require, class_forbid, dim_w,
+                                      queue_alloc, queue_limit, score_buf,
+                                      log_nodes, log_counts, log_len,
+                                      class_placed, job_placed, touched,
+                                      tlist, N, R, W, stream);
+                }
+                j = e;
+                continue;
+            }
+        }
         const VamdJobDesc& job = jobs[j];
         int nc = job.class_end - job.class_begin;
         bool single = (nc == 1);
@@ -134,5 +230,8 @@ extern "C" void vamd_run_cycle(
                                  stream);
             }
         }
+        ++j;
     }
+    if (chain_blob != nullptr)
+        (void)hipFreeAsync(chain_blob, stream);
 }

@@ -36,6 +36,88 @@ constexpr long long BIG_CAP = 1 << 30;
 // ---------------------------------------------------------------------------
 // core of K1+K2: score/cap for nodes {start, start+step, ...} — shared
 // by the grid-stride kernel and the fused megacycle (block stride)
+// per-node score/cap/feasibility — the ONE implementation shared by the
+// grid-stride kernel, the batched chain scorer, and the chain select's
+// live re-scoring of touched nodes.  Float op order is fixed: every
+// caller gets bit-identical values for identical inputs (the decision-
+// exactness contract with the torch oracle).
+struct NodeScore { float s; int cap; };
+
+DEVINL NodeScore node_score_one(
+    int i,
+    const float* __restrict__ alloc,
+    const float* __restrict__ used,
+    const float* __restrict__ extra,
+    const uint8_t* __restrict__ ready,
+    const int64_t* __restrict__ taints,
+    const int64_t* __restrict__ planes,
+    const float* __restrict__ req,
+    int64_t tolerated,
+    const int64_t* __restrict__ require,
+    const int64_t* __restrict__ forbid,
+    float w_least, float w_most, float w_bal,
+    const float* __restrict__ dim_w, float wsum,
+    const float* __restrict__ bias,
+    int N, int R, int W)
+{
+    bool feasible = ready[i] != 0;
+    feasible = feasible && ((taints[i] & ~tolerated) == 0);
+    for (int w = 0; w < W; ++w) {
+        int64_t p = planes[(size_t)w * N + i];
+        feasible = feasible && ((p & require[w]) == require[w]);
+        feasible = feasible && ((p & forbid[w]) == 0);
+    }
+
+    long long cap = BIG_CAP;
+    float least = 0.f, most = 0.f, mean = 0.f;
+    for (int r = 0; r < R; ++r) {
+        size_t off = (size_t)r * N + i;
+        float a = alloc[off];
+        float u = used[off];
+        float e = extra ? extra[off] : 0.f;
+        float avail = a - u + e;
+        float rq = req[r];
+        if (rq > EPS) {
+            feasible = feasible && (avail + EPS >= rq);
+            float cc = (avail + EPS) / rq;     // clamp before the int
+            if (cc > (float)BIG_CAP) cc = (float)BIG_CAP;  // cast: UB on inf
+            long long c = (long long)floorf(cc);
+            cap = min(cap, max(c, 0ll));
+        }
+        float f = fminf((u + rq) / fmaxf(a, EPS), 1.0f);
+        float dw = dim_w[r];
+        least += (1.0f - f) * dw;
+        most += f * dw;
+        mean += f;
+    }
+    mean /= (float)R;
+    // balanced-allocation variance: second pass RECOMPUTES f instead
+    // of staging a per-thread frac[] array (a fixed-size array caps R
+    // and spills VGPRs; inputs are L2-resident so the re-read is
+    // cheap, and the float op sequence is identical — bit-equal to
+    // the torch oracle).  Skipped entirely when the weight is 0.
+    float bal = 0.f;
+    if (w_bal != 0.f) {
+        float var = 0.f;
+        for (int r = 0; r < R; ++r) {
+            size_t off = (size_t)r * N + i;
+            float f = fminf((used[off] + req[r]) /
+                            fmaxf(alloc[off], EPS), 1.0f);
+            float d = f - mean;
+            var += d * d;
+        }
+        bal = 1.0f - sqrtf(var / (float)R);
+    }
+    float s = w_least * least / wsum + w_most * most / wsum + w_bal * bal;
+    if (bias) s += bias[i];
+
+    bool ok = feasible && cap > 0;
+    NodeScore out;
+    out.s = ok ? s : NEG_INF;
+    out.cap = ok ? (int)min(cap, BIG_CAP) : 0;
+    return out;
+}
+
 __device__ void score_pass(
     int start, int step,
     const float* __restrict__ alloc,
@@ -61,60 +143,12 @@ __device__ void score_pass(
     wsum = fmaxf(wsum, EPS);
 
     for (int i = start; i < N; i += step) {
-        bool feasible = ready[i] != 0;
-        feasible = feasible && ((taints[i] & ~tolerated) == 0);
-        for (int w = 0; w < W; ++w) {
-            int64_t p = planes[(size_t)w * N + i];
-            feasible = feasible && ((p & require[w]) == require[w]);
-            feasible = feasible && ((p & forbid[w]) == 0);
-        }
-
-        long long cap = BIG_CAP;
-        float least = 0.f, most = 0.f, mean = 0.f;
-        for (int r = 0; r < R; ++r) {
-            size_t off = (size_t)r * N + i;
-            float a = alloc[off];
-            float u = used[off];
-            float e = extra ? extra[off] : 0.f;
-            float avail = a - u + e;
-            float rq = req[r];
-            if (rq > EPS) {
-                feasible = feasible && (avail + EPS >= rq);
-                float cc = (avail + EPS) / rq;     // clamp before the int
-                if (cc > (float)BIG_CAP) cc = (float)BIG_CAP;  // cast: UB on inf
-                long long c = (long long)floorf(cc);
-                cap = min(cap, max(c, 0ll));
-            }
-            float f = fminf((u + rq) / fmaxf(a, EPS), 1.0f);
-            float dw = dim_w[r];
-            least += (1.0f - f) * dw;
-            most += f * dw;
-            mean += f;
-        }
-        mean /= (float)R;
-        // balanced-allocation variance: second pass RECOMPUTES f instead
-        // of staging a per-thread frac[] array (a fixed-size array caps R
-        // and spills VGPRs; inputs are L2-resident so the re-read is
-        // cheap, and the float op sequence is identical — bit-equal to
-        // the torch oracle).  Skipped entirely when the weight is 0.
-        float bal = 0.f;
-        if (w_bal != 0.f) {
-            float var = 0.f;
-            for (int r = 0; r < R; ++r) {
-                size_t off = (size_t)r * N + i;
-                float f = fminf((used[off] + req[r]) /
-                                fmaxf(alloc[off], EPS), 1.0f);
-                float d = f - mean;
-                var += d * d;
-            }
-            bal = 1.0f - sqrtf(var / (float)R);
-        }
-        float s = w_least * least / wsum + w_most * most / wsum + w_bal * bal;
-        if (bias) s += bias[i];
-
-        bool ok = feasible && cap > 0;
-        score_out[i] = ok ? s : NEG_INF;
-        cap_out[i] = ok ? (int)min(cap, BIG_CAP) : 0;
+        NodeScore ns = node_score_one(
+            i, alloc, used, extra, ready, taints, planes, req, tolerated,
+            require, forbid, w_least, w_most, w_bal, dim_w, wsum, bias,
+            N, R, W);
+        score_out[i] = ns.s;
+        if (cap_out) cap_out[i] = ns.cap;
     }
 }
 
@@ -702,6 +736,243 @@ megacycle_kernel(
     }
 }
 
+// ---------------------------------------------------------------------------
+// Chain path for many-small-class plans (heterogeneous mixes).  The
+// per-class pipeline pays 2 launches + a serial under-occupied score
+// pass per class (~10k classes in a mixed inventory).  Here a CHUNK of
+// consecutive single-class jobs runs as TWO launches:
+//
+//   1. batch_score_kernel — scores ALL classes of the chunk against the
+//      usage at chunk start, one class per blockIdx.y, filling the chip
+//      (the [C,R]x[R,N] pass; L2-resident node planes make it bandwidth-
+//      trivial).
+//   2. select_chain_kernel — one workgroup replays the selects in job
+//      order with LAZY EXACT re-scoring: a node whose usage changed
+//      since the chunk snapshot ("touched") is re-scored live via the
+//      same node_score_one; untouched nodes' snapshot scores ARE their
+//      live scores.  The merged argmax is therefore bit-identical to
+//      re-scoring every class against live usage — the chain is a pure
+//      optimization, decisions match the per-class path and the torch
+//      oracle exactly.
+//
+// Reference semantics covered: allocate.go:719-866 per-job sequential
+// order; statement.go gang revert fused per single-class job.
+// ---------------------------------------------------------------------------
+__global__ void batch_score_kernel(
+    const VamdClassDesc* __restrict__ classes,   // [C] device
+    int c0,
+    const float* __restrict__ alloc, const float* __restrict__ used,
+    const float* __restrict__ extra,
+    const uint8_t* __restrict__ ready, const int64_t* __restrict__ taints,
+    const int64_t* __restrict__ planes,
+    const float* __restrict__ bias, const float* __restrict__ bias_rows,
+    const float* __restrict__ class_req, const int64_t* __restrict__ class_tol,
+    const int64_t* __restrict__ class_require,
+    const int64_t* __restrict__ class_forbid,
+    const float* __restrict__ dim_w,
+    float* __restrict__ score_buf,               // [chunk, N]
+    int N, int R, int W)
+{
+    const int c = c0 + blockIdx.y;
+    const VamdClassDesc cd = classes[c];
+    const float* ext = (cd.flags & 1) ? extra : nullptr;
+    const float* b = (cd.bias_row >= 0 && bias_rows)
+        ? bias_rows + (size_t)cd.bias_row * N : bias;
+    score_pass(blockIdx.x * blockDim.x + threadIdx.x,
+               gridDim.x * blockDim.x,
+               alloc, used, ext, ready, taints, planes,
+               class_req + (size_t)c * R, class_tol[c],
+               class_require + (size_t)c * W,
+               class_forbid + (size_t)c * W,
+               cd.w_least, cd.w_most, cd.w_bal, dim_w, b,
+               score_buf + (size_t)blockIdx.y * N, nullptr, N, R, W);
+}
+
+__global__ void __launch_bounds__(SC_THREADS)
+select_chain_kernel(
+    const VamdClassDesc* __restrict__ classes,   // [C] device
+    const VamdJobDesc* __restrict__ jobs,        // [J] device
+    int c0, int c1,
+    const float* __restrict__ alloc, float* __restrict__ used,
+    const float* __restrict__ extra,
+    const uint8_t* __restrict__ ready, const int64_t* __restrict__ taints,
+    const int64_t* __restrict__ planes,
+    const float* __restrict__ bias, const float* __restrict__ bias_rows,
+    const float* __restrict__ class_req, const int64_t* __restrict__ class_tol,
+    const int64_t* __restrict__ class_require,
+    const int64_t* __restrict__ class_forbid,
+    const float* __restrict__ dim_w,
+    float* __restrict__ queue_alloc, const float* __restrict__ queue_limit,
+    float* __restrict__ score_buf,               // [c1-c0, N] snapshot scores
+    int* __restrict__ log_nodes, int* __restrict__ log_counts,
+    int* __restrict__ log_len, int* __restrict__ class_placed,
+    int* __restrict__ job_placed,
+    uint8_t* __restrict__ touched,               // [N] scratch (reset here)
+    int* __restrict__ touched_list,              // [N] scratch
+    int N, int R, int W)
+{
+    __shared__ ValIdx s_wave[SC_WAVES];
+    __shared__ ValIdx s_best;
+    __shared__ int s_remaining;
+    __shared__ int s_cursor;
+    __shared__ int s_take;
+    __shared__ int s_tcount;
+    __shared__ long long s_budget;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & (WAVE - 1);
+    const int wid = tid / WAVE;
+
+    // touched set is relative to THIS chunk's score snapshot
+    for (int i = tid; i < N; i += SC_THREADS) touched[i] = 0;
+    if (tid == 0) s_tcount = 0;
+    __syncthreads();
+
+    float wsum = 0.f;
+    for (int r = 0; r < R; ++r) wsum += dim_w[r];
+    wsum = fmaxf(wsum, EPS);
+
+    for (int c = c0; c < c1; ++c) {
+        const VamdClassDesc cd = classes[c];
+        const VamdJobDesc jd = jobs[cd.job_idx];
+        const float* req = class_req + (size_t)c * R;
+        const float* ext = (cd.flags & 1) ? extra : nullptr;
+        const float* b = (cd.bias_row >= 0 && bias_rows)
+            ? bias_rows + (size_t)cd.bias_row * N : bias;
+        const float* row = score_buf + (size_t)(c - c0) * N;
+        float* qa = queue_alloc + (size_t)cd.queue_idx * R;
+        const float* ql = queue_limit + (size_t)cd.queue_idx * R;
+        int* ln = log_nodes + cd.log_off;
+        int* lc = log_counts + cd.log_off;
+        const int K = cd.log_cap;
+        int fuse = jd.min_available - jd.occupied;
+        if (cd.min_needed > fuse) fuse = cd.min_needed;
+        if (fuse < 0) fuse = 0;
+
+        if (tid == 0) {
+            long long quota = BIG_CAP;
+            for (int r = 0; r < R; ++r) {
+                float rq = req[r];
+                if (rq > EPS) {
+                    float head = ql[r] - qa[r];
+                    float qq = (head + EPS) / rq;
+                    if (qq > (float)BIG_CAP) qq = (float)BIG_CAP;
+                    long long q = (long long)floorf(qq);
+                    quota = min(quota, max(q, 0ll));
+                }
+            }
+            s_budget = min((long long)cd.ntasks, quota);
+            s_remaining = (int)min(s_budget, (long long)INT32_MAX);
+            s_cursor = 0;
+        }
+        __syncthreads();
+
+        while (s_remaining > 0 && s_cursor < K) {
+            // candidate = max over (stale scores of untouched nodes) ∪
+            // (live scores of touched nodes) — exactly the live argmax
+            ValIdx loc; loc.v = NEG_INF; loc.i = INT32_MAX;
+            for (int i = tid; i < N; i += SC_THREADS) {
+                if (!touched[i]) {
+                    ValIdx cnd; cnd.v = row[i]; cnd.i = i;
+                    loc = better(loc, cnd);
+                }
+            }
+            const int tc = s_tcount;
+            const int cur = s_cursor;
+            for (int k = tid; k < tc; k += SC_THREADS) {
+                int i = touched_list[k];
+                bool consumed = false;       // already taken by THIS class
+                for (int e = 0; e < cur; ++e)
+                    if (ln[e] == i) { consumed = true; break; }
+                if (consumed) continue;
+                NodeScore ns = node_score_one(
+                    i, alloc, used, ext, ready, taints, planes, req,
+                    class_tol[c], class_require + (size_t)c * W,
+                    class_forbid + (size_t)c * W,
+                    cd.w_least, cd.w_most, cd.w_bal, dim_w, wsum, b,
+                    N, R, W);
+                ValIdx cnd; cnd.v = ns.s; cnd.i = i;
+                loc = better(loc, cnd);
+            }
+            ValIdx w = wave_reduce(loc);
+            if (lane == 0) s_wave[wid] = w;
+            __syncthreads();
+            if (wid == 0) {
+                ValIdx bb; bb.v = NEG_INF; bb.i = INT32_MAX;
+                if (lane < SC_WAVES) bb = s_wave[lane];
+                bb = wave_reduce(bb);
+                if (lane == 0) s_best = bb;
+            }
+            __syncthreads();
+            ValIdx best = s_best;
+            if (best.v == NEG_INF) break;
+
+            if (tid == 0) {
+                // live capacity of the chosen node (untouched nodes keep
+                // their snapshot capacity — used unchanged by definition)
+                long long cap = BIG_CAP;
+                for (int r = 0; r < R; ++r) {
+                    float rq = req[r];
+                    if (rq > EPS) {
+                        size_t off = (size_t)r * N + best.i;
+                        float avail = alloc[off] - used[off]
+                            + (ext ? ext[off] : 0.f);
+                        float cc = (avail + EPS) / rq;
+                        if (cc > (float)BIG_CAP) cc = (float)BIG_CAP;
+                        long long cl = (long long)floorf(cc);
+                        cap = min(cap, max(cl, 0ll));
+                    }
+                }
+                int take = (int)min(cap, (long long)s_remaining);
+                s_take = take;
+                if (take > 0) {
+                    ln[s_cursor] = best.i;
+                    lc[s_cursor] = take;
+                    s_cursor += 1;
+                    s_remaining -= take;
+                }
+            }
+            __syncthreads();
+            int take = s_take;
+            if (take > 0) {
+                if (tid < R)
+                    used[(size_t)tid * N + best.i] += (float)take * req[tid];
+            }
+            if (tid == 0 && !touched[best.i]) {
+                // mark even on take==0 (defensive: a zero-cap winner must
+                // never win again; cannot occur when snapshot caps hold)
+                touched[best.i] = 1;
+                touched_list[s_tcount] = best.i;
+                s_tcount += 1;
+            }
+            __syncthreads();
+        }
+
+        __syncthreads();
+        int total = (int)s_budget - s_remaining;
+        if (total < fuse) {
+            // single-class gang revert (statement.go:375 Discard)
+            for (int e = wid; e < s_cursor; e += SC_WAVES) {
+                int node = ln[e];
+                int cnt = lc[e];
+                if (lane < R)
+                    used[(size_t)lane * N + node] -= (float)cnt * req[lane];
+                if (lane == 0) lc[e] = 0;
+            }
+            __syncthreads();
+            if (tid == 0) { log_len[c] = s_cursor; class_placed[c] = 0; }
+        } else {
+            if (tid < R) qa[tid] += (float)total * req[tid];
+            if (tid == 0) {
+                log_len[c] = s_cursor;
+                class_placed[c] = total;
+                job_placed[cd.job_idx] += total;
+            }
+        }
+        __syncthreads();
+    }
+}
+
 }  // namespace vamd
 
 // ---------------------------------------------------------------------------
@@ -767,6 +1038,52 @@ void vamd_cond_revert(
     hipLaunchKernelGGL(vamd::cond_revert_kernel, dim3(1), dim3(1024), 0,
                        stream, flag, log_nodes, log_counts, log_len, req,
                        used, queue_alloc, placed, job_placed, N, R);
+}
+
+void vamd_batch_score(
+    const VamdClassDesc* classes_dev, int c0, int count,
+    const float* alloc, const float* used, const float* extra,
+    const uint8_t* ready, const int64_t* taints, const int64_t* planes,
+    const float* bias, const float* bias_rows,
+    const float* class_req, const int64_t* class_tol_dev,
+    const int64_t* class_require, const int64_t* class_forbid,
+    const float* dim_w, float* score_buf,
+    int N, int R, int W, hipStream_t stream)
+{
+    int threads = 256;
+    int nx = (N + threads - 1) / threads;
+    if (nx > 1024) nx = 1024;
+    if (nx < 1) nx = 1;
+    hipLaunchKernelGGL(vamd::batch_score_kernel, dim3(nx, count),
+                       dim3(threads), 0, stream, classes_dev, c0, alloc,
+                       used, extra, ready, taints, planes, bias, bias_rows,
+                       class_req, class_tol_dev, class_require, class_forbid,
+                       dim_w, score_buf, N, R, W);
+}
+
+void vamd_select_chain(
+    const VamdClassDesc* classes_dev, const VamdJobDesc* jobs_dev,
+    int c0, int c1,
+    const float* alloc, float* used, const float* extra,
+    const uint8_t* ready, const int64_t* taints, const int64_t* planes,
+    const float* bias, const float* bias_rows,
+    const float* class_req, const int64_t* class_tol_dev,
+    const int64_t* class_require, const int64_t* class_forbid,
+    const float* dim_w,
+    float* queue_alloc, const float* queue_limit,
+    float* score_buf,
+    int* log_nodes, int* log_counts, int* log_len,
+    int* class_placed, int* job_placed,
+    uint8_t* touched, int* touched_list,
+    int N, int R, int W, hipStream_t stream)
+{
+    hipLaunchKernelGGL(vamd::select_chain_kernel, dim3(1), dim3(SC_THREADS),
+                       0, stream, classes_dev, jobs_dev, c0, c1, alloc, used,
+                       extra, ready, taints, planes, bias, bias_rows,
+                       class_req, class_tol_dev, class_require, class_forbid,
+                       dim_w, queue_alloc, queue_limit, score_buf, log_nodes,
+                       log_counts, log_len, class_placed, job_placed,
+                       touched, touched_list, N, R, W);
 }
 
 void vamd_megacycle(

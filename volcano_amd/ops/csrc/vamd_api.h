@@ -60,6 +60,37 @@ struct VamdJobDesc {
 // finalize/revert.  Single-class jobs fuse the gang check into
 // select_commit (2 launches per job).  Everything stays on `stream`;
 // NO host synchronisation happens here.
+// Chain path (heterogeneous mixes): batch-score a chunk of consecutive
+// single-class jobs against chunk-start usage, then replay their selects
+// in one launch with lazy exact re-scoring of touched nodes.  Decisions
+// are bit-identical to the per-class path (see scheduler_kernels.hip).
+// Descriptor/taint arrays must be DEVICE pointers.
+void vamd_batch_score(
+    const VamdClassDesc* classes_dev, int c0, int count,
+    const float* alloc, const float* used, const float* extra,
+    const uint8_t* ready, const int64_t* taints, const int64_t* planes,
+    const float* bias, const float* bias_rows,
+    const float* class_req, const int64_t* class_tol_dev,
+    const int64_t* class_require, const int64_t* class_forbid,
+    const float* dim_w, float* score_buf,
+    int N, int R, int W, hipStream_t stream);
+
+void vamd_select_chain(
+    const VamdClassDesc* classes_dev, const VamdJobDesc* jobs_dev,
+    int c0, int c1,
+    const float* alloc, float* used, const float* extra,
+    const uint8_t* ready, const int64_t* taints, const int64_t* planes,
+    const float* bias, const float* bias_rows,
+    const float* class_req, const int64_t* class_tol_dev,
+    const int64_t* class_require, const int64_t* class_forbid,
+    const float* dim_w,
+    float* queue_alloc, const float* queue_limit,
+    float* score_buf,
+    int* log_nodes, int* log_counts, int* log_len,
+    int* class_placed, int* job_placed,
+    uint8_t* touched, int* touched_list,
+    int N, int R, int W, hipStream_t stream);
+
 // One launch for a whole small-class plan (heterogeneous shapes):
 // descriptor/taint arrays must be DEVICE pointers here.
 void vamd_megacycle(
