@@ -176,21 +176,14 @@ class AllocateAction:
                     b.min_needed = gang_min
                 return
             close_bundle()
-            # the merged class keeps ONE representative task (handlers and
-            # constraint builders read tasks[0]); the instance count lives
-            # in ntasks_override — no 100k-element merged list
-            merged = TaskClass(signature=cp.tclass.signature,
-                               role=cp.tclass.role,
-                               request=cp.tclass.request,
-                               tasks=[tasks[0]],
-                               priority=cp.tclass.priority)
-            cp = ClassPlan(
-                tclass=merged, job_key=job.key, queue_idx=qi, req=cp.req,
-                tolerated=cp.tolerated, require=cp.require,
-                forbid=cp.forbid, min_needed=gang_min,
-                w_least=cp.w_least, w_most=cp.w_most, w_bal=cp.w_bal,
-                ntasks_override=len(tasks),
-                bundle=[BundleEntry(job.key, tasks, len(tasks), gang_min)])
+            # open the bundle on the ORIGINAL ClassPlan: the instance
+            # count lives in ntasks_override, per-job boundaries in the
+            # bundle list; tclass stays the first job's (its tasks[0] is
+            # the representative for handlers/constraint builders)
+            cp.job_key = job.key
+            cp.min_needed = gang_min
+            cp.ntasks_override = len(tasks)
+            cp.bundle = [BundleEntry(job.key, tasks, len(tasks), gang_min)]
             open_bundle, open_key = cp, key
 
         _MISS = object()
