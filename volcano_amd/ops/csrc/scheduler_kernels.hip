@@ -788,7 +788,13 @@ __global__ void batch_score_kernel(
                score_buf + (size_t)blockIdx.y * N, nullptr, N, R, W);
 }
 
-__global__ void __launch_bounds__(SC_THREADS)
+// ONE WAVE (64 lanes): the chain is inherently sequential (each select
+// must see its predecessors' staged usage), so wider blocks only add
+// barrier latency — a wave executes in lockstep, reductions are 6
+// shuffles, and the only __syncthreads (single-wave: compiles to a
+// memory wait, no real barrier) order the cross-lane used/touched
+// stores against the next pop's loads.
+__global__ void __launch_bounds__(WAVE)
 select_chain_kernel(
     const VamdClassDesc* __restrict__ classes,   // [C] device
     const VamdJobDesc* __restrict__ jobs,        // [J] device
@@ -811,21 +817,11 @@ select_chain_kernel(
     int* __restrict__ touched_list,              // [N] scratch
     int N, int R, int W)
 {
-    __shared__ ValIdx s_wave[SC_WAVES];
-    __shared__ ValIdx s_best;
-    __shared__ int s_remaining;
-    __shared__ int s_cursor;
-    __shared__ int s_take;
-    __shared__ int s_tcount;
-    __shared__ long long s_budget;
-
-    const int tid = threadIdx.x;
-    const int lane = tid & (WAVE - 1);
-    const int wid = tid / WAVE;
+    const int lane = threadIdx.x;
 
     // touched set is relative to THIS chunk's score snapshot
-    for (int i = tid; i < N; i += SC_THREADS) touched[i] = 0;
-    if (tid == 0) s_tcount = 0;
+    for (int i = lane; i < N; i += WAVE) touched[i] = 0;
+    int tcount = 0;                 // wave-uniform (all lanes mirror it)
     __syncthreads();
 
     float wsum = 0.f;
@@ -849,40 +845,36 @@ select_chain_kernel(
         if (cd.min_needed > fuse) fuse = cd.min_needed;
         if (fuse < 0) fuse = 0;
 
-        if (tid == 0) {
-            long long quota = BIG_CAP;
-            for (int r = 0; r < R; ++r) {
-                float rq = req[r];
-                if (rq > EPS) {
-                    float head = ql[r] - qa[r];
-                    float qq = (head + EPS) / rq;
-                    if (qq > (float)BIG_CAP) qq = (float)BIG_CAP;
-                    long long q = (long long)floorf(qq);
-                    quota = min(quota, max(q, 0ll));
-                }
+        // queue quota — every lane computes the same scalar (uniform)
+        long long quota = BIG_CAP;
+        for (int r = 0; r < R; ++r) {
+            float rq = req[r];
+            if (rq > EPS) {
+                float head = ql[r] - qa[r];
+                float qq = (head + EPS) / rq;
+                if (qq > (float)BIG_CAP) qq = (float)BIG_CAP;
+                long long q = (long long)floorf(qq);
+                quota = min(quota, max(q, 0ll));
             }
-            s_budget = min((long long)cd.ntasks, quota);
-            s_remaining = (int)min(s_budget, (long long)INT32_MAX);
-            s_cursor = 0;
         }
-        __syncthreads();
+        const long long budget = min((long long)cd.ntasks, quota);
+        int remaining = (int)min(budget, (long long)INT32_MAX);
+        int cursor = 0;
 
-        while (s_remaining > 0 && s_cursor < K) {
+        while (remaining > 0 && cursor < K) {
             // candidate = max over (stale scores of untouched nodes) ∪
             // (live scores of touched nodes) — exactly the live argmax
             ValIdx loc; loc.v = NEG_INF; loc.i = INT32_MAX;
-            for (int i = tid; i < N; i += SC_THREADS) {
+            for (int i = lane; i < N; i += WAVE) {
                 if (!touched[i]) {
                     ValIdx cnd; cnd.v = row[i]; cnd.i = i;
                     loc = better(loc, cnd);
                 }
             }
-            const int tc = s_tcount;
-            const int cur = s_cursor;
-            for (int k = tid; k < tc; k += SC_THREADS) {
+            for (int k = lane; k < tcount; k += WAVE) {
                 int i = touched_list[k];
                 bool consumed = false;       // already taken by THIS class
-                for (int e = 0; e < cur; ++e)
+                for (int e = 0; e < cursor; ++e)
                     if (ln[e] == i) { consumed = true; break; }
                 if (consumed) continue;
                 NodeScore ns = node_score_one(
@@ -894,77 +886,64 @@ select_chain_kernel(
                 ValIdx cnd; cnd.v = ns.s; cnd.i = i;
                 loc = better(loc, cnd);
             }
-            ValIdx w = wave_reduce(loc);
-            if (lane == 0) s_wave[wid] = w;
-            __syncthreads();
-            if (wid == 0) {
-                ValIdx bb; bb.v = NEG_INF; bb.i = INT32_MAX;
-                if (lane < SC_WAVES) bb = s_wave[lane];
-                bb = wave_reduce(bb);
-                if (lane == 0) s_best = bb;
-            }
-            __syncthreads();
-            ValIdx best = s_best;
+            ValIdx best = wave_reduce(loc);
+            best.v = __shfl(best.v, 0);
+            best.i = __shfl(best.i, 0);
             if (best.v == NEG_INF) break;
 
-            if (tid == 0) {
-                // live capacity of the chosen node (untouched nodes keep
-                // their snapshot capacity — used unchanged by definition)
-                long long cap = BIG_CAP;
-                for (int r = 0; r < R; ++r) {
-                    float rq = req[r];
-                    if (rq > EPS) {
-                        size_t off = (size_t)r * N + best.i;
-                        float avail = alloc[off] - used[off]
-                            + (ext ? ext[off] : 0.f);
-                        float cc = (avail + EPS) / rq;
-                        if (cc > (float)BIG_CAP) cc = (float)BIG_CAP;
-                        long long cl = (long long)floorf(cc);
-                        cap = min(cap, max(cl, 0ll));
-                    }
-                }
-                int take = (int)min(cap, (long long)s_remaining);
-                s_take = take;
-                if (take > 0) {
-                    ln[s_cursor] = best.i;
-                    lc[s_cursor] = take;
-                    s_cursor += 1;
-                    s_remaining -= take;
+            // live capacity of the chosen node (uniform scalar compute;
+            // untouched nodes keep their snapshot capacity by definition)
+            long long cap = BIG_CAP;
+            for (int r = 0; r < R; ++r) {
+                float rq = req[r];
+                if (rq > EPS) {
+                    size_t off = (size_t)r * N + best.i;
+                    float avail = alloc[off] - used[off]
+                        + (ext ? ext[off] : 0.f);
+                    float cc = (avail + EPS) / rq;
+                    if (cc > (float)BIG_CAP) cc = (float)BIG_CAP;
+                    long long cl = (long long)floorf(cc);
+                    cap = min(cap, max(cl, 0ll));
                 }
             }
-            __syncthreads();
-            int take = s_take;
+            int take = (int)min(cap, (long long)remaining);
             if (take > 0) {
-                if (tid < R)
-                    used[(size_t)tid * N + best.i] += (float)take * req[tid];
+                if (lane == 0) {
+                    ln[cursor] = best.i;
+                    lc[cursor] = take;
+                }
+                cursor += 1;
+                remaining -= take;
+                if (lane < R)
+                    used[(size_t)lane * N + best.i] += (float)take * req[lane];
             }
-            if (tid == 0 && !touched[best.i]) {
+            if (!touched[best.i]) {
                 // mark even on take==0 (defensive: a zero-cap winner must
                 // never win again; cannot occur when snapshot caps hold)
-                touched[best.i] = 1;
-                touched_list[s_tcount] = best.i;
-                s_tcount += 1;
+                if (lane == 0) {
+                    touched[best.i] = 1;
+                    touched_list[tcount] = best.i;
+                }
+                tcount += 1;
             }
-            __syncthreads();
+            __syncthreads();    // order used/touched stores vs next loads
         }
 
-        __syncthreads();
-        int total = (int)s_budget - s_remaining;
+        int total = (int)budget - remaining;
         if (total < fuse) {
             // single-class gang revert (statement.go:375 Discard)
-            for (int e = wid; e < s_cursor; e += SC_WAVES) {
+            for (int e = 0; e < cursor; ++e) {
                 int node = ln[e];
                 int cnt = lc[e];
                 if (lane < R)
                     used[(size_t)lane * N + node] -= (float)cnt * req[lane];
                 if (lane == 0) lc[e] = 0;
             }
-            __syncthreads();
-            if (tid == 0) { log_len[c] = s_cursor; class_placed[c] = 0; }
+            if (lane == 0) { log_len[c] = cursor; class_placed[c] = 0; }
         } else {
-            if (tid < R) qa[tid] += (float)total * req[tid];
-            if (tid == 0) {
-                log_len[c] = s_cursor;
+            if (lane < R) qa[lane] += (float)total * req[lane];
+            if (lane == 0) {
+                log_len[c] = cursor;
                 class_placed[c] = total;
                 job_placed[cd.job_idx] += total;
             }
@@ -1077,7 +1056,7 @@ void vamd_select_chain(
     uint8_t* touched, int* touched_list,
     int N, int R, int W, hipStream_t stream)
 {
-    hipLaunchKernelGGL(vamd::select_chain_kernel, dim3(1), dim3(SC_THREADS),
+    hipLaunchKernelGGL(vamd::select_chain_kernel, dim3(1), dim3(64),
                        0, stream, classes_dev, jobs_dev, c0, c1, alloc, used,
                        extra, ready, taints, planes, bias, bias_rows,
                        class_req, class_tol_dev, class_require, class_forbid,
