@@ -512,7 +512,11 @@ def test_cycle_equivalence_chain_path(hip):
         return dict(binder.binds)
 
     cpu = build("cpu", False)
-    gpu_chain = build("cuda", True)
+    os.environ["VAMD_CHAIN_CHUNK"] = "128"   # force the chain at small N
+    try:
+        gpu_chain = build("cuda", True)
+    finally:
+        del os.environ["VAMD_CHAIN_CHUNK"]
     assert len(cpu) > 500
     assert cpu == gpu_chain, (
         f"chain path diverges from oracle: {len(cpu)} vs {len(gpu_chain)}; "

@@ -85,6 +85,10 @@ extern "C" void vamd_run_cycle(
     // of touched nodes — scheduler_kernels.hip).  VAMD_NO_CHAIN=1 kills it.
     const int CHAIN_MIN = 32;
     int CHUNK_MAX = 512;
+    {
+        const char* e = getenv("VAMD_CHAIN_CHUNK");
+        if (e != nullptr && atoi(e) > 0) CHUNK_MAX = atoi(e);
+    }
     auto chainable = [&](int j) {
         const VamdJobDesc& jb = jobs[j];
         if (jb.class_end - jb.class_begin != 1) return false;
@@ -94,6 +98,17 @@ extern "C" void vamd_run_cycle(
     {
         const char* e = getenv("VAMD_NO_CHAIN");
         bool disabled = e != nullptr && atoi(e) != 0;
+        // Measured OFF by default (like the megacycle): the batched
+        // score + topk phases are cheap, but the serial select chain
+        // degrades under placement CONCENTRATION — when every class
+        // prefers the same nodes, the top-K candidate lists exhaust and
+        // each pop pays a full-row fallback scan (mix bench: 199-291 ms
+        // vs 157 ms per-class at 10k nodes; 1.9 s vs 0.7 s at 50k —
+        // gpurun_out/bench_r2e-g*).  The async per-class pipeline keeps
+        // the GPU busy at ~12.5 us/class.  VAMD_CHAIN_CHUNK=<n> opts in
+        // (the equivalence test exercises it; decisions are exact).
+        if (getenv("VAMD_CHAIN_CHUNK") == nullptr)
+            disabled = true;
         if (!disabled) {
             int run = 0;
             for (int j = 0; j < n_jobs; ++j) {
@@ -110,6 +125,8 @@ extern "C" void vamd_run_cycle(
     const int64_t* tdev = nullptr;
     float* score_buf = nullptr;
     int* tlist = nullptr;
+    float* topk_vals = nullptr;
+    int* topk_ids = nullptr;
     uint8_t* touched = nullptr;
     if (chain_on) {
         size_t per_class = (size_t)N * sizeof(float);
@@ -123,8 +140,11 @@ extern "C" void vamd_run_cycle(
         size_t sz_s = (size_t)CHUNK_MAX * per_class;
         size_t sz_l = (size_t)N * sizeof(int);
         size_t sz_b = (size_t)N;
+        size_t sz_tv = (size_t)CHUNK_MAX * 16 * sizeof(float);
+        size_t sz_ti = (size_t)CHUNK_MAX * 16 * sizeof(int);
         void* p = nullptr;
-        if (hipMallocAsync(&p, sz_c + sz_j + sz_t + sz_s + sz_l + sz_b,
+        if (hipMallocAsync(&p, sz_c + sz_j + sz_t + sz_s + sz_l + sz_b
+                           + sz_tv + sz_ti,
                            stream) == hipSuccess && p != nullptr) {
             chain_blob = (char*)p;
             char* q = chain_blob;
@@ -139,6 +159,8 @@ extern "C" void vamd_run_cycle(
             tdev = (const int64_t*)q; q += sz_t;
             score_buf = (float*)q; q += sz_s;
             tlist = (int*)q; q += sz_l;
+            topk_vals = (float*)q; q += sz_tv;
+            topk_ids = (int*)q; q += sz_ti;
             touched = (uint8_t*)q;
         } else {
             chain_on = false;
@@ -162,11 +184,14 @@ extern "C" void vamd_run_cycle(
                                      class_req, tdev, class_require,
                                      class_forbid, dim_w, score_buf,
                                      N, R, W, stream);
+                    vamd_topk(score_buf, cend - cc, topk_vals, topk_ids, N,
+                              stream);
                     vamd_select_chain(cdev, jdev, cc, cend, alloc, used,
                                       extra, ready, taints, planes, bias,
                                       bias_rows, class_req, tdev,
                                       class_require, class_forbid, dim_w,
                                       queue_alloc, queue_limit, score_buf,
+                                      topk_vals, topk_ids,
                                       log_nodes, log_counts, log_len,
                                       class_placed, job_placed, touched,
                                       tlist, N, R, W, stream);

@@ -788,13 +788,77 @@ __global__ void batch_score_kernel(
                score_buf + (size_t)blockIdx.y * N, nullptr, N, R, W);
 }
 
-// ONE WAVE (64 lanes): the chain is inherently sequential (each select
-// must see its predecessors' staged usage), so wider blocks only add
-// barrier latency — a wave executes in lockstep, reductions are 6
-// shuffles, and the only __syncthreads (single-wave: compiles to a
-// memory wait, no real barrier) order the cross-lane used/touched
-// stores against the next pop's loads.
-__global__ void __launch_bounds__(WAVE)
+#define CH_THREADS 256
+#define CH_WAVES (CH_THREADS / WAVE)
+#define CH_TOPK 16
+
+// Per-class top-K candidates by snapshot score (desc, ties → lower
+// index), computed in PARALLEL across the chunk (one block per class)
+// so the serial select chain never scans the full node row in the
+// common case: the best untouched node is the first untouched entry of
+// the candidate list.  Chosen entries are knocked out of the row
+// (-inf) — the chain handles them via the candidate list / touched set,
+// and the rare fallback scan only needs the remaining nodes.
+__global__ void __launch_bounds__(SC_THREADS)
+topk_kernel(
+    float* __restrict__ score_buf,        // [chunk, N] (chosen -> -inf)
+    float* __restrict__ topk_vals,        // [chunk, CH_TOPK]
+    int* __restrict__ topk_ids,           // [chunk, CH_TOPK]
+    int N)
+{
+    __shared__ ValIdx s_wave[SC_WAVES];
+    __shared__ ValIdx s_best;
+    float* row = score_buf + (size_t)blockIdx.x * N;
+    float* tv = topk_vals + (size_t)blockIdx.x * CH_TOPK;
+    int* ti = topk_ids + (size_t)blockIdx.x * CH_TOPK;
+    const int tid = threadIdx.x;
+    const int lane = tid & (WAVE - 1);
+    const int wid = tid / WAVE;
+
+    ValIdx loc; loc.v = NEG_INF; loc.i = INT32_MAX;
+    for (int i = tid; i < N; i += SC_THREADS) {
+        ValIdx c; c.v = row[i]; c.i = i;
+        loc = better(loc, c);
+    }
+    for (int k = 0; k < CH_TOPK; ++k) {
+        ValIdx w = wave_reduce(loc);
+        if (lane == 0) s_wave[wid] = w;
+        __syncthreads();
+        if (wid == 0) {
+            ValIdx b; b.v = NEG_INF; b.i = INT32_MAX;
+            if (lane < SC_WAVES) b = s_wave[lane];
+            b = wave_reduce(b);
+            if (lane == 0) s_best = b;
+        }
+        __syncthreads();
+        ValIdx best = s_best;
+        if (tid == 0) { tv[k] = best.v; ti[k] = best.i; }
+        if (best.v == NEG_INF) {
+            // fill the tail so the chain's walk can stop at -inf
+            if (tid == 0)
+                for (int t = k; t < CH_TOPK; ++t) { tv[t] = NEG_INF; ti[t] = INT32_MAX; }
+            return;
+        }
+        if (tid == (best.i & (SC_THREADS - 1))) {
+            row[best.i] = NEG_INF;
+            loc.v = NEG_INF; loc.i = INT32_MAX;
+            for (int i = tid; i < N; i += SC_THREADS) {
+                ValIdx c; c.v = row[i]; c.i = i;
+                loc = better(loc, c);
+            }
+        }
+        __syncthreads();
+    }
+}
+
+// 256 threads / 4 waves: enough waves to hide the L2 latency of the
+// touched-list re-scoring (one wave measured 5x slower — latency
+// bound), small enough that the two per-pop barriers stay cheap.
+// Scalar decisions (budget/cap/take) are computed REDUNDANTLY by every
+// thread from the same memory — uniform by construction, no broadcast
+// barrier needed; the cross-wave argmax merge is one LDS write + one
+// barrier + a 4-entry sequential merge on every thread.
+__global__ void __launch_bounds__(CH_THREADS)
 select_chain_kernel(
     const VamdClassDesc* __restrict__ classes,   // [C] device
     const VamdJobDesc* __restrict__ jobs,        // [J] device
@@ -810,6 +874,8 @@ select_chain_kernel(
     const float* __restrict__ dim_w,
     float* __restrict__ queue_alloc, const float* __restrict__ queue_limit,
     float* __restrict__ score_buf,               // [c1-c0, N] snapshot scores
+    const float* __restrict__ topk_vals,         // [c1-c0, CH_TOPK]
+    const int* __restrict__ topk_ids,            // [c1-c0, CH_TOPK]
     int* __restrict__ log_nodes, int* __restrict__ log_counts,
     int* __restrict__ log_len, int* __restrict__ class_placed,
     int* __restrict__ job_placed,
@@ -817,11 +883,14 @@ select_chain_kernel(
     int* __restrict__ touched_list,              // [N] scratch
     int N, int R, int W)
 {
-    const int lane = threadIdx.x;
+    __shared__ ValIdx s_wave[CH_WAVES];
+    const int tid = threadIdx.x;
+    const int lane = tid & (WAVE - 1);
+    const int wid = tid / WAVE;
 
     // touched set is relative to THIS chunk's score snapshot
-    for (int i = lane; i < N; i += WAVE) touched[i] = 0;
-    int tcount = 0;                 // wave-uniform (all lanes mirror it)
+    for (int i = tid; i < N; i += CH_THREADS) touched[i] = 0;
+    int tcount = 0;                 // block-uniform (all threads mirror it)
     __syncthreads();
 
     float wsum = 0.f;
@@ -845,33 +914,71 @@ select_chain_kernel(
         if (cd.min_needed > fuse) fuse = cd.min_needed;
         if (fuse < 0) fuse = 0;
 
-        // queue quota — every lane computes the same scalar (uniform)
-        long long quota = BIG_CAP;
-        for (int r = 0; r < R; ++r) {
-            float rq = req[r];
+        // queue quota — per-dim candidates on lanes < R, wave min-reduce
+        // (sequential dependent L2 loads cost ~300 ns each; the wave does
+        // them in one round).  Every wave computes the same values from
+        // the same memory → block-uniform without a barrier.
+        long long qcand = BIG_CAP;
+        if (lane < R) {
+            float rq = req[lane];
             if (rq > EPS) {
-                float head = ql[r] - qa[r];
+                float head = ql[lane] - qa[lane];
                 float qq = (head + EPS) / rq;
                 if (qq > (float)BIG_CAP) qq = (float)BIG_CAP;
                 long long q = (long long)floorf(qq);
-                quota = min(quota, max(q, 0ll));
+                qcand = max(q, 0ll);
             }
         }
+        for (int off = WAVE / 2; off > 0; off >>= 1)
+            qcand = min(qcand, (long long)__shfl_down((long long)qcand, off));
+        const long long quota = __shfl((long long)qcand, 0);
         const long long budget = min((long long)cd.ntasks, quota);
         int remaining = (int)min(budget, (long long)INT32_MAX);
         int cursor = 0;
+        const float* tv = topk_vals + (size_t)(c - c0) * CH_TOPK;
+        const int* ti = topk_ids + (size_t)(c - c0) * CH_TOPK;
 
         while (remaining > 0 && cursor < K) {
-            // candidate = max over (stale scores of untouched nodes) ∪
-            // (live scores of touched nodes) — exactly the live argmax
+            // stale side: first UNTOUCHED candidate of the top-K list is
+            // exactly the best untouched node (every node outside the
+            // list scores <= the last candidate).  Lanes load the list in
+            // parallel; a ballot finds the first untouched entry — the
+            // serial walk paid ~16 dependent L2 loads here.
+            float kv = NEG_INF;
+            int ki = INT32_MAX;
+            bool feas = false, ok = false;
+            if (lane < CH_TOPK) {
+                kv = tv[lane];
+                ki = ti[lane];
+                feas = (kv != NEG_INF);
+                ok = feas && !touched[ki];
+            }
+            uint64_t m_ok = __ballot(ok);
+            uint64_t m_feas = __ballot(feas);
+            ValIdx stale; stale.v = NEG_INF; stale.i = INT32_MAX;
+            bool fallback = false;
+            if (m_ok != 0) {
+                int pos = __ffsll((unsigned long long)m_ok) - 1;
+                stale.v = __shfl(kv, pos);
+                stale.i = __shfl(ki, pos);
+            } else if ((m_feas & ((1ull << CH_TOPK) - 1))
+                       == ((1ull << CH_TOPK) - 1)) {
+                // all CH_TOPK candidates feasible AND touched: the best
+                // untouched (if any) is below the list — full row scan
+                fallback = true;
+            }
             ValIdx loc; loc.v = NEG_INF; loc.i = INT32_MAX;
-            for (int i = lane; i < N; i += WAVE) {
-                if (!touched[i]) {
-                    ValIdx cnd; cnd.v = row[i]; cnd.i = i;
-                    loc = better(loc, cnd);
+            if (fallback) {
+                for (int i = tid; i < N; i += CH_THREADS) {
+                    if (!touched[i]) {
+                        ValIdx cnd; cnd.v = row[i]; cnd.i = i;
+                        loc = better(loc, cnd);
+                    }
                 }
             }
-            for (int k = lane; k < tcount; k += WAVE) {
+            loc = better(loc, stale);   // uniform: merges identically
+            // touched side: live re-score (usage changed since snapshot)
+            for (int k = tid; k < tcount; k += CH_THREADS) {
                 int i = touched_list[k];
                 bool consumed = false;       // already taken by THIS class
                 for (int e = 0; e < cursor; ++e)
@@ -886,47 +993,54 @@ select_chain_kernel(
                 ValIdx cnd; cnd.v = ns.s; cnd.i = i;
                 loc = better(loc, cnd);
             }
-            ValIdx best = wave_reduce(loc);
-            best.v = __shfl(best.v, 0);
-            best.i = __shfl(best.i, 0);
+            ValIdx w = wave_reduce(loc);
+            if (lane == 0) s_wave[wid] = w;
+            __syncthreads();                      // barrier 1
+            ValIdx best = s_wave[0];              // 4-entry merge, uniform
+            for (int q = 1; q < CH_WAVES; ++q) best = better(best, s_wave[q]);
             if (best.v == NEG_INF) break;
 
-            // live capacity of the chosen node (uniform scalar compute;
-            // untouched nodes keep their snapshot capacity by definition)
-            long long cap = BIG_CAP;
-            for (int r = 0; r < R; ++r) {
-                float rq = req[r];
+            // live capacity of the chosen node: per-dim candidates on
+            // lanes < R, wave min-reduce (uniform across waves; untouched
+            // nodes keep their snapshot capacity by definition)
+            long long ccand = BIG_CAP;
+            if (lane < R) {
+                float rq = req[lane];
                 if (rq > EPS) {
-                    size_t off = (size_t)r * N + best.i;
+                    size_t off = (size_t)lane * N + best.i;
                     float avail = alloc[off] - used[off]
                         + (ext ? ext[off] : 0.f);
                     float cc = (avail + EPS) / rq;
                     if (cc > (float)BIG_CAP) cc = (float)BIG_CAP;
                     long long cl = (long long)floorf(cc);
-                    cap = min(cap, max(cl, 0ll));
+                    ccand = max(cl, 0ll);
                 }
             }
+            for (int off = WAVE / 2; off > 0; off >>= 1)
+                ccand = min(ccand, (long long)__shfl_down((long long)ccand,
+                                                          off));
+            const long long cap = __shfl((long long)ccand, 0);
             int take = (int)min(cap, (long long)remaining);
             if (take > 0) {
-                if (lane == 0) {
+                if (tid == 0) {
                     ln[cursor] = best.i;
                     lc[cursor] = take;
                 }
                 cursor += 1;
                 remaining -= take;
-                if (lane < R)
-                    used[(size_t)lane * N + best.i] += (float)take * req[lane];
+                if (tid < R)
+                    used[(size_t)tid * N + best.i] += (float)take * req[tid];
             }
             if (!touched[best.i]) {
                 // mark even on take==0 (defensive: a zero-cap winner must
                 // never win again; cannot occur when snapshot caps hold)
-                if (lane == 0) {
+                if (tid == 0) {
                     touched[best.i] = 1;
                     touched_list[tcount] = best.i;
                 }
                 tcount += 1;
             }
-            __syncthreads();    // order used/touched stores vs next loads
+            __syncthreads();    // barrier 2: order stores vs next loads
         }
 
         int total = (int)budget - remaining;
@@ -935,14 +1049,14 @@ select_chain_kernel(
             for (int e = 0; e < cursor; ++e) {
                 int node = ln[e];
                 int cnt = lc[e];
-                if (lane < R)
-                    used[(size_t)lane * N + node] -= (float)cnt * req[lane];
-                if (lane == 0) lc[e] = 0;
+                if (tid < R)
+                    used[(size_t)tid * N + node] -= (float)cnt * req[tid];
+                if (tid == 0) lc[e] = 0;
             }
-            if (lane == 0) { log_len[c] = cursor; class_placed[c] = 0; }
+            if (tid == 0) { log_len[c] = cursor; class_placed[c] = 0; }
         } else {
-            if (lane < R) qa[lane] += (float)total * req[lane];
-            if (lane == 0) {
+            if (tid < R) qa[tid] += (float)total * req[tid];
+            if (tid == 0) {
                 log_len[c] = cursor;
                 class_placed[c] = total;
                 job_placed[cd.job_idx] += total;
@@ -1040,6 +1154,13 @@ void vamd_batch_score(
                        dim_w, score_buf, N, R, W);
 }
 
+void vamd_topk(float* score_buf, int count, float* topk_vals, int* topk_ids,
+               int N, hipStream_t stream)
+{
+    hipLaunchKernelGGL(vamd::topk_kernel, dim3(count), dim3(SC_THREADS), 0,
+                       stream, score_buf, topk_vals, topk_ids, N);
+}
+
 void vamd_select_chain(
     const VamdClassDesc* classes_dev, const VamdJobDesc* jobs_dev,
     int c0, int c1,
@@ -1050,17 +1171,18 @@ void vamd_select_chain(
     const int64_t* class_require, const int64_t* class_forbid,
     const float* dim_w,
     float* queue_alloc, const float* queue_limit,
-    float* score_buf,
+    float* score_buf, const float* topk_vals, const int* topk_ids,
     int* log_nodes, int* log_counts, int* log_len,
     int* class_placed, int* job_placed,
     uint8_t* touched, int* touched_list,
     int N, int R, int W, hipStream_t stream)
 {
-    hipLaunchKernelGGL(vamd::select_chain_kernel, dim3(1), dim3(64),
+    hipLaunchKernelGGL(vamd::select_chain_kernel, dim3(1), dim3(CH_THREADS),
                        0, stream, classes_dev, jobs_dev, c0, c1, alloc, used,
                        extra, ready, taints, planes, bias, bias_rows,
                        class_req, class_tol_dev, class_require, class_forbid,
-                       dim_w, queue_alloc, queue_limit, score_buf, log_nodes,
+                       dim_w, queue_alloc, queue_limit, score_buf,
+                       topk_vals, topk_ids, log_nodes,
                        log_counts, log_len, class_placed, job_placed,
                        touched, touched_list, N, R, W);
 }

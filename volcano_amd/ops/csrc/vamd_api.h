@@ -75,6 +75,9 @@ void vamd_batch_score(
     const float* dim_w, float* score_buf,
     int N, int R, int W, hipStream_t stream);
 
+void vamd_topk(float* score_buf, int count, float* topk_vals, int* topk_ids,
+               int N, hipStream_t stream);
+
 void vamd_select_chain(
     const VamdClassDesc* classes_dev, const VamdJobDesc* jobs_dev,
     int c0, int c1,
@@ -85,7 +88,7 @@ void vamd_select_chain(
     const int64_t* class_require, const int64_t* class_forbid,
     const float* dim_w,
     float* queue_alloc, const float* queue_limit,
-    float* score_buf,
+    float* score_buf, const float* topk_vals, const int* topk_ids,
     int* log_nodes, int* log_counts, int* log_len,
     int* class_placed, int* job_placed,
     uint8_t* touched, int* touched_list,
