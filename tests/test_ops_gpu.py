@@ -527,3 +527,77 @@ def test_cycle_equivalence_chain_path(hip):
     finally:
         del os.environ["VAMD_NO_CHAIN"]
     assert gpu_legacy == gpu_chain, "chain vs per-class kernel divergence"
+
+
+def test_cycle_equivalence_topology_preempt(hip):
+    """Topology-aware preemption (domain dry-run trials) produces the
+    SAME evictions/pipelines/binds on the CPU oracle and the HIP
+    decision plane across a multi-cycle preempt-then-bind sequence."""
+    from volcano_amd.api.objects import (HyperNode, HyperNodeMember,
+                                         MemberSelector, ObjectMeta)
+    from volcano_amd.scheduler import FakeBinder, Scheduler, SchedulerCache, \
+        default_config
+    from volcano_amd.scheduler.config import PluginOption
+    from volcano_amd.store import ObjectStore
+    from volcano_amd.utils import synth
+
+    GI = 1024 ** 3
+
+    def mk_hn(name, tier, nodes=None, children=None):
+        members = []
+        if nodes:
+            members.append(HyperNodeMember(
+                type="Node", selector=MemberSelector(exact_match=nodes)))
+        if children:
+            members.append(HyperNodeMember(
+                type="HyperNode",
+                selector=MemberSelector(exact_match=children)))
+        return HyperNode(meta=ObjectMeta(name=name), tier=tier,
+                         members=members)
+
+    def build(device, use_hip):
+        store = ObjectStore()
+        binder = FakeBinder()
+        cache = SchedulerCache(store=store, binder=binder, device=device)
+        config = default_config()
+        config.actions = ["enqueue", "allocate", "preempt", "backfill"]
+        config.use_hip = use_hip
+        config.device = device
+        config.tiers[1].plugins.append(
+            PluginOption("network-topology-aware"))
+        sched = Scheduler(cache, config)
+        for i in range(8):
+            store.create("Node", synth.make_node(
+                f"n{i}", cpu_milli=4000, mem=16 * GI))
+        store.create("HyperNode", mk_hn("rack-a", 1,
+                                        nodes=[f"n{i}" for i in range(4)]))
+        store.create("HyperNode", mk_hn("rack-b", 1,
+                                        nodes=[f"n{i}" for i in range(4, 8)]))
+        store.create("HyperNode", mk_hn("spine", 2,
+                                        children=["rack-a", "rack-b"]))
+        store.create("Queue", synth.make_queue("default"))
+        synth.make_gang(store, "filler", replicas=8, min_member=1,
+                        cpu_milli=3000, mem=GI, priority=1)
+        sched.run_once()
+        pg = synth.make_podgroup("net", min_member=3)
+        pg.spec.network_topology = {"mode": "hard", "highestTierAllowed": 1}
+        store.create("PodGroup", pg)
+        for i in range(3):
+            store.create("Pod", synth.make_pod(
+                f"net-w-{i}", "net", cpu_milli=3000, mem=GI, priority=50))
+        sched.run_once()
+        evicted = sorted(binder.evictions)
+        # let evicted pods terminate, then settle
+        for key in evicted:
+            ns, name = key.split("/")
+            if store.get("Pod", ns, name) is not None:
+                store.delete("Pod", ns, name)
+        for _ in range(3):
+            sched.run_once()
+        return evicted, dict(binder.binds)
+
+    ev_cpu, binds_cpu = build("cpu", False)
+    ev_gpu, binds_gpu = build("cuda", True)
+    assert len(ev_cpu) >= 1
+    assert ev_cpu == ev_gpu
+    assert binds_cpu == binds_gpu

@@ -150,3 +150,62 @@ def test_hypernode_controller_fabric_discovery(tmp_path):
     assert tree.members["spine-sp0"] == {"n0", "n1", "n2"}
     assert tree.lca_tier("n0", "n1") == 1
     assert tree.lca_tier("n0", "n2") == 2
+
+
+def test_topology_aware_preemption_into_domain():
+    """a hard-topology gang whose domains are all
+    occupied by lower-priority work preempts its way INTO one domain via
+    dry-run trials (reference preempt.go:479 topologyAwarePreempt), and
+    the pipelined tasks carry node nominations across cycles."""
+    store, binder, cache, sched = topo_world()
+    sched2 = None
+    # add preempt to the action list
+    from volcano_amd.scheduler.config import default_config as dc
+    config = dc()
+    config.actions = ["enqueue", "allocate", "preempt", "backfill"]
+    config.tiers[1].plugins.append(PluginOption("network-topology-aware"))
+    sched = Scheduler(cache, config)
+
+    # low-prio filler occupies both racks (2 cpu free per node: 4-node
+    # cluster, each node 4 cpu, filler takes 3 cpu/node)
+    synth.make_gang(store, "filler-a", replicas=4, min_member=1,
+                    cpu_milli=3000, mem=GI, priority=1)
+    sched.run_once()
+    assert binder.bound_count == 4
+    # bound tasks are preemptable victims (ALLOCATED_STATUSES)
+
+    # hard-topology gang needs 2x3 cpu in ONE rack — infeasible without
+    # eviction (each node has only 1 cpu free)
+    pg = synth.make_podgroup("net", min_member=2)
+    pg.spec.network_topology = {"mode": "hard", "highestTierAllowed": 1}
+    store.create("PodGroup", pg)
+    for i in range(2):
+        store.create("Pod", synth.make_pod(
+            f"net-w-{i}", "net", cpu_milli=3000, mem=GI, priority=100))
+    sched.run_once()
+
+    job = cache.jobs["default/net"]
+    # the gang is pipelined into exactly one rack with victims evicted
+    assert len(binder.evictions) >= 1
+    pipelined = job.tasks_with_status(
+        __import__("volcano_amd.api.types", fromlist=["TaskStatus"])
+        .TaskStatus.PIPELINED)
+    assert len(pipelined) == 2
+    nodes = {t.node_name for t in pipelined}
+    assert nodes <= {"n0", "n1"} or nodes <= {"n2", "n3"}, nodes
+
+    # evicted filler pods terminate; the nomination fast path (or the
+    # normal scored path) binds the gang into the SAME domain
+    for key in list(binder.evictions):
+        ns, name = key.split("/")
+        pod = store.get("Pod", ns, name)
+        if pod is not None:
+            store.delete("Pod", ns, name)
+    for _ in range(3):
+        sched.run_once()
+        got = {k: v for k, v in binder.binds.items() if k.startswith("default/net")}
+        if len(got) == 2:
+            break
+    assert len(got) == 2, binder.binds
+    bound_nodes = set(got.values())
+    assert bound_nodes <= {"n0", "n1"} or bound_nodes <= {"n2", "n3"}

@@ -75,7 +75,94 @@ class PreemptAction:
                         if ssn.job_starving(j) and j.pending_tasks
                         and ssn.job_valid(j)]
             for job in ssn.sorted_jobs(starving):
+                # hard-network-topology gangs preempt INTO a domain: the
+                # whole gang must land in one hypernode, so victims are
+                # selected per candidate domain via dry-run simulation
+                # (reference preempt.go:479 topologyAwarePreempt +
+                # actions/utils/simulate.go:40 BuildNominationPlanInDomain)
+                if self._job_hard_topology(ssn, job):
+                    if self._topology_preempt_for_job(ssn, job,
+                                                      same_queue=True):
+                        continue
                 self._preempt_for_job(ssn, job, same_queue=True)
+
+    @staticmethod
+    def _job_hard_topology(ssn, job: JobInfo) -> bool:
+        if getattr(ssn, "hypernode_tree", None) is None:
+            return False
+        spec = job.podgroup.spec.network_topology if job.podgroup else None
+        return bool(spec) and spec.get("mode", "hard") == "hard"
+
+    def _topology_preempt_for_job(self, ssn, job: JobInfo, same_queue: bool,
+                                  victim_filter=None) -> bool:
+        """Domain-gradient preemption: walk candidate hypernodes in tier
+        order (closest domains first — reference HyperNodeGradient), and
+        for each run a DRY-RUN trial (Statement applied to session state,
+        discarded on failure): evict just-enough in-domain victims, then
+        pipeline the gang's pending tasks onto in-domain nodes.  The
+        first domain whose trial pipelines the whole gang wins; its
+        statement commits and the pipelined tasks carry their node
+        nominations into the next cycle (cache._demote_pipelined →
+        allocate._try_nominated).  Mirrors preempt.go:479-673 +
+        simulate.go:40-300 runSimulateTrialAtHyperNode on the plan
+        machinery."""
+        tree = ssn.hypernode_tree
+        spec = job.podgroup.spec.network_topology
+        max_tier = spec.get("highestTierAllowed")
+        running_nodes = {t.node_name for t in job.tasks.values()
+                         if t.node_name}
+        for hname in tree.domains_by_tier(max_tier):
+            members = tree.members[hname]
+            if running_nodes and not running_nodes <= members:
+                continue   # gang fragments must share the domain
+            if self._trial_in_domain(ssn, job, members, same_queue,
+                                     victim_filter):
+                return True
+        return False
+
+    def _trial_in_domain(self, ssn, job: JobInfo, members, same_queue: bool,
+                         victim_filter) -> bool:
+        """One simulate trial: evictions + pipelines applied to session
+        state through a Statement; commit iff the gang reaches pipelined
+        state inside THIS domain, else reverse-order discard (the
+        reference's tmpStmt merge/discard, preempt.go:505-519)."""
+        nt = ssn.node_tensors
+        predicates = getattr(ssn, "predicates", None)
+        stmt = Statement(ssn)
+        still_needed = job.min_available - job.occupied_count \
+            - job.waiting_count
+        if still_needed <= 0:
+            return False
+        placed = 0
+        for tc in job.pending_classes():
+            if placed >= still_needed:
+                break
+            req_vec = nt.req_vector(tc.tasks[0])
+            if req_vec is None:
+                continue
+            constraints = predicates.class_constraints(
+                tc, job, skip_topology=True) if predicates \
+                else (-1, np.zeros(max(nt.labels.words, 1), dtype=np.int64),
+                      np.zeros(max(nt.labels.words, 1), dtype=np.int64))
+            scan = self._candidate_scan(ssn, job, tc, req_vec, constraints,
+                                        same_queue, victim_filter,
+                                        node_filter=members)
+            if scan is None:
+                continue
+            order, victims_by_node = scan
+            for task in tc.tasks:
+                if placed >= still_needed:
+                    break
+                node = self._preempt_one(ssn, stmt, job, task, req_vec,
+                                         order, victims_by_node)
+                if node is not None:
+                    placed += 1
+        if placed >= still_needed and ssn.job_pipelined(job) \
+                and job.is_pipelined():
+            stmt.commit()
+            return True
+        stmt.discard()
+        return False
 
     # shared with reclaim (cross-queue variant)
     def _preempt_for_job(self, ssn, job: JobInfo, same_queue: bool,
@@ -120,9 +207,11 @@ class PreemptAction:
             stmt.discard()
 
     def _candidate_scan(self, ssn, job, tc, req_vec, constraints,
-                        same_queue, victim_filter):
+                        same_queue, victim_filter, node_filter=None):
         """One fused feasibility pass for the whole class: evictable
-        resources per node become the kernel's future-credit plane."""
+        resources per node become the kernel's future-credit plane.
+        ``node_filter`` (a set of node names) restricts both victims and
+        candidate nodes — the hypernode-domain trial path."""
         import torch
         from ...ops import reference as ref
         nt = ssn.node_tensors
@@ -133,6 +222,8 @@ class PreemptAction:
         candidates = []
         for ni in ssn.nodes.values():
             if not ni.ready:
+                continue
+            if node_filter is not None and ni.name not in node_filter:
                 continue
             for t in ni.tasks.values():
                 if t.status not in self.victim_statuses:
@@ -182,6 +273,10 @@ class PreemptAction:
         if nodes_sorted is None or len(nodes_sorted) != len(ssn.nodes):
             nodes_sorted = sorted(ssn.nodes.values(), key=lambda n: n.name)
         order = [nodes_sorted[i] for i in order_idx if feasible[i]]
+        if node_filter is not None:
+            # domain trials must not pipeline OUTSIDE the domain even
+            # when an out-of-domain node would fit without evictions
+            order = [ni for ni in order if ni.name in node_filter]
         return order, victims_by_node
 
     def _preempt_one(self, ssn, stmt: Statement, job: JobInfo,

@@ -126,6 +126,7 @@ class NetworkTopologyAwarePlugin(Plugin):
                 set_plane_bit(require, impossible_bit)
             # soft + no feasible domain: unconstrained
 
+        hook.is_topology = True       # preempt domain trials strip it
         domain_sets = {h: set(ids) for h, ids in self.domain_ids.items()}
 
         def on_allocate(tclass, node_ids, counts, tasks=None):

@@ -30,8 +30,9 @@ class PredicatesPlugin(Plugin):
         self._nt = ssn.node_tensors
         self._memo = {}
 
-    def class_constraints(self, tclass,
-                          job=None) -> Tuple[int, np.ndarray, np.ndarray]:
+    def class_constraints(self, tclass, job=None,
+                          skip_topology=False
+                          ) -> Tuple[int, np.ndarray, np.ndarray]:
         """(tolerated taint mask, require planes, forbid planes) for a class.
 
         All tasks of a class share constraints by construction
@@ -40,9 +41,16 @@ class PredicatesPlugin(Plugin):
         (nodegroup/tdm/usage) extend the bits through
         ``ssn.class_constraint_hooks``."""
         hooks = getattr(self._ssn, "class_constraint_hooks", [])
+        if skip_topology:
+            # hypernode-domain preemption trials restrict nodes via an
+            # explicit domain filter; the topology hook's own domain
+            # choice (made BEFORE the trial's evictions) must not leak in
+            hooks = [h for h in hooks
+                     if not getattr(h, "is_topology", False)]
         # per-class topology overrides (SubGroupPolicy) choose a domain
         # PER SUBGROUP — never share a memo entry across subgroups
-        memoize = getattr(tclass, "topology", None) is None
+        memoize = getattr(tclass, "topology", None) is None \
+            and not skip_topology
         key = (tclass.signature, job.queue if (job and hooks) else None)
         if memoize:
             got = self._memo.get(key)
