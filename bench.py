@@ -173,6 +173,7 @@ def main():
 
     cache = SchedulerCache(store=None, binder=FakeBinder(), device=device)
     sched = Scheduler(cache, config)
+    coord = None
     if dist and args.shard_mode == "soft":
         from volcano_amd.parallel.softshard import SoftShardCoordinator
         from volcano_amd.scheduler.actions.allocate import AllocateAction
@@ -306,6 +307,22 @@ def main():
         elapsed = float(te.item())
         bound = int(tb.item())
 
+    # soft-shard reconcile stats (conflict rate in the
+    # bench JSON): sum of lost-node incidents across ranks / reconciles
+    soft_stats = None
+    if coord is not None:
+        cs = torch.tensor([float(coord.conflict_nodes),
+                           float(coord.reconciles)])
+        if use_gpu and torch.distributed.get_backend() == "nccl":
+            cs = cs.cuda()
+        torch.distributed.all_reduce(cs, op=torch.distributed.ReduceOp.SUM)
+        soft_stats = {
+            "conflict_nodes_total": int(cs[0].item()),
+            "reconciles_total": int(cs[1].item()),
+            "conflict_rate": round(
+                float(cs[0].item()) / max(float(cs[1].item()), 1.0), 4),
+        }
+
     expected = args.steps * total_pods
     if bound != expected and not args.churn:
         # report honestly; a shortfall means capacity/plan bug, not a perf win
@@ -350,6 +367,8 @@ def main():
             # second half of the BASELINE metric ("pods scheduled/sec +
             # p99 scheduling-cycle latency")
             "p99_cycle_ms": round(p99 * 1000.0, 2),
+            "per_rank_pods_per_sec": round(value / max(world, 1), 2),
+            "soft_shard": soft_stats,
             "p99_pod_latency_ms": round(
                 (sorted(churn_state["latencies"])[
                     max(0, int(len(churn_state["latencies"]) * 0.99) - 1)]

@@ -45,6 +45,10 @@ class SoftShardCoordinator:
         self.world = world
         self._stagger = None
         self._repel = {}               # node_id -> penalty strength
+        # measured-mode stats (bench JSON): reconcile rounds and nodes
+        # this rank lost in the cross-rank admission
+        self.reconciles = 0
+        self.conflict_nodes = 0
 
     def stagger_bias(self, nt) -> torch.Tensor:
         """Rank-staggered tie-break: an epsilon-scale score bias that makes
@@ -103,6 +107,8 @@ class SoftShardCoordinator:
         self._gathered_sum = stack.sum(dim=0)
         ids = torch.nonzero(lost, as_tuple=False).flatten()
         out = frozenset(int(i) for i in ids.cpu())
+        self.reconciles += 1
+        self.conflict_nodes += len(out)
         self._note_losses(out)
         return out
 
