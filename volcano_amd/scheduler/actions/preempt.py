@@ -268,7 +268,11 @@ class PreemptAction:
                       1.0, 0.0, 0.0, torch.ones(nt.r, device=dev), None,
                       score, cap)
         feasible = (score > float("-inf")).cpu().numpy()
-        order_idx = torch.argsort(score, descending=True).cpu().numpy()
+        # stable: equal scores keep index order (CUDA argsort is otherwise
+        # tie-unstable, which would pair tasks to nodes differently than
+        # the CPU path — found by the topology-preempt equivalence test)
+        order_idx = torch.argsort(score, descending=True,
+                                  stable=True).cpu().numpy()
         nodes_sorted = getattr(ssn.cache, "nodes_sorted", None)
         if nodes_sorted is None or len(nodes_sorted) != len(ssn.nodes):
             nodes_sorted = sorted(ssn.nodes.values(), key=lambda n: n.name)
