@@ -42,7 +42,7 @@ def create_app(store: ObjectStore, with_admission: bool = True):
 
     @app.get("/metrics", response_class=PlainTextResponse)
     def metrics():
-        return METRICS.export_text()
+        return METRICS.export_prometheus()
 
     @app.get("/apis/{kind}")
     def list_objs(kind: str, namespace: Optional[str] = None):
