@@ -459,6 +459,12 @@ class NodeInfo:
             self.oversubscription.q[MEMORY] = float(
                 ann["volcano.sh/oversubscription-memory"])
         self.allocatable.add(self.oversubscription)
+        # DRA: device-class capacity advertised per node (the
+        # ResourceSlice analog) — becomes the dense dim dra:<class>
+        for k, v in ann.items():
+            if k.startswith("dra.volcano.sh/"):
+                self.allocatable.q[f"dra:{k[len('dra.volcano.sh/'):]}"] = \
+                    float(v)
         # pre-adoption local truth (ledger adoption copies these in)
         self._used = Resource()
         self._releasing = Resource()
@@ -643,15 +649,18 @@ class QueueInfo:
 
     @property
     def capability(self) -> Resource:
-        return self.queue.spec.capability
+        from .resource import normalize_dra_keys
+        return normalize_dra_keys(self.queue.spec.capability)
 
     @property
     def guarantee(self) -> Resource:
-        return self.queue.spec.guarantee
+        from .resource import normalize_dra_keys
+        return normalize_dra_keys(self.queue.spec.guarantee)
 
     @property
     def deserved_spec(self) -> Resource:
-        return self.queue.spec.deserved
+        from .resource import normalize_dra_keys
+        return normalize_dra_keys(self.queue.spec.deserved)
 
     @property
     def reclaimable(self) -> bool:

@@ -117,6 +117,9 @@ class Pod:
     volumes: List[str] = field(default_factory=list)   # PVC names (same ns)
     host_ports: List[int] = field(default_factory=list)  # requested hostPorts
     image: str = ""                                    # container image
+    # DRA (k8s dynamic resource allocation): ResourceClaim names (same
+    # namespace) the pod consumes — each claim names a DeviceClass
+    resource_claims: List[str] = field(default_factory=list)
 
     def __post_init__(self):
         if self.request.is_empty():
@@ -465,6 +468,32 @@ class HyperJob:
 
 
 @dataclass
+class DeviceClass:
+    """resource.k8s.io DeviceClass (DRA): a named class of devices.
+    Scheduling rides the dense dims — a class becomes the synthetic
+    resource dim ``dra:<name>``; nodes advertise capacity via the
+    ``dra.volcano.sh/<name>`` annotation (the ResourceSlice analog).
+    Reference: plugins/predicates dynamicresources wrap
+    (predicates.go:34-47) + capacity DRA quotas (capacity.go:107-197)."""
+
+    meta: ObjectMeta = field(default_factory=ObjectMeta)
+    driver: str = ""                      # informational (device plugin)
+    config: Dict[str, Any] = field(default_factory=dict)
+
+
+@dataclass
+class ResourceClaim:
+    """resource.k8s.io ResourceClaim (DRA): a pod's claim for N devices
+    of a DeviceClass."""
+
+    meta: ObjectMeta = field(default_factory=ObjectMeta)
+    device_class_name: str = ""
+    count: int = 1
+    # filled at bind: node the devices were allocated on
+    allocated_node: str = ""
+
+
+@dataclass
 class CronJob:
     """batch/v1alpha1 CronJob (reference cronjob controller)."""
 
@@ -532,4 +561,5 @@ KINDS = {
     "ResourceQuota": ResourceQuota, "HyperJob": HyperJob,
     "PersistentVolume": PersistentVolume,
     "PersistentVolumeClaim": PersistentVolumeClaim,
+    "DeviceClass": DeviceClass, "ResourceClaim": ResourceClaim,
 }

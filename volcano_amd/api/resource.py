@@ -100,6 +100,18 @@ class ResourceDims:
         return out
 
 
+def normalize_dra_keys(r: "Resource") -> "Resource":
+    """Translate the reference's DRA quota keys (``count/<deviceClass>``
+    — capacity.go DeviceClassCountPrefix) into the dense device-class
+    dims (``dra:<class>``) so queue capability/deserved/guarantee rows
+    enforce device-class quotas through the SAME in-kernel clamp as any
+    other resource."""
+    if not any(k.startswith("count/") for k in r.q):
+        return r
+    return Resource({(f"dra:{k[6:]}" if k.startswith("count/") else k): v
+                     for k, v in r.q.items()})
+
+
 class Resource:
     """Small dict-backed resource vector for control-plane code.
 

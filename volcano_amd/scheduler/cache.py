@@ -233,9 +233,20 @@ class SchedulerCache:
 
         if ev.type == EventType.DELETED:
             return
+        # DRA: resolve the pod's ResourceClaims into dra:<class> request
+        # dims (predicates + queue quotas then ride the dense machinery)
+        if pod.resource_claims and self.store is not None:
+            for cname in pod.resource_claims:
+                claim = self.store.get("ResourceClaim",
+                                       pod.meta.namespace, cname)
+                if claim is not None and claim.device_class_name:
+                    dim = f"dra:{claim.device_class_name}"
+                    pod.request.q[dim] = pod.request.q.get(dim, 0.0) + \
+                        float(claim.count)
         task = TaskInfo.from_pod(pod, key)
         for k in task.request.q:
-            if k.startswith(("paa:", "hp:", "nvl:")) and k not in self.dims:
+            if k.startswith(("paa:", "hp:", "nvl:", "dra:")) \
+                    and k not in self.dims:
                 self.dims.add(k)
                 self._tensors_dirty = True
         self._job_for(key).add_task(task)

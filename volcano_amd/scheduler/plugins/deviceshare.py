@@ -80,3 +80,31 @@ class DeviceSharePlugin(Plugin):
                                           staticmethod(on_allocate)})()
         ssn.class_constraint_hooks.append(hook)
         ssn.event_handlers.append(handler)
+
+        # GPU binpack/spread node scoring (reference deviceshare.go:233-285
+        # NodeOrder): classes carrying a device request get a bias plane
+        # favoring fuller (binpack, default) or emptier (spread) nodes'
+        # device pools — per-card fill averaged per node.
+        import numpy as np
+        mode = str(self.args.get("deviceshare.schedule-policy", "binpack"))
+        weight = float(self.args.get("deviceshare.gpu-score-weight", 0.05))
+        if weight > 0:
+            N = nt.n
+            fill = np.zeros(N, dtype=np.float32)
+            pooled = np.zeros(N, dtype=bool)
+            for name, pool in pools.items():
+                ni = ssn.nodes.get(name)
+                if ni is None or ni.node_id < 0 or not pool.cards:
+                    continue
+                f = sum((c.mem_used / max(c.mem_total, 1)) if not c.exclusive
+                        else 1.0 for c in pool.cards) / len(pool.cards)
+                fill[ni.node_id] = f
+                pooled[ni.node_id] = True
+            bias_row = (weight * fill if mode == "binpack"
+                        else -weight * fill).astype(np.float32)
+            bias_row[~pooled] = 0.0
+
+            def gpu_bias(tclass, job):
+                return bias_row if request_of(tclass) is not None else None
+
+            ssn.class_bias_fns.append(gpu_bias)

@@ -35,6 +35,8 @@ GVK: Dict[str, Tuple[str, str, str]] = {
     "PersistentVolumeClaim": ("", "v1", "persistentvolumeclaims"),
     "ResourceQuota": ("", "v1", "resourcequotas"),
     "PodDisruptionBudget": ("policy", "v1", "poddisruptionbudgets"),
+    "DeviceClass": ("resource.k8s.io", "v1", "deviceclasses"),
+    "ResourceClaim": ("resource.k8s.io", "v1", "resourceclaims"),
 }
 
 # (group, version, plural) → kind
@@ -130,7 +132,7 @@ def crd_manifest(kind: str) -> Optional[Dict[str, Any]]:
     validation lives in the admission chain, as in the reference's
     webhook-manager."""
     g, v, plural = GVK[kind]
-    if not g or g == "policy":
+    if not g or g in ("policy", "resource.k8s.io"):
         return None
     return {
         "apiVersion": "apiextensions.k8s.io/v1",
