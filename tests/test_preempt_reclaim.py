@@ -189,3 +189,16 @@ def test_nominated_node_fast_path_after_preempt():
     placed = sorted(t.node_name for t in high.tasks.values())
     assert placed == nominated                   # stayed on freed nodes
     assert high.occupied_count == 2
+
+
+def test_preemption_storm_converges():
+    """Scaled-down storm (benchmark/storm.py at full size on GPU):
+    hundreds of starving high-prio gangs preempt a saturated cluster
+    within a few cycles with drift-free accounting."""
+    import sys, os
+    sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
+    from benchmark.storm import run
+    placed, cycles, bad = run(nodes=100, use_gpu=False, max_cycles=6)
+    assert bad == 0
+    assert placed == 100, f"storm placed {placed}/100 in {cycles} cycles"
+    assert cycles <= 5

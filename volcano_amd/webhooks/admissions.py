@@ -31,35 +31,144 @@ def mutate_job(store, job, op) -> None:
         job.spec.min_available = sum(t.min_needed for t in job.spec.tasks)
 
 
+_DNS1123 = None
+
+
+def _dns1123(name: str) -> bool:
+    global _DNS1123
+    if _DNS1123 is None:
+        import re
+        _DNS1123 = re.compile(r"^[a-z0-9]([-a-z0-9]*[a-z0-9])?$")
+    return bool(_DNS1123.fullmatch(name)) and len(name) <= 63
+
+
 def validate_job(store, job, op) -> None:
-    """reference admission/jobs/validate (spec legality)."""
+    """reference admission/jobs/validate (admit_job.go:120-230): spec
+    legality incl. plugin names, MPI master presence, policy sets,
+    networkTopology, task DAG, pod-name length and leaf-queue rules."""
     if not job.meta.name:
         raise AdmissionError("job name required")
+    if not _dns1123(job.meta.name):
+        raise AdmissionError(
+            f"job name {job.meta.name!r} must be a DNS-1123 label")
     if not job.spec.tasks:
-        raise AdmissionError("job must define at least one task")
+        raise AdmissionError("No task specified in job spec")
     names = [t.name for t in job.spec.tasks]
     if len(names) != len(set(names)):
         raise AdmissionError(f"duplicated task names: {names}")
     total = job.spec.total_replicas
-    if job.spec.min_available is not None and job.spec.min_available > total:
-        raise AdmissionError(
-            f"minAvailable {job.spec.min_available} > total replicas {total}")
-    for ts in job.spec.tasks:
+    if job.spec.min_available is not None:
+        if job.spec.min_available < 0:
+            raise AdmissionError("minAvailable must be >= 0")
+        if job.spec.min_available > total:
+            raise AdmissionError(
+                f"minAvailable {job.spec.min_available} > "
+                f"total replicas {total}")
+    if job.spec.min_success is not None and job.spec.min_success < 0:
+        raise AdmissionError("minSuccess must be >= 0")
+    if job.spec.ttl_seconds_after_finished is not None \
+            and job.spec.ttl_seconds_after_finished < 0:
+        raise AdmissionError("ttlSecondsAfterFinished must be >= 0")
+    for i, ts in enumerate(job.spec.tasks):
+        if ts.name and not _dns1123(ts.name):
+            raise AdmissionError(
+                f"task name {ts.name!r} must be a DNS-1123 label")
+        # generated pod names must stay within the k8s 253-char bound
+        if len(f"{job.meta.name}-{ts.name}-{ts.replicas}") > 253:
+            raise AdmissionError(f"task {ts.name}: pod name too long")
+        if ts.replicas < 0:
+            raise AdmissionError(f"task {ts.name}: replicas must be >= 0")
         if ts.min_available is not None and ts.min_available > ts.replicas:
             raise AdmissionError(
                 f"task {ts.name}: minAvailable > replicas")
-        for pol in ts.policies:
-            _validate_policy(pol, f"task {ts.name}")
-    for pol in job.spec.policies:
-        _validate_policy(pol, "job")
+        _validate_policy_set(ts.policies, f"task {ts.name}")
+    _validate_policy_set(job.spec.policies, "job")
     if job.spec.max_retry < 0:
         raise AdmissionError("maxRetry must be >= 0")
+    _validate_network_topology(job.spec.network_topology)
+    # job plugins must be registered (admit_job.go "unable to find job
+    # plugin"); the MPI plugin additionally needs its master task present
+    if job.spec.plugins:
+        from ..controllers.jobplugins import JOB_PLUGINS
+        for pname in job.spec.plugins:
+            if pname not in JOB_PLUGINS:
+                raise AdmissionError(f"unable to find job plugin: {pname}")
+        if "mpi" in job.spec.plugins:
+            # reference: the configured mpi master task must exist
+            # (admit_job.go:134-141); enforced when explicitly configured
+            args = job.spec.plugins.get("mpi") or []
+            master = next((a.split("=", 1)[1] for a in args
+                           if a.startswith("master=")), None)
+            if master is not None and \
+                    all(t.name != master for t in job.spec.tasks):
+                raise AdmissionError(
+                    "the specified mpi master task was not found")
+    # task dependsOn must form a DAG (admit_job.go topoSort)
+    if any(ts.depends_on for ts in job.spec.tasks):
+        _validate_task_dag(job.spec.tasks)
     if store is not None and job.spec.queue:
         q = store.get("Queue", "default", job.spec.queue)
         if q is None:
             raise AdmissionError(f"queue {job.spec.queue!r} does not exist")
         if q.status.state != QueueState.OPEN.value:
             raise AdmissionError(f"queue {job.spec.queue!r} is not open")
+        if q.meta.name == "root":
+            raise AdmissionError("can not submit job to root queue")
+        children = [c for c in store.list("Queue")
+                    if c.spec.parent == q.meta.name]
+        if children:
+            raise AdmissionError(
+                f"can only submit job to leaf queue; queue "
+                f"{q.meta.name!r} has {len(children)} child queues")
+
+
+def _validate_network_topology(spec) -> None:
+    if not spec:
+        return
+    if spec.get("mode", "hard") not in ("hard", "soft"):
+        raise AdmissionError("networkTopology.mode must be hard|soft")
+    t = spec.get("highestTierAllowed")
+    if t is not None and int(t) < 1:
+        raise AdmissionError("networkTopology.highestTierAllowed must be >= 1")
+
+
+def _validate_task_dag(tasks) -> None:
+    names = {t.name for t in tasks}
+    state = {}
+
+    def visit(n):
+        if state.get(n) == 1:
+            raise AdmissionError(
+                "job task dependencies do not form a DAG")
+        if state.get(n) == 2:
+            return
+        state[n] = 1
+        ts = next(t for t in tasks if t.name == n)
+        for d in ts.depends_on:
+            if d not in names:
+                raise AdmissionError(
+                    f"task {n}: unknown dependency {d!r}")
+            visit(d)
+        state[n] = 2
+
+    for t in tasks:
+        visit(t.name)
+
+
+def _validate_policy_set(policies, where: str) -> None:
+    """validatePolicies (admit_job.go): per-policy legality + duplicate
+    events + '*' exclusivity."""
+    seen = set()
+    for pol in policies:
+        _validate_policy(pol, where)
+        evs = pol.events or ([pol.event] if pol.event else [])
+        for e in evs:
+            if e in seen:
+                raise AdmissionError(f"{where}: duplicated policy event {e}")
+            seen.add(e)
+    if "*" in seen and len(seen) > 1:
+        raise AdmissionError(
+            f"{where}: '*' policy cannot be combined with other events")
 
 
 def _validate_policy(pol, where: str) -> None:
