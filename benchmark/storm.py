@@ -93,6 +93,7 @@ if __name__ == "__main__":
     args = ap.parse_args()
     placed, cycles, bad = run(args.nodes, args.low_jobs, args.storm_jobs,
                               use_gpu=(False if args.cpu else None))
+    want = args.storm_jobs if args.storm_jobs is not None else args.nodes
     assert bad == 0, "accounting drift"
-    assert placed == args.storm_jobs, "storm did not converge"
+    assert placed == want, "storm did not converge"
     print("STORM OK")

@@ -74,6 +74,7 @@ class PreemptAction:
             starving = [j for j in jobs_in_q
                         if ssn.job_starving(j) and j.pending_tasks
                         and ssn.job_valid(j)]
+            scan_memo = {}   # per-cycle: identical preemptor shapes
             for job in ssn.sorted_jobs(starving):
                 # hard-network-topology gangs preempt INTO a domain: the
                 # whole gang must land in one hypernode, so victims are
@@ -84,7 +85,8 @@ class PreemptAction:
                     if self._topology_preempt_for_job(ssn, job,
                                                       same_queue=True):
                         continue
-                self._preempt_for_job(ssn, job, same_queue=True)
+                self._preempt_for_job(ssn, job, same_queue=True,
+                                      memo=scan_memo)
 
     @staticmethod
     def _job_hard_topology(ssn, job: JobInfo) -> bool:
@@ -166,7 +168,7 @@ class PreemptAction:
 
     # shared with reclaim (cross-queue variant)
     def _preempt_for_job(self, ssn, job: JobInfo, same_queue: bool,
-                         victim_filter=None) -> None:
+                         victim_filter=None, memo=None) -> None:
         nt = ssn.node_tensors
         stmt = Statement(ssn)
         predicates = getattr(ssn, "predicates", None)
@@ -188,8 +190,25 @@ class PreemptAction:
             # K5 tensor pass: global victim filter once per
             # class, per-node evictable capacity as an extra-credit plane,
             # ONE fused score_cap over all nodes → ranked candidate nodes.
-            scan = self._candidate_scan(ssn, job, tc, req_vec, constraints,
-                                        same_queue, victim_filter)
+            # storm regime: thousands of identical
+            # starving preemptors would each pay the full victim
+            # collection + plugin intersection + scoring pass.  Identical
+            # preemptor shapes (queue, priority, class signature) share
+            # ONE scan per cycle; the walk itself stays exact because
+            # _preempt_one revalidates fit and skips RELEASING victims.
+            mkey = None
+            if memo is not None:
+                mkey = (job.queue, tc.tasks[0].priority, tc.signature)
+                scan = memo.get(mkey, False)
+                if scan is False:
+                    scan = self._candidate_scan(
+                        ssn, job, tc, req_vec, constraints, same_queue,
+                        victim_filter)
+                    memo[mkey] = scan
+            else:
+                scan = self._candidate_scan(ssn, job, tc, req_vec,
+                                            constraints, same_queue,
+                                            victim_filter)
             if scan is None:
                 continue
             order, victims_by_node = scan
