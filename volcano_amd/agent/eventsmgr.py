@@ -30,14 +30,26 @@ class UsageProbe:
     (volcano.sh/cpu-usage, volcano.sh/memory-usage — same keys the usage
     plugin and metrics pipeline use)."""
 
-    def __init__(self, store: ObjectStore, node_name: str):
+    def __init__(self, store: ObjectStore, node_name: str,
+                 usage_getter=None):
         self.store = store
         self.node_name = node_name
         self.injected: Optional[NodeUsage] = None
+        # metriccollect-backed source (ResourceUsageGetter): real local
+        # counters when running as a node daemon
+        self.usage_getter = usage_getter
 
     def sample(self) -> Optional[NodeUsage]:
         if self.injected is not None:
             return self.injected
+        if self.usage_getter is not None:
+            pods = self.store.list(
+                "Pod", selector=lambda p: p.node_name == self.node_name
+                and p.phase in ("Running", "Pending"))
+            return NodeUsage(node_name=self.node_name,
+                             cpu_pct=self.usage_getter.cpu_pct(),
+                             mem_pct=self.usage_getter.memory_pct(),
+                             pod_count=len(pods))
         node = self.store.get("Node", "default", self.node_name)
         if node is None:
             return None

@@ -26,10 +26,15 @@ def _offline(pod) -> bool:
 
 
 class CpuQosHandler:
-    """Offline pods' cpu quota shrinks as node pressure rises."""
+    """Offline pods' cpu quota shrinks as node pressure rises.  With a
+    CgroupDriver the quota is written to the pod slice's cpu.max /
+    cpu.cfs_quota_us (the reference's kernel contract); the annotation
+    stays as the observable report."""
 
-    def __init__(self, low: float = 50.0, high: float = 80.0):
+    def __init__(self, low: float = 50.0, high: float = 80.0,
+                 cgroup=None):
         self.low, self.high = low, high
+        self.cgroup = cgroup
 
     def handle(self, usage) -> None:
         for p in self.store.list(
@@ -46,14 +51,21 @@ class CpuQosHandler:
             if p.meta.annotations.get(ANN_CPU_QUOTA) != val:
                 p.meta.annotations[ANN_CPU_QUOTA] = val
                 self.store.update("Pod", p)
+                if self.cgroup is not None:
+                    uid = p.meta.uid or p.meta.key
+                    self.cgroup.set_cpu_quota(uid, True, quota)
+                    self.cgroup.set_cpu_burst(uid, True,
+                                              int(quota * 100))
 
 
 class MemoryQosHandler:
     """memory.high for offline pods = request × factor (cgroup v2
-    memoryqosv2 analog)."""
+    memoryqosv2 analog); with a CgroupDriver the value lands in
+    memory.high (and guaranteed pods would get memory.low/min)."""
 
-    def __init__(self, factor: float = 1.2):
+    def __init__(self, factor: float = 1.2, cgroup=None):
         self.factor = factor
+        self.cgroup = cgroup
 
     def handle(self, usage) -> None:
         for p in self.store.list(
@@ -63,6 +75,10 @@ class MemoryQosHandler:
             if p.meta.annotations.get(ANN_MEM_HIGH) != val:
                 p.meta.annotations[ANN_MEM_HIGH] = val
                 self.store.update("Pod", p)
+                if self.cgroup is not None:
+                    uid = p.meta.uid or p.meta.key
+                    self.cgroup.set_memory_high(
+                        uid, True, p.request.memory * self.factor)
 
 
 class NetworkQosHandler:
@@ -70,9 +86,11 @@ class NetworkQosHandler:
     eBPF map values ONLINE_BANDWIDTH_WATERMARK etc.); offline share is
     squeezed when online traffic needs the headroom."""
 
-    def __init__(self, total_bps: float = 25e9, offline_share: float = 0.3):
+    def __init__(self, total_bps: float = 25e9, offline_share: float = 0.3,
+                 enforcer=None):
         self.total = total_bps
         self.offline_share = offline_share
+        self.enforcer = enforcer      # NetQoSEnforcer (tc/eBPF layer)
 
     def handle(self, usage) -> None:
         node = self.store.get("Node", "default", self.node_name)
@@ -84,3 +102,10 @@ class NetworkQosHandler:
         if node.meta.annotations.get(ANN_NET_LIMIT) != val:
             node.meta.annotations[ANN_NET_LIMIT] = val
             self.store.update("Node", node)
+        if self.enforcer is not None:
+            if not self.enforcer.attached:
+                self.enforcer.attach()
+            # online pressure approximated from node cpu utilization
+            online = int(self.enforcer.config.total_bps
+                         * min(usage.cpu_pct, 100.0) / 100.0)
+            self.enforcer.adjust(online)
