@@ -25,6 +25,9 @@ class SchedulerConfiguration:
     actions: List[str] = field(default_factory=lambda: ["enqueue", "allocate", "backfill"])
     tiers: List[Tier] = field(default_factory=list)
     configurations: Dict[str, Dict[str, object]] = field(default_factory=dict)
+    # reference conf `metrics:` section (type/address/interval) — feeds
+    # node real-usage into the cache (cache.setMetricsData analog)
+    metrics: Dict[str, object] = field(default_factory=dict)
     schedule_period: float = 1.0
     use_hip: bool = False          # decision plane on GPU (HIP kernels)
     device: str = "cpu"
@@ -56,6 +59,8 @@ class SchedulerConfiguration:
             conf.tiers = tiers
         for c in data.get("configurations", []) or []:
             conf.configurations[c.get("name", "")] = c.get("arguments", {}) or {}
+        if isinstance(data.get("metrics"), dict):
+            conf.metrics = dict(data["metrics"])
         fg = data.get("feature_gates") or data.get("featureGates") or {}
         if isinstance(fg, dict):
             conf.feature_gates = {str(k): bool(v) for k, v in fg.items()}

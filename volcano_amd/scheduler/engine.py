@@ -26,6 +26,21 @@ class Scheduler:
             features.set_gates(self.config.feature_gates)
         self._stop = threading.Event()
         self._actions = [actions_mod.new_action(a) for a in self.config.actions]
+        # real-usage metrics source (reference cache.setMetricsData,
+        # cache.go:1808): periodically publishes node usage annotations
+        # the usage plugin / oversubscription consume
+        self._metrics_source = None
+        self._metrics_interval = 30.0
+        self._metrics_last = 0.0
+        m = dict(self.config.metrics or {})
+        if m.get("type") and m.get("type") != "annotation":
+            from .metrics_source import new_source
+            self._metrics_interval = float(m.pop("interval", 30.0))
+            kind = m.pop("type")
+            try:
+                self._metrics_source = new_source(kind, **m)
+            except Exception:
+                self._metrics_source = None
 
     # -- session lifecycle (framework.go:34-71) ------------------------------
     def open_session(self) -> Session:
@@ -61,9 +76,31 @@ class Scheduler:
         if hasattr(ssn, "job_enqueued_fns"):
             ssn.job_enqueued_fns.clear()
 
+    def _sync_metrics(self) -> None:
+        src = self._metrics_source
+        if src is None or self.cache.store is None:
+            return
+        now = time.monotonic()
+        if now - self._metrics_last < self._metrics_interval:
+            return
+        self._metrics_last = now
+        from .metrics_source import ANN_CPU_USAGE, ANN_MEM_USAGE
+        nodes = self.cache.store.list("Node")
+        usage = src.node_usage(nodes)
+        for node in nodes:
+            u = usage.get(node.meta.name)
+            if not u:
+                continue
+            ann = node.meta.annotations
+            new = (str(u["cpu"]), str(u["memory"]))
+            if (ann.get(ANN_CPU_USAGE), ann.get(ANN_MEM_USAGE)) != new:
+                ann[ANN_CPU_USAGE], ann[ANN_MEM_USAGE] = new
+                self.cache.store.update("Node", node)
+
     # -- cycle ---------------------------------------------------------------
     def run_once(self) -> Session:
         t0 = time.perf_counter()
+        self._sync_metrics()
         ssn = self.open_session()
         timings = {}
         for action in self._actions:
