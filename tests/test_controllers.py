@@ -341,3 +341,79 @@ def test_queue_close_drains_then_closes():
     cm.sync_until_quiet()
     assert store.get("Queue", "default",
                      "default").status.state == "Open"
+
+
+def test_job_controller_sharded_workers_and_error_backoff():
+    """Reference job_controller.go: hashed worker shards + errTasks
+    rate-limited resync."""
+    import time
+
+    from volcano_amd.controllers.job import JobController
+    from volcano_amd.store import ObjectStore
+    from volcano_amd.utils import synth
+
+    store = ObjectStore()
+    store.create("Queue", synth.make_queue("default"))
+    ctl = JobController()
+    ctl.initialize(store)
+    # distinct jobs land on distinct shards deterministically
+    shards = {JobController._shard_of(f"default/j{i}", 4)
+              for i in range(32)}
+    assert len(shards) > 1
+    for i in range(8):
+        store.create("Job", mk_job(f"shard-{i}", replicas=2))
+    ctl.sync_once()
+    # all jobs reconciled (pods created) despite multi-shard execution
+    for i in range(8):
+        pods = [p for p in store.list("Pod")
+                if p.meta.labels.get("volcano.sh/job-name") == f"shard-{i}"]
+        assert len(pods) == 2
+
+    # error backoff: a failing sync re-queues with increasing delay
+    fails = {"n": 0}
+    orig = ctl.sync_job
+
+    def flaky(job):
+        if job.meta.name == "shard-0" and fails["n"] < 2:
+            fails["n"] += 1
+            raise RuntimeError("transient")
+        return orig(job)
+
+    ctl.sync_job = flaky
+    ctl._dirty.add("default/shard-0")
+    ctl.sync_once()
+    assert fails["n"] == 1
+    assert "default/shard-0" in ctl._err_queue
+    ctl.sync_once()                      # backoff not yet expired
+    assert fails["n"] == 1
+    time.sleep(0.06)
+    ctl.sync_once()                      # first retry (fails again)
+    assert fails["n"] == 2
+    time.sleep(0.11)
+    ctl.sync_once()                      # second retry succeeds
+    assert "default/shard-0" not in ctl._err_queue
+
+
+def test_jobtemplate_controller_bookkeeping():
+    """Standalone jobtemplate controller (reference
+    pkg/controllers/jobtemplate): status tracks jobs created from the
+    template."""
+    from volcano_amd.api.objects import (FlowStep, JobFlow, JobTemplate,
+                                         ObjectMeta)
+    from volcano_amd.controllers import ControllerManager
+    from volcano_amd.store import ObjectStore
+    from volcano_amd.utils import synth
+
+    store = ObjectStore()
+    store.create("Queue", synth.make_queue("default"))
+    store.create("JobTemplate", JobTemplate(
+        meta=ObjectMeta(name="step-a", namespace="default"),
+        spec=mk_job("ignored", replicas=1).spec))
+    store.create("JobFlow", JobFlow(
+        meta=ObjectMeta(name="f1", namespace="default"),
+        flows=[FlowStep(name="step-a")]))
+    cm = ControllerManager(store, ["jobflow", "jobtemplate", "job",
+                                   "podgroup"])
+    cm.sync_until_quiet()
+    tmpl = store.get("JobTemplate", "default", "step-a")
+    assert tmpl.status.get("jobDependsOnList") == ["f1-step-a"]

@@ -453,6 +453,9 @@ class JobFlow:
 class JobTemplate:
     meta: ObjectMeta = field(default_factory=ObjectMeta)
     spec: JobSpec = field(default_factory=JobSpec)
+    # reference jobtemplate controller bookkeeping: names of Jobs
+    # created from this template (status.jobDependsOnList)
+    status: Dict[str, Any] = field(default_factory=dict)
 
 
 @dataclass

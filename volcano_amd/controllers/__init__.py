@@ -3,6 +3,7 @@ from .job import JobController
 from .podgroup import PodGroupController
 from .queue import QueueController
 from .jobflow import JobFlowController
+from .jobtemplate import JobTemplateController
 from .cronjob import CronJobController
 from .garbagecollector import GarbageCollector
 from .hypernode import HyperNodeController

@@ -30,6 +30,15 @@ FINAL_PHASES = (JobPhase.COMPLETED.value, JobPhase.FAILED.value,
 class JobController(Controller):
     watch_kinds = ("Job", "Pod", "Command")
 
+    #: sharded work queues (reference job_controller.go:139-186,313-325:
+    #: a job key is hashed to ONE of N workers so per-job ordering holds
+    #: while distinct jobs reconcile concurrently)
+    workers = 4
+    #: error-resync backoff (reference job_controller.go:145 errTasks
+    #: rate-limited queue): base delay, doubled per consecutive failure
+    err_backoff_base = 0.05
+    err_backoff_max = 5.0
+
     def initialize(self, store) -> None:
         super().initialize(store)
         self._dirty: set = set()
@@ -37,6 +46,13 @@ class JobController(Controller):
         # queued event shows the pod's LATEST state at drain time — fire
         # lifecycle policies only on an observed phase *transition*
         self._pod_phase_seen: Dict[str, str] = {}
+        # key → (retry deadline, consecutive failures)
+        self._err_queue: Dict[str, tuple] = {}
+
+    @staticmethod
+    def _shard_of(key: str, workers: int) -> int:
+        import zlib
+        return zlib.crc32(key.encode()) % max(workers, 1)
 
     # -- event routing --------------------------------------------------------
     def handle(self, ev) -> None:
@@ -64,12 +80,53 @@ class JobController(Controller):
             self._command(ev)
 
     def resync(self) -> None:
+        import time as _time
         dirty, self._dirty = self._dirty, set()
+        # retryable failures whose backoff deadline passed re-enter the
+        # worksheet (errTasks resync)
+        now = _time.monotonic()
+        for key, (deadline, _) in list(self._err_queue.items()):
+            if deadline <= now:
+                dirty.add(key)
+        if not dirty:
+            return
+        # shard by key hash — per-job ordering inside a shard, shards
+        # reconcile concurrently (the reference's N worker goroutines)
+        shards: List[List[str]] = [[] for _ in range(self.workers)]
         for key in dirty:
-            ns, name = key.split("/", 1)
-            job = self.store.get("Job", ns, name)
-            if job is not None:
-                self.sync_job(job)
+            shards[self._shard_of(key, self.workers)].append(key)
+
+        def run_shard(keys: List[str]) -> None:
+            for key in keys:
+                ns, name = key.split("/", 1)
+                job = self.store.get("Job", ns, name)
+                if job is None:
+                    self._err_queue.pop(key, None)
+                    continue
+                try:
+                    self.sync_job(job)
+                except Exception:
+                    _, fails = self._err_queue.get(key, (0.0, 0))
+                    fails += 1
+                    delay = min(self.err_backoff_base * (2 ** (fails - 1)),
+                                self.err_backoff_max)
+                    self._err_queue[key] = (_time.monotonic() + delay,
+                                            fails)
+                else:
+                    self._err_queue.pop(key, None)
+
+        live = [sh for sh in shards if sh]
+        if len(live) <= 1:
+            for sh in live:
+                run_shard(sh)
+        else:
+            import threading
+            ts = [threading.Thread(target=run_shard, args=(sh,))
+                  for sh in live]
+            for t in ts:
+                t.start()
+            for t in ts:
+                t.join()
 
     # -- lifecycle events → actions (bus/v1alpha1) ----------------------------
     def _pod_event(self, job_key: str, pod: Pod) -> None:

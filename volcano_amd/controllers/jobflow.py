@@ -81,6 +81,8 @@ class JobFlowController(Controller):
             self.store.create("Job", Job(
                 meta=ObjectMeta(name=self._job_name(flow, step.name),
                                 namespace=flow.meta.namespace,
+                                labels={"flow.volcano.sh/jobtemplate":
+                                        step.name},
                                 owner=f"JobFlow/{flow.meta.key}"),
                 spec=spec))
             states[step.name] = JobPhase.PENDING.value
