@@ -410,6 +410,9 @@ class NumaZone:
     id: int = 0
     cpu_milli: float = 0.0       # allocatable millicores in this NUMA node
     memory: float = 0.0
+    # explicit CPU id list for cpuset pinning (reference numaaware
+    # providers); empty → derived as a contiguous range from cpu_milli
+    cpus: List[int] = field(default_factory=list)
 
 
 @dataclass

@@ -19,6 +19,33 @@ from .base import Plugin, register
 
 ANN_POLICY = "volcano.sh/numa-topology-policy"
 ANN_PINNED = "volcano.sh/numa-node"
+ANN_CPUSET = "volcano.sh/cpuset"
+
+
+class CpusetProvider:
+    """cpuset assignment provider (reference ``numaaware/provider``):
+    hands out SPECIFIC CPU ids from a pinned zone's pool — the contract
+    the node-side cpumanager enforces via cgroup cpuset.cpus.  Whole
+    cores only (ceil of millicores/1000); released ids return to the
+    pool (cache._release_devices analog is the pod-delete path)."""
+
+    def __init__(self, topo):
+        self.pools: Dict[int, List[int]] = {}
+        for z in topo.zones:
+            cpus = list(z.cpus) if z.cpus else list(range(
+                z.id * 1000, z.id * 1000 + int(z.cpu_milli // 1000)))
+            self.pools[z.id] = cpus
+
+    def assign(self, zone: int, cpu_milli: float) -> List[int]:
+        need = max(1, int(-(-cpu_milli // 1000)))     # ceil cores
+        pool = self.pools.get(zone, [])
+        if len(pool) < need:
+            return []
+        out, self.pools[zone] = pool[:need], pool[need:]
+        return out
+
+    def release(self, zone: int, cpus: List[int]) -> None:
+        self.pools.setdefault(zone, []).extend(cpus)
 
 
 class NumaState:
@@ -26,6 +53,7 @@ class NumaState:
 
     def __init__(self, topo):
         self.free: List[float] = [z.cpu_milli for z in topo.zones]
+        self.cpuset = CpusetProvider(topo)
 
     def best_zone(self, cpu_milli: float):
         best = None
@@ -102,7 +130,12 @@ class NumaAwarePlugin(Plugin):
                     if zone is not None:
                         st.take(zone, cpu)
                         if task.pod is not None:
-                            task.pod.meta.annotations[ANN_PINNED] = str(zone)
+                            ann = task.pod.meta.annotations
+                            ann[ANN_PINNED] = str(zone)
+                            ids = st.cpuset.assign(zone, cpu)
+                            if ids:
+                                ann[ANN_CPUSET] = ",".join(
+                                    str(c) for c in ids)
 
         handler = type("NumaHandler", (), {"on_allocate":
                                            staticmethod(on_allocate)})()
