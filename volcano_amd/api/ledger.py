@@ -115,6 +115,16 @@ class NodeLedger:
         np.add.at(self.planes[USED, :, :len(req)], rows, vals)
         self.version += 1
 
+    def add_used_rows(self, rows: np.ndarray, vals: np.ndarray) -> None:
+        """used[rows] += vals ([K, R] per-piece deltas — the whole apply
+        pass in ONE np.add.at; duplicate rows accumulate)."""
+        w = self.width
+        R = vals.shape[1]
+        if R > w:
+            self._widen(R)
+        np.add.at(self.planes[USED, :, :R], rows, vals)
+        self.version += 1
+
     def zero_usage(self) -> None:
         self.planes[:] = 0.0
         self.version += 1

@@ -433,18 +433,32 @@ class AllocateAction:
     def _gate_subgroup_families(self, plan, result) -> None:
         """minSubGroups: if fewer than the required number of a policy's
         subgroups fully placed, revert the ones that did (whole-family
-        atomicity above the per-subgroup gangs)."""
-        for real_key, min_subs, members in getattr(
-                plan, "subgroup_families", []):
-            placed = [k for k in members if result.job_committed.get(k)]
+        atomicity above the per-subgroup gangs).  Mutates the ARRAY
+        result (job_flag / log_counts) — the apply path reads those."""
+        families = getattr(plan, "subgroup_families", [])
+        if not families:
+            return
+        class_of = {cp.job_key: c for c, cp in enumerate(plan.classes)}
+        jf = result.job_flag
+        cls_job = plan.class_job
+        for real_key, min_subs, members in families:
+            placed = [k for k in members
+                      if k in class_of
+                      and jf[cls_job[class_of[k]]]
+                      and result.class_placed[class_of[k]] > 0]
             if len(placed) >= min_subs:
                 continue
-            for c, cp in enumerate(plan.classes):
-                if cp.job_key in placed:
-                    cres = result.class_results[c]
-                    self._revert_pieces(plan, cp, cres.placements)
-                    cres.placements = []
-                    result.job_committed[cp.job_key] = False
+            for k in placed:
+                c = class_of[k]
+                cp = plan.classes[c]
+                ln, lc = result.class_entries(c)
+                self._revert_pieces(
+                    plan, cp,
+                    [(int(n), int(cnt)) for n, cnt in zip(ln, lc)
+                     if cnt > 0])
+                lc[:] = 0
+                result.class_placed[c] = 0
+                jf[cls_job[c]] = 0
 
     def _try_nominated(self, ssn, job, qi: int, predicates) -> bool:
         """Commit a fully-nominated gang onto its nominated nodes if they
@@ -577,16 +591,23 @@ class AllocateAction:
         # fire event handlers only when someone registered one — building
         # the per-piece argument lists for nobody was measurable at 10k jobs
         fire = ssn.fire_allocate if ssn.event_handlers else None
+        jf = result.job_flag
+        cls_job = plan.class_job
+        ln_all = result.log_nodes
+        lc_all = result.log_counts
+        ll = result.log_len
+        # per-piece ledger accumulation across ALL classes — one
+        # np.add.at after the walk (was one bulk op per class)
+        acc_rows: List[int] = []
+        acc_cnts: List[int] = []
+        acc_cls: List[int] = []
 
-        def commit_pieces(job, cp, pieces, led_rows, led_counts):
+        def commit_pieces(job, cp, c, pieces):
             """Assign (node_id, count) pieces to the next tasks of `job`.
-            Node accounting is ledger-vectorized per CLASS (led_rows/
-            led_counts accumulate here, one np.add.at after the walk);
-            node→task membership defers into NodeInfo._batches (folded
-            lazily by cold readers).  Status moves once, PENDING→BOUND,
-            inside bind_tasks (the reference's Allocated→Binding→Bound
-            pipeline compressed — nothing observes the intermediate
-            states between plan apply and bind here)."""
+            Node accounting accumulates per piece for the single end-of-
+            apply np.add.at; node→task membership defers into
+            NodeInfo._batches (folded lazily by cold readers).  Status
+            moves once, PENDING→BOUND, inside bind_tasks."""
             jl = bind_by_job.setdefault(job.key, [])
             for nid, count, tasks in pieces:
                 ni = nodes_sorted[nid]
@@ -594,35 +615,46 @@ class AllocateAction:
                 for t in tasks:
                     t.node_name = name
                 ni._batches.append(tasks)
-                led_rows.append(nid)
-                led_counts.append(count)
+                acc_rows.append(nid)
+                acc_cnts.append(count)
+                acc_cls.append(c)
                 to_bind.extend(tasks)
                 jl.extend(tasks)
             committed_jobs.add(job.key)
 
-        for c, (cp, cres) in enumerate(zip(plan.classes, result.class_results)):
-            if not result.job_committed.get(cp.job_key, False):
+        for c, cp in enumerate(plan.classes):
+            if not jf[cls_job[c]]:
                 continue
-            if not cres.placements:
+            n = int(ll[c])
+            if n == 0:
                 continue
-            led_rows: List[int] = []
-            led_counts: List[int] = []
+            off = cp.log_off
             if cp.bundle is None:
                 job = ssn.jobs.get(cp.job_key) or \
                     ssn.jobs[getattr(plan, "job_alias", {})[cp.job_key]]
                 tasks = iter(cp.tclass.tasks)
-                pieces = [(nid, cnt, [next(tasks) for _ in range(cnt)])
-                          for nid, cnt in cres.placements]
-                if bad_nodes and any(nid in bad_nodes
-                                     for nid, _ in cres.placements):
+                pieces = []
+                hit_bad = False
+                for e in range(off, off + n):
+                    cnt = int(lc_all[e])
+                    if cnt <= 0:
+                        continue
+                    nid = int(ln_all[e])
+                    if bad_nodes and nid in bad_nodes:
+                        hit_bad = True
+                    pieces.append((nid, cnt,
+                                   [next(tasks) for _ in range(cnt)]))
+                if not pieces:
+                    continue
+                if hit_bad:
                     self._revert_pieces(plan, cp,
                                         [(nid, cnt) for nid, cnt, _ in pieces])
                     continue
-                commit_pieces(job, cp, pieces, led_rows, led_counts)
+                commit_pieces(job, cp, c, pieces)
                 if fire is not None:
                     fire(cp.tclass,
-                         [nid for nid, _ in cres.placements],
-                         [cnt for _, cnt in cres.placements],
+                         [nid for nid, _, _ in pieces],
+                         [cnt for _, cnt, _ in pieces],
                          [t for _, _, ts in pieces for t in ts])
             else:
                 # -- bundle: walk jobs over the placement stream.  An entry
@@ -631,17 +663,16 @@ class AllocateAction:
                 # place; an entry touching a lost node (soft-shard
                 # conflict) reverts its slots outright.  Unclaimed slots
                 # revert at the end.
-                stream = cres.placements
                 ei, eoff = 0, 0
-                nstream = len(stream)
                 reverted: List[tuple] = []
                 for be in cp.bundle:
                     ei0, eoff0 = ei, eoff
                     pieces = []
                     need = be.ntasks
                     toff = 0
-                    while need > 0 and ei < nstream:
-                        nid, cnt = stream[ei]
+                    while need > 0 and ei < n:
+                        nid = int(ln_all[off + ei])
+                        cnt = int(lc_all[off + ei])
                         avail = cnt - eoff
                         take = avail if avail < need else need
                         pieces.append((nid, take, be.tasks[toff:toff + take]))
@@ -659,30 +690,36 @@ class AllocateAction:
                         reverted.extend((nid, cnt) for nid, cnt, _ in pieces)
                         continue
                     job = ssn.jobs[be.job_key]
-                    commit_pieces(job, cp, pieces, led_rows, led_counts)
+                    commit_pieces(job, cp, c, pieces)
                     if fire is not None:
                         fire(cp.tclass, [p[0] for p in pieces],
                              [p[1] for p in pieces],
                              [t for _, _, ts in pieces for t in ts])
-                if ei < nstream:
+                if ei < n:
                     # slots the walk never claimed
                     if eoff:
-                        reverted.append((stream[ei][0], stream[ei][1] - eoff))
+                        reverted.append((int(ln_all[off + ei]),
+                                         int(lc_all[off + ei]) - eoff))
                         ei += 1
-                    reverted.extend(stream[ei:])
+                    reverted.extend((int(ln_all[off + e]),
+                                     int(lc_all[off + e]))
+                                    for e in range(ei, n))
                 if reverted:
                     self._revert_pieces(plan, cp, reverted)
-            if led_rows:
-                if ledger is not None:
-                    ledger.add_used_bulk(
-                        np.asarray(led_rows, dtype=np.int64),
-                        np.asarray(led_counts, dtype=np.int64),
-                        cp.req)
-                else:   # no ledger (bare-cache tests): per-piece Resource math
-                    for nid, cnt in zip(led_rows, led_counts):
-                        nodes_sorted[nid]._acct(
-                            cp.tclass.request.clone().multi(float(cnt)),
-                            1, 0, 0)
+
+        if acc_rows:
+            rows = np.asarray(acc_rows, dtype=np.int64)
+            cnts = np.asarray(acc_cnts, dtype=np.float64)
+            vals = cnts[:, None] * plan.req_np[
+                np.asarray(acc_cls, dtype=np.int64)].astype(np.float64)
+            if ledger is not None:
+                ledger.add_used_rows(rows, vals)
+            else:   # bare-cache tests: per-piece Resource math
+                for k in range(len(acc_rows)):
+                    cp = plan.classes[acc_cls[k]]
+                    nodes_sorted[acc_rows[k]]._acct(
+                        cp.tclass.request.clone().multi(float(acc_cnts[k])),
+                        1, 0, 0)
 
         if to_bind:
             ssn.cache.bind_tasks(to_bind, by_job=bind_by_job)

@@ -104,11 +104,54 @@ class ClassResult:
     placed: int
 
 
-@dataclass
 class CycleResult:
-    class_results: List[ClassResult]
-    job_placed: Dict[str, int]
-    job_committed: Dict[str, bool]
+    """Array-backed cycle outcome (one readback): per-class undo logs,
+    placed counts, per-job commit flags.  The apply path consumes the
+    numpy arrays directly; the dict/object views materialize lazily for
+    cold consumers (subgroup gating, tests)."""
+
+    def __init__(self, plan, log_nodes, log_counts, log_len, class_placed,
+                 job_placed, job_flag):
+        self.plan = plan
+        self.log_nodes = log_nodes
+        self.log_counts = log_counts
+        self.log_len = log_len
+        self.class_placed = class_placed
+        self.job_placed_arr = job_placed
+        self.job_flag = job_flag
+        self._committed = None
+        self._class_results = None
+
+    @property
+    def job_committed(self) -> Dict[str, bool]:
+        if self._committed is None:
+            self._committed = {jp.job_key: bool(self.job_flag[j])
+                               for j, jp in enumerate(self.plan.jobs)}
+        return self._committed
+
+    @property
+    def job_placed(self) -> Dict[str, int]:
+        return {jp.job_key: int(self.job_placed_arr[j])
+                for j, jp in enumerate(self.plan.jobs)}
+
+    def class_entries(self, c: int):
+        """(nodes, counts) numpy views of class c's undo log."""
+        cp = self.plan.classes[c]
+        n = int(self.log_len[c])
+        sl = slice(cp.log_off, cp.log_off + n)
+        return self.log_nodes[sl], self.log_counts[sl]
+
+    @property
+    def class_results(self) -> List[ClassResult]:
+        if self._class_results is None:
+            out = []
+            for c in range(len(self.plan.classes)):
+                ln, lc = self.class_entries(c)
+                entries = [(int(n), int(k)) for n, k in zip(ln, lc)
+                           if k > 0]
+                out.append(ClassResult(entries, int(self.class_placed[c])))
+            self._class_results = out
+        return self._class_results
 
 
 class CyclePlan:
@@ -135,9 +178,10 @@ class CyclePlan:
             occupied=job.occupied_count, min_available=job.min_available))
 
     def finalize(self) -> None:
-        """Assign undo-log slots (one contiguous region per class) and
-        normalize constraint widths (the label-bit registry may have grown
-        while classes were built)."""
+        """Assign undo-log slots (one contiguous region per class),
+        normalize constraint widths (the label-bit registry may have
+        grown while classes were built), and stage the per-class job
+        index + request matrix both runners and the apply path share."""
         # per-class bias rows REPLACE the plan bias in the kernel — fold
         # the plan-wide plane (score_bias / soft-shard stagger) into them
         # so biased classes don't lose it.  Shared rows fold once.
@@ -159,7 +203,9 @@ class CyclePlan:
             else max(self.nt.labels.words, 1)
         off = 0
         n = self.nt.n
-        for cp in self.classes:
+        C = len(self.classes)
+        self.req_np = np.empty((C, self.nt.r), dtype=np.float32)
+        for c, cp in enumerate(self.classes):
             cp.log_off = off
             cp.log_cap = max(1, min(cp.ntasks, n))
             off += cp.log_cap
@@ -167,7 +213,13 @@ class CyclePlan:
                 cp.require = np.pad(cp.require, (0, W - len(cp.require)))
             if len(cp.forbid) < W:
                 cp.forbid = np.pad(cp.forbid, (0, W - len(cp.forbid)))
+            self.req_np[c] = cp.req
         self.log_total = off
+        lens = np.fromiter((jp.class_end - jp.class_begin
+                            for jp in self.jobs), dtype=np.int64,
+                           count=len(self.jobs))
+        self.class_job = np.repeat(
+            np.arange(len(self.jobs), dtype=np.int32), lens)
 
     @property
     def n_classes(self) -> int:
@@ -250,8 +302,10 @@ def run_plan_torch(plan: CyclePlan) -> CycleResult:
                                 plan.queue_alloc[cp.queue_idx],
                                 class_placed[c], job_placed[j])
 
-    return _collect(plan, log_nodes, log_counts, log_len, class_placed,
-                    job_placed, job_flag)
+    return CycleResult(plan, log_nodes.cpu().numpy(),
+                      log_counts.cpu().numpy(), log_len.cpu().numpy(),
+                      class_placed.cpu().numpy(), job_placed.cpu().numpy(),
+                      job_flag.cpu().numpy())
 
 
 def run_plan_hip(plan: CyclePlan) -> CycleResult:
@@ -279,16 +333,13 @@ def run_plan_hip(plan: CyclePlan) -> CycleResult:
     # binary-compatible with VamdClassDesc/VamdJobDesc).
     J = len(plan.jobs)
     cds = np.empty(C, dtype=CLASS_DT)
-    lens = np.empty(J, dtype=np.int64)
     jds = np.empty(J, dtype=JOB_DT)
     for j, jp in enumerate(plan.jobs):
-        lens[j] = jp.class_end - jp.class_begin
         jds[j] = (jp.class_begin, jp.class_end, jp.occupied,
                   jp.min_available)
-    # job_idx via vectorized expansion of each job's class range
-    job_idx = np.repeat(np.arange(J, dtype=np.int32), lens)
+    job_idx = plan.class_job
 
-    req_np = np.empty((C, R), dtype=np.float32)
+    req_np = plan.req_np
     require_np = np.empty((C, W), dtype=np.int64)
     forbid_np = np.empty((C, W), dtype=np.int64)
     tol_np = np.empty(C, dtype=np.int64)
@@ -309,7 +360,6 @@ def run_plan_hip(plan: CyclePlan) -> CycleResult:
                   cp.min_needed, cp.log_off, cp.log_cap,
                   1 if cp.use_future else 0, brow,
                   cp.w_least, cp.w_most, cp.w_bal, 0.0)
-        req_np[c] = cp.req
         require_np[c] = cp.require
         forbid_np[c] = cp.forbid
         tol_np[c] = cp.tolerated
@@ -351,29 +401,12 @@ def run_plan_hip(plan: CyclePlan) -> CycleResult:
         log_nodes, log_counts, log_len, class_placed, job_placed, job_flag,
         sort_scratch)
 
-    res = _collect(plan, log_nodes.cpu(), log_counts.cpu(), log_len.cpu(),
-                   class_placed.cpu(), job_placed.cpu(), job_flag.cpu())
+    res = CycleResult(plan, log_nodes.cpu().numpy(),
+                      log_counts.cpu().numpy(), log_len.cpu().numpy(),
+                      class_placed.cpu().numpy(), job_placed.cpu().numpy(),
+                      job_flag.cpu().numpy())
     # the cycle mutated device queue_alloc; reflect back to the host copy
     plan.queue_alloc.copy_(q_alloc.cpu())
     return res
 
 
-def _collect(plan, log_nodes, log_counts, log_len, class_placed, job_placed,
-             job_flag) -> CycleResult:
-    ln = log_nodes.cpu().numpy()
-    lc = log_counts.cpu().numpy()
-    ll = log_len.cpu().numpy()
-    cpn = class_placed.cpu().numpy()
-    jpn = job_placed.cpu().numpy()
-    jf = job_flag.cpu().numpy()
-
-    class_results: List[ClassResult] = []
-    for c, cp in enumerate(plan.classes):
-        entries = []
-        for e in range(cp.log_off, cp.log_off + int(ll[c])):
-            if lc[e] > 0:
-                entries.append((int(ln[e]), int(lc[e])))
-        class_results.append(ClassResult(entries, int(cpn[c])))
-    job_placed_map = {jp.job_key: int(jpn[j]) for j, jp in enumerate(plan.jobs)}
-    job_committed = {jp.job_key: bool(jf[j]) for j, jp in enumerate(plan.jobs)}
-    return CycleResult(class_results, job_placed_map, job_committed)
