@@ -114,7 +114,9 @@ class AllocateAction:
                         continue
                     qi = ssn.queue_index[q.name]
                     rows = ssn.ordered_job_rows(jt, sel[qis == qi])
-                    ordered_jobs.extend((q, jjobs[int(k)]) for k in rows)
+                    # tolist(): native ints, no per-row numpy scalar boxing
+                    ordered_jobs.extend((q, jjobs[k])
+                                        for k in rows.tolist())
         else:
             by_queue = {}
             for job in ssn.jobs.values():
@@ -188,6 +190,9 @@ class AllocateAction:
 
         _MISS = object()
         req_memo = {}
+        # (sig, qi) → (req, tolerated, require, forbid): one lookup per
+        # job covers both the request vector and the constraint planes
+        cons_memo = {}
         from ...api.types import TaskStatus as _TS
         w_least = w.get("least", 1.0)
         w_most = w.get("most", 0.0)
@@ -254,23 +259,36 @@ class AllocateAction:
                     if gang_min < b.min_needed:
                         b.min_needed = gang_min
                     continue
-                req = req_memo.get(sig, _MISS)
-                if req is _MISS:
+                ckey = (sig, qi)
+                got = cons_memo.get(ckey, _MISS)
+                if got is _MISS:
                     req = nt.req_vector(first)
-                    req_memo[sig] = req
-                if req is None:
+                    if req is None:
+                        got = None
+                    else:
+                        tc0 = TaskClass(signature=sig, role=role,
+                                        request=request, tasks=[first],
+                                        priority=priority)
+                        if predicates is not None:
+                            tol, require, forbid = \
+                                predicates.class_constraints(tc0, job)
+                        else:
+                            tol = -1
+                            W = max(nt.labels.words, 1)
+                            require = np.zeros(W, dtype=np.int64)
+                            forbid = np.zeros(W, dtype=np.int64)
+                        got = (req, tol, require, forbid)
+                    cons_memo[ckey] = got
+                if got is None:
                     continue      # asks for a resource no node offers
+                req, tol, require, forbid = got
                 tc = TaskClass(signature=sig, role=role, request=request,
                                tasks=list(pend.values()), priority=priority)
-                if predicates is not None:
-                    tol, require, forbid = predicates.class_constraints(tc, job)
-                else:
-                    tol = -1
-                    W = max(nt.labels.words, 1)
-                    require = np.zeros(W, dtype=np.int64)
-                    forbid = np.zeros(W, dtype=np.int64)
-                gang_min = max(job.min_available,
-                               job.min_task_member.get(role, 0))
+                mtm = job._mtm
+                if mtm is None:
+                    mtm = job.min_task_member
+                gang_min = job.min_available if not mtm else \
+                    max(job.min_available, mtm.get(role, 0))
                 cp = ClassPlan(tclass=tc, job_key=job.key, queue_idx=qi,
                                req=req, tolerated=tol, require=require,
                                forbid=forbid, min_needed=gang_min,
