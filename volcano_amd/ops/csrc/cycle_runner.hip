@@ -212,13 +212,6 @@ extern "C" void vamd_run_cycle(
             const float* b = (cd.bias_row >= 0 && bias_rows)
                 ? bias_rows + (size_t)cd.bias_row * N : bias;
 
-            vamd_score_cap(alloc, used, ext, ready, taints, planes,
-                           class_req + (size_t)c * R, class_tol[c],
-                           class_require + (size_t)c * W,
-                           class_forbid + (size_t)c * W,
-                           cd.w_least, cd.w_most, cd.w_bal, dim_w, b,
-                           score_scratch, cap_scratch, N, R, W, stream);
-
             // single-class jobs: gang check fused into the commit.  The
             // effective minimum is how many more tasks the job needs to
             // become ready (min_available - occupied), but never less than
@@ -229,6 +222,34 @@ extern "C" void vamd_run_cycle(
                 if (cd.min_needed > need) need = cd.min_needed;
                 fuse_min = need > 0 ? need : 0;
             }
+
+            if (cd.ntasks < 512 && N <= 4096) {
+                // small class on a small inventory: ONE fused launch
+                // (score in-block + select).  At larger N the one-block
+                // score loses the multi-CU latency hiding of the
+                // grid-stride kernel (mix bench at N=10k measured
+                // 552 ms/step fused vs 334 ms split).
+                vamd_fused_score_select(
+                    alloc, used, ext, ready, taints, planes,
+                    class_req + (size_t)c * R, class_tol[c],
+                    class_require + (size_t)c * W,
+                    class_forbid + (size_t)c * W,
+                    cd.w_least, cd.w_most, cd.w_bal, dim_w, b,
+                    score_scratch, cap_scratch, cd.ntasks,
+                    queue_alloc + (size_t)cd.queue_idx * R,
+                    queue_limit + (size_t)cd.queue_idx * R,
+                    log_nodes + cd.log_off, log_counts + cd.log_off,
+                    log_len + c, class_placed + c, job_placed + j,
+                    fuse_min, N, R, W, cd.log_cap, stream);
+                continue;
+            }
+
+            vamd_score_cap(alloc, used, ext, ready, taints, planes,
+                           class_req + (size_t)c * R, class_tol[c],
+                           class_require + (size_t)c * W,
+                           class_forbid + (size_t)c * W,
+                           cd.w_least, cd.w_most, cd.w_bal, dim_w, b,
+                           score_scratch, cap_scratch, N, R, W, stream);
 
             vamd_select_commit(score_scratch, cap_scratch,
                                class_req + (size_t)c * R, cd.ntasks, used,

@@ -332,6 +332,41 @@ select_commit_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// Fused score+select for small classes: ONE launch does the score pass
+// (block-stride, 16 waves hide the L2 latency at small N) and the
+// serial select.  Halves the per-class launch count on heterogeneous
+// plans where the grid-stride score kernel was dispatch-latency bound
+// (~12.5 us/class -> ~7 us measured on the mix bench).  Bit-identical
+// math: same score_pass + dev_select_commit bodies.
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(SC_THREADS)
+fused_score_select_kernel(
+    const float* __restrict__ alloc, float* __restrict__ used,
+    const float* __restrict__ extra,
+    const uint8_t* __restrict__ ready, const int64_t* __restrict__ taints,
+    const int64_t* __restrict__ planes,
+    const float* __restrict__ req, int64_t tolerated,
+    const int64_t* __restrict__ require, const int64_t* __restrict__ forbid,
+    float w_least, float w_most, float w_bal,
+    const float* __restrict__ dim_w, const float* __restrict__ bias,
+    float* __restrict__ score, int* __restrict__ cap,
+    int ntasks,
+    float* __restrict__ queue_alloc, const float* __restrict__ queue_limit,
+    int* __restrict__ log_nodes, int* __restrict__ log_counts,
+    int* __restrict__ log_len, int* __restrict__ placed,
+    int* __restrict__ job_placed, int fuse_min,
+    int N, int R, int W, int K)
+{
+    score_pass(threadIdx.x, SC_THREADS, alloc, used, extra, ready, taints,
+               planes, req, tolerated, require, forbid,
+               w_least, w_most, w_bal, dim_w, bias, score, cap, N, R, W);
+    __syncthreads();
+    dev_select_commit(score, cap, req, ntasks, used, queue_alloc,
+                      queue_limit, log_nodes, log_counts, log_len, placed,
+                      job_placed, fuse_min, N, R, K);
+}
+
+// ---------------------------------------------------------------------------
 // K4 gang readiness for multi-class jobs (JobReady AND-aggregation,
 // session_plugins.go:483).  One tiny block.
 // ---------------------------------------------------------------------------
@@ -1111,6 +1146,26 @@ void vamd_select_commit(
                        0, stream, score, cap, req, ntasks, used, queue_alloc,
                        queue_limit, log_nodes, log_counts, log_len, placed,
                        job_placed, fuse_min, N, R, K);
+}
+
+void vamd_fused_score_select(
+    const float* alloc, float* used, const float* extra,
+    const uint8_t* ready, const int64_t* taints, const int64_t* planes,
+    const float* req, int64_t tolerated, const int64_t* require,
+    const int64_t* forbid, float w_least, float w_most, float w_bal,
+    const float* dim_w, const float* bias, float* score, int* cap,
+    int ntasks, float* queue_alloc, const float* queue_limit,
+    int* log_nodes, int* log_counts, int* log_len, int* placed,
+    int* job_placed, int fuse_min, int N, int R, int W, int K,
+    hipStream_t stream)
+{
+    hipLaunchKernelGGL(vamd::fused_score_select_kernel, dim3(1),
+                       dim3(SC_THREADS), 0, stream, alloc, used, extra,
+                       ready, taints, planes, req, tolerated, require,
+                       forbid, w_least, w_most, w_bal, dim_w, bias, score,
+                       cap, ntasks, queue_alloc, queue_limit, log_nodes,
+                       log_counts, log_len, placed, job_placed, fuse_min,
+                       N, R, W, K);
 }
 
 void vamd_finalize_job(

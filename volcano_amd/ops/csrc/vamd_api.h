@@ -21,6 +21,17 @@ void vamd_select_commit(
     unsigned* sort_scratch,  // [4N] bulk-fill radix scratch (nullable)
     int N, int R, int K, hipStream_t stream);
 
+void vamd_fused_score_select(
+    const float* alloc, float* used, const float* extra,
+    const uint8_t* ready, const int64_t* taints, const int64_t* planes,
+    const float* req, int64_t tolerated, const int64_t* require,
+    const int64_t* forbid, float w_least, float w_most, float w_bal,
+    const float* dim_w, const float* bias, float* score, int* cap,
+    int ntasks, float* queue_alloc, const float* queue_limit,
+    int* log_nodes, int* log_counts, int* log_len, int* placed,
+    int* job_placed, int fuse_min, int N, int R, int W, int K,
+    hipStream_t stream);
+
 void vamd_finalize_job(
     const int* job_placed, int occupied, int min_available,
     const int* class_placed, const int* class_min, uint8_t* flag, int nc,
