@@ -115,7 +115,7 @@ class AllocateAction:
                     qi = ssn.queue_index[q.name]
                     rows = ssn.ordered_job_rows(jt, sel[qis == qi])
                     # tolist(): native ints, no per-row numpy scalar boxing
-                    ordered_jobs.extend((q, jjobs[k])
+                    ordered_jobs.extend((qi, jjobs[k])
                                         for k in rows.tolist())
         else:
             by_queue = {}
@@ -137,8 +137,9 @@ class AllocateAction:
                 gate = ssn.allocatable(q, None) if jobs_q else True
                 if not gate:
                     continue
+                qi = ssn.queue_index[q.name]
                 for job in ssn.sorted_jobs(jobs_q):
-                    ordered_jobs.append((q, job))
+                    ordered_jobs.append((qi, job))
 
         # Gang bundling: a run of consecutive jobs whose single pending
         # class has identical (queue, signature) is fused into ONE kernel
@@ -207,8 +208,7 @@ class AllocateAction:
                     out = b if out is None else out + b
             return out
 
-        for q, job in ordered_jobs:
-            qi = ssn.queue_index[q.name]
+        for qi, job in ordered_jobs:
 
             # -- SubGroupPolicy (reference types.go:218 SubGroupPolicySpec
             # + allocate.go allocateForSubJob): matching pods partition
@@ -603,7 +603,6 @@ class AllocateAction:
             nodes_sorted = sorted(ssn.nodes.values(), key=lambda n: n.name)
         ledger = getattr(ssn.cache, "ledger", None)
 
-        to_bind = []
         bind_by_job: Dict[str, List] = {}
         committed_jobs = set()
         # fire event handlers only when someone registered one — building
@@ -636,7 +635,6 @@ class AllocateAction:
                 acc_rows.append(nid)
                 acc_cnts.append(count)
                 acc_cls.append(c)
-                to_bind.extend(tasks)
                 jl.extend(tasks)
             committed_jobs.add(job.key)
 
@@ -739,8 +737,8 @@ class AllocateAction:
                         cp.tclass.request.clone().multi(float(acc_cnts[k])),
                         1, 0, 0)
 
-        if to_bind:
-            ssn.cache.bind_tasks(to_bind, by_job=bind_by_job)
+        if bind_by_job:
+            ssn.cache.bind_tasks(None, by_job=bind_by_job)
 
         # flip gang-ready podgroups to Running (job_updater analog)
         seen = set()

@@ -407,19 +407,28 @@ class SchedulerCache:
         self._used_dirty = True
 
     # -- commit pipeline ------------------------------------------------------
-    def bind_tasks(self, tasks: List[TaskInfo],
+    def bind_tasks(self, tasks: Optional[List[TaskInfo]],
                    by_job: Optional[Dict[str, List[TaskInfo]]] = None) -> None:
         """Async in the reference (cache.go:1343 AddBindTask → 20 ms drain);
         here a batched call — the binder itself may thread if it wants.
-        Callers that already have the per-job grouping pass it in."""
-        if self._watch is not None:
-            # incarnation bookkeeping only matters when store events flow
-            self._task_node.update((t.key, t.node_name) for t in tasks)
-        self.binder.bind(tasks)
+        Callers that already have the per-job grouping may pass ONLY
+        by_job (tasks=None): the flat list is not materialized."""
         if by_job is None:
             by_job = {}
             for t in tasks:
                 by_job.setdefault(t.job_key, []).append(t)
+        if tasks is None:
+            if self._watch is not None:
+                for ts in by_job.values():
+                    self._task_node.update((t.key, t.node_name)
+                                           for t in ts)
+            for ts in by_job.values():
+                self.binder.bind(ts)
+        else:
+            if self._watch is not None:
+                # incarnation bookkeeping only matters when store events flow
+                self._task_node.update((t.key, t.node_name) for t in tasks)
+            self.binder.bind(tasks)
         for key, ts in by_job.items():
             job = self.jobs.get(key)
             if job is not None:
