@@ -119,6 +119,9 @@ def main():
     ap.add_argument("--shard-mode", choices=["hard", "soft"], default="hard",
                     help="hard: nodes+jobs sharded (conflict-free); "
                          "soft: nodes replicated, delta all-reduce")
+    ap.add_argument("--extra-actions", default="",
+                    help="comma list appended to the action pipeline "
+                         "(e.g. preempt,reclaim — BASELINE config 4)")
     ap.add_argument("--extra-plugins", default="",
                     help="comma list of extra plugins for the last tier "
                          "(e.g. task-topology,numaaware — BASELINE config 5)")
@@ -164,6 +167,10 @@ def main():
         for name in args.extra_plugins.split(","):
             if name.strip():
                 config.tiers[-1].plugins.append(PluginOption(name.strip()))
+    if args.extra_actions:
+        for name in args.extra_actions.split(","):
+            if name.strip() and name.strip() not in config.actions:
+                config.actions.append(name.strip())
     config.use_hip = use_gpu
     config.device = device
     if use_gpu:
