@@ -104,3 +104,78 @@ def test_conflict_repulsion_decays():
     for _ in range(40):
         coord._note_losses(())
     assert torch.equal(coord.stagger_bias(NT), base)
+
+
+def _storm_rank(rank, world, port, q):
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    try:
+        from volcano_amd.parallel import (DistributedScheduler,
+                                          init_distributed)
+        from volcano_amd.scheduler import FakeBinder, SchedulerCache
+        from volcano_amd.store import ObjectStore
+        from volcano_amd.utils import synth
+
+        GI = 1024 ** 3
+        policy = init_distributed(backend="gloo")
+        store = ObjectStore()
+        # adversarial same-score storm: 8 identical nodes, every rank
+        # owns identical gangs that all prefer the same nodes — maximal
+        # conflict pressure on the rank-prefix admission + repulsion
+        for n in synth.make_nodes(8, cpu_milli=4000, mem=16 * GI):
+            store.create("Node", n)
+        store.create("Queue", synth.make_queue("default"))
+        for j in range(16):
+            synth.make_gang(store, f"st-{j}", replicas=2, cpu_milli=1000,
+                            mem=GI)
+        binder = FakeBinder()
+        cache = SchedulerCache(store=store, binder=binder)
+        ds = DistributedScheduler(cache, policy=policy, mode="soft")
+        # schedule-to-completion: every owned gang must land despite
+        # cross-rank collisions (capacity: 8*4 cpu = exactly 16 gangs)
+        for _ in range(world + 6):
+            ds.run_once()
+        owned = [f"default/st-{j}" for j in range(16)
+                 if policy.owns_job(f"default/st-{j}")]
+        bound_jobs = {k.rsplit("-", 2)[0].replace("default/", "default/")
+                      for k in binder.binds}
+        my_bound = sorted({k for k in binder.binds})
+        over = [n for n, ni in cache.nodes.items()
+                if ni.used.milli_cpu > ni.allocatable.milli_cpu + 0.5]
+        coord = ds.coordinator if hasattr(ds, "coordinator") else None
+        q.put((rank, len(owned),
+               sum(1 for k in my_bound), over))
+        torch.distributed.destroy_process_group()
+    except Exception:
+        import traceback
+        q.put((rank, "ERROR", traceback.format_exc(), None))
+
+
+def test_soft_shard_world8_conflict_storm():
+    """world-8 conflict storm (gloo semantics) — all
+    gangs land, no node oversubscribed, despite every rank preferring
+    the same nodes."""
+    world = 8
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_storm_rank, args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=180) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=30)
+    total_owned = 0
+    total_bound = 0
+    for rank, owned, bound, over in sorted(results):
+        assert owned != "ERROR", bound
+        assert over == [], f"rank {rank} oversubscribed nodes: {over}"
+        total_owned += owned
+        total_bound += bound
+    assert total_owned == 16
+    assert total_bound == 32      # 16 gangs x 2 pods, each bound exactly once

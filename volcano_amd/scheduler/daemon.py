@@ -104,6 +104,37 @@ def main(argv=None) -> int:
 
     signal.signal(signal.SIGUSR1, on_usr1)
 
+    # SIGUSR2: the --enable-pprof analog (reference cmd/scheduler/app/
+    # server.go:162) — profile the NEXT cycle with cProfile and write
+    # pstats + a text summary next to the cache dumps
+    def on_usr2(sig, frame):
+        import cProfile
+        import io
+        import pstats
+        stamp = int(time.time())
+        orig = sched.run_once
+
+        def profiled_once():
+            sched.run_once = orig       # one-shot
+            pr = cProfile.Profile()
+            pr.enable()
+            out = orig()
+            pr.disable()
+            base = f"{args.dump_dir}/volcano-amd-profile-{stamp}"
+            pr.dump_stats(base + ".pstats")
+            buf = io.StringIO()
+            pstats.Stats(pr, stream=buf).sort_stats(
+                "cumulative").print_stats(40)
+            with open(base + ".txt", "w") as f:
+                f.write(buf.getvalue())
+            print(f"cycle profile written to {base}.txt",
+                  file=sys.stderr, flush=True)
+            return out
+
+        sched.run_once = profiled_once
+
+    signal.signal(signal.SIGUSR2, on_usr2)
+
     if args.once:
         sched.run_once()
         store.save(args.state)
