@@ -173,7 +173,7 @@ class AllocateAction:
             if open_bundle is not None and key == open_key:
                 b = open_bundle
                 b.bundle.append(BundleEntry(job.key, tasks, len(tasks),
-                                            gang_min))
+                                            gang_min, job))
                 b.ntasks_override += len(tasks)
                 if gang_min < b.min_needed:
                     b.min_needed = gang_min
@@ -186,7 +186,8 @@ class AllocateAction:
             cp.job_key = job.key
             cp.min_needed = gang_min
             cp.ntasks_override = len(tasks)
-            cp.bundle = [BundleEntry(job.key, tasks, len(tasks), gang_min)]
+            cp.bundle = [BundleEntry(job.key, tasks, len(tasks), gang_min,
+                                     job)]
             open_bundle, open_key = cp, key
 
         _MISS = object()
@@ -254,7 +255,8 @@ class AllocateAction:
                         max(job.min_available, mtm.get(role, 0))
                     b = open_bundle
                     b.bundle.append(
-                        BundleEntry(job.key, tasks, len(tasks), gang_min))
+                        BundleEntry(job.key, tasks, len(tasks), gang_min,
+                                    job))
                     b.ntasks_override += len(tasks)
                     if gang_min < b.min_needed:
                         b.min_needed = gang_min
@@ -604,7 +606,7 @@ class AllocateAction:
         ledger = getattr(ssn.cache, "ledger", None)
 
         bind_by_job: Dict[str, List] = {}
-        committed_jobs = set()
+        committed_jobs: Dict[str, object] = {}     # key -> JobInfo
         # fire event handlers only when someone registered one — building
         # the per-piece argument lists for nobody was measurable at 10k jobs
         fire = ssn.fire_allocate if ssn.event_handlers else None
@@ -636,7 +638,7 @@ class AllocateAction:
                 acc_cnts.append(count)
                 acc_cls.append(c)
                 jl.extend(tasks)
-            committed_jobs.add(job.key)
+            committed_jobs[job.key] = job
 
         for c, cp in enumerate(plan.classes):
             if not jf[cls_job[c]]:
@@ -705,7 +707,8 @@ class AllocateAction:
                                          for nid, _, _ in pieces):
                         reverted.extend((nid, cnt) for nid, cnt, _ in pieces)
                         continue
-                    job = ssn.jobs[be.job_key]
+                    job = be.job if be.job is not None \
+                        else ssn.jobs[be.job_key]
                     commit_pieces(job, cp, c, pieces)
                     if fire is not None:
                         fire(cp.tclass, [p[0] for p in pieces],
@@ -740,28 +743,29 @@ class AllocateAction:
         if bind_by_job:
             ssn.cache.bind_tasks(None, by_job=bind_by_job)
 
-        # flip gang-ready podgroups to Running (job_updater analog)
-        seen = set()
+        # flip gang-ready podgroups to Running (job_updater analog) —
+        # walk the committed set directly (was a re-walk of every plan
+        # class/bundle entry)
         alias = getattr(plan, "job_alias", {})
         jr = ssn.job_ready_fns
         fast_ready = len(jr) == 1 and getattr(jr[0], "is_gang", False)
         running = PodGroupPhase.RUNNING.value
         store = getattr(ssn.cache, "store", None)
-        for cp in plan.classes:
-            keys = [be.job_key for be in cp.bundle] if cp.bundle else [cp.job_key]
-            for key in keys:
-                key = alias.get(key, key)
-                if key in seen or key not in committed_jobs:
+        seen = set()
+        for key, job in committed_jobs.items():
+            real = alias.get(key)
+            if real is not None:
+                if real in seen:
                     continue
-                seen.add(key)
-                job = ssn.jobs[key]
-                ready = (job.is_ready() and job.roles_ready()) if fast_ready \
-                    else ssn.job_ready(job)
-                pg = job.podgroup
-                if ready and pg is not None and pg.status.phase != running:
-                    pg.status.phase = running
-                    if store is not None:
-                        ssn.cache.update_podgroup(job)
+                seen.add(real)
+                job = ssn.jobs[real]
+            ready = (job.is_ready() and job.roles_ready()) if fast_ready \
+                else ssn.job_ready(job)
+            pg = job.podgroup
+            if ready and pg is not None and pg.status.phase != running:
+                pg.status.phase = running
+                if store is not None:
+                    ssn.cache.update_podgroup(job)
 
     @staticmethod
     def _revert_pieces(plan: CyclePlan, cp, pieces) -> None:

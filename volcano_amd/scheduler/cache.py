@@ -156,6 +156,9 @@ class SchedulerCache:
             if store else None
         self._tensors_dirty = True      # full repack (node set/labels changed)
         self._used_dirty = False        # dynamic planes only (usage changed)
+        # jobs holding PIPELINED reservations (noted by Statement.pipeline)
+        # — _demote_pipelined scans only these instead of every job
+        self._pipelined_jobs: set = set()
 
     # -- event ingestion (reference cache/event_handlers.go) -----------------
     def sync(self) -> int:
@@ -339,6 +342,9 @@ class SchedulerCache:
         self.job_table.refresh(self, self.node_tensors, ssn.queue_index)
         ssn.job_table = self.job_table
 
+    def note_pipelined(self, job_key: str) -> None:
+        self._pipelined_jobs.add(job_key)
+
     def _demote_pipelined(self) -> None:
         """Pipelined reservations live one cycle: at the next snapshot they
         re-enter Pending and compete with the capacity their evictions
@@ -346,8 +352,16 @@ class SchedulerCache:
         PodGroupOldState — session.go:77-79; the one-cycle reservation is
         the plan-design equivalent: the gang stays pipelined through the
         session in which its evictions were committed, and converges to
-        Bound the following cycle.)"""
-        for job in self.jobs.values():
+        Bound the following cycle.)  Only jobs noted by
+        Statement.pipeline are scanned (plus none at all in the common
+        no-preemption cycle)."""
+        if not self._pipelined_jobs:
+            return
+        noted, self._pipelined_jobs = self._pipelined_jobs, set()
+        for key in noted:
+            job = self.jobs.get(key)
+            if job is None:
+                continue
             bucket = job.task_status_index.get(TaskStatus.PIPELINED)
             if not bucket:
                 continue

@@ -50,6 +50,7 @@ class BundleEntry:
     tasks: list                # the job's pending TaskInfos (class order)
     ntasks: int
     min_needed: int            # gang minimum for THIS job
+    job: object = None         # JobInfo ref (skips the apply-time lookup)
 
 
 @dataclass(slots=True)

@@ -40,6 +40,9 @@ class Statement:
         self.ops.append(("evict", victim, prev, reason))
 
     def pipeline(self, task: TaskInfo, node_name: str) -> None:
+        note = getattr(self.ssn.cache, "note_pipelined", None)
+        if note is not None:
+            note(task.job_key)
         job = self.ssn.jobs.get(task.job_key)
         node = self.ssn.nodes.get(node_name)
         task.node_name = node_name
