@@ -233,3 +233,32 @@ def test_topology_spread_across_zones():
     # with maxSkew 1 the four pods alternate: 2 in each zone
     assert placed_zones.count("z1") == 2
     assert placed_zones.count("z2") == 2
+
+
+def test_rescheduling_high_node_utilization():
+    """highNodeUtilization strategy: offline pods on overloaded nodes
+    become shuffle victims (reference rescheduling strategy registry)."""
+    from volcano_amd.scheduler.config import PluginOption
+
+    store, binder, cache, sched = mk(
+        actions=["enqueue", "allocate", "shuffle"])
+    sched_cfg = sched.config
+    sched_cfg.tiers[1].plugins.append(PluginOption(
+        "rescheduling",
+        arguments={"interval": 0.0,
+                   "strategies": "highNodeUtilization",
+                   "highNodeUtilization": {"cpu": 50.0}}))
+    import volcano_amd.scheduler.plugins.rescheduling as rs
+    rs.ReschedulingPlugin._last_run = 0.0
+    for n in synth.make_nodes(1, cpu_milli=4000, mem=16 * GI):
+        store.create("Node", n)
+    store.create("Queue", synth.make_queue("default"))
+    pg = synth.make_podgroup("hot", min_member=1)
+    store.create("PodGroup", pg)
+    pod = synth.make_pod("hot-w-0", "hot", cpu_milli=3000, mem=GI)
+    pod.meta.annotations["volcano.sh/preemptable"] = "true"
+    store.create("Pod", pod)
+    sched.run_once()          # binds; node at 75% cpu > 50% threshold
+    rs.ReschedulingPlugin._last_run = 0.0
+    sched.run_once()          # shuffle evicts the offline pod
+    assert "default/hot-w-0" in binder.evictions

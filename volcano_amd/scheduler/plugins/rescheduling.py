@@ -1,8 +1,13 @@
 """Rescheduling plugin (reference ``plugins/rescheduling``): periodic
 re-balancing — strategies nominate running tasks to move; the shuffle
-action evicts them.  Strategy implemented: ``lowNodeUtilization``
-(evacuate nodes whose requested utilization sits below thresholds so
-they can be drained/binpacked)."""
+action evicts them.  Strategies (the reference registry
+rescheduling.go:45-61 is extensible, with lowNodeUtilization in-tree):
+
+* ``lowNodeUtilization`` — evacuate nodes whose requested utilization
+  sits below thresholds so they can be drained/binpacked;
+* ``highNodeUtilization`` — relieve overloaded nodes by moving their
+  offline (preemptable) pods elsewhere.
+"""
 
 from __future__ import annotations
 
@@ -24,10 +29,16 @@ class ReschedulingPlugin(Plugin):
         cpu_t = float(thresholds.get("cpu", 20.0))
         mem_t = float(thresholds.get("memory", 20.0))
 
+        high = self.args.get("highNodeUtilization", {}) or {}
+        cpu_hi = float(high.get("cpu", 85.0))
+        mem_hi = float(high.get("memory", 85.0))
+
         now = time.time()
         if now - ReschedulingPlugin._last_run < interval:
             return
-        if "lowNodeUtilization" not in str(strategy):
+        strategies = [x.strip() for x in str(strategy).split(",")]
+        if not any(st in ("lowNodeUtilization", "highNodeUtilization")
+                   for st in strategies):
             return
 
         def victim_tasks(tasks):
@@ -38,8 +49,16 @@ class ReschedulingPlugin(Plugin):
                 if alloc.get(CPU) <= 0:
                     continue
                 cpu_pct = 100.0 * ni.used.get(CPU) / max(alloc.get(CPU), 1.0)
-                mem_pct = 100.0 * ni.used.get(MEMORY) / max(alloc.get(MEMORY), 1.0)
-                if 0 < cpu_pct < cpu_t and mem_pct < mem_t:
+                mem_pct = 100.0 * ni.used.get(MEMORY) / \
+                    max(alloc.get(MEMORY), 1.0)
+                take = False
+                if "lowNodeUtilization" in strategies and \
+                        0 < cpu_pct < cpu_t and mem_pct < mem_t:
+                    take = True
+                if "highNodeUtilization" in strategies and \
+                        (cpu_pct > cpu_hi or mem_pct > mem_hi):
+                    take = True
+                if take:
                     victims.extend(
                         t for t in ni.tasks.values()
                         if t.status in (TaskStatus.RUNNING, TaskStatus.BOUND)
