@@ -601,3 +601,57 @@ def test_cycle_equivalence_topology_preempt(hip):
     assert len(ev_cpu) >= 1
     assert ev_cpu == ev_gpu
     assert binds_cpu == binds_gpu
+
+
+def test_cycle_equivalence_dra_dims(hip):
+    """DRA device-class dims (dra:<class>) ride the dense kernels: the
+    CPU oracle and HIP plane make identical placements for claim-backed
+    gangs under node device capacity + queue class quotas."""
+    from volcano_amd.api.objects import (DeviceClass, ObjectMeta, Queue,
+                                         QueueSpec, ResourceClaim)
+    from volcano_amd.api.resource import CPU, Resource
+    from volcano_amd.scheduler import FakeBinder, Scheduler, SchedulerCache, \
+        default_config
+    from volcano_amd.store import ObjectStore
+    from volcano_amd.utils import synth
+
+    GI = 1024 ** 3
+
+    def build(device, use_hip):
+        store = ObjectStore()
+        binder = FakeBinder()
+        cache = SchedulerCache(store=store, binder=binder, device=device)
+        config = default_config()
+        config.use_hip = use_hip
+        config.device = device
+        sched = Scheduler(cache, config)
+        store.create("DeviceClass", DeviceClass(
+            meta=ObjectMeta(name="mi355x.amd.com"), driver="amdgpu"))
+        for i in range(6):
+            node = synth.make_node(f"n{i}", cpu_milli=32000, mem=128 * GI)
+            node.meta.annotations["dra.volcano.sh/mi355x.amd.com"] = "8"
+            store.create("Node", node)
+        store.create("Queue", Queue(
+            meta=ObjectMeta(name="ml"),
+            spec=QueueSpec(weight=1, capability=Resource(
+                {CPU: 1e9, "count/mi355x.amd.com": 24.0}))))
+        store.create("Queue", synth.make_queue("default"))
+        for j in range(8):
+            pg = synth.make_podgroup(f"d{j}", queue="ml", min_member=2)
+            store.create("PodGroup", pg)
+            for i in range(2):
+                store.create("ResourceClaim", ResourceClaim(
+                    meta=ObjectMeta(name=f"d{j}-{i}", namespace="default"),
+                    device_class_name="mi355x.amd.com", count=2))
+                pod = synth.make_pod(f"d{j}-w-{i}", f"d{j}", queue="ml",
+                                     cpu_milli=1000, mem=GI)
+                pod.resource_claims = [f"d{j}-{i}"]
+                store.create("Pod", pod)
+        sched.run_once()
+        return dict(binder.binds)
+
+    cpu = build("cpu", False)
+    gpu = build("cuda", True)
+    # 24-device quota admits exactly 6 gangs (4 devices each)
+    assert len(cpu) == 12
+    assert cpu == gpu
