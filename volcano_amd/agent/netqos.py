@@ -11,7 +11,7 @@ throttle adjustment loop — is exercised against the same contract."""
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, List
 
 # eBPF map keys (reference utils/ebpf throttling config map)

@@ -27,7 +27,7 @@ pkg/scheduler/api/node_info.go:52-97).
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 import numpy as np
 

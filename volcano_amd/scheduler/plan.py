@@ -227,10 +227,6 @@ class CyclePlan:
         return len(self.classes)
 
 
-def _flag_of(cp: ClassPlan) -> int:
-    return 1 if cp.use_future else 0
-
-
 def run_plan_torch(plan: CyclePlan) -> CycleResult:
     """Interpret the plan over the torch oracle ops — mirrors
     ``cycle_runner.hip`` statement for statement (CPU test tier)."""

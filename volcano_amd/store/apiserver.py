@@ -104,7 +104,6 @@ def create_app(store: ObjectStore, with_admission: bool = True):
     # JSON watch events (the k8s chunked-watch shape).
     import json as _json
 
-    from fastapi import Request
     from fastapi.responses import StreamingResponse
 
     from .k8s import kind_for, to_manifest, from_manifest, api_version
