@@ -90,3 +90,85 @@ def test_random_inventory_invariants(seed):
                 t.request.get(CPU) for t in job.tasks.values()
                 if t.status.occupies_node)
         assert sum(per_queue.values()) <= total_cpu + 1.0
+
+
+def test_ledger_accounting_matches_task_truth():
+    """Property fuzz over the columnar NodeLedger: after any sequence of
+    add/remove/evict/bulk ops, every node's ledger row equals the sum of
+    its tasks' requests per status bucket (the accounting invariant the
+    round-1 evict bug violated)."""
+    import random
+
+    from volcano_amd.api.info import JobInfo, NodeInfo, TaskInfo
+    from volcano_amd.api.resource import CPU, MEMORY
+    from volcano_amd.api.types import TaskStatus
+    from volcano_amd.scheduler import FakeBinder, SchedulerCache
+    from volcano_amd.utils import synth
+
+    GI = 1024 ** 3
+    rng = random.Random(1234)
+    cache = SchedulerCache(store=None, binder=FakeBinder())
+    nodes = []
+    for i in range(6):
+        ni = NodeInfo(synth.make_node(f"n{i}", cpu_milli=64000, mem=256 * GI))
+        cache.add_node_info(ni)
+        nodes.append(ni)
+    cache.ensure_packed()          # adopt into the ledger
+
+    live = []
+    jobs = {}
+    for step in range(400):
+        op = rng.random()
+        if op < 0.45 or not live:
+            j = rng.randrange(8)
+            key = f"default/j{j}"
+            job = jobs.get(key)
+            if job is None:
+                job = jobs[key] = JobInfo(key, synth.make_podgroup(f"j{j}"))
+                cache.add_job_info(job)
+            name = f"j{j}-w-{step}"
+            pod = synth.make_pod(name, f"j{j}",
+                                 cpu_milli=rng.choice([500, 1000, 2000]),
+                                 mem=rng.choice([1, 2]) * GI)
+            t = TaskInfo.from_pod(pod, key)
+            t.status = rng.choice([TaskStatus.BOUND, TaskStatus.RUNNING,
+                                   TaskStatus.PIPELINED])
+            ni = rng.choice(nodes)
+            t.node_name = ni.name
+            job.add_task(t)
+            ni.add_task(t)
+            live.append(t)
+        elif op < 0.75:
+            t = live.pop(rng.randrange(len(live)))
+            ni = cache.nodes[t.node_name]
+            ni.remove_task(t)
+            jobs[t.job_key].remove_task(t.key)
+        else:
+            t = rng.choice(live)
+            if t.status != TaskStatus.RELEASING:
+                cache.evict_task(t, "fuzz")
+
+    for ni in nodes:
+        want_used = {}
+        want_rel = {}
+        want_pip = {}
+        for t in ni.tasks.values():
+            tgt = None
+            if t.status.occupies_node:
+                tgt = want_used
+            elif t.status == TaskStatus.RELEASING:
+                for k, v in t.request.q.items():
+                    want_used[k] = want_used.get(k, 0.0) + v
+                tgt = want_rel
+            elif t.status == TaskStatus.PIPELINED:
+                tgt = want_pip
+            if tgt is None:
+                continue
+            for k, v in t.request.q.items():
+                tgt[k] = tgt.get(k, 0.0) + v
+        for name, want in (("used", want_used), ("releasing", want_rel),
+                           ("pipelined", want_pip)):
+            got = getattr(ni, name).q
+            for k in set(want) | set(got):
+                assert abs(want.get(k, 0.0) - got.get(k, 0.0)) < 0.5, (
+                    ni.name, name, k, want.get(k), got.get(k))
