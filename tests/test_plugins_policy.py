@@ -115,15 +115,20 @@ def test_tdm_revocable_window():
         "rev", cpu_milli=4000, mem=16 * GI,
         labels={"volcano.sh/revocable-zone": "z1"}))
     store.create("Queue", synth.make_queue("default"))
-    # non-preemptable job cannot use the revocable node
-    synth.make_gang(store, "np", replicas=1, cpu_milli=1000, mem=GI)
+    # EXPLICITLY non-preemptable job cannot use the revocable node
+    # (reference GetPodPreemptable: unset defaults to preemptable=true,
+    # so plain pods ARE revocable-eligible — explicit "false" is not)
+    pg_np = synth.make_podgroup("np", min_member=1)
+    store.create("PodGroup", pg_np)
+    pod_np = synth.make_pod("np-w-0", "np", cpu_milli=1000, mem=GI)
+    pod_np.meta.annotations["volcano.sh/preemptable"] = "false"
+    store.create("Pod", pod_np)
     sched.run_once()
     assert binder.binds == {}
-    # preemptable job can (active window)
+    # default (unset ⇒ preemptable) job can (active window)
     pg = synth.make_podgroup("pp", min_member=1)
     store.create("PodGroup", pg)
     pod = synth.make_pod("pp-w-0", "pp", cpu_milli=1000, mem=GI)
-    pod.meta.annotations["volcano.sh/preemptable"] = "true"
     store.create("Pod", pod)
     sched.run_once()
     assert "default/pp-w-0" in binder.binds

@@ -1,0 +1,139 @@
+"""Reference preempt-action table cases through uthelper (reference
+actions/preempt/preempt_test.go:109-240): idle-resources short-circuit,
+pipelined short-circuit, evict-one, evict-just-enough, queue-capability
+pressure, and the explicit preemptable=false protection (#2232)."""
+
+from volcano_amd.utils import synth
+from volcano_amd.utils.uthelper import TestCommonStruct
+
+GI = 1024 ** 3
+G = 10 ** 9
+PREEMPT = ["enqueue", "allocate", "preempt", "backfill"]
+
+
+def pod(ns, name, pg_name, cpu, mem, node="", phase="Pending",
+        preemptable=None, prio=0):
+    p = synth.make_pod(name, pg_name, namespace=ns, cpu_milli=cpu, mem=mem,
+                       node_name=node, phase=phase, priority=prio)
+    if preemptable is not None:
+        p.meta.annotations["volcano.sh/preemptable"] = preemptable
+    return p
+
+
+def pg(ns, name, queue, min_member, phase="Inqueue", prio=0):
+    g = synth.make_podgroup(name, queue=queue, namespace=ns,
+                            min_member=min_member)
+    g.status.phase = phase
+    if prio:
+        g.meta.annotations["priority"] = str(prio)
+    return g
+
+
+def q(name, cap=None):
+    return synth.make_queue(name, capability=cap)
+
+
+def test_no_preempt_when_idle_resources():
+    t = TestCommonStruct(
+        name="do not preempt if there are enough idle resources",
+        podgroups=[pg("c1", "pg1", "q1", 3)],
+        pods=[pod("c1", "preemptee1", "pg1", 1000, G, "n1", "Running"),
+              pod("c1", "preemptee2", "pg1", 1000, G, "n1", "Running"),
+              pod("c1", "preemptor1", "pg1", 1000, G)],
+        nodes=[synth.make_node("n1", cpu_milli=10000, mem=10 * G, pods=10)],
+        queues=[q("q1")],
+        actions=PREEMPT,
+        expect_evicted=[],
+    ).run()
+    t.check_all()
+
+
+def test_no_preempt_when_job_pipelined():
+    t = TestCommonStruct(
+        name="do not preempt if job is pipelined",
+        podgroups=[pg("c1", "pg1", "q1", 1), pg("c1", "pg2", "q1", 1)],
+        pods=[pod("c1", "preemptee1", "pg1", 1000, G, "n1", "Running"),
+              pod("c1", "preemptee2", "pg1", 1000, G, "n1", "Running"),
+              pod("c1", "preemptee3", "pg2", 1000, G, "n1", "Running"),
+              pod("c1", "preemptor2", "pg2", 1000, G)],
+        nodes=[synth.make_node("n1", cpu_milli=3000, mem=3 * G, pods=10)],
+        queues=[q("q1")],
+        actions=PREEMPT,
+        expect_evicted=[],
+    ).run()
+    t.check_all()
+
+
+def test_preempt_one_task_of_lower_priority_job():
+    t = TestCommonStruct(
+        name="preempt one task of different job to fit both jobs",
+        podgroups=[pg("c1", "pg1", "q1", 1, prio=1),
+                   pg("c1", "pg2", "q1", 1, prio=100)],
+        pods=[pod("c1", "preemptee1", "pg1", 1000, G, "n1", "Running",
+                  preemptable="true", prio=1),
+              pod("c1", "preemptee2", "pg1", 1000, G, "n1", "Running",
+                  preemptable="false", prio=1),
+              pod("c1", "preemptor1", "pg2", 1000, G, prio=100),
+              pod("c1", "preemptor2", "pg2", 1000, G, prio=100)],
+        nodes=[synth.make_node("n1", cpu_milli=2000, mem=2 * G, pods=10)],
+        queues=[q("q1")],
+        actions=PREEMPT,
+        expect_evicted=["c1/preemptee1"],
+    ).run()
+    t.check_all()
+
+
+def test_preempt_enough_tasks_for_large_preemptor():
+    t = TestCommonStruct(
+        name="preempt enough tasks to fit large task",
+        podgroups=[pg("c1", "pg1", "q1", 1, prio=1),
+                   pg("c1", "pg2", "q1", 1, prio=100)],
+        pods=[pod("c1", "preemptee1", "pg1", 1000, G, "n1", "Running",
+                  preemptable="true", prio=1),
+              pod("c1", "preemptee2", "pg1", 1000, G, "n1", "Running",
+                  preemptable="true", prio=1),
+              pod("c1", "preemptee3", "pg1", 1000, G, "n1", "Running",
+                  preemptable="false", prio=1),
+              pod("c1", "preemptor1", "pg2", 5000, 5 * G, prio=100)],
+        nodes=[synth.make_node("n1", cpu_milli=6000, mem=6 * G, pods=10)],
+        queues=[q("q1")],
+        actions=PREEMPT,
+        expect_evicted=["c1/preemptee1", "c1/preemptee2"],
+    ).run()
+    t.check_all()
+
+
+def test_preempt_under_queue_capability_pressure():
+    """#3161: queue at capability — the low-prio job's task must go."""
+    t = TestCommonStruct(
+        name="preempt low priority job in same queue",
+        podgroups=[pg("c1", "pg1", "q1", 0, prio=1),
+                   pg("c1", "pg2", "q1", 1, prio=100)],
+        pods=[pod("c1", "preemptee1", "pg1", 3000, 3 * G, "n1", "Running",
+                  preemptable="true", prio=1),
+              pod("c1", "preemptor1", "pg2", 3000, 3 * G, prio=100)],
+        nodes=[synth.make_node("n1", cpu_milli=12000, mem=12 * G, pods=10)],
+        queues=[q("q1", cap={"cpu": 4000.0, "memory": 4.0 * G})],
+        actions=PREEMPT,
+        expect_evicted=["c1/preemptee1"],
+    ).run()
+    t.check_all()
+
+
+def test_no_preempt_when_queue_has_headroom():
+    """#3161 inverse: capability 6 — allocate places the preemptor
+    without any eviction."""
+    t = TestCommonStruct(
+        name="allocatable and has enough resource, don't preempt",
+        podgroups=[pg("c1", "pg1", "q1", 1, prio=1),
+                   pg("c1", "pg2", "q1", 1, prio=100)],
+        pods=[pod("c1", "preemptee1", "pg1", 3000, 3 * G, "n1", "Running",
+                  preemptable="true", prio=1),
+              pod("c1", "preemptor1", "pg2", 3000, 3 * G, prio=100)],
+        nodes=[synth.make_node("n1", cpu_milli=12000, mem=12 * G, pods=10)],
+        queues=[q("q1", cap={"cpu": 6000.0, "memory": 6.0 * G})],
+        actions=PREEMPT,
+        expect_evicted=[],
+        expect_bind_count=1,
+    ).run()
+    t.check_all()

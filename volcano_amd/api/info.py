@@ -71,6 +71,7 @@ class TaskInfo:
         p = self.pod
         sig = (
             self.role,
+            self.preemptable,       # tdm/rescheduling constraints differ
             tuple(sorted(self.request.q.items())),
             tuple(sorted((p.node_selector or {}).items())) if p else (),
             tuple((t.key, t.operator, t.value, t.effect)
@@ -127,7 +128,9 @@ class TaskInfo:
             node_name=pod.node_name,
             priority=pod.priority,
             best_effort=pod.best_effort,
-            preemptable=pod.meta.annotations.get(ANN_PREEMPTABLE, "") == "true",
+            # reference GetPodPreemptable (api/pod_info.go:176): explicit
+            # annotation wins, label fallback, DEFAULT TRUE
+            preemptable=_pod_preemptable(pod),
             gated=bool(pod.scheduling_gates),
             pod=pod,
         )
@@ -152,6 +155,13 @@ class TaskClass:
     @property
     def count(self) -> int:
         return len(self.tasks)
+
+
+def _pod_preemptable(pod) -> bool:
+    val = pod.meta.annotations.get(ANN_PREEMPTABLE)
+    if val is None:
+        val = pod.meta.labels.get(ANN_PREEMPTABLE)
+    return True if val is None else val == "true"
 
 
 _OCCUPIED_STATUSES = tuple(ALLOCATED_STATUSES) + (TaskStatus.SUCCEEDED,)

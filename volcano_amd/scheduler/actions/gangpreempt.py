@@ -65,9 +65,7 @@ class GangPreemptAction:
             if ann is not None and ann != "true":
                 return False
         for v in victims:
-            p = v.pod
-            if p is not None and (p.meta.annotations.get(
-                    "volcano.sh/preemptable") == "false"):
+            if not v.preemptable:   # explicit volcano.sh/preemptable=false
                 return False
         fns = ssn.preemptable_fns if self.same_queue else ssn.reclaimable_fns
         allowed = victims

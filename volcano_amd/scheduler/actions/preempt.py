@@ -176,6 +176,15 @@ class PreemptAction:
         if still_needed <= 0:
             return
 
+        # queue-quota headroom (reference #3161 cases: a queue AT its
+        # capability must evict its own lower-priority work to admit the
+        # preemptor — node-level fit alone is not admission)
+        headroom = None
+        qi = ssn.queue_index.get(job.queue)
+        if same_queue and ssn.queue_limit is not None and qi is not None:
+            headroom = (ssn.queue_limit[qi] - ssn.queue_alloc[qi]) \
+                .numpy().astype(np.float64).copy()
+
         placed = 0
         for tc in job.pending_classes():
             if placed >= still_needed:
@@ -215,15 +224,44 @@ class PreemptAction:
             for task in tc.tasks:
                 if placed >= still_needed:
                     break
+                if headroom is not None and \
+                        (req_vec > headroom + 0.1).any():
+                    # over queue quota: evict same-queue victims to free
+                    # quota (they also free their nodes)
+                    if not self._evict_for_quota(ssn, stmt, nt, req_vec,
+                                                 headroom, victims_by_node):
+                        break
                 node = self._preempt_one(ssn, stmt, job, task, req_vec,
                                          order, victims_by_node)
                 if node is not None:
                     placed += 1
+                    if headroom is not None:
+                        headroom -= req_vec
 
         if ssn.job_pipelined(job) and job.is_pipelined():
             stmt.commit()
         else:
             stmt.discard()
+
+    @staticmethod
+    def _evict_for_quota(ssn, stmt, nt, req_vec, headroom,
+                         victims_by_node) -> bool:
+        """Evict lowest-value same-queue victims until the queue headroom
+        covers one preemptor task (in-place headroom update).  Returns
+        False when even evicting every candidate cannot free the
+        quota."""
+        flat = sorted((v for vs in victims_by_node.values() for v in vs),
+                      key=lambda v: victim_sort_key(ssn, v))
+        from ...api.types import TaskStatus
+        for v in flat:
+            if not (req_vec > headroom + 0.1).any():
+                return True
+            if v.status == TaskStatus.RELEASING:
+                continue
+            stmt.evict(v)
+            headroom += nt.req_vector(v) if nt.req_vector(v) is not None \
+                else 0.0
+        return not (req_vec > headroom + 0.1).any()
 
     def _candidate_scan(self, ssn, job, tc, req_vec, constraints,
                         same_queue, victim_filter, node_filter=None):
@@ -247,6 +285,8 @@ class PreemptAction:
             for t in ni.tasks.values():
                 if t.status not in self.victim_statuses:
                     continue
+                if not t.preemptable:
+                    continue   # explicit volcano.sh/preemptable=false
                 vjob = ssn.jobs.get(t.job_key)
                 if vjob is None:
                     continue
