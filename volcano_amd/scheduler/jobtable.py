@@ -55,6 +55,13 @@ class JobTable:
         self.ctime: Optional[np.ndarray] = None     # [J] f64
         self.qi: Optional[np.ndarray] = None        # [J] i64 (-1 unknown)
         self.keys: Optional[np.ndarray] = None      # [J] object (job keys)
+        # interned plan-atom signature id (-1: multi-class job).  Used as
+        # a tie-break BELOW every plugin order key: jobs the plugins left
+        # unordered sort by signature, so identical gangs run back to
+        # back and the allocate bundler fuses them (the reference leaves
+        # tie order unspecified — sort.Slice is unstable).
+        self.sigid: Optional[np.ndarray] = None     # [J] i64
+        self._sig_ids: Dict = {}
         # dynamic columns (refreshed every cycle)
         self.phase: Optional[np.ndarray] = None     # [J] i8
         self.occ: Optional[np.ndarray] = None       # [J] i64
@@ -95,7 +102,8 @@ class JobTable:
         self.minres = np.zeros((J, R), dtype=np.float64)
         self.totreq = np.zeros((J, R), dtype=np.float64)
         minav, ntasks, prio, ctime, vers = [], [], [], [], []
-        qnames, keys = [], []
+        qnames, keys, sigids = [], [], []
+        sid = self._sig_ids
         for k, job in enumerate(jobs):
             job._jrow = k
             self.minres[k] = job.minres_vec(nt)
@@ -107,6 +115,9 @@ class JobTable:
             vers.append(job._tver)
             qnames.append(job.queue)
             keys.append(job.key)
+            atom = job.plan_atom()
+            sigids.append(sid.setdefault(atom[0], len(sid)) if atom else -1)
+        self.sigid = np.array(sigids, dtype=np.int64)
         self.minav = np.array(minav, dtype=np.int64)
         self.ntasks = np.array(ntasks, dtype=np.int64)
         self.prio = np.array(prio, dtype=np.int64)
@@ -128,6 +139,9 @@ class JobTable:
             qidx = self._queue_index
             if qidx is not None:
                 self.qi[k] = qidx.get(job.queue, -1)
+            atom = job.plan_atom()
+            self.sigid[k] = self._sig_ids.setdefault(
+                atom[0], len(self._sig_ids)) if atom else -1
             self._vers[k] = job._tver
 
     def _refresh_dynamic(self) -> None:

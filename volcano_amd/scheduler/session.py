@@ -271,8 +271,13 @@ class Session:
                 arrs.append(np.asarray(c(jt, rows)))
             else:
                 arrs.append(np.array([k(jt.jobs[int(i)]) for i in rows]))
-        order = np.lexsort(tuple([jt.keys[rows], jt.ctime[rows]]
-                                 + arrs[::-1]))
+        # sigid sits BELOW every plugin key and ABOVE ctime: jobs the
+        # plugins left tied group by plan-atom signature, so identical
+        # gangs run consecutively and the allocate bundler fuses them
+        # into single kernel passes (heterogeneous-mix shape).  Legal:
+        # the reference's sort.Slice leaves tie order unspecified.
+        order = np.lexsort(tuple([jt.keys[rows], jt.ctime[rows],
+                                  jt.sigid[rows]] + arrs[::-1]))
         return rows[order]
 
     # -- queue tensor rows ----------------------------------------------------
