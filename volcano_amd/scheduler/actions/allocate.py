@@ -81,6 +81,10 @@ class AllocateAction:
         if jt is not None and len(jt.jobs) != len(ssn.jobs):
             jt = None
         ordered_jobs = []
+        # run metadata (parallel to ordered_jobs, columnar path only):
+        # maximal runs of consecutive atomic same-signature jobs emit as
+        # ONE fused bundle without per-job plan_atom/signature work
+        elig_l = sig_l = gm_l = qi_l = None
         if jt is not None:
             from ..jobtable import PH_INQUEUE, PH_RUNNING
             mask = ((jt.phase == PH_INQUEUE) | (jt.phase == PH_RUNNING)) \
@@ -104,6 +108,9 @@ class AllocateAction:
                 queues = [ssn.queues[qname_by_qi[int(i)]] for i in present
                           if qname_by_qi.get(int(i)) in ssn.queues]
                 jjobs = jt.jobs
+                track_runs = not ssn.class_bias_fns
+                if track_runs:
+                    elig_l, sig_l, gm_l, qi_l = [], [], [], []
                 for q in ssn.sorted_queues(queues):
                     if not q.is_open or ssn.queue_overused(q):
                         continue
@@ -117,6 +124,13 @@ class AllocateAction:
                     # tolist(): native ints, no per-row numpy scalar boxing
                     ordered_jobs.extend((qi, jjobs[k])
                                         for k in rows.tolist())
+                    if track_runs:
+                        elig_l.extend(((jt.sigid[rows] >= 0)
+                                       & (jt.npend[rows] == jt.ntasks[rows])
+                                       & ~jt.subpol[rows]).tolist())
+                        sig_l.extend(jt.sigid[rows].tolist())
+                        gm_l.extend(jt.gangmin[rows].tolist())
+                        qi_l.extend([qi] * len(rows))
         else:
             by_queue = {}
             for job in ssn.jobs.values():
@@ -209,7 +223,100 @@ class AllocateAction:
                     out = b if out is None else out + b
             return out
 
-        for qi, job in ordered_jobs:
+        # nominated gangs take the host fast path — exclude them from runs
+        if elig_l is not None and self.coordinator is None and \
+                getattr(ssn.cache, "had_nominations", False):
+            for ix, (_, jb) in enumerate(ordered_jobs):
+                if jb._nom:
+                    elig_l[ix] = False
+
+        L = len(ordered_jobs)
+        skip_until = -1
+        for i, (qi, job) in enumerate(ordered_jobs):
+            if i < skip_until:
+                continue
+
+            # -- fused run: consecutive atomic jobs with one interned
+            # signature in one queue emit as a single bundled ClassPlan —
+            # no per-job signature/constraint work (the dominant shape
+            # after signature-grouped ordering)
+            if elig_l is not None and elig_l[i]:
+                s = sig_l[i]
+                j = i + 1
+                while j < L and elig_l[j] and sig_l[j] == s \
+                        and qi_l[j] == qi:
+                    j += 1
+                if j > i + 1:
+                    skip_until = j
+                    atom = job.plan_atom()
+                    sig, role, request, priority = atom
+                    ckey = (sig, qi)
+                    got = cons_memo.get(ckey, _MISS)
+                    if got is _MISS:
+                        req = nt.req_vector(
+                            next(iter(job.task_status_index.get(
+                                _TS.PENDING).values())))
+                        if req is None:
+                            got = None
+                        else:
+                            tc0 = TaskClass(
+                                signature=sig, role=role, request=request,
+                                tasks=[next(iter(job.task_status_index.get(
+                                    _TS.PENDING).values()))],
+                                priority=priority)
+                            if predicates is not None:
+                                tol, require, forbid = \
+                                    predicates.class_constraints(tc0, job)
+                            else:
+                                tol = -1
+                                W = max(nt.labels.words, 1)
+                                require = np.zeros(W, dtype=np.int64)
+                                forbid = np.zeros(W, dtype=np.int64)
+                            got = (req, tol, require, forbid)
+                        cons_memo[ckey] = got
+                    if got is None:
+                        continue  # asks for a resource no node offers
+                    req, tol, require, forbid = got
+                    entries: List[BundleEntry] = []
+                    total = 0
+                    mn = 1 << 60
+                    for k in range(i, j):
+                        jb = ordered_jobs[k][1]
+                        pend = jb.task_status_index.get(_TS.PENDING)
+                        if not pend:
+                            continue
+                        if next(iter(pend.values())).gated:
+                            continue
+                        tasks = list(pend.values())
+                        gm = gm_l[k]
+                        entries.append(BundleEntry(jb.key, tasks,
+                                                   len(tasks), gm, jb))
+                        total += len(tasks)
+                        if gm < mn:
+                            mn = gm
+                    if not entries:
+                        continue
+                    close_bundle()
+                    e0 = entries[0]
+                    tc = TaskClass(signature=sig, role=role,
+                                   request=request, tasks=e0.tasks,
+                                   priority=priority)
+                    cp = ClassPlan(tclass=tc, job_key=e0.job_key,
+                                   queue_idx=qi, req=req, tolerated=tol,
+                                   require=require, forbid=forbid,
+                                   min_needed=mn, w_least=w_least,
+                                   w_most=w_most, w_bal=w_bal)
+                    cp.ntasks_override = total
+                    cp.bundle = entries
+                    plan.jobs.append(JobPlan(
+                        job_key=e0.job_key,
+                        class_begin=len(plan.classes),
+                        class_end=len(plan.classes) + 1,
+                        occupied=0, min_available=mn))
+                    plan.classes.append(cp)
+                    continue
+                # singleton run: fall through to the per-job path (it
+                # may open/continue a bundle with neighbours)
 
             # -- SubGroupPolicy (reference types.go:218 SubGroupPolicySpec
             # + allocate.go allocateForSubJob): matching pods partition

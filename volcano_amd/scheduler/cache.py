@@ -159,6 +159,7 @@ class SchedulerCache:
         # jobs holding PIPELINED reservations (noted by Statement.pipeline)
         # — _demote_pipelined scans only these instead of every job
         self._pipelined_jobs: set = set()
+        self.had_nominations = False
 
     # -- event ingestion (reference cache/event_handlers.go) -----------------
     def sync(self) -> int:
@@ -325,6 +326,7 @@ class SchedulerCache:
     # -- snapshot -------------------------------------------------------------
     def snapshot_into(self, ssn) -> None:
         self.sync()
+        self.had_nominations = False
         self._demote_pipelined()
         if "default" not in self.queues:
             self.queues["default"] = QueueInfo(
@@ -375,6 +377,7 @@ class SchedulerCache:
                 job.update_task_status(t, TaskStatus.PENDING)
             if pipelined:
                 job._nom = True       # allocate checks this before the
+                self.had_nominations = True
                 self._used_dirty = True
 
     def ensure_packed(self) -> None:

@@ -62,6 +62,11 @@ class JobTable:
         # tie order unspecified — sort.Slice is unstable).
         self.sigid: Optional[np.ndarray] = None     # [J] i64
         self._sig_ids: Dict = {}
+        # effective gang minimum for atomic jobs: max(minAvailable,
+        # minTaskMember[role]) — what the allocate fast path computes
+        # per job; here once per static refresh
+        self.gangmin: Optional[np.ndarray] = None   # [J] i64
+        self.subpol: Optional[np.ndarray] = None    # [J] bool
         # dynamic columns (refreshed every cycle)
         self.phase: Optional[np.ndarray] = None     # [J] i8
         self.occ: Optional[np.ndarray] = None       # [J] i64
@@ -102,13 +107,14 @@ class JobTable:
         self.minres = np.zeros((J, R), dtype=np.float64)
         self.totreq = np.zeros((J, R), dtype=np.float64)
         minav, ntasks, prio, ctime, vers = [], [], [], [], []
-        qnames, keys, sigids = [], [], []
+        qnames, keys, sigids, gangmin, subpol = [], [], [], [], []
         sid = self._sig_ids
         for k, job in enumerate(jobs):
             job._jrow = k
             self.minres[k] = job.minres_vec(nt)
             self.totreq[k] = job.total_req_vec(nt)
-            minav.append(job.min_available)
+            ma = job.min_available
+            minav.append(ma)
             ntasks.append(len(job.tasks))
             prio.append(job.priority)
             ctime.append(job.creation_timestamp)
@@ -116,8 +122,18 @@ class JobTable:
             qnames.append(job.queue)
             keys.append(job.key)
             atom = job.plan_atom()
-            sigids.append(sid.setdefault(atom[0], len(sid)) if atom else -1)
+            if atom:
+                sigids.append(sid.setdefault(atom[0], len(sid)))
+                mtm = job.min_task_member
+                gangmin.append(max(ma, mtm.get(atom[1], 0)) if mtm else ma)
+            else:
+                sigids.append(-1)
+                gangmin.append(ma)
+            pg = job.podgroup
+            subpol.append(pg is not None and bool(pg.spec.sub_group_policy))
         self.sigid = np.array(sigids, dtype=np.int64)
+        self.gangmin = np.array(gangmin, dtype=np.int64)
+        self.subpol = np.array(subpol, dtype=bool)
         self.minav = np.array(minav, dtype=np.int64)
         self.ntasks = np.array(ntasks, dtype=np.int64)
         self.prio = np.array(prio, dtype=np.int64)
@@ -140,8 +156,19 @@ class JobTable:
             if qidx is not None:
                 self.qi[k] = qidx.get(job.queue, -1)
             atom = job.plan_atom()
-            self.sigid[k] = self._sig_ids.setdefault(
-                atom[0], len(self._sig_ids)) if atom else -1
+            if atom:
+                self.sigid[k] = self._sig_ids.setdefault(
+                    atom[0], len(self._sig_ids))
+                mtm = job.min_task_member
+                self.gangmin[k] = max(job.min_available,
+                                      mtm.get(atom[1], 0)) if mtm \
+                    else job.min_available
+            else:
+                self.sigid[k] = -1
+                self.gangmin[k] = job.min_available
+            pg = job.podgroup
+            self.subpol[k] = pg is not None and \
+                bool(pg.spec.sub_group_policy)
             self._vers[k] = job._tver
 
     def _refresh_dynamic(self) -> None:
