@@ -80,11 +80,14 @@ class AllocateAction:
         jt = getattr(ssn, "job_table", None)
         if jt is not None and len(jt.jobs) != len(ssn.jobs):
             jt = None
-        ordered_jobs = []
-        # run metadata (parallel to ordered_jobs, columnar path only):
-        # maximal runs of consecutive atomic same-signature jobs emit as
-        # ONE fused bundle without per-job plan_atom/signature work
-        elig_l = sig_l = gm_l = qi_l = None
+        # parallel worksheets: jobs_l[i] is the i-th job in plan order,
+        # qi_l[i] its queue index (no per-job tuple boxing)
+        jobs_l: List = []
+        qi_l: List[int] = []
+        # run metadata (parallel to jobs_l, columnar path only): maximal
+        # runs of consecutive atomic same-signature jobs emit as ONE
+        # fused bundle without per-job plan_atom/signature work
+        elig_l = sig_l = gm_l = None
         if jt is not None:
             from ..jobtable import PH_INQUEUE, PH_RUNNING
             mask = ((jt.phase == PH_INQUEUE) | (jt.phase == PH_RUNNING)) \
@@ -110,7 +113,7 @@ class AllocateAction:
                 jjobs = jt.jobs
                 track_runs = not ssn.class_bias_fns
                 if track_runs:
-                    elig_l, sig_l, gm_l, qi_l = [], [], [], []
+                    elig_l, sig_l, gm_l = [], [], []
                 for q in ssn.sorted_queues(queues):
                     if not q.is_open or ssn.queue_overused(q):
                         continue
@@ -122,15 +125,15 @@ class AllocateAction:
                     qi = ssn.queue_index[q.name]
                     rows = ssn.ordered_job_rows(jt, sel[qis == qi])
                     # tolist(): native ints, no per-row numpy scalar boxing
-                    ordered_jobs.extend((qi, jjobs[k])
-                                        for k in rows.tolist())
+                    rl = rows.tolist()
+                    jobs_l.extend(jjobs[k] for k in rl)
+                    qi_l.extend([qi] * len(rl))
                     if track_runs:
                         elig_l.extend(((jt.sigid[rows] >= 0)
                                        & (jt.npend[rows] == jt.ntasks[rows])
                                        & ~jt.subpol[rows]).tolist())
                         sig_l.extend(jt.sigid[rows].tolist())
                         gm_l.extend(jt.gangmin[rows].tolist())
-                        qi_l.extend([qi] * len(rows))
         else:
             by_queue = {}
             for job in ssn.jobs.values():
@@ -152,8 +155,9 @@ class AllocateAction:
                 if not gate:
                     continue
                 qi = ssn.queue_index[q.name]
-                for job in ssn.sorted_jobs(jobs_q):
-                    ordered_jobs.append((qi, job))
+                sj = ssn.sorted_jobs(jobs_q)
+                jobs_l.extend(sj)
+                qi_l.extend([qi] * len(sj))
 
         # Gang bundling: a run of consecutive jobs whose single pending
         # class has identical (queue, signature) is fused into ONE kernel
@@ -226,15 +230,17 @@ class AllocateAction:
         # nominated gangs take the host fast path — exclude them from runs
         if elig_l is not None and self.coordinator is None and \
                 getattr(ssn.cache, "had_nominations", False):
-            for ix, (_, jb) in enumerate(ordered_jobs):
+            for ix, jb in enumerate(jobs_l):
                 if jb._nom:
                     elig_l[ix] = False
 
-        L = len(ordered_jobs)
+        L = len(jobs_l)
         skip_until = -1
-        for i, (qi, job) in enumerate(ordered_jobs):
+        for i in range(L):
             if i < skip_until:
                 continue
+            job = jobs_l[i]
+            qi = qi_l[i]
 
             # -- fused run: consecutive atomic jobs with one interned
             # signature in one queue emit as a single bundled ClassPlan —
@@ -281,7 +287,7 @@ class AllocateAction:
                     total = 0
                     mn = 1 << 60
                     for k in range(i, j):
-                        jb = ordered_jobs[k][1]
+                        jb = jobs_l[k]
                         pend = jb.task_status_index.get(_TS.PENDING)
                         if not pend:
                             continue
