@@ -326,6 +326,30 @@ class JobInfo:
             dst[t.key] = t
         self._alloc_vec = None
 
+    def finish_bind(self, tasks: List[TaskInfo]) -> None:
+        """Index repair after a caller-fused bind: the apply walk already
+        set ``t.status = BOUND`` on formerly-PENDING tasks in the same
+        pass that set ``node_name`` (one task walk instead of two).  Only
+        the bucket move and occupancy remain — wholesale when the batch
+        is the whole pending bucket (the gang case)."""
+        src = self.task_status_index.get(TaskStatus.PENDING)
+        n = len(tasks)
+        if src is not None and len(src) == n:
+            dst = self.task_status_index.get(TaskStatus.BOUND)
+            if dst:
+                dst.update(src)
+            else:
+                self.task_status_index[TaskStatus.BOUND] = src
+            self.task_status_index[TaskStatus.PENDING] = {}
+        else:
+            dst = self.task_status_index.setdefault(TaskStatus.BOUND, {})
+            for t in tasks:
+                if src is not None:
+                    src.pop(t.key, None)
+                dst[t.key] = t
+        self._occ += n          # BOUND occupies, PENDING does not
+        self._alloc_vec = None
+
     # -- dense vector caches (hot: plugin tensor builds) ---------------------
     def alloc_vec(self, nt):
         """Allocated-resource vector over nt.dims (cached)."""

@@ -27,6 +27,7 @@ from ...api.info import TaskClass
 from ...api.types import PodGroupPhase, TaskStatus
 
 PENDING_S = TaskStatus.PENDING
+BOUND_S = TaskStatus.BOUND
 from ...utils.metrics import METRICS
 from ..plan import (BundleEntry, ClassPlan, CyclePlan, JobPlan, run_plan_hip,
                     run_plan_torch)
@@ -746,6 +747,7 @@ class AllocateAction:
                 name = ni.name
                 for t in tasks:
                     t.node_name = name
+                    t.status = BOUND_S    # index repaired by finish_bind
                 ni._batches.append(tasks)
                 acc_rows.append(nid)
                 acc_cnts.append(count)
@@ -854,7 +856,7 @@ class AllocateAction:
                         1, 0, 0)
 
         if bind_by_job:
-            ssn.cache.bind_tasks(None, by_job=bind_by_job)
+            ssn.cache.bind_tasks(None, by_job=bind_by_job, preset=True)
 
         # flip gang-ready podgroups to Running (job_updater analog) —
         # walk the committed set directly (was a re-walk of every plan

@@ -425,7 +425,8 @@ class SchedulerCache:
 
     # -- commit pipeline ------------------------------------------------------
     def bind_tasks(self, tasks: Optional[List[TaskInfo]],
-                   by_job: Optional[Dict[str, List[TaskInfo]]] = None) -> None:
+                   by_job: Optional[Dict[str, List[TaskInfo]]] = None,
+                   preset: bool = False) -> None:
         """Async in the reference (cache.go:1343 AddBindTask → 20 ms drain);
         here a batched call — the binder itself may thread if it wants.
         Callers that already have the per-job grouping may pass ONLY
@@ -446,10 +447,16 @@ class SchedulerCache:
                 # incarnation bookkeeping only matters when store events flow
                 self._task_node.update((t.key, t.node_name) for t in tasks)
             self.binder.bind(tasks)
+        # ``preset``: the caller's task walk already wrote BOUND statuses
+        # (allocate._apply fuses it with the node_name pass); only the
+        # bucket move remains
         for key, ts in by_job.items():
             job = self.jobs.get(key)
             if job is not None:
-                job.move_tasks_status(ts, TaskStatus.BOUND)
+                if preset:
+                    job.finish_bind(ts)
+                else:
+                    job.move_tasks_status(ts, TaskStatus.BOUND)
 
     def evict_task(self, task: TaskInfo, reason: str = "") -> None:
         self.binder.evict(task, reason)

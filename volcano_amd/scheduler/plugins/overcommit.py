@@ -24,9 +24,17 @@ class OvercommitPlugin(Plugin):
         # Inqueue-phase podgroups; Running jobs hold real allocations and
         # no longer occupy admission headroom)
         inqueue = np.zeros(nt.r, dtype=np.float64)
-        for job in ssn.jobs.values():
-            if job.phase == PodGroupPhase.INQUEUE.value:
-                inqueue += job.minres_vec(nt)
+        jt = getattr(ssn, "job_table", None)
+        if jt is not None and len(jt.jobs) == len(ssn.jobs) \
+                and jt.minres.shape[1] == nt.r:
+            from ..jobtable import PH_INQUEUE
+            sel = jt.phase == PH_INQUEUE
+            if sel.any():
+                inqueue += jt.minres[sel].sum(axis=0)
+        else:
+            for job in ssn.jobs.values():
+                if job.phase == PodGroupPhase.INQUEUE.value:
+                    inqueue += job.minres_vec(nt)
 
         def job_enqueueable(job) -> int:
             head = inqueue[mask] + job.minres_vec(nt)[mask]
