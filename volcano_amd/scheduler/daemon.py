@@ -170,6 +170,12 @@ def main(argv=None) -> int:
 
         sched.run_once = run_once_with_reload
     try:
+        # long-lived inventory (node/pod mirrors, tensors) out of the
+        # cyclic collector's gen-2 scan set: avoids multi-hundred-ms GC
+        # stalls under sustained churn (young garbage still collects)
+        import gc
+        gc.collect()
+        gc.freeze()
         sched.run(period=args.period)
     except KeyboardInterrupt:
         pass
