@@ -69,22 +69,40 @@ class TaskInfo:
         if sig is not None:
             return sig
         p = self.pod
-        sig = (
-            self.role,
-            self.preemptable,       # tdm/rescheduling constraints differ
-            tuple(sorted(self.request.q.items())),
-            tuple(sorted((p.node_selector or {}).items())) if p else (),
-            tuple((t.key, t.operator, t.value, t.effect)
-                  for t in (p.tolerations if p else [])),
-            repr(p.affinity) if p is not None and p.affinity else "",
-            self.priority,
-            tuple(sorted(p.scheduling_gates)) if p else (),
-            tuple(sorted(k for k in (p.meta.annotations if p else {})
-                         if k.startswith("volcano.sh/gpu")
-                         or k.startswith("volcano.sh/vgpu")
-                         or k == "volcano.sh/numa-topology-policy")),
-            tuple(sorted(p.volumes)) if p is not None and p.volumes else (),
-        )
+        # empty-guard every optional facet: the common pod (no selector,
+        # tolerations, gates, device annotations or volumes) skips all
+        # eight sorted()/genexpr constructions — the signature build was
+        # the dominant arrival-path cost under sustained churn
+        if p is not None:
+            sel = p.node_selector
+            tol = p.tolerations
+            gates = p.scheduling_gates
+            anns = p.meta.annotations
+            vols = p.volumes
+            sig = (
+                self.role,
+                self.preemptable,   # tdm/rescheduling constraints differ
+                tuple(sorted(self.request.q.items())),
+                tuple(sorted(sel.items())) if sel else (),
+                tuple((t.key, t.operator, t.value, t.effect)
+                      for t in tol) if tol else (),
+                repr(p.affinity) if p.affinity else "",
+                self.priority,
+                tuple(sorted(gates)) if gates else (),
+                tuple(sorted(k for k in anns
+                             if k.startswith("volcano.sh/gpu")
+                             or k.startswith("volcano.sh/vgpu")
+                             or k == "volcano.sh/numa-topology-policy"))
+                if anns else (),
+                tuple(sorted(vols)) if vols else (),
+            )
+        else:
+            sig = (
+                self.role,
+                self.preemptable,
+                tuple(sorted(self.request.q.items())),
+                (), (), "", self.priority, (), (), (),
+            )
         # intern: equal signatures share ONE tuple object, so downstream
         # equality checks (bundle-continuation keys, memo dicts) hit the
         # identity fast path instead of deep tuple compares — measurable
