@@ -397,6 +397,22 @@ class JobInfo:
         self._minres_vec_c = (nt.r, vec)
         return vec
 
+    def minres_unrepresented(self, nt) -> bool:
+        """True if minResources demands a resource NO dim represents
+        (nothing in the cluster offers or accounts it).  Dense vectors
+        silently drop unknown dims, so admission gates must reject these
+        explicitly — the reference computes realCapability 0 for an
+        unknown resource and refuses the enqueue (proportion.go
+        jobEnqueueableFn minReq ≤ realCapability)."""
+        pg = self.podgroup
+        if pg is None or not pg.spec.min_resources.q:
+            return False
+        didx = nt.dims.index
+        for k, v in pg.spec.min_resources.q.items():
+            if v > 0 and k not in didx:
+                return True
+        return False
+
     def tasks_with_status(self, *statuses: TaskStatus) -> List[TaskInfo]:
         out: List[TaskInfo] = []
         for s in statuses:

@@ -41,6 +41,10 @@ class EnqueueAction:
         import numpy as np
         from ..jobtable import PH_PENDING
         sel = np.nonzero(jt.phase == PH_PENDING)[0]
+        if len(sel) and ssn.job_enqueueable_fns:
+            # minResources naming a resource no dim represents can never
+            # fit realCapability (reference proportion jobEnqueueableFn)
+            sel = sel[~jt.minres_bad[sel]]
         if not len(sel):
             return
         qis = jt.qi[sel]
@@ -58,6 +62,11 @@ class EnqueueAction:
     def _admit_queue(self, ssn, q, jobs_q, rows=None, jt=None) -> None:
         if jobs_q is None:
             jobs_q = [jt.jobs[int(k)] for k in rows]
+        elif ssn.job_enqueueable_fns and ssn.node_tensors is not None:
+            # non-columnar path: same unrepresented-resource admission
+            # veto the columnar path applies via jt.minres_bad
+            jobs_q = [j for j in jobs_q
+                      if not j.minres_unrepresented(ssn.node_tensors)]
         if not jobs_q:
             return
         fifo = q.queue.spec.dequeue_strategy == "fifo"

@@ -67,6 +67,9 @@ class JobTable:
         # per job; here once per static refresh
         self.gangmin: Optional[np.ndarray] = None   # [J] i64
         self.subpol: Optional[np.ndarray] = None    # [J] bool
+        # minResources names a resource no dim represents → admission
+        # gates must reject (dense projection would silently drop it)
+        self.minres_bad: Optional[np.ndarray] = None  # [J] bool
         # dynamic columns (refreshed every cycle)
         self.phase: Optional[np.ndarray] = None     # [J] i8
         self.occ: Optional[np.ndarray] = None       # [J] i64
@@ -108,6 +111,7 @@ class JobTable:
         self.totreq = np.zeros((J, R), dtype=np.float64)
         minav, ntasks, prio, ctime, vers = [], [], [], [], []
         qnames, keys, sigids, gangmin, subpol = [], [], [], [], []
+        mrbad = []
         sid = self._sig_ids
         for k, job in enumerate(jobs):
             job._jrow = k
@@ -131,6 +135,8 @@ class JobTable:
                 gangmin.append(ma)
             pg = job.podgroup
             subpol.append(pg is not None and bool(pg.spec.sub_group_policy))
+            mrbad.append(job.minres_unrepresented(nt))
+        self.minres_bad = np.array(mrbad, dtype=bool)
         self.sigid = np.array(sigids, dtype=np.int64)
         self.gangmin = np.array(gangmin, dtype=np.int64)
         self.subpol = np.array(subpol, dtype=bool)
@@ -169,6 +175,7 @@ class JobTable:
             pg = job.podgroup
             self.subpol[k] = pg is not None and \
                 bool(pg.spec.sub_group_policy)
+            self.minres_bad[k] = job.minres_unrepresented(nt)
             self._vers[k] = job._tver
 
     def _refresh_dynamic(self) -> None:

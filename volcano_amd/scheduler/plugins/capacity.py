@@ -174,8 +174,11 @@ class CapacityPlugin(Plugin):
             qi = qi_of.get(job.queue)
             if qi is None:
                 return REJECT
+            minres = job.minres_vec(nt)
+            if (minres[~mask] > 0.1).any():
+                return REJECT     # demand on a zero-capacity dim
             head = (alloc_np[qi][mask].astype(np.float64)
-                    + inqueue_np[qi][mask] + job.minres_vec(nt)[mask])
+                    + inqueue_np[qi][mask] + minres[mask])
             des = des_np[qi][mask].astype(np.float64)
             return PERMIT if bool((head <= des + 0.1 + 1e-6 * des).all()) \
                 else REJECT
@@ -212,6 +215,8 @@ class CapacityPlugin(Plugin):
                 demand = np.zeros(R, dtype=np.float64)
                 for j in jobs:
                     demand += j.minres_vec(nt)
+            if (demand[~mask] > 0.1).any():
+                return None       # demand on a zero-capacity dim
             head = (alloc_np[qi][mask].astype(np.float64)
                     + inqueue_np[qi][mask] + demand[mask])
             des = des_np[qi][mask].astype(np.float64)

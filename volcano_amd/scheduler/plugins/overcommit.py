@@ -37,7 +37,10 @@ class OvercommitPlugin(Plugin):
                     inqueue += job.minres_vec(nt)
 
         def job_enqueueable(job) -> int:
-            head = inqueue[mask] + job.minres_vec(nt)[mask]
+            minres = job.minres_vec(nt)
+            if (minres[~mask] > 0.1).any():
+                return REJECT     # demand on a zero-capacity dim
+            head = inqueue[mask] + minres[mask]
             ok = bool((head <= total[mask] + 0.1 + 1e-6 * total[mask]).all())
             return PERMIT if ok else REJECT
 
@@ -55,6 +58,8 @@ class OvercommitPlugin(Plugin):
                 demand = np.zeros(nt.r, dtype=np.float64)
                 for j in jobs:
                     demand += j.minres_vec(nt)
+            if (demand[~mask] > 0.1).any():
+                return None       # demand on a zero-capacity dim
             head = inqueue[mask] + demand[mask]
             if bool((head <= total[mask] + 0.1 + 1e-6 * total[mask]).all()):
                 return lambda: inqueue.__iadd__(demand)

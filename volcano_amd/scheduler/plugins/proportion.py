@@ -133,6 +133,8 @@ class ProportionPlugin(Plugin):
             if qi is None:
                 return REJECT
             minres = job.minres_vec(nt)
+            if (minres[~mask] > 0.1).any():
+                return REJECT     # demand on a zero-capacity dim
             head = (alloc_np[qi][mask].astype(np.float64)
                     + inqueue_np[qi][mask] + minres[mask])
             des = des_np[qi][mask].astype(np.float64)
@@ -175,6 +177,8 @@ class ProportionPlugin(Plugin):
                 demand = np.zeros(R, dtype=np.float64)
                 for j in jobs:
                     demand += j.minres_vec(nt)
+            if (demand[~mask] > 0.1).any():
+                return None       # demand on a zero-capacity dim
             head = (alloc_np[qi][mask].astype(np.float64)
                     + inqueue_np[qi][mask] + demand[mask])
             des = des_np[qi][mask].astype(np.float64)
