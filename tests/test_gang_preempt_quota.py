@@ -154,3 +154,105 @@ def test_evict_task_credits_releasing():
     assert node.used.get(CPU) == used_before    # still held until released
     fi = node.future_idle
     assert fi.get(CPU) == 4000 - used_before + 1000
+
+
+def test_gangpreempt_safe_bundle_keeps_victim_gang_alive():
+    """Reference bundle.go BundleSafe: the victim gang has surplus above
+    minAvailable — gangpreempt evicts ONLY the surplus and the victim
+    stays a live gang."""
+    from volcano_amd.utils import synth
+    from volcano_amd.utils.uthelper import TestCommonStruct
+    G = 10 ** 9
+
+    def pg(name, mm, prio, phase="Inqueue"):
+        g = synth.make_podgroup(name, queue="q1", namespace="c1",
+                                min_member=mm)
+        g.status.phase = phase
+        g.meta.annotations["priority"] = str(prio)
+        return g
+
+    pods = [synth.make_pod(f"low-{i}", "lowpg", namespace="c1",
+                           cpu_milli=1000, mem=G, node_name="n1",
+                           phase="Running", priority=1)
+            for i in range(4)]
+    pods.append(synth.make_pod("high-0", "highpg", namespace="c1",
+                               cpu_milli=2000, mem=2 * G, priority=100))
+    t = TestCommonStruct(
+        name="safe bundle evicts surplus only",
+        podgroups=[pg("lowpg", 2, 1, phase="Running"),
+                   pg("highpg", 1, 100)],
+        pods=pods,
+        nodes=[synth.make_node("n1", cpu_milli=4000, mem=4 * G, pods=10)],
+        queues=[synth.make_queue("q1")],
+        actions=["enqueue", "gangpreempt"],
+        expect_evicted=["c1/low-0", "c1/low-1"],
+    ).run()
+    t.check_all()
+    low = t.cache.jobs["c1/lowpg"]
+    assert low.occupied_count >= low.min_available   # gang survived
+
+
+def test_gangpreempt_allow_whole_bundle_false_blocks_whole_job():
+    """Reference gangpreempt.go:253: with allowWholeBundle=false a victim
+    at its minAvailable (whole-bundle only) is never evicted."""
+    from volcano_amd.utils import synth
+    from volcano_amd.utils.uthelper import TestCommonStruct
+    G = 10 ** 9
+
+    def scenario(allow):
+        g1 = synth.make_podgroup("lowpg", queue="q1", namespace="c1",
+                                 min_member=2)
+        g1.status.phase = "Running"
+        g1.meta.annotations["priority"] = "1"
+        g2 = synth.make_podgroup("highpg", queue="q1", namespace="c1",
+                                 min_member=1)
+        g2.status.phase = "Inqueue"
+        g2.meta.annotations["priority"] = "100"
+        pods = [synth.make_pod(f"low-{i}", "lowpg", namespace="c1",
+                               cpu_milli=2000, mem=2 * G, node_name="n1",
+                               phase="Running", priority=1)
+                for i in range(2)]       # at min: no surplus
+        pods.append(synth.make_pod("high-0", "highpg", namespace="c1",
+                                   cpu_milli=3000, mem=3 * G, priority=100))
+        t = TestCommonStruct(
+            podgroups=[g1, g2], pods=pods,
+            nodes=[synth.make_node("n1", cpu_milli=4000, mem=4 * G,
+                                   pods=10)],
+            queues=[synth.make_queue("q1")],
+            actions=["enqueue", "gangpreempt"],
+            cycles=0)           # build only — configure, then cycle
+        t.run()
+        return t
+
+    t = scenario(False)
+    t.scheduler.config.configurations = {
+        "gangpreempt": {"allowWholeBundle": False}}
+    t.scheduler.run_once()
+    assert t.binder.evictions == [], t.binder.evictions
+
+    t2 = scenario(True)     # default allowWholeBundle=true evicts both
+    t2.scheduler.run_once()
+    assert sorted(t2.binder.evictions) == ["c1/low-0", "c1/low-1"], \
+        t2.binder.evictions
+
+
+def test_gangpreempt_parse_arguments_defaults():
+    """Reference gangpreempt_test.go TestParseArguments +
+    InvalidMaxDomainsFallsBackToDefault."""
+    from volcano_amd.scheduler.actions.gangpreempt import (
+        DEFAULT_MAX_DOMAINS, GangPreemptAction)
+
+    class FakeSsn:
+        class config:
+            configurations = {"gangpreempt": {"maxDomains": 3,
+                                              "allowWholeBundle": False}}
+
+    a = GangPreemptAction()
+    a._parse_arguments(FakeSsn)
+    assert a.max_domains == 3 and a.allow_whole_bundle is False
+
+    for bad in (0, -1, "x", None):
+        FakeSsn.config.configurations = {"gangpreempt": {"maxDomains": bad}}
+        a._parse_arguments(FakeSsn)
+        assert a.max_domains == DEFAULT_MAX_DOMAINS
+        assert a.allow_whole_bundle is True

@@ -16,17 +16,49 @@ from ...api.info import JobInfo
 from ...api.types import ALLOCATED_STATUSES, PodGroupPhase, TaskStatus
 from ..statement import Statement
 
-VICTIM_STATUSES = ALLOCATED_STATUSES + (TaskStatus.RUNNING,)
+# ALLOCATED_STATUSES already includes RUNNING — appending it again made
+# tasks_with_status yield running victims twice (double eviction)
+VICTIM_STATUSES = ALLOCATED_STATUSES
+assert TaskStatus.RUNNING in VICTIM_STATUSES
+
+
+DEFAULT_MAX_DOMAINS = 8     # reference gangpreempt.go:36
 
 
 class GangPreemptAction:
     name = "gangpreempt"
     same_queue = True
 
+    def __init__(self):
+        self.max_domains = DEFAULT_MAX_DOMAINS
+        self.allow_whole_bundle = True
+
+    def _parse_arguments(self, ssn) -> None:
+        """Action knobs (reference gangpreempt.go:63-77 parseArguments):
+        ``maxDomains`` caps the candidate-domain scan (invalid/<=0 falls
+        back to the default; domain iteration itself lives in the
+        topology preempt path — actions/preempt.py), ``allowWholeBundle``
+        toggles whole-job victim bundles."""
+        args = getattr(ssn.config, "configurations", {}).get(self.name, {})
+        md = args.get("maxDomains")
+        try:
+            md = int(md) if md is not None else 0
+        except (TypeError, ValueError):
+            md = 0
+        self.max_domains = md if md > 0 else DEFAULT_MAX_DOMAINS
+        awb = args.get("allowWholeBundle")
+        if isinstance(awb, bool):
+            self.allow_whole_bundle = awb
+        elif isinstance(awb, str):
+            self.allow_whole_bundle = awb.lower() != "false"
+        else:
+            self.allow_whole_bundle = True
+
     def execute(self, ssn) -> None:
         nt = ssn.node_tensors
         if nt is None or nt.n == 0 or not ssn.preemptable_fns:
             return
+        self._parse_arguments(ssn)
         for q in ssn.sorted_queues():
             jobs_in_q = [j for j in ssn.jobs.values() if j.queue == q.name
                          and j.phase in (PodGroupPhase.INQUEUE.value,
@@ -78,6 +110,65 @@ class GangPreemptAction:
                 return False    # some task vetoed → the bundle is not free
         return True
 
+    def _victim_bundles(self, ssn, rep, pj: JobInfo, vj: JobInfo):
+        """Split a victim job's occupied tasks into a SAFE bundle
+        (evictable surplus above the job/role minimums — the victim gang
+        survives) and a WHOLE bundle (the remainder; taking it strands
+        the gang, so the whole-job semantics of :meth:`_victim_ok`
+        apply) — reference ``actions/utils/bundle.go:96-193``
+        BuildVictimBundles.  Safe victims pass the FULL plugin
+        intersection (including the gang keep-above-min filter: they are
+        above-min by construction); whole bundles are gated by
+        ``allowWholeBundle`` and the bundle-exempt intersection."""
+        victims = vj.tasks_with_status(*VICTIM_STATUSES)
+        if not victims or vj.priority >= pj.priority:
+            return []
+        nt = ssn.node_tensors
+        out = []
+        surplus = vj.occupied_count - vj.min_available
+        mtm = vj.min_task_member or {}
+        role_surplus = {r: vj.role_occupied(r) - m for r, m in mtm.items()}
+        safe = []
+        safe_ids = set()
+        for t in sorted(victims, key=lambda t: t.priority):
+            rs = role_surplus.get(t.role)
+            if surplus > 0 and (rs is None or rs > 0) and t.preemptable:
+                safe.append(t)
+                safe_ids.add(t.uid)
+                surplus -= 1
+                if rs is not None:
+                    role_surplus[t.role] = rs - 1
+        if safe:
+            fns = ssn.preemptable_fns if self.same_queue \
+                else ssn.reclaimable_fns
+            allowed = safe
+            for fn in fns:
+                keep = {t.uid for t in fn(rep, allowed)}
+                allowed = [t for t in allowed if t.uid in keep]
+                if not allowed:
+                    break
+            if allowed:
+                gvec = np.zeros(nt.r, dtype=np.float64)
+                for t in allowed:
+                    v = nt.req_vector(t)
+                    if v is not None:
+                        gvec += v
+                out.append(("safe", vj, allowed, gvec))
+                safe_ids = {t.uid for t in allowed}
+            else:
+                safe_ids = set()
+        if self.allow_whole_bundle and self._victim_ok(ssn, rep, pj, vj):
+            # disjoint from the safe bundle (evict lists never overlap)
+            rest = [t for t in victims if t.uid not in safe_ids]
+            if rest:
+                gvec = np.zeros(nt.r, dtype=np.float64)
+                for t in rest:
+                    v = nt.req_vector(t)
+                    if v is not None:
+                        gvec += v
+                out.append(("whole", vj, rest, gvec))
+        return out
+
     def _gang_preempt(self, ssn, job: JobInfo) -> None:
         nt = ssn.node_tensors
         need_tasks = job.min_available - job.occupied_count - job.waiting_count
@@ -100,9 +191,10 @@ class GangPreemptAction:
             if ni.ready:
                 free += nt.resource_vector(ni.future_idle)
 
-        # candidate victim bundles, ROI order: lowest priority first, then
-        # smallest job (fewest evictions)
-        cands: List[JobInfo] = []
+        # candidate victim bundles: SAFE bundles first (no gang cost),
+        # then WHOLE, ROI order within kind (lowest priority, fewest
+        # evictions) — reference utils.SortBundlesForPreempt
+        bundles = []
         for vj in ssn.jobs.values():
             if vj.key == job.key:
                 continue
@@ -110,23 +202,23 @@ class GangPreemptAction:
                 continue
             if not self.same_queue and vj.queue == job.queue:
                 continue
-            if self._victim_ok(ssn, rep, job, vj):
-                cands.append(vj)
-        cands.sort(key=lambda j: (j.priority, len(j.tasks)))
+            bundles.extend(self._victim_bundles(ssn, rep, job, vj))
+        bundles.sort(key=lambda b: (b[0] != "safe",
+                                    b[1].priority, len(b[2])))
 
-        chosen: List[JobInfo] = []
+        chosen = []
         gain = free.copy()
-        for vj in cands:
+        for kind, vj, victims, gvec in bundles:
             if (gain + 0.1 >= need_vec).all():
                 break
-            chosen.append(vj)
-            gain += nt.resource_vector(vj.allocated_resource())
+            chosen.append((vj, victims))
+            gain += gvec
         if not (gain + 0.1 >= need_vec).all():
             return     # even evicting every bundle would not fit: do nothing
 
         stmt = Statement(ssn)
-        for vj in chosen:
-            for v in vj.tasks_with_status(*VICTIM_STATUSES):
+        for vj, victims in chosen:
+            for v in victims:
                 stmt.evict(v, reason="gang-preempted")
         # pipeline the preemptor tasks onto freed capacity (node-level
         # placement happens next cycle when the evictions have landed;

@@ -61,7 +61,7 @@ class PreemptAction:
     name = "preempt"
 
     #: victims eligible: running/bound tasks that are preemptable targets
-    victim_statuses = ALLOCATED_STATUSES + (TaskStatus.RUNNING,)
+    victim_statuses = ALLOCATED_STATUSES    # includes RUNNING
 
     def execute(self, ssn) -> None:
         nt = ssn.node_tensors
