@@ -417,3 +417,45 @@ def test_jobtemplate_controller_bookkeeping():
     cm.sync_until_quiet()
     tmpl = store.get("JobTemplate", "default", "step-a")
     assert tmpl.status.get("jobDependsOnList") == ["f1-step-a"]
+
+
+def test_datasource_claim_binding_and_reclaim():
+    """datadependency/v1alpha1 (reference staging types.go:32-201):
+    claims bind by (system, type, name); the source mirrors claimRefs;
+    reclaimPolicy=Delete removes the source when the last claim goes."""
+    from volcano_amd.api.objects import DataSource, DataSourceClaim, ObjectMeta
+    from volcano_amd.controllers import ControllerManager
+    from volcano_amd.store import ObjectStore
+
+    store = ObjectStore()
+    mgr = ControllerManager(store, controllers=["datadependency"])
+    src = DataSource(meta=ObjectMeta(name="sales", namespace=""),
+                     system="hive", type="table", name="db.sales",
+                     cluster_names=["c1"], reclaim_policy="Delete")
+    store.create("DataSource", src)
+    claim = DataSourceClaim(
+        meta=ObjectMeta(name="train-input", namespace="ml"),
+        system="hive", data_source_type="table",
+        data_source_name="db.sales",
+        workload={"apiVersion": "batch.volcano.sh/v1alpha1",
+                  "kind": "Job", "name": "train"})
+    store.create("DataSourceClaim", claim)
+    mgr.sync_once()
+
+    c = store.get("DataSourceClaim", "ml", "train-input")
+    assert c.phase == "Bound" and c.bound_data_source == "sales"
+    s = store.get("DataSource", "", "sales")
+    assert s.bound_claims == 1 and s.claim_refs == ["ml/train-input"]
+
+    # unmatched claim stays Pending
+    orphan = DataSourceClaim(
+        meta=ObjectMeta(name="orphan", namespace="ml"),
+        system="s3", data_source_type="bucket", data_source_name="nope")
+    store.create("DataSourceClaim", orphan)
+    mgr.sync_once()
+    assert store.get("DataSourceClaim", "ml", "orphan").phase == "Pending"
+
+    # last claim released → Delete reclaim removes the source
+    store.delete("DataSourceClaim", "ml", "train-input")
+    mgr.sync_once()
+    assert store.get("DataSource", "", "sales") is None

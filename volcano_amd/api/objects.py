@@ -500,6 +500,42 @@ class ResourceClaim:
 
 
 @dataclass
+class DataSource:
+    """datadependency/v1alpha1 DataSource: a cached, cluster-scoped
+    record of where a logical data source (hive table, s3 bucket, hdfs
+    path ...) is available.  Reference:
+    staging/.../datadependency/v1alpha1/types.go:32-103."""
+
+    meta: ObjectMeta = field(default_factory=ObjectMeta)
+    system: str = ""                    # "hive" | "s3" | "hdfs" | ...
+    type: str = ""                      # "table" | "bucket" | ...
+    name: str = ""                      # logical name within the system
+    cluster_names: List[str] = field(default_factory=list)  # locality
+    attributes: Dict[str, str] = field(default_factory=dict)
+    reclaim_policy: str = "Retain"      # Retain | Delete
+    # status
+    claim_refs: List[str] = field(default_factory=list)   # bound claims
+    bound_claims: int = 0
+
+
+@dataclass
+class DataSourceClaim:
+    """datadependency/v1alpha1 DataSourceClaim: a workload's request for
+    a DataSource, matched by (system, dataSourceType, dataSourceName).
+    Reference: staging/.../datadependency/v1alpha1/types.go:124-201."""
+
+    meta: ObjectMeta = field(default_factory=ObjectMeta)
+    system: str = ""
+    data_source_type: str = ""
+    data_source_name: str = ""
+    workload: Dict[str, str] = field(default_factory=dict)  # WorkloadRef
+    attributes: Dict[str, str] = field(default_factory=dict)
+    # status
+    phase: str = "Pending"              # Pending | Bound | Lost
+    bound_data_source: str = ""
+
+
+@dataclass
 class CronJob:
     """batch/v1alpha1 CronJob (reference cronjob controller)."""
 
@@ -568,4 +604,5 @@ KINDS = {
     "PersistentVolume": PersistentVolume,
     "PersistentVolumeClaim": PersistentVolumeClaim,
     "DeviceClass": DeviceClass, "ResourceClaim": ResourceClaim,
+    "DataSource": DataSource, "DataSourceClaim": DataSourceClaim,
 }

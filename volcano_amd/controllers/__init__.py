@@ -10,3 +10,4 @@ from .hypernode import HyperNodeController
 from .sharding import ShardingController
 from .colocationconfig import ColocationConfigController
 from .hyperjob import HyperJobController
+from .datadependency import DataDependencyController

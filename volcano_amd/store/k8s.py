@@ -37,6 +37,9 @@ GVK: Dict[str, Tuple[str, str, str]] = {
     "PodDisruptionBudget": ("policy", "v1", "poddisruptionbudgets"),
     "DeviceClass": ("resource.k8s.io", "v1", "deviceclasses"),
     "ResourceClaim": ("resource.k8s.io", "v1", "resourceclaims"),
+    "DataSource": ("datadependency.volcano.sh", "v1alpha1", "datasources"),
+    "DataSourceClaim": ("datadependency.volcano.sh", "v1alpha1",
+                        "datasourceclaims"),
 }
 
 # (group, version, plural) → kind
