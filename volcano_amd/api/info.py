@@ -716,6 +716,18 @@ class QueueInfo:
         return max(int(self.queue.spec.weight), 1)
 
     @property
+    def hierarchy(self) -> str:
+        """Slash path in the fair-share tree (reference queue_info.go:48,
+        annotation ``volcano.sh/hierarchy``, e.g. ``root/eng/dev``)."""
+        return self.queue.meta.annotations.get("volcano.sh/hierarchy", "")
+
+    @property
+    def hierarchy_weights(self) -> str:
+        """Slash weights along :attr:`hierarchy` (queue_info.go:45)."""
+        return self.queue.meta.annotations.get(
+            "volcano.sh/hierarchy-weights", "")
+
+    @property
     def capability(self) -> Resource:
         from .resource import normalize_dra_keys
         return normalize_dra_keys(self.queue.spec.capability)
