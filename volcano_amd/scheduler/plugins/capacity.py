@@ -188,17 +188,120 @@ class CapacityPlugin(Plugin):
             if qi is not None:
                 inqueue_np[qi] += job.minres_vec(nt)
 
+        # -- reclaim eligibility with ancestor levels ------------------------
+        # (reference capacity.go:500-600 AddReclaimableFn + the level
+        # semantics of capacity_test.go cases 1-11)
+        anc_level = int(self.args.get("ancestorReclaimLevel", 0) or 0)
+        des_spec_np = deserved_spec.numpy().astype(np.float64)
+        gua_np = guarantee.numpy().astype(np.float64)
+        # ancestors chain per queue, root-first, excluding self
+        # ("" = virtual root, never counted)
+        anc_chain: Dict[str, List[str]] = {}
+        for q in queues:
+            chain, cur, seen = [], q.queue.spec.parent, set()
+            while cur and cur in by_name and cur not in seen:
+                seen.add(cur)
+                chain.append(cur)
+                cur = by_name[cur].queue.spec.parent
+            anc_chain[q.name] = chain[::-1]       # root → parent
+
+        desc_rows: Dict[str, List[int]] = {q.name: [qi_of[q.name]]
+                                           for q in queues}
+        for q in queues:
+            for a in anc_chain[q.name]:
+                desc_rows.setdefault(a, []).append(qi_of[q.name])
+
+        def _is_root(qn: str) -> bool:
+            q = by_name.get(qn)
+            return q is None or not q.queue.spec.parent
+
+        def _anc_at(qname: str, level: int):
+            chain = anc_chain.get(qname, ())
+            return chain[len(chain) - level] if 0 < level <= len(chain) \
+                else None
+
+        def _shares_nonroot_anc(a: str, b: str) -> bool:
+            aa = {x for lv in range(1, anc_level + 1)
+                  if (x := _anc_at(a, lv)) is not None and not _is_root(x)}
+            bb = {x for lv in range(1, anc_level + 1)
+                  if (x := _anc_at(b, lv)) is not None and not _is_root(x)}
+            return bool(aa & bb)
+
         def reclaimable(reclaimer, candidates):
+            rjob = ssn.jobs.get(reclaimer.job_key)
+            rq = rjob.queue if rjob is not None else ""
+            rreq = nt.req_vector(reclaimer)
+            rreq = rreq.astype(np.float64) if rreq is not None else None
+            rdes = des_spec_np[qi_of[rq]] if rq in qi_of else None
+            allocations: Dict[str, np.ndarray] = {}
+
+            def alloc_of(qn: str) -> np.ndarray:
+                a = allocations.get(qn)
+                if a is None:
+                    rows = desc_rows.get(qn, [])
+                    a = alloc_np[rows].astype(np.float64).sum(axis=0)
+                    allocations[qn] = a
+                return a
+
             out = []
             for v in candidates:
                 job = ssn.jobs.get(v.job_key)
                 if job is None:
                     continue
                 q = ssn.queues.get(job.queue)
-                if q is None or not q.reclaimable:
+                if q is None or not q.reclaimable or job.queue not in qi_of:
                     continue
-                if overused(q):
-                    out.append(v)
+                vreq = nt.req_vector(v)
+                if vreq is None:
+                    continue
+                vreq = vreq.astype(np.float64)
+                # no intersecting dims with the reclaimer → irrelevant
+                if rreq is not None and not ((vreq > 0) & (rreq > 0)).any():
+                    continue
+                qi = qi_of[job.queue]
+                alloc = alloc_of(job.queue)
+                # guarantee floor: evicting must not dip below guarantee
+                if ((alloc - vreq) < gua_np[qi] - 1e-9).any():
+                    continue
+                des = des_spec_np[qi]
+                relevant = (vreq > 0) & (des > 0)
+                if not relevant.any():
+                    # immediate victim (leaf deserved silent on its dims)
+                    if anc_level > 0 and rq and \
+                            _shares_nonroot_anc(rq, job.queue) and \
+                            not (rreq is not None and rdes is not None
+                                 and ((rreq > 0) & (rdes > 0)).any()):
+                        continue    # no real contention under shared scope
+                elif not (alloc[relevant] > des[relevant] + 0.1).any():
+                    continue        # within deserved on every relevant dim
+                # ancestor gates up to the configured level
+                ok = True
+                touched = []
+                for lv in range(1, anc_level + 1):
+                    anc = _anc_at(job.queue, lv)
+                    if anc is None or _is_root(anc) \
+                            or anc == _anc_at(rq, lv):
+                        continue    # reference skips the tree root and
+                        # shared ancestors (getReclaimeeAncestorToCheck)
+                    a_alloc = alloc_of(anc)
+                    a_des = des_spec_np[qi_of[anc]] if anc in qi_of else None
+                    if a_des is None:
+                        ok = False
+                        break
+                    rel = (vreq > 0) & (a_des > 0)
+                    if not rel.any():
+                        touched.append(a_alloc)
+                        continue    # immediate victim vs this ancestor
+                    if not (a_alloc[rel] > a_des[rel] + 0.1).any():
+                        ok = False
+                        break
+                    touched.append(a_alloc)
+                if not ok:
+                    continue
+                alloc -= vreq
+                for a in touched:
+                    a -= vreq
+                out.append(v)
             return out
 
         ssn.add_queue_order_fn(

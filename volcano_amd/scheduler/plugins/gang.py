@@ -74,5 +74,9 @@ class GangPlugin(Plugin):
         ssn.job_ready_fns.append(job_ready)
         ssn.job_pipelined_fns.append(job_pipelined)
         ssn.job_starving_fns.append(job_starving)
-        ssn.preemptable_fns.append(preemptable)
-        ssn.reclaimable_fns.append(preemptable)
+        # per-function enable flags (reference conf.PluginOption
+        # EnabledPreemptable/EnabledReclaimable — default on)
+        if self.args.get("enabledPreemptable", True) is not False:
+            ssn.preemptable_fns.append(preemptable)
+        if self.args.get("enabledReclaimable", True) is not False:
+            ssn.reclaimable_fns.append(preemptable)

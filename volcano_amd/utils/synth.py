@@ -39,12 +39,16 @@ def make_nodes(n: int, prefix: str = "node", **kw) -> List[Node]:
 def make_queue(name: str, weight: int = 1,
                capability: Optional[Dict[str, float]] = None,
                guarantee: Optional[Dict[str, float]] = None,
-               reclaimable: bool = True) -> Queue:
+               reclaimable: bool = True,
+               parent: str = "",
+               deserved: Optional[Dict[str, float]] = None) -> Queue:
     return Queue(meta=ObjectMeta(name=name, namespace="default"),
                  spec=QueueSpec(weight=weight,
                                 capability=Resource(capability or {}),
                                 guarantee=Resource(guarantee or {}),
-                                reclaimable=reclaimable))
+                                reclaimable=reclaimable,
+                                parent=parent,
+                                deserved=Resource(deserved or {})))
 
 
 def make_podgroup(name: str, queue: str = "default", min_member: int = 1,
