@@ -137,3 +137,81 @@ def test_no_preempt_when_queue_has_headroom():
         expect_bind_count=1,
     ).run()
     t.check_all()
+
+
+def test_preempt_protects_nonpreemptable_and_higher_priority_pods():
+    """preempt_test.go: queue at capability 3 — only the explicitly
+    preemptable, lower-priority victim goes; preemptable=false and
+    equal/higher pod priority are both protected."""
+    t = TestCommonStruct(
+        name="not pod with preemptable=false or higher priority",
+        podgroups=[pg("c1", "pg1", "q1", 1, prio=1),
+                   pg("c1", "pg2", "q1", 1, prio=100)],
+        pods=[pod("c1", "preemptee1", "pg1", 1000, G, "n1", "Running",
+                  preemptable="false", prio=1),
+              pod("c1", "preemptee2", "pg1", 1000, G, "n1", "Running",
+                  preemptable="true", prio=1),
+              pod("c1", "preemptee3", "pg1", 1000, G, "n1", "Running",
+                  preemptable="true", prio=100),
+              pod("c1", "preemptor1", "pg2", 1000, G, prio=100)],
+        nodes=[synth.make_node("n1", cpu_milli=12000, mem=12 * G, pods=10)],
+        queues=[q("q1", cap={"cpu": 3000.0, "memory": 3.0 * G})],
+        actions=PREEMPT,
+        expect_evicted=["c1/preemptee2"],
+    ).run()
+    t.check_all()
+
+
+def test_unbesteffort_preempts_besteffort_pod_slot():
+    """#3335: the node has ONE pod slot held by a best-effort pod; the
+    resourceful high-priority preemptor evicts it on the pods dim."""
+    t = TestCommonStruct(
+        name="unBestEffort preempts BestEffort",
+        podgroups=[pg("c1", "pg1", "q1", 0, prio=1),
+                   pg("c1", "pg2", "q1", 1, prio=100)],
+        pods=[pod("c1", "preemptee1", "pg1", 0, 0, "n1", "Running",
+                  preemptable="true", prio=1),
+              pod("c1", "preemptor1", "pg2", 3000, 3 * G, prio=100)],
+        nodes=[synth.make_node("n1", cpu_milli=12000, mem=12 * G, pods=1)],
+        queues=[q("q1", cap={"cpu": 6000.0, "memory": 6.0 * G})],
+        actions=PREEMPT,
+        expect_evicted=["c1/preemptee1"],
+    ).run()
+    t.check_all()
+
+
+def test_besteffort_preempts_besteffort_pod_slot():
+    """#3335: both best-effort — still contend on the pods dim."""
+    t = TestCommonStruct(
+        name="BestEffort preempts BestEffort",
+        podgroups=[pg("c1", "pg1", "q1", 0, prio=1),
+                   pg("c1", "pg2", "q1", 1, prio=100)],
+        pods=[pod("c1", "preemptee1", "pg1", 0, 0, "n1", "Running",
+                  preemptable="true", prio=1),
+              pod("c1", "preemptor1", "pg2", 0, 0, prio=100)],
+        nodes=[synth.make_node("n1", cpu_milli=12000, mem=12 * G, pods=1)],
+        queues=[q("q1", cap={"cpu": 6000.0, "memory": 6.0 * G})],
+        actions=PREEMPT,
+        expect_evicted=["c1/preemptee1"],
+    ).run()
+    t.check_all()
+
+
+def test_preemption_policy_never_blocks_preemption():
+    """#3642 (preempt.go:441): a preemptor with preemptionPolicy=Never
+    waits instead of displacing the running pod."""
+    never = pod("c1", "preemptor1", "pg2", 1000, G, prio=100)
+    never.preemption_policy = "Never"
+    t = TestCommonStruct(
+        name="task preemption policy never",
+        podgroups=[pg("c1", "pg1", "q1", 0, prio=1),
+                   pg("c1", "pg2", "q1", 1, prio=100)],
+        pods=[pod("c1", "preemptee1", "pg1", 1000, G, "n1", "Running",
+                  preemptable="true", prio=1),
+              never],
+        nodes=[synth.make_node("n1", cpu_milli=1000, mem=G, pods=1)],
+        queues=[q("q1")],
+        actions=PREEMPT,
+        expect_evicted=[],
+    ).run()
+    t.check_all()

@@ -115,6 +115,10 @@ class Pod:
     best_effort: bool = field(default=False)
     scheduling_gates: List[str] = field(default_factory=list)
     volumes: List[str] = field(default_factory=list)   # PVC names (same ns)
+    # k8s pod.spec.preemptionPolicy: "" (PreemptLowerPriority) | "Never" —
+    # a Never preemptor may wait but must not displace anything
+    # (reference preempt.go:441 taskEligibleToPreempt, #3642)
+    preemption_policy: str = ""
     host_ports: List[int] = field(default_factory=list)  # requested hostPorts
     image: str = ""                                    # container image
     # DRA (k8s dynamic resource allocation): ResourceClaim names (same

@@ -139,7 +139,12 @@ class PreemptAction:
         for tc in job.pending_classes():
             if placed >= still_needed:
                 break
-            req_vec = nt.req_vector(tc.tasks[0])
+            rep0 = tc.tasks[0]
+            if rep0.pod is not None and \
+                    rep0.pod.preemption_policy == "Never":
+                continue    # reference preempt.go:441 (#3642): a Never
+                # preemptor may wait but must not displace anything
+            req_vec = nt.req_vector(rep0)
             if req_vec is None:
                 continue
             constraints = predicates.class_constraints(
@@ -189,7 +194,12 @@ class PreemptAction:
         for tc in job.pending_classes():
             if placed >= still_needed:
                 break
-            req_vec = nt.req_vector(tc.tasks[0])
+            rep0 = tc.tasks[0]
+            if rep0.pod is not None and \
+                    rep0.pod.preemption_policy == "Never":
+                continue    # reference preempt.go:441 (#3642): a Never
+                # preemptor may wait but must not displace anything
+            req_vec = nt.req_vector(rep0)
             if req_vec is None:
                 continue
             constraints = predicates.class_constraints(tc, job) if predicates \
