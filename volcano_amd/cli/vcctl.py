@@ -182,6 +182,16 @@ def pod_list(backend, args):
     print(_fmt_table(rows, ["NAME", "PHASE", "NODE", "PODGROUP"]))
 
 
+def podgroup_list(backend, args):
+    """reference pkg/cli/podgroup list-podgroup.go"""
+    rows = []
+    for g in backend.list("PodGroup", namespace=args.namespace or None):
+        rows.append((g.meta.namespace, g.meta.name, g.spec.queue,
+                     g.spec.min_member, g.status.phase))
+    print(_fmt_table(rows, ["NAMESPACE", "NAME", "QUEUE", "MINMEMBER",
+                            "PHASE"]))
+
+
 def jobflow_list(backend, args):
     rows = []
     for f in backend.list("JobFlow"):
@@ -271,6 +281,12 @@ def build_parser() -> argparse.ArgumentParser:
     pl = pod.add_parser("list")
     pl.add_argument("--namespace", "-n", default=None)
     pl.set_defaults(fn=pod_list)
+
+    pgp = sub.add_parser("podgroup").add_subparsers(dest="cmd",
+                                                    required=True)
+    pgl = pgp.add_parser("list")
+    pgl.add_argument("--namespace", "-n", default=None)
+    pgl.set_defaults(fn=podgroup_list)
 
     jf = sub.add_parser("jobflow").add_subparsers(dest="cmd", required=True)
     jf.add_parser("list").set_defaults(fn=jobflow_list)
