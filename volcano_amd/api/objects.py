@@ -126,7 +126,10 @@ class Pod:
     resource_claims: List[str] = field(default_factory=list)
 
     def __post_init__(self):
-        if self.request.is_empty():
+        # BestEffort = no resource requests (reference pod_info.go); the
+        # implicit one-pod slot every pod occupies doesn't count
+        if not any(v > 1e-9 for k, v in self.request.q.items()
+                   if k != "pods"):
             self.best_effort = True
 
     @property

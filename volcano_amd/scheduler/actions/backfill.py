@@ -29,20 +29,27 @@ class BackfillAction:
         predicates = getattr(ssn, "predicates", None)
 
         from ...api.types import TaskStatus
-        for job in ssn.jobs.values():
-            if not job.task_status_index.get(TaskStatus.PENDING):
-                continue      # emptiness check before any property work
-            if job.phase not in (PodGroupPhase.INQUEUE.value,
-                                 PodGroupPhase.RUNNING.value):
-                continue
+        # job-order tiers decide who backfills first when pod slots are
+        # scarce (reference backfill.go: ssn.JobOrderFn over pending
+        # BestEffort owners; tasks within a job by TaskOrder/priority)
+        eligible = [j for j in ssn.jobs.values()
+                    if j.task_status_index.get(TaskStatus.PENDING)
+                    and j.phase in (PodGroupPhase.INQUEUE.value,
+                                    PodGroupPhase.RUNNING.value)
+                    and ssn.queue_index.get(j.queue) is not None]
+        for job in ssn.sorted_jobs(eligible):
             classes: List[ClassPlan] = []
             qi = ssn.queue_index.get(job.queue)
-            if qi is None:
-                continue
-            for tc in job.pending_classes():
+            for tc in sorted(job.pending_classes(),
+                             key=lambda c: -c.priority):
                 if not tc.tasks[0].best_effort:
                     continue
-                req = np.zeros(nt.r, dtype=np.float32)
+                # a BestEffort pod still occupies its one-pod slot — use
+                # the real (pods-only) request so node pod capacity and
+                # queue pod quotas clamp the fill
+                req = nt.req_vector(tc.tasks[0])
+                if req is None:
+                    req = np.zeros(nt.r, dtype=np.float32)
                 if predicates is not None:
                     tol, require, forbid = predicates.class_constraints(tc, job)
                 else:
