@@ -655,3 +655,64 @@ def test_cycle_equivalence_dra_dims(hip):
     # 24-device quota admits exactly 6 gangs (4 devices each)
     assert len(cpu) == 12
     assert cpu == gpu
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("seed", [11, 23, 47])
+def test_cycle_equivalence_randomized_sweep(hip, seed):
+    """Seeded random inventories (heterogeneous gangs, queues, selectors,
+    priorities, partial occupancy) run two cycles on CPU oracle and HIP —
+    bind maps must match exactly.  The randomized counterpart of the
+    hand-built feature-rich case."""
+    from volcano_amd.scheduler import FakeBinder, Scheduler, SchedulerCache, \
+        default_config
+    from volcano_amd.store import ObjectStore
+    from volcano_amd.utils import synth
+
+    GI = 1024 ** 3
+
+    def build(device, use_hip):
+        rng = np.random.RandomState(seed)
+        store = ObjectStore()
+        for i in range(200):
+            labels = {"zone": f"z{i % 3}"} if rng.rand() < 0.5 else {}
+            store.create("Node", synth.make_node(
+                f"n-{i:04d}", cpu_milli=float(rng.choice([4000, 8000,
+                                                          16000])),
+                mem=float(rng.choice([8, 16, 32])) * GI, labels=labels))
+        for qn, w in (("qa", 2), ("qb", 1), ("qc", 3)):
+            store.create("Queue", synth.make_queue(qn, weight=w))
+        for j in range(60):
+            q = ["qa", "qb", "qc"][int(rng.randint(3))]
+            size = int(rng.choice([1, 2, 4, 8]))
+            sel = {"zone": f"z{int(rng.randint(3))}"} \
+                if rng.rand() < 0.25 else None
+            synth.make_gang(store, f"r{j:03d}", replicas=size, queue=q,
+                            cpu_milli=float(rng.choice([250, 500, 1000,
+                                                        2000])),
+                            mem=float(rng.choice([1, 2, 4])) * GI,
+                            priority=int(rng.randint(0, 8)),
+                            node_selector=sel)
+        config = default_config()
+        config.use_hip = use_hip
+        config.device = device
+        binder = FakeBinder()
+        cache = SchedulerCache(store=store, binder=binder, device=device)
+        sched = Scheduler(cache, config)
+        sched.run_once()
+        # second wave arrives over the partially-occupied cluster
+        rng2 = np.random.RandomState(seed + 1)
+        for j in range(20):
+            synth.make_gang(store, f"w{j:03d}", replicas=int(
+                rng2.choice([1, 2, 4])), queue="qb",
+                cpu_milli=float(rng2.choice([500, 1000])), mem=GI,
+                priority=int(rng2.randint(0, 8)))
+        sched.run_once()
+        return binder.binds
+
+    cpu = build("cpu", False)
+    gpu = build("cuda", True)
+    assert len(cpu) > 100
+    assert cpu == gpu, (
+        f"decision divergence at seed {seed}: {len(cpu)} vs {len(gpu)}; "
+        f"diff={sorted(set(cpu.items()) ^ set(gpu.items()))[:6]}")
