@@ -52,3 +52,32 @@ def test_backfill_skips_resourceful_pending():
     ).run()
     t.check_all()
     assert list(t.binder.binds) == ["default/thin"]
+
+
+def test_allocate_skips_besteffort_backfill_places_it():
+    """Reference allocate.go:265: allocate never places BestEffort
+    tasks; the same cycle's backfill does."""
+    t = TestCommonStruct(
+        name="allocate skips best-effort",
+        podgroups=[pg("pg1", 1)],
+        pods=[synth.make_pod("fat", "pg1", cpu_milli=1000, mem=G),
+              be_pod("thin", "pg1", 1)],
+        nodes=[synth.make_node("n1", cpu_milli=12000, mem=12 * G, pods=5)],
+        queues=[synth.make_queue("q1")],
+        actions=["enqueue", "allocate"],
+        expect_bind_count=1,       # only the resourceful task
+    ).run()
+    t.check_all()
+    assert list(t.binder.binds) == ["default/fat"]
+
+    t2 = TestCommonStruct(
+        name="backfill completes the job",
+        podgroups=[pg("pg1", 1)],
+        pods=[synth.make_pod("fat", "pg1", cpu_milli=1000, mem=G),
+              be_pod("thin", "pg1", 1)],
+        nodes=[synth.make_node("n1", cpu_milli=12000, mem=12 * G, pods=5)],
+        queues=[synth.make_queue("q1")],
+        actions=["enqueue", "allocate", "backfill"],
+        expect_bind_count=2,
+    ).run()
+    t2.check_all()

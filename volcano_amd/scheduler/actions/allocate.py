@@ -257,6 +257,10 @@ class AllocateAction:
                     skip_until = j
                     atom = job.plan_atom()
                     sig, role, request, priority = atom
+                    rep0 = next(iter(job.task_status_index.get(
+                        _TS.PENDING).values()))
+                    if rep0.best_effort:
+                        continue  # BestEffort is backfill's (allocate.go:265)
                     ckey = (sig, qi)
                     got = cons_memo.get(ckey, _MISS)
                     if got is _MISS:
@@ -352,8 +356,8 @@ class AllocateAction:
             if atom and pend and len(pend) == len(job.tasks):
                 sig, role, request, priority = atom
                 first = next(iter(pend.values()))
-                if first.gated:
-                    continue
+                if first.gated or first.best_effort:
+                    continue    # BestEffort is backfill's (allocate.go:265)
                 # bundle continuation: identical (queue, signature) run —
                 # append the gang entry directly, skipping the
                 # TaskClass/ClassPlan/constraint work bundle_in would
@@ -421,6 +425,8 @@ class AllocateAction:
             classes: List[ClassPlan] = []
             skipped = False
             for tc in job.pending_classes():
+                if tc.tasks[0].best_effort:
+                    continue    # BestEffort is backfill's (allocate.go:265)
                 req = nt.req_vector(tc.tasks[0])
                 if req is None:
                     skipped = True    # asks for a resource no node offers
