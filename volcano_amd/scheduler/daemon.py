@@ -62,6 +62,9 @@ def main(argv=None) -> int:
     ap.add_argument("--api-port", type=int, default=8343)
     ap.add_argument("--dump-dir", default="/tmp")
     ap.add_argument("--once", action="store_true")
+    ap.add_argument("--plugins-dir", default=None,
+                    help="directory of out-of-tree plugin modules "
+                         "(the reference's --plugins-dir)")
     args = ap.parse_args(argv)
 
     import torch
@@ -79,6 +82,13 @@ def main(argv=None) -> int:
         store = ObjectStore.load(args.state)
     except FileNotFoundError:
         store = ObjectStore()
+
+    if args.plugins_dir:
+        from .plugins import load_plugins_dir
+        loaded = load_plugins_dir(args.plugins_dir)
+        if loaded:
+            print(f"loaded out-of-tree plugins: {', '.join(loaded)}",
+                  file=sys.stderr, flush=True)
 
     if args.conf:
         with open(args.conf) as f:
