@@ -287,6 +287,9 @@ class QueueSpec:
     # fifo: a job that cannot enqueue blocks the jobs behind it;
     # traverse (default): keep trying the rest (reference dequeueStrategy)
     dequeue_strategy: str = "traverse"
+    # multi-cluster dispatch targets (reference Queue spec extendClusters,
+    # types.go:461-508 — incubating; carried for API-surface parity)
+    extend_clusters: List[Dict[str, Any]] = field(default_factory=list)
 
 
 @dataclass
