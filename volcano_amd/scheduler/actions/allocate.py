@@ -727,6 +727,8 @@ class AllocateAction:
 
         bind_by_job: Dict[str, List] = {}
         committed_jobs: Dict[str, object] = {}     # key -> JobInfo
+        full_keys = set()       # jobs whose ENTIRE task set placed —
+        # gang-ready by construction, no is_ready() walk needed
         # fire event handlers only when someone registered one — building
         # the per-piece argument lists for nobody was measurable at 10k jobs
         fire = ssn.fire_allocate if ssn.event_handlers else None
@@ -831,6 +833,10 @@ class AllocateAction:
                     job = be.job if be.job is not None \
                         else ssn.jobs[be.job_key]
                     commit_pieces(job, cp, c, pieces)
+                    if need == 0:
+                        # bundle entries are whole-job (all tasks were
+                        # pending): zero remainder ⇒ fully placed
+                        full_keys.add(be.job_key)
                     if fire is not None:
                         fire(cp.tclass, [p[0] for p in pieces],
                              [p[1] for p in pieces],
@@ -880,8 +886,11 @@ class AllocateAction:
                     continue
                 seen.add(real)
                 job = ssn.jobs[real]
-            ready = (job.is_ready() and job.roles_ready()) if fast_ready \
-                else ssn.job_ready(job)
+            if fast_ready and key in full_keys:
+                ready = True        # whole gang placed this cycle
+            else:
+                ready = (job.is_ready() and job.roles_ready()) \
+                    if fast_ready else ssn.job_ready(job)
             pg = job.podgroup
             if ready and pg is not None and pg.status.phase != running:
                 pg.status.phase = running
