@@ -97,3 +97,35 @@ def test_equal_priority_victim_yields_no_bundles():
     pj, rep = make_preemptor(prio=100)
     a = GangPreemptAction()
     assert a._victim_bundles(StubSsn(), rep, pj, vj) == []
+
+
+def test_bundle_invariants_property():
+    """Property sweep: for random victim jobs, the safe bundle never
+    dips the gang below job or role minimums, and safe/whole partition
+    the victims disjointly."""
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    @settings(max_examples=60, deadline=None)
+    @given(n=st.integers(1, 12), mm=st.integers(0, 12),
+           role_min=st.integers(0, 12), prio=st.integers(0, 5))
+    def check(n, mm, role_min, prio):
+        vj = make_victim_job(n, mm,
+                             role_min={"worker": role_min} if role_min
+                             else None, prio=prio)
+        pj, rep = make_preemptor(prio=100)
+        a = GangPreemptAction()
+        bundles = a._victim_bundles(StubSsn(), rep, pj, vj)
+        safe = [b for b in bundles if b[0] == "safe"]
+        whole = [b for b in bundles if b[0] == "whole"]
+        assert len(safe) <= 1 and len(whole) <= 1
+        s_ids = {t.uid for b in safe for t in b[2]}
+        w_ids = {t.uid for b in whole for t in b[2]}
+        assert not (s_ids & w_ids)
+        assert len(s_ids) + len(w_ids) == n
+        # safe eviction keeps the gang at/above both minimums
+        assert n - len(s_ids) >= min(mm, n)
+        if role_min:
+            assert n - len(s_ids) >= min(role_min, n)
+
+    check()
