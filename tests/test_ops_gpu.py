@@ -716,3 +716,59 @@ def test_cycle_equivalence_randomized_sweep(hip, seed):
     assert cpu == gpu, (
         f"decision divergence at seed {seed}: {len(cpu)} vs {len(gpu)}; "
         f"diff={sorted(set(cpu.items()) ^ set(gpu.items()))[:6]}")
+
+
+@pytest.mark.gpu
+def test_cycle_equivalence_hdrf_caps(hip):
+    """Hierarchical-DRF equilibrium caps feed the in-kernel queue clamp:
+    CPU oracle and HIP must bind identically under enableHierarchy."""
+    from volcano_amd.scheduler import FakeBinder, Scheduler, SchedulerCache, \
+        default_config
+    from volcano_amd.scheduler.config import PluginOption
+    from volcano_amd.store import ObjectStore
+    from volcano_amd.utils import synth
+
+    GI = 1024 ** 3
+    HIER = "volcano.sh/hierarchy"
+    HIERW = "volcano.sh/hierarchy-weights"
+
+    def build(device, use_hip):
+        store = ObjectStore()
+        for i in range(50):
+            store.create("Node", synth.make_node(
+                f"n-{i:03d}", cpu_milli=8000, mem=16 * GI))
+        for qn, h, w in (("sci", "root/sci", "100/50"),
+                         ("dev", "root/eng/dev", "100/50/30"),
+                         ("prod", "root/eng/prod", "100/50/70")):
+            q = synth.make_queue(qn)
+            q.meta.annotations[HIER] = h
+            q.meta.annotations[HIERW] = w
+            store.create("Queue", q)
+        rng = np.random.RandomState(7)
+        for j in range(45):
+            qn = ("sci", "dev", "prod")[j % 3]
+            synth.make_gang(store, f"h{j:03d}",
+                            replicas=int(rng.choice([2, 4, 8])), queue=qn,
+                            cpu_milli=float(rng.choice([500, 1000, 2000])),
+                            mem=float(rng.choice([1, 2])) * GI,
+                            priority=int(rng.randint(0, 4)))
+        config = default_config()
+        config.use_hip = use_hip
+        config.device = device
+        for tier in config.tiers:
+            for p in tier.plugins:
+                if p.name == "drf":
+                    p.arguments = {"enableHierarchy": True}
+        binder = FakeBinder()
+        cache = SchedulerCache(store=store, binder=binder, device=device)
+        sched = Scheduler(cache, config)
+        sched.run_once()
+        sched.run_once()
+        return binder.binds
+
+    cpu = build("cpu", False)
+    gpu = build("cuda", True)
+    assert len(cpu) > 50
+    assert cpu == gpu, (
+        f"HDRF decision divergence: {len(cpu)} vs {len(gpu)}; "
+        f"diff={sorted(set(cpu.items()) ^ set(gpu.items()))[:6]}")
