@@ -25,11 +25,15 @@ from ..statement import Statement
 
 
 def victim_sort_key(ssn, victim: TaskInfo):
-    """Lowest-value victims first: priority asc, youngest first
-    (reference victim ordering, session_plugins victim order)."""
+    """Lowest-value victims first: victim queue priority asc (reclaim
+    gives back from the least important queue first — reference
+    reclaim_test.go "different queue priority"), then task priority asc,
+    youngest first."""
     job = ssn.jobs.get(victim.job_key)
     created = job.creation_timestamp if job else 0.0
-    return (victim.priority, -created)
+    q = ssn.queues.get(job.queue) if job else None
+    qprio = q.priority if q is not None else 0
+    return (qprio, victim.priority, -created)
 
 
 def _node_fits(nt, ni: NodeInfo, req_vec: np.ndarray, extra: float = 0.1) -> bool:

@@ -22,6 +22,9 @@ class ReclaimAction(PreemptAction):
         if nt is None or nt.n == 0 or not ssn.reclaimable_fns:
             return
         for q in ssn.sorted_queues():
+            if not q.is_open:
+                continue        # closed queues cannot reclaim
+                # (reference reclaim.go queue state check)
             if ssn.queue_overused(q):
                 continue        # only under-served queues reclaim
             jobs_in_q = [j for j in ssn.jobs.values() if j.queue == q.name

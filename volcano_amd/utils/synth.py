@@ -41,14 +41,20 @@ def make_queue(name: str, weight: int = 1,
                guarantee: Optional[Dict[str, float]] = None,
                reclaimable: bool = True,
                parent: str = "",
-               deserved: Optional[Dict[str, float]] = None) -> Queue:
-    return Queue(meta=ObjectMeta(name=name, namespace="default"),
-                 spec=QueueSpec(weight=weight,
-                                capability=Resource(capability or {}),
-                                guarantee=Resource(guarantee or {}),
-                                reclaimable=reclaimable,
-                                parent=parent,
-                                deserved=Resource(deserved or {})))
+               deserved: Optional[Dict[str, float]] = None,
+               priority: int = 0,
+               state: Optional[str] = None) -> Queue:
+    q = Queue(meta=ObjectMeta(name=name, namespace="default"),
+              spec=QueueSpec(weight=weight,
+                             capability=Resource(capability or {}),
+                             guarantee=Resource(guarantee or {}),
+                             reclaimable=reclaimable,
+                             parent=parent,
+                             priority=priority,
+                             deserved=Resource(deserved or {})))
+    if state is not None:
+        q.status.state = state
+    return q
 
 
 def make_podgroup(name: str, queue: str = "default", min_member: int = 1,
