@@ -180,3 +180,49 @@ def test_reclaim_counts_node_idle_toward_need():
         expect_evicted=["c1/preemptee1-1"],
     ).run()
     t.check_all()
+
+
+def test_reclaim_second_task_when_first_is_preempt_never():
+    """reclaim_test.go:262 — per-TASK eligibility: the Never-policy task
+    waits, but its gang-mate without the policy still reclaims."""
+    never = pod("c1", "preemptor-task1", "pg2", 1000, G, prio=1000)
+    never.preemption_policy = "Never"
+    t = TestCommonStruct(
+        name="second task reclaims when first is PreemptNever",
+        podgroups=[pg("c1", "pg1", "q1", 1, prio=100, phase="Running"),
+                   pg("c1", "pg2", "q2", 2, prio=1000)],
+        pods=[pod("c1", "victim-pod-no", "pg1", 1000, G, "n1", "Running",
+                  preemptable="false", prio=100),
+              pod("c1", "victim-pod", "pg1", 1000, G, "n1", "Running",
+                  preemptable="true", prio=100),
+              never,
+              pod("c1", "preemptor-task2", "pg2", 1000, G, prio=900)],
+        nodes=[synth.make_node("n1", cpu_milli=2000, mem=2 * G, pods=10)],
+        queues=[synth.make_queue("q1"), synth.make_queue("q2")],
+        actions=RECLAIM,
+        tiers=[["conformance", "gang", "proportion", "priority"]],
+        plugin_args={"gang": {"enabledJobPipelined": False}},
+        expect_evicted=["c1/victim-pod"],
+    ).run()
+    t.check_all()
+
+
+def test_reclaim_no_eviction_when_node_idle_suffices():
+    """reclaim_test.go:312 regression — FutureIdle already fits the
+    preemptor: zero evictions committed even though the victim's queue
+    is over its deserved share."""
+    t = TestCommonStruct(
+        name="no eviction when idle suffices",
+        podgroups=[pg("c1", "pg1", "q1", 0, prio=100, phase="Running"),
+                   pg("c1", "pg2", "q2", 1, prio=100)],
+        pods=[pod("c1", "victim-n1", "pg1", 2000, 2 * G, "n1", "Running",
+                  preemptable="true", prio=100),
+              pod("c1", "preemptor1", "pg2", 3000, 3 * G, prio=100)],
+        nodes=[synth.make_node("n1", cpu_milli=10000, mem=10 * G, pods=10)],
+        queues=[synth.make_queue("q1", weight=1),
+                synth.make_queue("q2", weight=9)],
+        actions=RECLAIM,
+        expect_evicted=[],
+        expect_pipelined=None,
+    ).run()
+    t.check_all()

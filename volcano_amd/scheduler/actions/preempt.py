@@ -252,7 +252,11 @@ class PreemptAction:
                     if headroom is not None:
                         headroom -= req_vec
 
-        if ssn.job_pipelined(job) and job.is_pipelined():
+        # commit on the JobPipelined VOTE alone (reference reclaim.go:163 /
+        # preempt.go): with the gang plugin's pipelined gate enabled this
+        # equals the occupied+waiting>=min check; with it disabled (conf
+        # flags) partial progress commits — reference semantics
+        if stmt.ops and ssn.job_pipelined(job):
             stmt.commit()
         else:
             stmt.discard()

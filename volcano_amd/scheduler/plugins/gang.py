@@ -72,7 +72,8 @@ class GangPlugin(Plugin):
                              col=lambda jt, rows:
                                  jt.occ[rows] >= jt.minav[rows])
         ssn.job_ready_fns.append(job_ready)
-        ssn.job_pipelined_fns.append(job_pipelined)
+        if self.args.get("enabledJobPipelined", True) is not False:
+            ssn.job_pipelined_fns.append(job_pipelined)
         ssn.job_starving_fns.append(job_starving)
         # per-function enable flags (reference conf.PluginOption
         # EnabledPreemptable/EnabledReclaimable — default on)
