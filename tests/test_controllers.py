@@ -459,3 +459,47 @@ def test_datasource_claim_binding_and_reclaim():
     store.delete("DataSourceClaim", "ml", "train-input")
     mgr.sync_once()
     assert store.get("DataSource", "", "sales") is None
+
+
+def test_job_state_action_matrix():
+    """Reference job_state_test.go per-state action sets: finished jobs
+    ignore every action; Aborted acts only on Resume (→ Restarting with
+    a retry bump); live phases accept the full action set."""
+    from volcano_amd.api.types import Action
+    store, cm, sched, kubelet = mk_world()
+    jc = next(c for c in cm.controllers if c.__class__.__name__
+              == "JobController")
+
+    def job_in(phase):
+        j = mk_job("sm", replicas=1)
+        j.status.phase = phase
+        store.create("Job", j)
+        return j
+
+    # finished states: every action is a no-op
+    for phase in ("Completed", "Failed", "Terminated"):
+        j = job_in(phase)
+        for act in ("AbortJob", "RestartJob", "CompleteJob",
+                    "TerminateJob", "ResumeJob"):
+            jc.execute_action(j, act)
+            assert j.status.phase == phase, (phase, act)
+        store.delete("Job", "default", "sm")
+
+    # Aborted: only Resume acts — Restarting with retry bump
+    j = job_in("Aborted")
+    for act in ("AbortJob", "RestartJob", "CompleteJob", "TerminateJob"):
+        jc.execute_action(j, act)
+        assert j.status.phase == "Aborted", act
+    jc.execute_action(j, "ResumeJob")
+    assert j.status.phase == "Restarting"
+    assert j.status.retry_count == 1
+    store.delete("Job", "default", "sm")
+
+    # live phases take the whole action set
+    for act, want in (("AbortJob", "Aborting"),
+                      ("TerminateJob", "Terminating"),
+                      ("CompleteJob", "Completing")):
+        j = job_in("Running")
+        jc.execute_action(j, act)
+        assert j.status.phase == want, act
+        store.delete("Job", "default", "sm")

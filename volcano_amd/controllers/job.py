@@ -164,7 +164,22 @@ class JobController(Controller):
 
     def execute_action(self, job: Job, action: str, task_name: str = "",
                        pod: Optional[Pod] = None) -> None:
-        """bus/v1alpha1 actions (actions.go:19-39) → state transitions."""
+        """bus/v1alpha1 actions (actions.go:19-39) → state transitions.
+        Per-state action sets follow the reference state machines
+        (controllers/job/state/*.go, job_state_test.go): finished jobs
+        ignore every action; an Aborted job acts only on Resume (which
+        moves it to Restarting with a retry bump — aborted.go:29-40)."""
+        phase = job.status.phase
+        # Aborted is resumable, so it is NOT in the finished set here
+        if phase in FINAL_PHASES and phase != JobPhase.ABORTED.value:
+            return                      # finishedState: no-op
+        if phase == JobPhase.ABORTED.value:
+            if action == Action.RESUME_JOB.value:
+                job.status.retry_count += 1
+                job.status.phase = JobPhase.RESTARTING.value
+                self.store.update("Job", job)
+                self._dirty.add(job.meta.key)
+            return                      # abortedState: other actions no-op
         if action == Action.ABORT_JOB.value:
             job.status.phase = JobPhase.ABORTING.value
         elif action == Action.TERMINATE_JOB.value:
