@@ -107,8 +107,8 @@ def reset_cluster(cache: SchedulerCache, jobs):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=3)
-    ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=5)
+    ap.add_argument("--warmup", type=int, default=2)
     ap.add_argument("--nodes", type=int, default=10000)
     ap.add_argument("--jobs", type=int, default=10000)
     ap.add_argument("--pods-per-job", type=int, default=10)
