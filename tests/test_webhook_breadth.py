@@ -122,3 +122,61 @@ def test_root_queue_rejected():
     chain = default_chain(store)
     with pytest.raises(AdmissionError, match="root queue"):
         chain.admit("Job", mk_job(queue="root"), "CREATE")
+
+
+def test_queue_validate_depth():
+    """Reference validate_queue_test.go: state legality, guarantee ≤
+    deserved ≤ capability, hierarchy/weights alignment, and delete
+    protection for default/root and parents with children."""
+    import pytest
+
+    from volcano_amd.utils import synth
+    from volcano_amd.webhooks import AdmissionError, default_chain
+    from volcano_amd.store import ObjectStore
+
+    store = ObjectStore()
+    chain = default_chain(store)
+    g = chain.guard(store)
+    G = 1024 ** 3
+
+    # state legality
+    bad = synth.make_queue("q-badstate", state="abnormal-case")
+    with pytest.raises(AdmissionError, match="state"):
+        g.create("Queue", bad)
+
+    # deserved >= guarantee required (deserved missing)
+    with pytest.raises(AdmissionError, match="deserved"):
+        g.create("Queue", synth.make_queue("q-g", guarantee={"cpu": 2000.0}))
+
+    # capability >= deserved
+    with pytest.raises(AdmissionError, match="capability"):
+        g.create("Queue", synth.make_queue(
+            "q-cd", capability={"cpu": 1000.0}, deserved={"cpu": 2000.0}))
+
+    # legal tree
+    g.create("Queue", synth.make_queue(
+        "q-ok", capability={"cpu": 4000.0}, deserved={"cpu": 2000.0},
+        guarantee={"cpu": 1000.0}))
+
+    # hierarchy / weights mismatch
+    hq = synth.make_queue("q-h")
+    hq.meta.annotations["volcano.sh/hierarchy"] = "root/sci"
+    hq.meta.annotations["volcano.sh/hierarchy-weights"] = "100/50/25"
+    with pytest.raises(AdmissionError, match="depth"):
+        g.create("Queue", hq)
+    hq.meta.annotations["volcano.sh/hierarchy-weights"] = "100/-1"
+    with pytest.raises(AdmissionError, match="positive"):
+        g.create("Queue", hq)
+    hq.meta.annotations["volcano.sh/hierarchy-weights"] = "100/50"
+    g.create("Queue", hq)
+
+    # delete protection
+    g.create("Queue", synth.make_queue("default"))
+    with pytest.raises(AdmissionError, match="can not be deleted"):
+        g.delete("Queue", "default", "default")
+    g.create("Queue", synth.make_queue("parent-q"))
+    g.create("Queue", synth.make_queue("child-q", parent="parent-q"))
+    with pytest.raises(AdmissionError, match="child queues"):
+        g.delete("Queue", "default", "parent-q")
+    g.delete("Queue", "default", "child-q")
+    g.delete("Queue", "default", "parent-q")      # drains bottom-up

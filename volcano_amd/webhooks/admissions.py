@@ -190,6 +190,58 @@ def mutate_queue(store, queue, op) -> None:
 def validate_queue(store, queue, op) -> None:
     if not queue.meta.name:
         raise AdmissionError("queue name required")
+    if op == "DELETE":
+        # validate_queue.go validateQueueDeleting: default/root are
+        # permanent; parents with children must be drained bottom-up
+        if queue.meta.name in ("default", "root"):
+            raise AdmissionError(
+                f"`{queue.meta.name}` queue can not be deleted")
+        if store is not None:
+            kids = [q.meta.name for q in store.list("Queue")
+                    if q.spec.parent == queue.meta.name]
+            if kids:
+                raise AdmissionError(
+                    f"queue {queue.meta.name} can not be deleted because "
+                    f"it has {len(kids)} child queues: {', '.join(kids)}")
+        return
+    # state legality (validateStateOfQueue)
+    if queue.status.state not in ("", "Open", "Closed"):
+        raise AdmissionError(
+            "queue state must be in [Open Closed]")
+    # guarantee ≤ deserved ≤ capability, dimension-wise where set
+    # (validateResourceQuantityOfQueue)
+    cap = queue.spec.capability.q
+    des = queue.spec.deserved.q
+    gua = queue.spec.guarantee.q
+    for r, g in gua.items():
+        d = des.get(r)
+        if d is None or d < g:
+            raise AdmissionError(
+                f"deserved[{r}] must be >= guarantee[{r}]={g:g}")
+    for r, d in des.items():
+        c = cap.get(r)
+        if c is not None and c < d:
+            raise AdmissionError(
+                f"capability[{r}]={c:g} must be >= deserved[{r}]={d:g}")
+    # hierarchy annotations: path and weights must align, weights > 0
+    # (validate_queue.go:150-176)
+    h = queue.meta.annotations.get("volcano.sh/hierarchy", "")
+    hw = queue.meta.annotations.get("volcano.sh/hierarchy-weights", "")
+    if h or hw:
+        paths = h.split("/")
+        weights = hw.split("/")
+        if len(paths) != len(weights):
+            raise AdmissionError(
+                f"hierarchy {h!r} and weights {hw!r} must have the "
+                "same depth")
+        for w in weights:
+            try:
+                ok = float(w) > 0
+            except ValueError:
+                ok = False
+            if not ok:
+                raise AdmissionError(
+                    f"hierarchy weight {w!r} must be a positive number")
     if queue.spec.parent and store is not None:
         parent = store.get("Queue", "default", queue.spec.parent)
         if parent is None:
@@ -301,7 +353,8 @@ def register_all(chain: AdmissionChain) -> None:
         AdmissionService("/jobs/mutate", "Job", mutate_job),
         AdmissionService("/jobs/validate", "Job", validate_job),
         AdmissionService("/queues/mutate", "Queue", mutate_queue),
-        AdmissionService("/queues/validate", "Queue", validate_queue),
+        AdmissionService("/queues/validate", "Queue", validate_queue,
+                         operations=("CREATE", "UPDATE", "DELETE")),
         AdmissionService("/podgroups/validate", "PodGroup", validate_podgroup),
         AdmissionService("/pods/mutate", "Pod", mutate_pod),
         AdmissionService("/jobflows/validate", "JobFlow", validate_jobflow),

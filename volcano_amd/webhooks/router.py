@@ -67,6 +67,12 @@ class GuardedStore:
         self._chain.admit(kind, obj, "UPDATE")
         return self._store.update(kind, obj)
 
+    def delete(self, kind: str, namespace: str, name: str):
+        obj = self._store.get(kind, namespace, name)
+        if obj is not None:
+            self._chain.admit(kind, obj, "DELETE")
+        return self._store.delete(kind, namespace, name)
+
     def apply(self, kind: str, obj):
         op = "UPDATE" if self._store.get(
             kind, obj.meta.namespace, obj.meta.name) else "CREATE"
