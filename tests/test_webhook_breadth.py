@@ -180,3 +180,32 @@ def test_queue_validate_depth():
         g.delete("Queue", "default", "parent-q")
     g.delete("Queue", "default", "child-q")
     g.delete("Queue", "default", "parent-q")      # drains bottom-up
+
+
+def test_pod_validate_jdb_annotations():
+    """Reference admit_pod.go: jdb-min-available / jdb-max-unavailable
+    must be positive int-or-percentage and mutually exclusive."""
+    import pytest
+
+    from volcano_amd.utils import synth
+    from volcano_amd.store import ObjectStore
+    from volcano_amd.webhooks import AdmissionError, default_chain
+
+    store = ObjectStore()
+    g = default_chain(store).guard(store)
+    g.create("Queue", synth.make_queue("default"))
+
+    ok = synth.make_pod("p-ok", "")
+    ok.meta.annotations["scheduling.volcano.sh/jdb-min-available"] = "30%"
+    g.create("Pod", ok)
+
+    bad = synth.make_pod("p-bad", "")
+    bad.meta.annotations["scheduling.volcano.sh/jdb-min-available"] = "zero"
+    with pytest.raises(AdmissionError, match="positive integer"):
+        g.create("Pod", bad)
+
+    both = synth.make_pod("p-both", "")
+    both.meta.annotations["scheduling.volcano.sh/jdb-min-available"] = "1"
+    both.meta.annotations["scheduling.volcano.sh/jdb-max-unavailable"] = "1"
+    with pytest.raises(AdmissionError, match="multiple annotations"):
+        g.create("Pod", both)

@@ -282,6 +282,40 @@ def validate_podgroup(store, pg, op) -> None:
 
 # -- pods ---------------------------------------------------------------------
 
+JDB_MIN_AVAILABLE = "scheduling.volcano.sh/jdb-min-available"
+JDB_MAX_UNAVAILABLE = "scheduling.volcano.sh/jdb-max-unavailable"
+
+
+def _int_or_percentage(key: str, value: str) -> None:
+    """validateIntPercentageStr (admit_pod.go): "3" or "25%", > 0."""
+    v = value.strip()
+    if v.endswith("%"):
+        v = v[:-1]
+    try:
+        ok = int(v) > 0
+    except ValueError:
+        ok = False
+    if not ok:
+        raise AdmissionError(
+            f"invalid value {value!r} for annotation {key}: must be a "
+            "positive integer or percentage")
+
+
+def validate_pod(store, pod, op) -> None:
+    """reference admission/pods/validate/admit_pod.go:99-135 — the job
+    disruption budget annotations must parse and are mutually
+    exclusive."""
+    anns = pod.meta.annotations
+    present = [k for k in (JDB_MIN_AVAILABLE, JDB_MAX_UNAVAILABLE)
+               if k in anns]
+    for k in present:
+        _int_or_percentage(k, anns[k])
+    if len(present) > 1:
+        raise AdmissionError(
+            f"not allow configure multiple annotations {present} at "
+            "same time")
+
+
 def mutate_pod(store, pod, op) -> None:
     """reference admission/pods/mutate: annotate the scheduler."""
     if not pod.scheduler_name:
@@ -357,6 +391,7 @@ def register_all(chain: AdmissionChain) -> None:
                          operations=("CREATE", "UPDATE", "DELETE")),
         AdmissionService("/podgroups/validate", "PodGroup", validate_podgroup),
         AdmissionService("/pods/mutate", "Pod", mutate_pod),
+        AdmissionService("/pods/validate", "Pod", validate_pod),
         AdmissionService("/jobflows/validate", "JobFlow", validate_jobflow),
         AdmissionService("/cronjobs/validate", "CronJob", validate_cronjob),
         AdmissionService("/hypernodes/validate", "HyperNode",
