@@ -209,3 +209,32 @@ def test_pod_validate_jdb_annotations():
     both.meta.annotations["scheduling.volcano.sh/jdb-max-unavailable"] = "1"
     with pytest.raises(AdmissionError, match="multiple annotations"):
         g.create("Pod", both)
+
+
+def test_podgroup_inherits_namespace_queue():
+    """Reference mutate_podgroup_test.go: a podgroup on the default
+    queue picks up its namespace's queue-name annotation; an explicit
+    queue wins."""
+    from volcano_amd.api.objects import Namespace, ObjectMeta
+    from volcano_amd.store import ObjectStore
+    from volcano_amd.utils import synth
+    from volcano_amd.webhooks import default_chain
+
+    store = ObjectStore()
+    g = default_chain(store).guard(store)
+    ns = Namespace(meta=ObjectMeta(name="team-ns", namespace=""))
+    ns.meta.annotations["scheduling.volcano.sh/queue-name"] = "ns-queue"
+    store.create("Namespace", ns)
+
+    pg = synth.make_podgroup("pg-default", namespace="team-ns")
+    g.create("PodGroup", pg)
+    assert pg.spec.queue == "ns-queue"
+
+    pg2 = synth.make_podgroup("pg-explicit", namespace="team-ns",
+                              queue="custom-queue")
+    g.create("PodGroup", pg2)
+    assert pg2.spec.queue == "custom-queue"
+
+    pg3 = synth.make_podgroup("pg-plain", namespace="other-ns")
+    g.create("PodGroup", pg3)
+    assert pg3.spec.queue == "default"

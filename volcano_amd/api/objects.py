@@ -99,6 +99,17 @@ class Node:
 
 
 @dataclass
+class Namespace:
+    """core/v1 Namespace — carried for the queue-defaulting admission
+    path (reference mutate_podgroup.go:104-125: a namespace may pin its
+    workloads to a queue via the ``scheduling.volcano.sh/queue-name``
+    annotation)."""
+
+    meta: ObjectMeta = field(default_factory=ObjectMeta)
+    phase: str = "Active"
+
+
+@dataclass
 class Pod:
     """core/v1 Pod reduced to scheduling-relevant fields."""
 
@@ -615,4 +626,5 @@ KINDS = {
     "PersistentVolumeClaim": PersistentVolumeClaim,
     "DeviceClass": DeviceClass, "ResourceClaim": ResourceClaim,
     "DataSource": DataSource, "DataSourceClaim": DataSourceClaim,
+    "Namespace": Namespace,
 }

@@ -30,6 +30,7 @@ GVK: Dict[str, Tuple[str, str, str]] = {
                          "colocationconfigs"),
     "HyperJob": ("training.volcano.sh", "v1alpha1", "hyperjobs"),
     "Pod": ("", "v1", "pods"),
+    "Namespace": ("", "v1", "namespaces"),
     "Node": ("", "v1", "nodes"),
     "PersistentVolume": ("", "v1", "persistentvolumes"),
     "PersistentVolumeClaim": ("", "v1", "persistentvolumeclaims"),

@@ -260,6 +260,19 @@ def validate_queue(store, queue, op) -> None:
 
 # -- podgroups ----------------------------------------------------------------
 
+def mutate_podgroup(store, pg, op) -> None:
+    """reference mutate_podgroup.go:104-125: a podgroup still on the
+    default queue inherits its namespace's queue-name annotation."""
+    if pg.spec.queue != DEFAULT_QUEUE or store is None:
+        return
+    ns = store.get("Namespace", "", pg.meta.namespace) or         store.get("Namespace", "default", pg.meta.namespace)
+    if ns is None:
+        return
+    q = ns.meta.annotations.get("scheduling.volcano.sh/queue-name")
+    if q:
+        pg.spec.queue = q
+
+
 def validate_podgroup(store, pg, op) -> None:
     if pg.spec.min_member < 0:
         raise AdmissionError("minMember must be >= 0")
@@ -389,6 +402,7 @@ def register_all(chain: AdmissionChain) -> None:
         AdmissionService("/queues/mutate", "Queue", mutate_queue),
         AdmissionService("/queues/validate", "Queue", validate_queue,
                          operations=("CREATE", "UPDATE", "DELETE")),
+        AdmissionService("/podgroups/mutate", "PodGroup", mutate_podgroup),
         AdmissionService("/podgroups/validate", "PodGroup", validate_podgroup),
         AdmissionService("/pods/mutate", "Pod", mutate_pod),
         AdmissionService("/pods/validate", "Pod", validate_pod),
