@@ -139,3 +139,40 @@ def test_manifest_round_trip_preserves_opaque_maps():
     assert back["spec"]["minMember"] == 2
     assert back["metadata"]["labels"] == {"volcano.sh/job-name": "x"}
     assert back["spec"]["networkTopology"]["highestTierAllowed"] == 1
+
+
+def test_datadependency_group_wire_paths():
+    """datadependency/v1alpha1 over the k8s wire: create a cluster-scoped
+    DataSource and a namespaced claim, list both back, and let the
+    controller bind them."""
+    store, client = mk_client()
+    r = client.post(
+        "/apis/datadependency.volcano.sh/v1alpha1/datasources", json={
+            "apiVersion": "datadependency.volcano.sh/v1alpha1",
+            "kind": "DataSource",
+            "metadata": {"name": "sales", "namespace": ""},
+            "system": "hive", "type": "table", "name": "db.sales",
+            "clusterNames": ["c1"],
+        })
+    assert r.status_code == 200, r.text
+    r = client.post(
+        "/apis/datadependency.volcano.sh/v1alpha1/namespaces/ml/"
+        "datasourceclaims", json={
+            "apiVersion": "datadependency.volcano.sh/v1alpha1",
+            "kind": "DataSourceClaim",
+            "metadata": {"name": "train-in", "namespace": "ml"},
+            "system": "hive", "dataSourceType": "table",
+            "dataSourceName": "db.sales",
+            "workload": {"apiVersion": "batch.volcano.sh/v1alpha1",
+                         "kind": "Job", "name": "train"},
+        })
+    assert r.status_code == 200, r.text
+    r = client.get(
+        "/apis/datadependency.volcano.sh/v1alpha1/datasources")
+    assert len(r.json()["items"]) == 1
+
+    from volcano_amd.controllers import ControllerManager
+    mgr = ControllerManager(store, controllers=["datadependency"])
+    mgr.sync_once()
+    c = store.get("DataSourceClaim", "ml", "train-in")
+    assert c.phase == "Bound" and c.bound_data_source == "sales"
