@@ -262,8 +262,9 @@ class GangReclaimAction(GangPreemptAction):
         nt = ssn.node_tensors
         if nt is None or nt.n == 0 or not ssn.reclaimable_fns:
             return
+        self._parse_arguments(ssn)      # gangreclaim shares the knobs
         for q in ssn.sorted_queues():
-            if ssn.queue_overused(q):
+            if not q.is_open or ssn.queue_overused(q):
                 continue
             jobs_in_q = [j for j in ssn.jobs.values() if j.queue == q.name
                          and j.phase in (PodGroupPhase.INQUEUE.value,
