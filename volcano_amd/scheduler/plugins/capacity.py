@@ -126,6 +126,10 @@ class CapacityPlugin(Plugin):
         limit = torch.where(total.unsqueeze(0) > 0, deserved,
                             torch.full_like(deserved, BIG_LIMIT))
         ssn.queue_limit = torch.maximum(limit, ssn.queue_alloc)
+        # hierarchical-DRF equilibrium caps (plugins/drf.py) must survive
+        # this row replacement regardless of plugin tier order
+        for _qi, _cap in getattr(ssn, "hdrf_cap_rows", {}).items():
+            ssn.queue_limit[_qi] = torch.minimum(ssn.queue_limit[_qi], _cap)
         self.deserved = deserved
 
         alloc_np = ssn.queue_alloc.numpy()

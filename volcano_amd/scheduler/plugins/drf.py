@@ -307,9 +307,20 @@ class DrfPlugin(Plugin):
                 if p[3] <= 0:
                     continue
                 need = p[2]
-                if ((need > free + 1e-9) & (need > 0)).any():
+                # relative tolerance: f32-sourced byte counts drift at
+                # the 1e-7 scale, absolute epsilons don't cover 1e9-byte
+                # dims
+                if ((need > free + 1e-9 + 1e-6 * need) & (need > 0)).any():
                     continue
-                if best is None or tree.compare_queues(p[1], best[1]) < 0:
+                if best is None:
+                    best = p
+                    continue
+                ret = tree.compare_queues(p[1], best[1])
+                # same-queue entries order by leaf (job) dominant share —
+                # the reference's within-queue DRF job order, so
+                # complementary-demand jobs in one queue alternate
+                if ret < 0 or (ret == 0
+                               and p[0].share < best[0].share - 1e-12):
                     best = p
             if best is None:
                 break
@@ -322,15 +333,23 @@ class DrfPlugin(Plugin):
             caps[qn] = quantum.copy() if prev is None else prev + quantum
         if caps and ssn.queue_limit is not None and ssn.queue_index:
             lim = ssn.queue_limit
+            hdrf_rows = {}
             for qn, extra in caps.items():
                 qi = ssn.queue_index.get(qn)
                 if qi is None:
                     continue
                 base = ssn.queue_alloc[qi] if ssn.queue_alloc is not None \
                     else torch.zeros(nt.r)
-                hdrf_cap = base + torch.from_numpy(
-                    extra.astype(np.float32))
+                # relative slack ≫ f32 rounding of big byte counts but
+                # ≪ one task quantum, so the clamp never eats a slot
+                hdrf_cap = (base + torch.from_numpy(
+                    extra.astype(np.float32))) * (1.0 + 1e-5)
                 lim[qi] = torch.minimum(lim[qi], hdrf_cap)
+                hdrf_rows[qi] = hdrf_cap
+            # proportion/capacity REPLACE ssn.queue_limit when they open
+            # after drf in the tier order — publish the equilibrium caps
+            # so they re-apply them (queue-limit row combination)
+            ssn.hdrf_cap_rows = hdrf_rows
 
     @staticmethod
     def _leaf(tree: HdrfTree, job_key: str):
