@@ -215,3 +215,48 @@ def test_preemption_policy_never_blocks_preemption():
         expect_evicted=[],
     ).run()
     t.check_all()
+
+
+def test_preempt_rollback_when_queue_capacity_fails():
+    """preempt_test.go "rollback evictions...": the node has idle room
+    and evictable victims, but the preemptor (4 CPU) exceeds the queue's
+    3-CPU capability even after evicting — no eviction may commit."""
+    t = TestCommonStruct(
+        name="rollback when preemptor exceeds queue capacity",
+        podgroups=[pg("c1", "pg1", "q1", 1, prio=1),
+                   pg("c1", "pg2", "q1", 1, prio=100)],
+        pods=[pod("c1", "preemptee1", "pg1", 1000, G, "n1", "Running",
+                  preemptable="true", prio=1),
+              pod("c1", "preemptee2", "pg1", 1000, G, "n1", "Running",
+                  preemptable="true", prio=1),
+              pod("c1", "preemptor1", "pg2", 4000, 4 * G, prio=100)],
+        nodes=[synth.make_node("n1", cpu_milli=10000, mem=10 * G, pods=10)],
+        queues=[q("q1", cap={"cpu": 3000.0, "memory": 3.0 * G})],
+        actions=PREEMPT,
+        expect_evicted=[],
+    ).run()
+    t.check_all()
+
+
+def test_preempt_commits_only_on_winning_node():
+    """preempt_test.go "only commit evictions on the node where
+    preemption succeeds": n1's small victim cannot bring the queue
+    under its capability, n2's large one can — only n2's eviction
+    commits."""
+    t = TestCommonStruct(
+        name="only commit on winning node",
+        podgroups=[pg("c1", "pg1", "q1", 0, prio=1),
+                   pg("c1", "pg2", "q1", 1, prio=100),
+                   pg("c1", "pg3", "q1", 0, prio=1)],
+        pods=[pod("c1", "preemptee1", "pg1", 1000, G, "n1", "Running",
+                  preemptable="true", prio=1),
+              pod("c1", "preemptee2", "pg3", 4000, 4 * G, "n2", "Running",
+                  preemptable="true", prio=1),
+              pod("c1", "preemptor1", "pg2", 3000, 3 * G, prio=100)],
+        nodes=[synth.make_node("n1", cpu_milli=3000, mem=3 * G, pods=10),
+               synth.make_node("n2", cpu_milli=4000, mem=4 * G, pods=10)],
+        queues=[q("q1", cap={"cpu": 6000.0, "memory": 6.0 * G})],
+        actions=PREEMPT,
+        expect_evicted=["c1/preemptee2"],
+    ).run()
+    t.check_all()

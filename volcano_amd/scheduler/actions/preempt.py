@@ -238,15 +238,9 @@ class PreemptAction:
             for task in tc.tasks:
                 if placed >= still_needed:
                     break
-                if headroom is not None and \
-                        (req_vec > headroom + 0.1).any():
-                    # over queue quota: evict same-queue victims to free
-                    # quota (they also free their nodes)
-                    if not self._evict_for_quota(ssn, stmt, nt, req_vec,
-                                                 headroom, victims_by_node):
-                        break
                 node = self._preempt_one(ssn, stmt, job, task, req_vec,
-                                         order, victims_by_node)
+                                         order, victims_by_node,
+                                         headroom=headroom)
                 if node is not None:
                     placed += 1
                     if headroom is not None:
@@ -260,26 +254,6 @@ class PreemptAction:
             stmt.commit()
         else:
             stmt.discard()
-
-    @staticmethod
-    def _evict_for_quota(ssn, stmt, nt, req_vec, headroom,
-                         victims_by_node) -> bool:
-        """Evict lowest-value same-queue victims until the queue headroom
-        covers one preemptor task (in-place headroom update).  Returns
-        False when even evicting every candidate cannot free the
-        quota."""
-        flat = sorted((v for vs in victims_by_node.values() for v in vs),
-                      key=lambda v: victim_sort_key(ssn, v))
-        from ...api.types import TaskStatus
-        for v in flat:
-            if not (req_vec > headroom + 0.1).any():
-                return True
-            if v.status == TaskStatus.RELEASING:
-                continue
-            stmt.evict(v)
-            headroom += nt.req_vector(v) if nt.req_vector(v) is not None \
-                else 0.0
-        return not (req_vec > headroom + 0.1).any()
 
     def _candidate_scan(self, ssn, job, tc, req_vec, constraints,
                         same_queue, victim_filter, node_filter=None):
@@ -361,27 +335,45 @@ class PreemptAction:
         return order, victims_by_node
 
     def _preempt_one(self, ssn, stmt: Statement, job: JobInfo,
-                     task: TaskInfo, req_vec, order, victims_by_node
-                     ) -> Optional[str]:
+                     task: TaskInfo, req_vec, order, victims_by_node,
+                     headroom=None) -> Optional[str]:
         """Place one preemptor task: walk the kernel-ranked candidate
-        nodes, evict just-enough of the node's (pre-sorted) victims."""
+        nodes, evict just-enough of the node's (pre-sorted) victims.
+        ``headroom`` (same-queue quota slack) joins the per-node trial:
+        a node whose victims cannot ALSO restore quota is skipped with
+        no eviction committed — the reference's per-node nodeStmt
+        isolation (preempt_test.go "only commit evictions on the node
+        where preemption succeeds").  On success the freed quota is
+        credited back into ``headroom`` in place."""
         nt = ssn.node_tensors
+        didx = nt.dims.index
+
+        def head_ok(gain):
+            if headroom is None:
+                return True
+            for name, i in didx.items():
+                if req_vec[i] > 0.1 and \
+                        headroom[i] + gain.get(name, 0.0) + 0.1 < req_vec[i]:
+                    return False
+            return True
 
         for ni in order:
-            if _node_fits(nt, ni, req_vec):
+            no_gain = {}
+            if _node_fits(nt, ni, req_vec) and head_ok(no_gain):
                 stmt.pipeline(task, ni.name)
                 return ni.name
             cands = victims_by_node.get(ni.name, [])
-            # evict-just-enough simulation over future idle
+            # evict-just-enough simulation over future idle + quota
             chosen = []
             fi = ni.future_idle
-            avail = {k: fi.get(k) for k in nt.dims.index}
+            avail = {k: fi.get(k) for k in didx}
+            gain = {}
 
             def fits():
-                for name, i in nt.dims.index.items():
+                for name, i in didx.items():
                     if req_vec[i] > 0.1 and avail[name] + 0.1 < req_vec[i]:
                         return False
-                return True
+                return head_ok(gain)
 
             for v in cands:
                 if fits():
@@ -392,10 +384,16 @@ class PreemptAction:
                 for name, val in v.request.q.items():
                     if name in avail:
                         avail[name] += val
+                    gain[name] = gain.get(name, 0.0) + val
             if not fits():
-                continue
+                continue               # per-node trial failed: no evictions
             for v in chosen:
                 stmt.evict(v)
+            if headroom is not None:
+                for name, val in gain.items():
+                    i = didx.get(name)
+                    if i is not None:
+                        headroom[i] += val
             stmt.pipeline(task, ni.name)
             return ni.name
         return None
