@@ -260,3 +260,25 @@ def test_preempt_commits_only_on_winning_node():
         expect_evicted=["c1/preemptee2"],
     ).run()
     t.check_all()
+
+
+def test_preempt_multi_queue_independent():
+    """preempt_test.go multi-queue regression: a non-starving q1 must
+    not interfere with q2's preemptor — pg3 evicts pg2's pod regardless
+    of queue iteration order."""
+    t = TestCommonStruct(
+        name="multi-queue preemptor isolation",
+        podgroups=[pg("c1", "pg1", "q1", 1),
+                   pg("c1", "pg2", "q2", 0, prio=1),
+                   pg("c1", "pg3", "q2", 1, prio=100)],
+        pods=[pod("c1", "q1-runner1", "pg1", 1000, G, "n1", "Running"),
+              pod("c1", "q2-preemptee1", "pg2", 1000, G, "n1", "Running",
+                  preemptable="true", prio=1),
+              pod("c1", "q2-preemptor1", "pg3", 1000, G, prio=100)],
+        nodes=[synth.make_node("n1", cpu_milli=2000, mem=2 * G, pods=10)],
+        queues=[q("q1"),
+                q("q2", cap={"cpu": 4000.0, "memory": 4.0 * G})],
+        actions=PREEMPT,
+        expect_evicted=["c1/q2-preemptee1"],
+    ).run()
+    t.check_all()
